@@ -1,0 +1,212 @@
+"""The gradient-accumulation + AdamWeightDecay engine (the reference's core IP).
+
+Maps the reference's ``tf.cond(step % K == 0, apply, accumulate)`` train_op
+(/root/reference/optimization.py:76-103) onto flat
+buffers and (on GPU) three hand-written CDNA4 HIP kernels:
+
+  every micro-step : K1  accum += grad; grad = 0          (one fused launch)
+  apply boundary   : [DP: RCCL all-reduce of the flat accum buffer -- ONCE
+                      per K micro-steps, not per micro-step as the
+                      reference's aggregation=SUM variable does (04:55);
+                      summation is linear so this is exact]
+                     K3  global squared-norm -> device scalar
+                     K4  fused  g = accum/K * clip_coef; Adam m,v; decoupled
+                         weight decay; p -= lr*u; bf16 write-back; accum = 0
+
+Semantics knobs:
+  * ``strict_reference_semantics=True`` reproduces the reference predicate
+    ``global_step % K == 0`` with step starting at 0 -- the step-0 apply sees
+    one micro-batch yet divides by K (documented off-by-one,
+    optimization.py:91). The corrected default applies at
+    ``(global_step+1) % K == 0`` so every window covers exactly K
+    micro-batches.
+  * ``global_step`` counts micro-steps and the LR schedule moves per
+    micro-step (optimization.py:99-103).
+  * Adam without bias correction, eps outside sqrt, decay-by-regex
+    (optimization.py:150-187).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Iterable, Optional, Sequence, Tuple
+
+import torch
+
+from .. import ops as ops_pkg
+from ..ops import eager as eager_ops
+from .flat import DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY, FlatState
+from .schedule import learning_rate
+
+
+class AccumEngine:
+    def __init__(
+        self,
+        named_params: Sequence[Tuple[str, torch.Tensor]],
+        *,
+        init_lr: float,
+        num_train_steps: int,
+        num_warmup_steps: int = 0,
+        gradient_accumulation_multiplier: int = 1,
+        clip_norm: Optional[float] = 1.0,
+        weight_decay: float = 0.01,
+        beta1: float = 0.9,
+        beta2: float = 0.999,
+        eps: float = 1e-6,
+        exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+        strict_reference_semantics: bool = False,
+        process_group=None,
+        allreduce_bucket_mb: int = 64,
+        backend: str = "auto",
+    ):
+        if gradient_accumulation_multiplier < 1:
+            raise ValueError("gradient_accumulation_multiplier must be >= 1")
+        self.state = FlatState(list(named_params), exclude_from_weight_decay)
+        self.K = int(gradient_accumulation_multiplier)
+        self.init_lr = float(init_lr)
+        self.num_train_steps = int(num_train_steps)
+        self.num_warmup_steps = int(num_warmup_steps)
+        self.clip_norm = clip_norm
+        self.weight_decay = float(weight_decay)
+        self.beta1, self.beta2, self.eps = float(beta1), float(beta2), float(eps)
+        self.strict = bool(strict_reference_semantics)
+        self.group = process_group
+        self.allreduce_bucket_elems = max(1, (allreduce_bucket_mb << 20) // 4)
+        self.global_step = 0
+        self.last_lr = 0.0
+
+        dev = self.state.device
+        if backend == "auto":
+            backend = "hip" if dev.type == "cuda" else "eager"
+        if backend == "hip":
+            self._hip = ops_pkg.require_hip()
+        elif backend == "eager":
+            self._hip = None
+        else:
+            raise ValueError(f"unknown backend {backend}")
+        self.backend = backend
+
+        on_dev = dev if dev.type == "cuda" else torch.device("cpu")
+        # lr travels through a device scalar so hipGraph capture replays with
+        # the schedule's current value (SURVEY.md section 7 hard parts).
+        self._lr_dev = torch.zeros(1, dtype=torch.float32, device=on_dev)
+        self._sqnorm_dev = torch.zeros(1, dtype=torch.float32, device=on_dev)
+
+    # ---- world size ----
+    @property
+    def world_size(self) -> int:
+        import torch.distributed as dist
+
+        if self.group is not None:
+            return dist.get_world_size(self.group)
+        if dist.is_available() and dist.is_initialized():
+            return dist.get_world_size()
+        return 1
+
+    # ---- predicate ----
+    def is_apply_step(self, step: Optional[int] = None) -> bool:
+        s = self.global_step if step is None else step
+        if self.strict:
+            return s % self.K == 0
+        return (s + 1) % self.K == 0
+
+    def lr_at(self, step: int) -> float:
+        return learning_rate(step, self.init_lr, self.num_train_steps, self.num_warmup_steps)
+
+    # ---- the per-micro-step body (the thing bench.py hipGraph-captures) ----
+    def accumulate(self) -> None:
+        st = self.state
+        if self._hip is not None:
+            self._hip.accumulate(st.accum, st.grads)
+        else:
+            eager_ops.accumulate(st.accum, st.grads)
+
+    def _allreduce_accum(self) -> None:
+        import torch.distributed as dist
+
+        if self.world_size <= 1:
+            return
+        accum = self.state.accum
+        n = accum.numel()
+        b = self.allreduce_bucket_elems
+        if n <= b:
+            dist.all_reduce(accum, group=self.group)
+            return
+        handles = []
+        for off in range(0, n, b):
+            handles.append(
+                dist.all_reduce(accum[off : min(off + b, n)], group=self.group, async_op=True)
+            )
+        for h in handles:
+            h.wait()
+
+    def apply(self, lr: Optional[float] = None) -> None:
+        st = self.state
+        if lr is None:
+            lr = self.lr_at(self.global_step)
+        self.last_lr = lr
+        inv_k = 1.0 / self.K
+        model = None if st.master is st.model else st.model
+        if self._hip is not None:
+            self._lr_dev.fill_(lr)
+            self._hip.fused_apply(
+                st.accum,
+                st.m,
+                st.v,
+                st.master,
+                model if model is not None else st.master,
+                model is not None,
+                self._lr_dev,
+                self._sqnorm_dev,
+                st.decay_boundary,
+                inv_k,
+                -1.0 if self.clip_norm is None else float(self.clip_norm),
+                self.weight_decay,
+                self.beta1,
+                self.beta2,
+                self.eps,
+            )
+        else:
+            eager_ops.fused_apply(
+                st.accum,
+                st.m,
+                st.v,
+                st.master,
+                model,
+                None,
+                st.decay_boundary,
+                lr=lr,
+                inv_k=inv_k,
+                clip_norm=self.clip_norm,
+                weight_decay=self.weight_decay,
+                beta1=self.beta1,
+                beta2=self.beta2,
+                eps=self.eps,
+            )
+
+    def micro_step(self) -> bool:
+        """One reference session.run: accumulate, maybe apply, step += 1.
+
+        Call after ``loss.backward()`` has filled the flat grad buffer.
+        Returns True if this micro-step applied an optimizer update.
+        """
+        self.accumulate()
+        applied = self.is_apply_step()
+        if applied:
+            self._allreduce_accum()
+            self.apply()
+        self.global_step += 1
+        return applied
+
+    # ---- checkpoint (accum + m/v + step all saved) ----
+    def state_dict(self) -> Dict:
+        d = self.state.state_dict()
+        d["global_step"] = self.global_step
+        d["K"] = self.K
+        d["strict"] = self.strict
+        return d
+
+    def load_state_dict(self, d: Dict) -> None:
+        self.state.load_state_dict(d)
+        self.global_step = int(d["global_step"])
+        if int(d.get("K", self.K)) != self.K:
+            raise ValueError("checkpoint K does not match engine K")

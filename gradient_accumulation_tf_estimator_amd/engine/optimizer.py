@@ -1,0 +1,98 @@
+"""``create_optimizer`` -- the reference's public L3 contract in PyTorch form.
+
+Reference: ``create_optimizer(loss, init_lr, num_train_steps,
+num_warmup_steps, use_tpu) -> train_op`` (/root/reference/optimization.py:25)
+builds LR schedule + AdamWeightDecayOptimizer + gradient accumulation +
+global-norm clip into one graph op. Here the same bundle becomes a
+``TrainOp`` object whose ``step(loss)`` is the ``session.run(train_op)``
+equivalent: backward -> accumulate-or-apply -> global_step += 1
+(SURVEY.md section 3.3).
+
+The non-clipping generic variant (another-example.py:126-155, 02:47-74,
+04:49-74) is ``clip_norm=None``.
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, Optional, Sequence, Tuple, Union
+
+import torch
+import torch.nn as nn
+
+from .accum import AccumEngine
+from .flat import DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY
+
+
+class TrainOp:
+    def __init__(self, engine: AccumEngine, *, ddp_scale_loss: bool = True):
+        self.engine = engine
+        self.ddp_scale_loss = ddp_scale_loss
+
+    @property
+    def global_step(self) -> int:
+        return self.engine.global_step
+
+    @property
+    def last_lr(self) -> float:
+        return self.engine.last_lr
+
+    def scale_loss(self, loss: torch.Tensor) -> torch.Tensor:
+        """Reference 04:46: pre-scale loss by 1/num_workers because the raw
+        gradient path has no implicit cross-replica averaging; the apply-step
+        all-reduce then SUMs, so the update equals the global-batch mean."""
+        w = self.engine.world_size
+        if self.ddp_scale_loss and w > 1:
+            return loss * (1.0 / w)
+        return loss
+
+    def step(self, loss: torch.Tensor) -> bool:
+        """One micro-step. Returns True when an optimizer update was applied."""
+        self.scale_loss(loss).backward()
+        return self.engine.micro_step()
+
+    def state_dict(self):
+        return self.engine.state_dict()
+
+    def load_state_dict(self, d):
+        self.engine.load_state_dict(d)
+
+
+def create_optimizer(
+    model_or_params: Union[nn.Module, Iterable[Tuple[str, torch.Tensor]]],
+    init_lr: float,
+    num_train_steps: int,
+    num_warmup_steps: int = 0,
+    *,
+    gradient_accumulation_multiplier: int = 1,
+    clip_norm: Optional[float] = 1.0,
+    weight_decay: float = 0.01,
+    beta1: float = 0.9,
+    beta2: float = 0.999,
+    eps: float = 1e-6,
+    exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+    strict_reference_semantics: bool = False,
+    process_group=None,
+    ddp_scale_loss: bool = True,
+    backend: str = "auto",
+) -> TrainOp:
+    if isinstance(model_or_params, nn.Module):
+        named = list(model_or_params.named_parameters())
+    else:
+        named = list(model_or_params)
+    engine = AccumEngine(
+        named,
+        init_lr=init_lr,
+        num_train_steps=num_train_steps,
+        num_warmup_steps=num_warmup_steps,
+        gradient_accumulation_multiplier=gradient_accumulation_multiplier,
+        clip_norm=clip_norm,
+        weight_decay=weight_decay,
+        beta1=beta1,
+        beta2=beta2,
+        eps=eps,
+        exclude_from_weight_decay=exclude_from_weight_decay,
+        strict_reference_semantics=strict_reference_semantics,
+        process_group=process_group,
+        backend=backend,
+    )
+    return TrainOp(engine, ddp_scale_loss=ddp_scale_loss)

@@ -1,0 +1,202 @@
+"""Unit tests of the accumulation-engine contract against a NumPy oracle.
+
+Covers SURVEY.md section 2.2 items 1-6: buffer accumulation, the apply/
+accumulate predicate (strict + corrected), divide-by-K then clip-after-
+normalize ordering, AdamWeightDecay with NO bias correction and eps OUTSIDE
+the sqrt, decoupled weight decay with regex exclusion, per-micro-step LR
+schedule, and the strict-mode step-0 off-by-one.
+"""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+from gradient_accumulation_tf_estimator_amd import create_optimizer, learning_rate
+from gradient_accumulation_tf_estimator_amd.engine.flat import (
+    build_layout,
+    use_weight_decay,
+)
+
+torch.manual_seed(0)
+
+
+class TinyNet(nn.Module):
+    """Two weights, a bias and a LayerNorm -> exercises the decay split."""
+
+    def __init__(self):
+        super().__init__()
+        self.fc1 = nn.Linear(7, 5)
+        self.LayerNorm = nn.LayerNorm(5)
+        self.fc2 = nn.Linear(5, 3)
+
+    def forward(self, x):
+        return self.fc2(self.LayerNorm(torch.relu(self.fc1(x))))
+
+
+class NumpyAdamWOracle:
+    """Straight transcription of optimization.py:150-171 + 76-103 in NumPy."""
+
+    def __init__(self, shapes, decay_flags, K, init_lr, num_train_steps, num_warmup_steps,
+                 clip_norm=1.0, wd=0.01, b1=0.9, b2=0.999, eps=1e-6, strict=False):
+        self.p = None  # set later
+        self.m = [np.zeros(s, np.float64) for s in shapes]
+        self.v = [np.zeros(s, np.float64) for s in shapes]
+        self.accum = [np.zeros(s, np.float64) for s in shapes]
+        self.decay_flags = decay_flags
+        self.K, self.clip_norm, self.wd = K, clip_norm, wd
+        self.b1, self.b2, self.eps = b1, b2, eps
+        self.step = 0
+        self.init_lr, self.nts, self.nws = init_lr, num_train_steps, num_warmup_steps
+        self.strict = strict
+
+    def lr(self):
+        s = min(self.step, self.nts)
+        lr = self.init_lr * (1 - s / self.nts)
+        if self.nws and self.step < self.nws:
+            lr = self.init_lr * self.step / self.nws
+        return lr
+
+    def micro_step(self, grads):
+        for a, g in zip(self.accum, grads):
+            a += g
+        applied = (self.step % self.K == 0) if self.strict else ((self.step + 1) % self.K == 0)
+        if applied:
+            norm_g = [a / self.K for a in self.accum]
+            if self.clip_norm is not None:
+                gn = math.sqrt(sum(float((g * g).sum()) for g in norm_g))
+                coef = self.clip_norm / max(gn, self.clip_norm)
+                norm_g = [g * coef for g in norm_g]
+            lr = self.lr()
+            for i, g in enumerate(norm_g):
+                self.m[i] = self.b1 * self.m[i] + (1 - self.b1) * g
+                self.v[i] = self.b2 * self.v[i] + (1 - self.b2) * g * g
+                u = self.m[i] / (np.sqrt(self.v[i]) + self.eps)
+                if self.decay_flags[i]:
+                    u = u + self.wd * self.p[i]
+                self.p[i] = self.p[i] - lr * u
+            self.accum = [np.zeros_like(a) for a in self.accum]
+        self.step += 1
+        return applied
+
+
+def run_pair(strict, K=3, steps=10, clip_norm=1.0):
+    torch.manual_seed(42)
+    net = TinyNet().double()
+    names = [n for n, _ in net.named_parameters()]
+    shapes = [tuple(p.shape) for _, p in net.named_parameters()]
+    decay = [use_weight_decay(n, ("LayerNorm", "layer_norm", "bias")) for n in names]
+    oracle = NumpyAdamWOracle(shapes, decay, K, 1e-2, 100, 5, clip_norm=clip_norm, strict=strict)
+    oracle.p = [p.detach().numpy().copy() for _, p in net.named_parameters()]
+
+    net_f = TinyNet()
+    net_f.load_state_dict({k: v.float() for k, v in net.state_dict().items()})
+    op = create_optimizer(
+        net_f, 1e-2, 100, 5,
+        gradient_accumulation_multiplier=K,
+        clip_norm=clip_norm,
+        strict_reference_semantics=strict,
+    )
+
+    xs = [torch.randn(4, 7) for _ in range(steps)]
+    applied_engine, applied_oracle = [], []
+    for x in xs:
+        out = net_f(x)
+        loss = (out * out).mean()
+        applied_engine.append(op.step(loss))
+
+        out64 = net.forward(x.double())
+        loss64 = (out64 * out64).mean()
+        net.zero_grad()
+        loss64.backward()
+        grads = [p.grad.numpy() for _, p in net.named_parameters()]
+        # oracle params must track engine's so fwd uses updated weights:
+        applied_oracle.append(oracle.micro_step(grads))
+        # sync fp64 net from oracle params
+        with torch.no_grad():
+            for (n, p), arr in zip(net.named_parameters(), oracle.p):
+                p.copy_(torch.from_numpy(arr))
+    return net_f, oracle, applied_engine, applied_oracle
+
+
+@pytest.mark.parametrize("strict", [False, True])
+def test_engine_matches_numpy_oracle(strict):
+    net_f, oracle, ae, ao = run_pair(strict)
+    assert ae == ao
+    for (n, p), arr in zip(net_f.named_parameters(), oracle.p):
+        np.testing.assert_allclose(p.detach().numpy(), arr, rtol=2e-4, atol=2e-5)
+
+
+def test_strict_step0_applies_immediately():
+    _, _, ae, _ = run_pair(strict=True, K=4, steps=9)
+    # strict reference: apply at micro-steps 0, 4, 8
+    assert ae == [True, False, False, False, True, False, False, False, True]
+
+
+def test_corrected_applies_after_full_window():
+    _, _, ae, _ = run_pair(strict=False, K=4, steps=9)
+    assert ae == [False, False, False, True, False, False, False, True, False]
+
+
+def test_no_clip_variant():
+    net_f, oracle, _, _ = run_pair(strict=False, K=2, steps=6, clip_norm=None)
+    for (n, p), arr in zip(net_f.named_parameters(), oracle.p):
+        np.testing.assert_allclose(p.detach().numpy(), arr, rtol=2e-4, atol=2e-5)
+
+
+def test_lr_schedule_reference_values():
+    # polynomial power=1 decay to 0 over 100 steps, 10 warmup (optimization.py:29-54)
+    assert learning_rate(0, 1.0, 100, 10) == 0.0
+    assert learning_rate(5, 1.0, 100, 10) == pytest.approx(0.5)
+    assert learning_rate(10, 1.0, 100, 10) == pytest.approx(0.9)
+    assert learning_rate(50, 1.0, 100, 10) == pytest.approx(0.5)
+    assert learning_rate(100, 1.0, 100, 10) == pytest.approx(0.0)
+    assert learning_rate(150, 1.0, 100, 10) == pytest.approx(0.0)
+    # no warmup
+    assert learning_rate(0, 2e-5, 200, 0) == pytest.approx(2e-5)
+
+
+def test_weight_decay_regex_exclusion():
+    excl = ("LayerNorm", "layer_norm", "bias")
+    assert use_weight_decay("fc1.weight", excl)
+    assert not use_weight_decay("fc1.bias", excl)
+    assert not use_weight_decay("encoder.LayerNorm.weight", excl)
+    assert not use_weight_decay("x.layer_norm.weight", excl)
+
+
+def test_layout_decay_first_and_aligned():
+    t = torch.zeros(10)
+    params = [("a.weight", torch.zeros(70)), ("a.bias", torch.zeros(10)),
+              ("b.weight", torch.zeros(3, 5))]
+    lay = build_layout(params)
+    assert [s.name for s in lay.slices] == ["a.weight", "b.weight", "a.bias"]
+    assert all(s.offset % 64 == 0 for s in lay.slices)
+    assert lay.decay_boundary == 128 + 64  # 70->128, 15->64
+    assert lay.total == 128 + 64 + 64
+
+
+def test_accum_buffer_checkpointed_mid_window():
+    torch.manual_seed(1)
+    net = TinyNet()
+    op = create_optimizer(net, 1e-2, 100, 0, gradient_accumulation_multiplier=4)
+    xs = [torch.randn(4, 7) for _ in range(6)]
+    for x in xs[:2]:
+        op.step((net(x) ** 2).mean())
+    sd = {k: (v.clone() if torch.is_tensor(v) else v) for k, v in op.state_dict().items()}
+
+    # continue original
+    for x in xs[2:]:
+        op.step((net(x) ** 2).mean())
+    ref = net.fc1.weight.detach().clone()
+
+    # rebuild + resume mid-accumulation-window -> bit-exact continuation
+    torch.manual_seed(1)
+    net2 = TinyNet()
+    op2 = create_optimizer(net2, 1e-2, 100, 0, gradient_accumulation_multiplier=4)
+    op2.load_state_dict(sd)
+    assert op2.global_step == 2
+    for x in xs[2:]:
+        op2.step((net2(x) ** 2).mean())
+    assert torch.equal(net2.fc1.weight.detach(), ref)
