@@ -1,0 +1,165 @@
+// MI355X (gfx950 / CDNA4) kernels for the gradient-accumulation engine.
+//
+// Replaces the implicit native op surface of the TF1 reference
+// (SURVEY.md section 2.3): per-variable AssignAdd / RealDiv /
+// clip_by_global_norm / Adam elementwise chains become three grid-stride
+// kernels over ONE flat, 256-B-aligned buffer set:
+//
+//   k_accum_*   : accum += grad (fp32 upcast); grad = 0      [every micro-step]
+//   k_sqnorm    : sum(accum^2) -> device scalar (wave64 shuffle reduce + LDS
+//                 tree + one atomic per block)               [apply step]
+//   k_apply_*   : g = accum/K * clip_coef; AdamWeightDecay (no bias
+//                 correction, eps outside sqrt, decoupled wd below the decay
+//                 boundary); p -= lr*u; optional bf16 write-back; accum = 0
+//                                                            [apply step]
+//
+// All kernels are HBM-bandwidth-bound: 16 B/lane vectorized access
+// (float4 / ushort4-as-bf16x4), grid-stride with a ~2048-workgroup cap
+// (256 CUs x 8 XCDs need >>256 workgroups; cdna_hip_programming.md G11).
+// lr and the squared norm travel through device scalars so the apply step
+// is hipGraph-capturable with a schedule-updated lr (no recapture).
+//
+// Every flat length is a multiple of 64 elements (engine/flat.py ALIGN), so
+// the float4 loops have no scalar tail and the decay boundary is float4-
+// uniform.
+
+#include <hip/hip_runtime.h>
+
+#define GA_THREADS 256
+#define GA_MAX_BLOCKS 2048
+
+static inline __device__ float bf16_to_f32(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = (unsigned int)u << 16;
+  return c.f;
+}
+
+// round-to-nearest-even f32 -> bf16, matching PyTorch's cast
+static inline __device__ unsigned short f32_to_bf16(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  unsigned int x = c.i;
+  if ((x & 0x7fffffffu) > 0x7f800000u) return (unsigned short)((x >> 16) | 0x0040u); // NaN
+  unsigned int round_bias = ((x >> 16) & 1u) + 0x7fffu;
+  return (unsigned short)((x + round_bias) >> 16);
+}
+
+extern "C" __global__ void k_accum_f32(float4* __restrict__ accum,
+                                       float4* __restrict__ grad,
+                                       long long n4) {
+  long long stride = (long long)gridDim.x * blockDim.x;
+  const float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = accum[i];
+    float4 g = grad[i];
+    a.x += g.x; a.y += g.y; a.z += g.z; a.w += g.w;
+    accum[i] = a;
+    grad[i] = z;
+  }
+}
+
+extern "C" __global__ void k_accum_bf16(float4* __restrict__ accum,
+                                        ushort4* __restrict__ grad,
+                                        long long n4) {
+  long long stride = (long long)gridDim.x * blockDim.x;
+  const ushort4 z = make_ushort4(0, 0, 0, 0);
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = accum[i];
+    ushort4 g = grad[i];
+    a.x += bf16_to_f32(g.x); a.y += bf16_to_f32(g.y);
+    a.z += bf16_to_f32(g.z); a.w += bf16_to_f32(g.w);
+    accum[i] = a;
+    grad[i] = z;
+  }
+}
+
+// Global squared norm of the flat accum buffer into out[0].
+// out must be zeroed before launch (engine does hipMemsetAsync-equivalent).
+extern "C" __global__ void k_sqnorm(const float4* __restrict__ accum,
+                                    long long n4,
+                                    float* __restrict__ out) {
+  long long stride = (long long)gridDim.x * blockDim.x;
+  float s = 0.f;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = accum[i];
+    s = fmaf(a.x, a.x, s); s = fmaf(a.y, a.y, s);
+    s = fmaf(a.z, a.z, s); s = fmaf(a.w, a.w, s);
+  }
+  // wave64 shuffle reduction
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  __shared__ float ws[GA_THREADS / 64];
+  int lane = threadIdx.x & 63;
+  int wid = threadIdx.x >> 6;
+  if (lane == 0) ws[wid] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+    for (int w = 0; w < GA_THREADS / 64; ++w) t += ws[w];
+    atomicAdd(out, t);  // device-scope, one per block
+  }
+}
+
+// Fused normalize + clip + AdamWeightDecay + (optional bf16 write-back) + zero.
+// Math contract: /root/reference/optimization.py:80-88 (apply branch order)
+// and :150-171 (AdamWeightDecay update), cited in SURVEY.md section 2.2.
+template <bool HAS_MODEL>
+static __device__ void apply_body(float4* __restrict__ accum,
+                                  float4* __restrict__ m,
+                                  float4* __restrict__ v,
+                                  float4* __restrict__ p,
+                                  ushort4* __restrict__ model,
+                                  const float* __restrict__ lr_p,
+                                  const float* __restrict__ sq_p,
+                                  long long n4, long long boundary4,
+                                  float inv_k, float clip, float wd,
+                                  float b1, float b2, float eps) {
+  const float lr = lr_p[0];
+  float coef = 1.f;
+  if (clip > 0.f) {
+    float norm = sqrtf(sq_p[0]) * inv_k;  // norm of accum/K (clip AFTER normalize)
+    coef = clip / fmaxf(norm, clip);      // tf.clip_by_global_norm scale
+  }
+  const float s = inv_k * coef;
+  const float omb1 = 1.f - b1, omb2 = 1.f - b2;
+  const float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+    float4 a = accum[i], mm = m[i], vv = v[i], pp = p[i];
+    const float dw = (i < boundary4) ? wd : 0.f;
+#define GA_C(c)                                        \
+    {                                                  \
+      float g = a.c * s;                               \
+      mm.c = fmaf(b1, mm.c, omb1 * g);                 \
+      vv.c = fmaf(b2, vv.c, omb2 * g * g);             \
+      float u = mm.c / (sqrtf(vv.c) + eps);            \
+      u = fmaf(dw, pp.c, u);                           \
+      pp.c = fmaf(-lr, u, pp.c);                       \
+    }
+    GA_C(x) GA_C(y) GA_C(z) GA_C(w)
+#undef GA_C
+    m[i] = mm; v[i] = vv; p[i] = pp; accum[i] = z;
+    if (HAS_MODEL) {
+      model[i] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
+                              f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+    }
+  }
+}
+
+extern "C" __global__ void k_apply_f32(float4* accum, float4* m, float4* v, float4* p,
+                                       const float* lr_p, const float* sq_p,
+                                       long long n4, long long boundary4,
+                                       float inv_k, float clip, float wd,
+                                       float b1, float b2, float eps) {
+  apply_body<false>(accum, m, v, p, nullptr, lr_p, sq_p, n4, boundary4,
+                    inv_k, clip, wd, b1, b2, eps);
+}
+
+extern "C" __global__ void k_apply_bf16(float4* accum, float4* m, float4* v, float4* p,
+                                        ushort4* model,
+                                        const float* lr_p, const float* sq_p,
+                                        long long n4, long long boundary4,
+                                        float inv_k, float clip, float wd,
+                                        float b1, float b2, float eps) {
+  apply_body<true>(accum, m, v, p, model, lr_p, sq_p, n4, boundary4,
+                   inv_k, clip, wd, b1, b2, eps);
+}

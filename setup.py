@@ -1,0 +1,37 @@
+"""Build the in-tree gfx950 HIP extension.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The .so lands at gradient_accumulation_tf_estimator_amd/ops/_ga_hip*.so so it
+travels with repo snapshots to GPU boxes (no JIT cache involved).
+"""
+
+import os
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from setuptools import setup  # noqa: E402
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+CSRC = os.path.join(HERE, "gradient_accumulation_tf_estimator_amd", "ops", "csrc")
+
+setup(
+    name="gradient_accumulation_tf_estimator_amd",
+    version="0.1.0",
+    packages=["gradient_accumulation_tf_estimator_amd"],
+    ext_modules=[
+        CUDAExtension(
+            name="gradient_accumulation_tf_estimator_amd.ops._ga_hip",
+            sources=[
+                os.path.join(CSRC, "ga_kernels.hip"),
+                os.path.join(CSRC, "ga_bindings.hip"),
+            ],
+            extra_compile_args={
+                "cxx": ["-O3"],
+                "nvcc": ["-O3", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension},
+)
