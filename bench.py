@@ -1,0 +1,245 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: BERT-Small seq128 effective-batch-32 gradient-accumulation
+training step on N MI355X GPUs (BASELINE.json metric: samples/sec/node).
+
+Launch (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+    # N>1 is launched via torch.distributed.run, one rank per GPU over RCCL.
+
+A "step" is one reference micro-step (fwd + bwd + accumulate; every
+``accum``-th step also runs RCCL all-reduce + fused clip/AdamW apply), i.e.
+one ``session.run(train_op)`` of the reference, so
+``samples/sec = micro_batch * steps * world / elapsed``. Synthetic token data,
+random-init weights, bf16 compute with fp32 master/accum (the reference's
+fp32 optimizer math).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=200, help="timed micro-steps")
+    p.add_argument("--warmup", type=int, default=40, help="untimed warmup micro-steps")
+    p.add_argument("--model", default="bert-small",
+                   choices=["bert-small", "bert-base", "bert-large"])
+    p.add_argument("--micro-batch", type=int, default=8)
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--accum", type=int, default=4)
+    p.add_argument("--lr", type=float, default=2e-5)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
+                   help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
+    p.add_argument("--allreduce-bucket-mb", type=int, default=64)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+
+    if world > 1:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group("nccl" if use_cuda else "gloo",
+                                rank=rank, world_size=world)
+    else:
+        dist = None
+
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+    from gradient_accumulation_tf_estimator_amd.models.bert import CONFIGS, BertForSequenceClassification
+
+    torch.manual_seed(1234 + rank)
+    cfg = CONFIGS[args.model]()
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and use_cuda) else torch.float32
+    model = BertForSequenceClassification(cfg).to(device=device, dtype=dtype)
+    model.train()
+
+    op = create_optimizer(
+        model, args.lr, num_train_steps=max(args.steps + args.warmup + 64, 1000),
+        num_warmup_steps=100,
+        gradient_accumulation_multiplier=args.accum,
+        clip_norm=1.0,
+        backend="hip" if use_cuda else "eager",
+    )
+    engine = op.engine
+
+    B, S, V = args.micro_batch, args.seq_len, cfg.vocab_size
+    POOL = 8
+    gen = torch.Generator(device="cpu").manual_seed(99 + rank)
+    pool_ids = torch.randint(0, V, (POOL, B, S), generator=gen).to(device)
+    pool_lab = torch.randint(0, cfg.num_labels, (POOL, B), generator=gen).to(device)
+
+    inv_world = 1.0 / world
+
+    def eager_micro_step(i):
+        ids, lab = pool_ids[i % POOL], pool_lab[i % POOL]
+        loss = model.loss(ids, lab)
+        if world > 1:
+            loss = loss * inv_world
+        loss.backward()
+        engine.micro_step()
+        return loss
+
+    use_graphs = use_cuda and (args.graphs == "on" or (args.graphs == "auto" and world == 1))
+    graphed = None
+    if use_graphs:
+        try:
+            graphed = capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world)
+        except Exception as e:
+            print(f"[bench] hipGraph capture failed, falling back to eager: {e}",
+                  file=sys.stderr)
+            graphed = None
+
+    def step(i):
+        if graphed is not None:
+            return graphed(i)
+        return eager_micro_step(i)
+
+    # ---- warmup ----
+    for i in range(args.warmup):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+        torch.cuda.reset_peak_memory_stats()
+    if dist:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+
+    # ---- timed region: exactly args.steps micro-steps ----
+    t0 = time.perf_counter()
+    for i in range(args.warmup, args.warmup + args.steps):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    samples = args.micro_batch * args.steps * world
+    value = samples / elapsed
+    peak_hbm_gb = (torch.cuda.max_memory_allocated() / 2**30) if use_cuda else 0.0
+
+    if rank == 0:
+        out = {
+            "metric": "samples_per_sec_per_node",
+            "value": round(value, 2),
+            "unit": "samples/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.micro_batch * args.accum * world,
+                "micro_batch": args.micro_batch,
+                "accum": args.accum,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{world}",
+                "graphs": graphed is not None,
+                "peak_hbm_gb": round(peak_hbm_gb, 3),
+            },
+        }
+        print(json.dumps(out))
+
+    if dist:
+        dist.destroy_process_group()
+
+
+def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
+    """hipGraph-capture the accumulate micro-step and the apply micro-step.
+
+    The tf.cond(apply, accumulate) of the reference (optimization.py:91-94)
+    becomes a host-side choice between two captured graphs.
+    lr reaches the apply kernel through a device scalar so the schedule
+    updates without re-capture.
+    """
+    if world > 1:
+        raise RuntimeError("graph capture currently enabled for 1 GPU only")
+    assert engine.backend == "hip"
+    static_ids = pool_ids[0].clone()
+    static_lab = pool_lab[0].clone()
+
+    def fwd_bwd_accum():
+        loss = model.loss(static_ids, static_lab)
+        loss.backward()
+        engine.accumulate()
+        return loss
+
+    # warmup on a side stream (torch.cuda.graphs requirement)
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(3):
+            fwd_bwd_accum()
+        engine.set_lr(engine.lr_at(0))
+        engine.apply_from_device()
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    # reset optimizer progress perturbed by warmup
+    engine.state.accum.zero_()
+
+    g_accum = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g_accum):
+        loss_accum = fwd_bwd_accum()
+    g_apply = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g_apply, pool=g_accum.pool()):
+        loss_apply = fwd_bwd_accum()
+        engine.apply_from_device()
+    torch.cuda.synchronize()
+
+    K = engine.K
+    POOL = pool_ids.shape[0]
+
+    def run(i):
+        static_ids.copy_(pool_ids[i % POOL])
+        static_lab.copy_(pool_lab[i % POOL])
+        if engine.is_apply_step():
+            engine.set_lr(engine.lr_at(engine.global_step))
+            g_apply.replay()
+            engine.global_step += 1
+            return loss_apply
+        g_accum.replay()
+        engine.global_step += 1
+        return loss_accum
+
+    return run
+
+
+if __name__ == "__main__":
+    main()

@@ -139,6 +139,37 @@ class AccumEngine:
         for h in handles:
             h.wait()
 
+    def set_lr(self, lr: float) -> None:
+        """Write the schedule's lr into the device scalar the apply kernel
+        reads -- the hipGraph-capture-safe path for a changing lr."""
+        self.last_lr = lr
+        self._lr_dev.fill_(lr)
+
+    def apply_from_device(self) -> None:
+        """Launch the apply kernels reading lr from the device scalar; no
+        host-side lr computation. Used inside hipGraph capture (bench.py)."""
+        if self._hip is None:
+            raise RuntimeError("apply_from_device requires the HIP backend")
+        st = self.state
+        model = None if st.master is st.model else st.model
+        self._hip.fused_apply(
+            st.accum,
+            st.m,
+            st.v,
+            st.master,
+            model if model is not None else st.master,
+            model is not None,
+            self._lr_dev,
+            self._sqnorm_dev,
+            st.decay_boundary,
+            1.0 / self.K,
+            -1.0 if self.clip_norm is None else float(self.clip_norm),
+            self.weight_decay,
+            self.beta1,
+            self.beta2,
+            self.eps,
+        )
+
     def apply(self, lr: Optional[float] = None) -> None:
         st = self.state
         if lr is None:

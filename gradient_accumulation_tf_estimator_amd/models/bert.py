@@ -1,0 +1,177 @@
+"""BERT family (Small/Base/Large) in PyTorch-ROCm, bf16-first.
+
+The reference fine-tunes Google BERT-Small with its patched optimizer
+(/root/reference/README.md:12-17,72); the model itself lives in the external
+google-research/bert repo. Here the architecture is implemented natively:
+post-LN transformer encoder with GELU FFN, learned position + token-type
+embeddings, pooler, and a CoLA-style sequence-classification head.
+
+Parameter naming deliberately uses ``LayerNorm`` / ``bias`` substrings so the
+engine's weight-decay regex exclusion (optimization.py:65,179-187) applies to
+the same parameter classes as the reference.
+
+Attention runs through ``torch.nn.functional.scaled_dot_product_attention``
+(MIOpen/CK flash path on ROCm); Linear layers hit hipBLASLt/rocBLAS GEMMs.
+Hand-written HIP kernels are reserved for the accumulation engine per the
+north star.
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class BertConfig:
+    vocab_size: int = 30522
+    hidden_size: int = 512
+    num_layers: int = 4
+    num_heads: int = 8
+    intermediate_size: int = 2048
+    max_position_embeddings: int = 512
+    type_vocab_size: int = 2
+    layer_norm_eps: float = 1e-12
+    dropout: float = 0.0  # 0 keeps the step hipGraph-capturable & deterministic
+    num_labels: int = 2
+    initializer_range: float = 0.02
+
+
+# BASELINE.json model configs
+def bert_small() -> BertConfig:
+    return BertConfig(hidden_size=512, num_layers=4, num_heads=8, intermediate_size=2048)
+
+
+def bert_base() -> BertConfig:
+    return BertConfig(hidden_size=768, num_layers=12, num_heads=12, intermediate_size=3072)
+
+
+def bert_large() -> BertConfig:
+    return BertConfig(hidden_size=1024, num_layers=24, num_heads=16, intermediate_size=4096)
+
+
+CONFIGS = {"bert-small": bert_small, "bert-base": bert_base, "bert-large": bert_large}
+
+
+class BertEmbeddings(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.position_embeddings = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
+        self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
+        self.LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.dropout = nn.Dropout(cfg.dropout)
+        self.register_buffer(
+            "position_ids", torch.arange(cfg.max_position_embeddings).unsqueeze(0), persistent=False
+        )
+
+    def forward(self, input_ids, token_type_ids=None):
+        S = input_ids.shape[1]
+        x = self.word_embeddings(input_ids)
+        x = x + self.position_embeddings(self.position_ids[:, :S])
+        if token_type_ids is not None:
+            x = x + self.token_type_embeddings(token_type_ids)
+        return self.dropout(self.LayerNorm(x))
+
+
+class BertSelfAttention(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.num_heads = cfg.num_heads
+        self.head_dim = cfg.hidden_size // cfg.num_heads
+        self.query = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.key = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.value = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.dropout_p = cfg.dropout
+
+    def forward(self, x, attn_mask=None):
+        B, S, H = x.shape
+        q = self.query(x).view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        k = self.key(x).view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        v = self.value(x).view(B, S, self.num_heads, self.head_dim).transpose(1, 2)
+        o = F.scaled_dot_product_attention(
+            q, k, v, attn_mask=attn_mask,
+            dropout_p=self.dropout_p if self.training else 0.0,
+        )
+        return o.transpose(1, 2).reshape(B, S, H)
+
+
+class BertLayer(nn.Module):
+    """Post-LN encoder layer (original BERT ordering)."""
+
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.attention = BertSelfAttention(cfg)
+        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.attention_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
+        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
+        self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.dropout = nn.Dropout(cfg.dropout)
+
+    def forward(self, x, attn_mask=None):
+        a = self.attention(x, attn_mask)
+        x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
+        h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))
+        return self.output_LayerNorm(x + self.dropout(h))
+
+
+class BertModel(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embeddings = BertEmbeddings(cfg)
+        self.encoder = nn.ModuleList(BertLayer(cfg) for _ in range(cfg.num_layers))
+        self.pooler = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.apply(self._init)
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=self.cfg.initializer_range)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
+        elif isinstance(m, nn.LayerNorm):
+            nn.init.ones_(m.weight)
+            nn.init.zeros_(m.bias)
+
+    def forward(self, input_ids, token_type_ids=None, attention_mask=None):
+        # attention_mask: [B, S] of 1/0 -> additive float mask for SDPA
+        mask = None
+        if attention_mask is not None:
+            mask = attention_mask[:, None, None, :].to(torch.bool)
+        x = self.embeddings(input_ids, token_type_ids)
+        for layer in self.encoder:
+            x = layer(x, mask)
+        pooled = torch.tanh(self.pooler(x[:, 0]))
+        return x, pooled
+
+
+class BertForSequenceClassification(nn.Module):
+    """CoLA-style head: pooled CLS -> num_labels logits, CE loss
+    (the reference's run_classifier task, README.md:72)."""
+
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.bert = BertModel(cfg)
+        self.classifier = nn.Linear(cfg.hidden_size, cfg.num_labels)
+        nn.init.normal_(self.classifier.weight, std=cfg.initializer_range)
+        nn.init.zeros_(self.classifier.bias)
+
+    def forward(self, input_ids, token_type_ids=None, attention_mask=None):
+        _, pooled = self.bert(input_ids, token_type_ids, attention_mask)
+        return self.classifier(pooled)
+
+    def loss(self, input_ids, labels, token_type_ids=None, attention_mask=None):
+        logits = self.forward(input_ids, token_type_ids, attention_mask)
+        return F.cross_entropy(logits.float(), labels)
+
+
+def build_model(name: str, **overrides) -> BertForSequenceClassification:
+    cfg = CONFIGS[name]()
+    for k, v in overrides.items():
+        setattr(cfg, k, v)
+    return BertForSequenceClassification(cfg)
