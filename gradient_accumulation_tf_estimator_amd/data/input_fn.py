@@ -1,0 +1,102 @@
+"""The reference's input_fn pipeline contract, PyTorch-side.
+
+Reference (01:6-18, identical in 02/03/04): shard (if input_context) ->
+shuffle(2*batch+1) -> batch -> repeat(num_epochs), shard BEFORE shuffle,
+no drop_remainder. ``InputContext`` mirrors tf.distribute.InputContext
+(03:101,111): num_input_pipelines / input_pipeline_id drive Dataset.shard.
+"""
+
+from __future__ import annotations
+
+import random
+from dataclasses import dataclass
+from typing import Any, Callable, Iterator, List, Optional, Sequence, Tuple
+
+import torch
+
+
+@dataclass
+class InputContext:
+    num_input_pipelines: int = 1
+    input_pipeline_id: int = 0
+
+
+class ArrayDataset:
+    """In-memory (features, labels) dataset with the reference pipeline ops.
+
+    features may be a tensor or a dict of tensors (housing feature columns).
+    """
+
+    def __init__(self, features, labels):
+        self.features = features
+        self.labels = labels
+        n = labels.shape[0]
+        self._indices = list(range(n))
+
+    def __len__(self):
+        return len(self._indices)
+
+    def _subset(self, idx: Sequence[int]) -> "ArrayDataset":
+        t = torch.as_tensor(list(idx), dtype=torch.long)
+        if isinstance(self.features, dict):
+            f = {k: v[t] for k, v in self.features.items()}
+        else:
+            f = self.features[t]
+        ds = ArrayDataset(f, self.labels[t])
+        return ds
+
+    def shard(self, num_shards: int, index: int) -> "ArrayDataset":
+        # tf.data.Dataset.shard: element i kept if i % num_shards == index
+        return self._subset([i for i in self._indices if i % num_shards == index])
+
+
+def input_fn_iterator(
+    dataset: ArrayDataset,
+    batch_size: int,
+    num_epochs: Optional[int] = None,
+    shuffle: bool = True,
+    shuffle_buffer: Optional[int] = None,
+    seed: Optional[int] = None,
+    input_context: Optional[InputContext] = None,
+) -> Iterator[Tuple[Any, torch.Tensor]]:
+    """shard -> shuffle(2*batch+1) -> batch -> repeat, like the reference.
+
+    A buffered shuffle (size 2*batch+1 by default, as in 01:16) rather than a
+    full permutation, to preserve the reference's sampling behavior.
+    """
+    if input_context and input_context.num_input_pipelines > 1:
+        dataset = dataset.shard(
+            input_context.num_input_pipelines, input_context.input_pipeline_id
+        )
+    n = len(dataset)
+    if n == 0:
+        return
+    buf_size = shuffle_buffer or (2 * batch_size + 1)
+    rng = random.Random(seed)
+
+    def epoch_indices():
+        if not shuffle:
+            yield from range(n)
+            return
+        buf: List[int] = []
+        for i in range(n):
+            buf.append(i)
+            if len(buf) >= buf_size:
+                j = rng.randrange(len(buf))
+                yield buf.pop(j)
+        while buf:
+            yield buf.pop(rng.randrange(len(buf)))
+
+    epoch = 0
+    while num_epochs is None or epoch < num_epochs:
+        batch: List[int] = []
+        for i in epoch_indices():
+            batch.append(i)
+            if len(batch) == batch_size:
+                ds = dataset._subset(batch)
+                yield ds.features, ds.labels
+                batch = []
+        if batch:  # no drop_remainder in the reference
+            ds = dataset._subset(batch)
+            yield ds.features, ds.labels
+        epoch += 1

@@ -1,0 +1,319 @@
+"""Estimator-style runner: the reference's L5 surface in eager PyTorch.
+
+Mirrors the tf.estimator API the reference configures
+(another-example.py:186-190,299-342; 01:83-111): ``Estimator(model_fn,
+config, params)``, ``ModeKeys``/``EstimatorSpec``, ``RunConfig``,
+``TrainSpec``/``EvalSpec``/``train_and_evaluate``.
+
+model_fn contract (L4): called ONCE per mode with a
+representative ``(features, labels)`` batch to build the module, and returns
+an ``EstimatorSpec`` whose callables run subsequent batches:
+
+    def model_fn(features, labels, mode, params):
+        model = MnistCNN().to(params.get("device", "cpu"))
+        if mode == ModeKeys.PREDICT:
+            return EstimatorSpec(mode, model=model,
+                                 predictions_fn=lambda f: model(f).argmax(-1))
+        loss_fn = lambda f, l: model.loss(f, l)
+        if mode == ModeKeys.EVAL:
+            return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
+                                 eval_metric_fns={"accuracy": ...})
+        train_op = create_optimizer(model, ..., gradient_accumulation_multiplier=K)
+        return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
+
+Like tf.estimator, evaluate()/predict() restore weights from the latest
+model_dir checkpoint rather than sharing live objects with train().
+"""
+
+from __future__ import annotations
+
+import inspect
+import os
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, Iterable, Iterator, Optional, Tuple
+
+import torch
+
+from ..engine.optimizer import TrainOp
+from ..utils import checkpoint as ckpt
+from ..utils.logging import StepLogger
+
+
+class ModeKeys:
+    TRAIN = "train"
+    EVAL = "eval"
+    PREDICT = "infer"
+
+
+@dataclass
+class EstimatorSpec:
+    mode: str
+    model: Optional[torch.nn.Module] = None
+    loss_fn: Optional[Callable] = None
+    train_op: Optional[TrainOp] = None
+    predictions_fn: Optional[Callable] = None
+    eval_metric_fns: Dict[str, Callable] = field(default_factory=dict)
+
+    def __post_init__(self):
+        if self.mode == ModeKeys.TRAIN and (self.loss_fn is None or self.train_op is None):
+            raise ValueError("TRAIN spec requires loss_fn and train_op")
+        if self.mode == ModeKeys.EVAL and self.loss_fn is None:
+            raise ValueError("EVAL spec requires loss_fn")
+        if self.mode == ModeKeys.PREDICT and self.predictions_fn is None:
+            raise ValueError("PREDICT spec requires predictions_fn")
+
+
+@dataclass
+class RunConfig:
+    model_dir: Optional[str] = None
+    save_checkpoints_steps: Optional[int] = None
+    keep_checkpoint_max: int = 5
+    log_step_count_steps: int = 100
+    tf_random_seed: Optional[int] = None  # reference name (01:77); seeds torch
+    device: Optional[str] = None
+    train_distribute: Optional[Any] = None  # process group / True for default
+
+
+@dataclass
+class TrainSpec:
+    input_fn: Callable
+    max_steps: Optional[int] = None
+
+
+@dataclass
+class EvalSpec:
+    input_fn: Callable
+    steps: Optional[int] = None
+    throttle_secs: float = 30.0
+
+
+def _call_input_fn(input_fn, mode):
+    sig = inspect.signature(input_fn)
+    kwargs = {}
+    if "mode" in sig.parameters:
+        kwargs["mode"] = mode
+    return input_fn(**kwargs)
+
+
+def _to_device(x, device):
+    if device is None:
+        return x
+    if torch.is_tensor(x):
+        return x.to(device)
+    if isinstance(x, dict):
+        return {k: _to_device(v, device) for k, v in x.items()}
+    return x
+
+
+class Estimator:
+    def __init__(self, model_fn: Callable, config: Optional[RunConfig] = None,
+                 params: Optional[Dict] = None):
+        self.model_fn = model_fn
+        self.config = config or RunConfig()
+        self.params = dict(params or {})
+        if self.config.device and "device" not in self.params:
+            self.params["device"] = self.config.device
+        self._train_spec: Optional[EstimatorSpec] = None
+        self._train_iter = None
+        self._train_iter_key = None
+        self._logger = StepLogger(self.config.model_dir)
+        if self.config.tf_random_seed is not None:
+            torch.manual_seed(self.config.tf_random_seed)
+
+    # ---- internal ----
+    def _build_spec(self, mode, features, labels) -> EstimatorSpec:
+        sig = inspect.signature(self.model_fn)
+        kwargs = {}
+        if "config" in sig.parameters:
+            kwargs["config"] = self.config
+        spec = self.model_fn(features, labels, mode, self.params, **kwargs)
+        if not isinstance(spec, EstimatorSpec):
+            raise TypeError("model_fn must return an EstimatorSpec")
+        return spec
+
+    def _restore(self, spec: EstimatorSpec, path: Optional[str] = None,
+                 with_engine: bool = False) -> Optional[int]:
+        path = path or (ckpt.latest(self.config.model_dir) if self.config.model_dir else None)
+        if path is None:
+            return None
+        data = ckpt.load(path)
+        if spec.model is not None:
+            spec.model.load_state_dict(data["model"])
+            # model params may be flat-buffer views owned by a train_op's
+            # engine; load_state_dict copies in-place so views stay intact.
+        if with_engine and spec.train_op is not None and "engine" in data:
+            eng = {
+                k: (v.to(self.device_of(spec)) if torch.is_tensor(v) else v)
+                for k, v in data["engine"].items()
+            }
+            spec.train_op.load_state_dict(eng)
+        return data["step"]
+
+    @staticmethod
+    def device_of(spec: EstimatorSpec):
+        if spec.model is not None:
+            try:
+                return next(spec.model.parameters()).device
+            except StopIteration:
+                pass
+        return torch.device("cpu")
+
+    def _save(self, spec: EstimatorSpec, step: int) -> None:
+        if not self.config.model_dir:
+            return
+        ckpt.save(
+            self.config.model_dir, step,
+            spec.model.state_dict() if spec.model is not None else {},
+            spec.train_op.state_dict() if spec.train_op is not None else {},
+            keep_max=self.config.keep_checkpoint_max,
+        )
+
+    # ---- public API ----
+    def train(self, input_fn, max_steps: Optional[int] = None,
+              steps: Optional[int] = None) -> Dict:
+        device = self.config.device
+        # keep one live iterator per input_fn so chunked train() calls
+        # (train_and_evaluate) continue the stream instead of restarting it
+        if self._train_iter is None or self._train_iter_key is not input_fn:
+            self._train_iter = iter(_call_input_fn(input_fn, ModeKeys.TRAIN))
+            self._train_iter_key = input_fn
+        it = self._train_iter
+        try:
+            first = next(it)
+        except StopIteration:
+            sp = self._train_spec
+            return {"global_step": sp.train_op.global_step if sp else 0, "loss": None}
+        features, labels = _to_device(first[0], device), _to_device(first[1], device)
+
+        if self._train_spec is None:
+            self._train_spec = self._build_spec(ModeKeys.TRAIN, features, labels)
+            self._restore(self._train_spec, with_engine=True)
+        spec = self._train_spec
+        op = spec.train_op
+
+        cfg = self.config
+        done_this_call = 0
+        t_last, s_last = time.perf_counter(), op.global_step
+        pending = (features, labels)
+        last_loss = None
+        while True:
+            step = op.global_step
+            if max_steps is not None and step >= max_steps:
+                break
+            if steps is not None and done_this_call >= steps:
+                break
+            if pending is None:
+                try:
+                    f, l = next(it)
+                except StopIteration:
+                    break
+                features, labels = _to_device(f, device), _to_device(l, device)
+            else:
+                features, labels = pending
+                pending = None
+            loss = spec.loss_fn(features, labels)
+            op.step(loss)
+            done_this_call += 1
+            last_loss = loss
+            step = op.global_step
+            if cfg.log_step_count_steps and step % cfg.log_step_count_steps == 0:
+                now = time.perf_counter()
+                rate = (step - s_last) / max(now - t_last, 1e-9)
+                self._logger.log(step=step, loss=float(loss.detach().float()),
+                                 lr=op.last_lr, steps_per_sec=round(rate, 3))
+                t_last, s_last = now, step
+            if cfg.save_checkpoints_steps and step % cfg.save_checkpoints_steps == 0:
+                self._save(spec, step)
+        self._save(spec, op.global_step)
+        return {
+            "global_step": op.global_step,
+            "loss": float(last_loss.detach().float()) if last_loss is not None else None,
+        }
+
+    def evaluate(self, input_fn, steps: Optional[int] = None,
+                 checkpoint_path: Optional[str] = None) -> Dict:
+        device = self.config.device
+        it = iter(_call_input_fn(input_fn, ModeKeys.EVAL))
+        try:
+            first = next(it)
+        except StopIteration:
+            return {}
+        features, labels = _to_device(first[0], device), _to_device(first[1], device)
+        spec = self._build_spec(ModeKeys.EVAL, features, labels)
+        restored_step = self._restore(spec, checkpoint_path)
+        if spec.model is not None:
+            spec.model.eval()
+
+        from ..utils.metrics import Mean
+
+        loss_m = Mean()
+        metric_means = {k: Mean() for k in spec.eval_metric_fns}
+        n_batches = 0
+
+        def run_batch(f, l):
+            nonlocal n_batches
+            with torch.no_grad():
+                loss = spec.loss_fn(f, l)
+                n = l.shape[0] if torch.is_tensor(l) else 1
+                loss_m.update(float(loss.detach().float()), n)
+                for k, fn in spec.eval_metric_fns.items():
+                    out = fn(f, l)
+                    v, cnt = out if isinstance(out, tuple) else (out, n)
+                    metric_means[k].update(float(v), cnt)
+            n_batches += 1
+
+        run_batch(features, labels)
+        for f, l in it:
+            if steps is not None and n_batches >= steps:
+                break
+            run_batch(_to_device(f, device), _to_device(l, device))
+
+        results = {"loss": loss_m.result(),
+                   "global_step": restored_step if restored_step is not None else 0}
+        results.update({k: m.result() for k, m in metric_means.items()})
+        self._logger.log(eval=True, **{k: v for k, v in results.items()})
+        return results
+
+    def predict(self, input_fn, checkpoint_path: Optional[str] = None) -> Iterator:
+        device = self.config.device
+        it = iter(_call_input_fn(input_fn, ModeKeys.PREDICT))
+        spec = None
+        for batch in it:
+            f = batch[0] if isinstance(batch, tuple) else batch
+            f = _to_device(f, device)
+            if spec is None:
+                spec = self._build_spec(ModeKeys.PREDICT, f, None)
+                self._restore(spec, checkpoint_path)
+                if spec.model is not None:
+                    spec.model.eval()
+            with torch.no_grad():
+                preds = spec.predictions_fn(f)
+            if torch.is_tensor(preds):
+                for p in preds:
+                    yield p
+            else:
+                yield preds
+
+
+def train_and_evaluate(estimator: Estimator, train_spec: TrainSpec,
+                       eval_spec: EvalSpec) -> Dict:
+    """Train to max_steps, evaluating at most every ``throttle_secs``
+    (the reference's cadence: 01:101, another-example.py:318)."""
+    last_eval = time.monotonic()
+    results: Dict = {}
+    chunk = estimator.config.save_checkpoints_steps or \
+        estimator.config.log_step_count_steps or 100
+    while True:
+        r = estimator.train(train_spec.input_fn, max_steps=train_spec.max_steps,
+                            steps=chunk)
+        step = r["global_step"]
+        if time.monotonic() - last_eval >= eval_spec.throttle_secs:
+            results = estimator.evaluate(eval_spec.input_fn, steps=eval_spec.steps)
+            last_eval = time.monotonic()
+        if train_spec.max_steps is not None and step >= train_spec.max_steps:
+            break
+        if r.get("loss") is None:  # input exhausted, nothing trained this chunk
+            break
+    results = estimator.evaluate(eval_spec.input_fn, steps=eval_spec.steps) or results
+    return results

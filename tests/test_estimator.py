@@ -1,0 +1,168 @@
+"""Estimator-layer tests: train/eval/predict cycle, checkpoint resume,
+train_and_evaluate, input_fn pipeline semantics (SURVEY.md sections 1, 3.2)."""
+
+import os
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from gradient_accumulation_tf_estimator_amd import create_optimizer
+from gradient_accumulation_tf_estimator_amd.data import synthetic
+from gradient_accumulation_tf_estimator_amd.data.input_fn import (
+    ArrayDataset,
+    InputContext,
+    input_fn_iterator,
+)
+from gradient_accumulation_tf_estimator_amd.estimator import (
+    Estimator,
+    EstimatorSpec,
+    EvalSpec,
+    ModeKeys,
+    RunConfig,
+    TrainSpec,
+    train_and_evaluate,
+)
+from gradient_accumulation_tf_estimator_amd.models.mnist import MnistCNN
+
+
+def mnist_model_fn(features, labels, mode, params):
+    torch.manual_seed(params.get("seed", 0))
+    model = MnistCNN()
+    if mode == ModeKeys.PREDICT:
+        return EstimatorSpec(mode, model=model,
+                             predictions_fn=lambda f: model(f).argmax(-1))
+    loss_fn = lambda f, l: model.loss(f, l)
+    if mode == ModeKeys.EVAL:
+        def acc(f, l):
+            return float((model(f).argmax(-1) == l).float().mean()), l.numel()
+        return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
+                             eval_metric_fns={"accuracy": acc})
+    train_op = create_optimizer(
+        model, params["learning_rate"], 10000, 0,
+        gradient_accumulation_multiplier=params["gradient_accumulation_multiplier"],
+        clip_norm=None,
+    )
+    return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
+
+
+def make_estimator(tmp_path, K=2):
+    cfg = RunConfig(model_dir=str(tmp_path / "model"), log_step_count_steps=50,
+                    save_checkpoints_steps=None, tf_random_seed=19830610)
+    return Estimator(mnist_model_fn, cfg,
+                     params={"learning_rate": 1e-3,
+                             "gradient_accumulation_multiplier": K, "seed": 0})
+
+
+def train_input_fn(mode=None):
+    ds = synthetic.mnist(n=512)
+    return input_fn_iterator(ds, batch_size=32, num_epochs=None, seed=1)
+
+
+def eval_input_fn(mode=None):
+    ds = synthetic.mnist(n=256, seed=4)
+    return input_fn_iterator(ds, batch_size=64, num_epochs=1, shuffle=False)
+
+
+def test_train_eval_predict_cycle(tmp_path):
+    est = make_estimator(tmp_path)
+    r = est.train(train_input_fn, max_steps=60)
+    assert r["global_step"] == 60
+    ev = est.evaluate(eval_input_fn)
+    assert "loss" in ev and "accuracy" in ev and ev["global_step"] == 60
+
+    preds = list(est.predict(lambda mode=None: (
+        (f, l) for f, l in eval_input_fn())))
+    assert len(preds) == 256
+    assert all(0 <= int(p) <= 9 for p in preds)
+
+
+def test_checkpoint_resume_continues_exactly(tmp_path):
+    est = make_estimator(tmp_path)
+    est.train(train_input_fn, max_steps=30)
+
+    # fresh estimator object (new process simulation) resumes from ckpt
+    est2 = make_estimator(tmp_path)
+    r = est2.train(train_input_fn, max_steps=30)
+    assert r["global_step"] == 30  # already done, no extra steps
+    r2 = est2.train(train_input_fn, max_steps=45)
+    assert r2["global_step"] == 45
+
+
+def test_train_and_evaluate(tmp_path):
+    est = make_estimator(tmp_path)
+    results = train_and_evaluate(
+        est,
+        TrainSpec(train_input_fn, max_steps=40),
+        EvalSpec(eval_input_fn, steps=2, throttle_secs=0.0),
+    )
+    assert "loss" in results
+
+
+def test_training_improves_model(tmp_path):
+    # eval on the training distribution (same data seed) -> loss must drop;
+    # held-out accuracy must rise too.
+    insample_fn = lambda mode=None: input_fn_iterator(
+        synthetic.mnist(n=512), batch_size=64, num_epochs=1, shuffle=False)
+    est = make_estimator(tmp_path, K=2)
+    est.train(train_input_fn, max_steps=5)
+    early = est.evaluate(insample_fn)
+    est.train(train_input_fn, max_steps=400)
+    late = est.evaluate(insample_fn)
+    assert late["loss"] < 0.5 * early["loss"]
+    held_out = est.evaluate(eval_input_fn)
+    assert held_out["accuracy"] > early["accuracy"]
+
+
+def test_input_fn_shard_shuffle_batch():
+    ds = synthetic.mnist(n=100)
+    # shard before shuffle: 2 pipelines see disjoint halves
+    seen = set()
+    for pid in (0, 1):
+        it = input_fn_iterator(ds, batch_size=10, num_epochs=1, seed=3,
+                               input_context=InputContext(2, pid))
+        labels = []
+        n = 0
+        for f, l in it:
+            n += l.shape[0]
+        assert n == 50
+    # no drop_remainder: 100/32 -> batches of 32,32,32,4
+    sizes = [l.shape[0] for _, l in input_fn_iterator(ds, 32, num_epochs=1, seed=0)]
+    assert sizes == [32, 32, 32, 4]
+    # repeat: 2 epochs doubles elements
+    total = sum(l.shape[0] for _, l in input_fn_iterator(ds, 32, num_epochs=2, seed=0))
+    assert total == 200
+
+
+def test_housing_estimator_end_to_end(tmp_path):
+    from gradient_accumulation_tf_estimator_amd.models.housing import HousingMLP
+
+    def housing_model_fn(features, labels, mode, params):
+        torch.manual_seed(0)
+        model = HousingMLP()
+        if mode == ModeKeys.PREDICT:
+            return EstimatorSpec(mode, model=model, predictions_fn=model.forward)
+        loss_fn = lambda f, l: model.loss(f, l)
+        if mode == ModeKeys.EVAL:
+            from gradient_accumulation_tf_estimator_amd.utils.metrics import mae
+            def mae_fn(f, l):
+                return mae(model(f), l.float())
+            return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
+                                 eval_metric_fns={"mae": mae_fn})
+        # housing example: stock Adam (no clip), K=3 (another-example.py:269,276)
+        train_op = create_optimizer(model, 1e-2, 10000, 0,
+                                    gradient_accumulation_multiplier=3,
+                                    clip_norm=None, weight_decay=0.0)
+        return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
+
+    ds = synthetic.housing(n=236)
+    cfg = RunConfig(model_dir=str(tmp_path / "housing"), log_step_count_steps=100)
+    est = Estimator(housing_model_fn, cfg)
+    fn = lambda mode=None: input_fn_iterator(ds, batch_size=59, num_epochs=None, seed=2)
+    est.train(fn, max_steps=300)
+    ev = est.evaluate(lambda mode=None: input_fn_iterator(ds, 59, num_epochs=1,
+                                                          shuffle=False))
+    assert ev["mae"] < 2.0
+    preds = list(est.predict(lambda mode=None: input_fn_iterator(ds, 59, num_epochs=1,
+                                                                 shuffle=False)))
+    assert len(preds) == 236
