@@ -39,7 +39,7 @@ def bert_batches(
     return ArrayDataset(ids, labels)
 
 
-def housing(n: int = 512, seed: int = 7) -> ArrayDataset:
+def housing(n: int = 512, seed: int = 7, label_seed: int = 321) -> ArrayDataset:
     from ..models.housing import CATEGORICAL_FEATURE, NUMERIC_FEATURES
 
     g = torch.Generator().manual_seed(seed)
@@ -47,7 +47,9 @@ def housing(n: int = 512, seed: int = 7) -> ArrayDataset:
         k: torch.randn(n, generator=g) for k in NUMERIC_FEATURES
     }
     feats[CATEGORICAL_FEATURE] = torch.randint(0, 2, (n,), generator=g)
-    w = torch.randn(len(NUMERIC_FEATURES), generator=g)
+    # target function FIXED across data seeds (see mnist above)
+    gw = torch.Generator().manual_seed(label_seed)
+    w = torch.randn(len(NUMERIC_FEATURES), generator=gw)
     x = torch.stack([feats[k] for k in NUMERIC_FEATURES], dim=1)
     y = x @ w + 0.5 * feats[CATEGORICAL_FEATURE].float() + 0.1 * torch.randn(n, generator=g)
     return ArrayDataset(feats, y)
