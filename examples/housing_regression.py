@@ -53,12 +53,12 @@ if __name__ == "__main__":
         model_fn,
         RunConfig(model_dir="/tmp/ga_amd_housing", log_step_count_steps=100,
                   tf_random_seed=19830610),
-        params={"learning_rate": 1e-2, "gradient_accumulation_multiplier": ACCUM},
+        params={"learning_rate": 1e-3, "gradient_accumulation_multiplier": ACCUM},
     )
     results = train_and_evaluate(
         est,
         TrainSpec(lambda mode=None: input_fn_iterator(
-            train_ds, BATCH_SIZE, num_epochs=200, seed=0), max_steps=1200),
+            train_ds, BATCH_SIZE, num_epochs=400, seed=0), max_steps=2400),
         EvalSpec(lambda mode=None: input_fn_iterator(
             eval_ds, BATCH_SIZE, num_epochs=1, shuffle=False), throttle_secs=30),
     )
