@@ -40,6 +40,9 @@ def parse_args():
     p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
                    help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
+    p.add_argument("--sdpa", default="auto",
+                   choices=["auto", "flash", "efficient", "math"],
+                   help="force a scaled_dot_product_attention backend")
     return p.parse_args()
 
 
@@ -63,6 +66,10 @@ def main():
     if use_cuda:
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
+        if args.sdpa != "auto":
+            torch.backends.cuda.enable_flash_sdp(args.sdpa == "flash")
+            torch.backends.cuda.enable_mem_efficient_sdp(args.sdpa == "efficient")
+            torch.backends.cuda.enable_math_sdp(args.sdpa == "math")
     else:
         device = torch.device("cpu")
 

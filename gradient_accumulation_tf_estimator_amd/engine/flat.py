@@ -162,6 +162,18 @@ class FlatState:
                 )
             self._params.append(p)
 
+        self._slice_by_param = {
+            id(p): s for s, p in zip(self.layout.slices, (by_name[s.name] for s in self.layout.slices))
+        }
+
+    def accum_view(self, param: torch.Tensor) -> torch.Tensor:
+        """Flat fp32 accumulation-buffer slice for a parameter -- the target
+        for ops whose backward accumulates gradients directly (ops/fused.py)."""
+        s = self._slice_by_param.get(id(param))
+        if s is None:
+            raise KeyError("parameter is not managed by this FlatState")
+        return self.accum[s.offset : s.offset + s.numel]
+
     @property
     def decay_boundary(self) -> int:
         return self.layout.decay_boundary

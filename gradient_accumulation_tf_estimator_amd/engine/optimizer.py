@@ -75,8 +75,9 @@ def create_optimizer(
     ddp_scale_loss: bool = True,
     backend: str = "auto",
 ) -> TrainOp:
-    if isinstance(model_or_params, nn.Module):
-        named = list(model_or_params.named_parameters())
+    model = model_or_params if isinstance(model_or_params, nn.Module) else None
+    if model is not None:
+        named = list(model.named_parameters())
     else:
         named = list(model_or_params)
     engine = AccumEngine(
@@ -95,4 +96,10 @@ def create_optimizer(
         process_group=process_group,
         backend=backend,
     )
+    if model is not None and engine.backend == "hip":
+        # fused modules (ops/fused.py) accumulate their param grads directly
+        # into the flat fp32 accum buffer, bypassing .grad
+        from ..ops.fused import bind_direct_grad
+
+        bind_direct_grad(model, engine)
     return TrainOp(engine, ddp_scale_loss=ddp_scale_loss)

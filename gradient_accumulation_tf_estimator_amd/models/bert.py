@@ -25,6 +25,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fused import FusedAddLayerNorm, FusedBiasGelu
+
 
 @dataclass
 class BertConfig:
@@ -63,7 +65,7 @@ class BertEmbeddings(nn.Module):
         self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.position_embeddings = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
         self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
-        self.LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.LayerNorm = FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
         self.dropout = nn.Dropout(cfg.dropout)
         self.register_buffer(
             "position_ids", torch.arange(cfg.max_position_embeddings).unsqueeze(0), persistent=False
@@ -106,23 +108,33 @@ class BertSelfAttention(nn.Module):
 
 
 class BertLayer(nn.Module):
-    """Post-LN encoder layer (original BERT ordering)."""
+    """Post-LN encoder layer (original BERT ordering), MI355X-fused:
+
+    The out-projection / FFN-output biases fold into the following fused
+    residual+LayerNorm kernel; the intermediate bias folds into the fused
+    bias+GELU kernel -- one launch each where eager PyTorch runs 2-3, and
+    their backwards feed parameter grads straight into the engine's fp32
+    accum buffer (ops/fused.py).
+    """
 
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.attention = BertSelfAttention(cfg)
-        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size)
-        self.attention_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
-        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
-        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
-        self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size, bias=False)
+        self.attention_LayerNorm = FusedAddLayerNorm(
+            cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
+        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+        self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
+        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+        self.output_LayerNorm = FusedAddLayerNorm(
+            cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
         self.dropout = nn.Dropout(cfg.dropout)
 
     def forward(self, x, attn_mask=None):
         a = self.attention(x, attn_mask)
-        x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
-        h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))
-        return self.output_LayerNorm(x + self.dropout(h))
+        x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
+        h = self.output(self.intermediate_act(self.intermediate(x)))
+        return self.output_LayerNorm(self.dropout(h), residual=x)
 
 
 class BertModel(nn.Module):

@@ -1,0 +1,292 @@
+// Fused residual-add + bias + LayerNorm and bias + GELU kernels (gfx950).
+//
+// Why these exist (profiles/r01_bench_bert_small_kernel_trace.md): at the
+// reference's micro-batch (8 x 128 tokens) the PyTorch step is a swarm of
+// small kernels -- separate residual adds, 2-kernel LN forward, 3-kernel LN
+// backward, per-bias reduce_kernels, and one AccumulateGrad add per
+// parameter per micro-step. Here:
+//
+//   k_addln_fwd     : h = x (+res) (+bias); y = LN(h)*gamma+beta  [1 launch]
+//   k_addln_bwd     : dh (to bf16) + per-block fp32 partials for
+//                     dgamma/dbeta/dbias                          [1 launch]
+//   k_biasgelu_fwd  : y = gelu_tanh(x + bias)                     [1 launch]
+//   k_biasgelu_bwd  : dx + per-block fp32 partials for dbias      [1 launch]
+//   k_colreduce_acc : reduce the partials over blocks and ADD the result
+//                     straight into the engine's flat fp32 accumulation
+//                     buffer slices (reference accum_grads semantics,
+//                     optimization.py:81,93) -- bypassing .grad entirely,
+//                     in fp32, one tiny launch for up to 3 params.
+//
+// Geometry: one wave64 per row, lane l owns columns [l*epl, (l+1)*epl),
+// epl = H/64 (H % 256 == 0 -> epl % 4 == 0; short4 = 8 B/lane vector
+// access). Row stats reduce with __shfl_xor across the wave -- no LDS on
+// the forward path. Backward combines the 4 waves' column partials in LDS
+// and writes one fp32 partial slab per block.
+
+#include <hip/hip_runtime.h>
+
+static inline __device__ float bf2f(unsigned short u) {
+  union { unsigned int i; float f; } c;
+  c.i = (unsigned int)u << 16;
+  return c.f;
+}
+
+static inline __device__ unsigned short f2bf(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  unsigned int x = c.i;
+  if ((x & 0x7fffffffu) > 0x7f800000u) return (unsigned short)((x >> 16) | 0x0040u);
+  return (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
+}
+
+static inline __device__ float wave_sum(float s) {
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+  return s;
+}
+
+#define MAX_EPL 16  // H <= 1024 for LN
+
+// ---------------- fused add + bias + LayerNorm forward ----------------
+extern "C" __global__ void k_addln_fwd(
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ res,
+    const unsigned short* __restrict__ bias,  // [H] or null
+    const unsigned short* __restrict__ gamma, const unsigned short* __restrict__ beta,
+    unsigned short* __restrict__ y, unsigned short* __restrict__ h_out,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    int R, int H, float eps) {
+  const int epl = H >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wid = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
+  const int nwaves = gridDim.x * (blockDim.x >> 6);
+  const int c0 = lane * epl;
+
+  float hv[MAX_EPL];
+  for (int row = wid; row < R; row += nwaves) {
+    const unsigned short* xr = x + (long long)row * H + c0;
+    const unsigned short* rr = res ? res + (long long)row * H + c0 : nullptr;
+    float s = 0.f;
+    for (int c = 0; c < epl; c += 4) {
+      ushort4 xv = *(const ushort4*)(xr + c);
+      hv[c] = bf2f(xv.x); hv[c + 1] = bf2f(xv.y);
+      hv[c + 2] = bf2f(xv.z); hv[c + 3] = bf2f(xv.w);
+      if (rr) {
+        ushort4 rv = *(const ushort4*)(rr + c);
+        hv[c] += bf2f(rv.x); hv[c + 1] += bf2f(rv.y);
+        hv[c + 2] += bf2f(rv.z); hv[c + 3] += bf2f(rv.w);
+      }
+      if (bias) {
+        ushort4 bv = *(const ushort4*)(bias + c0 + c);
+        hv[c] += bf2f(bv.x); hv[c + 1] += bf2f(bv.y);
+        hv[c + 2] += bf2f(bv.z); hv[c + 3] += bf2f(bv.w);
+      }
+      s += hv[c] + hv[c + 1] + hv[c + 2] + hv[c + 3];
+    }
+    const float mean = wave_sum(s) / H;
+    float sq = 0.f;
+    for (int c = 0; c < epl; ++c) {
+      const float d = hv[c] - mean;
+      sq = fmaf(d, d, sq);
+    }
+    const float rstd = rsqrtf(wave_sum(sq) / H + eps);
+    unsigned short* yr = y + (long long)row * H + c0;
+    unsigned short* hr = h_out + (long long)row * H + c0;
+    for (int c = 0; c < epl; c += 4) {
+      ushort4 gv = *(const ushort4*)(gamma + c0 + c);
+      ushort4 bv = *(const ushort4*)(beta + c0 + c);
+      ushort4 ho, yo;
+      ho.x = f2bf(hv[c]);     ho.y = f2bf(hv[c + 1]);
+      ho.z = f2bf(hv[c + 2]); ho.w = f2bf(hv[c + 3]);
+      yo.x = f2bf(fmaf((hv[c] - mean) * rstd, bf2f(gv.x), bf2f(bv.x)));
+      yo.y = f2bf(fmaf((hv[c + 1] - mean) * rstd, bf2f(gv.y), bf2f(bv.y)));
+      yo.z = f2bf(fmaf((hv[c + 2] - mean) * rstd, bf2f(gv.z), bf2f(bv.z)));
+      yo.w = f2bf(fmaf((hv[c + 3] - mean) * rstd, bf2f(gv.w), bf2f(bv.w)));
+      *(ushort4*)(hr + c) = ho;
+      *(ushort4*)(yr + c) = yo;
+    }
+    if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+  }
+}
+
+// ---------------- fused LayerNorm backward ----------------
+// dh = rstd * (dxh - mean(dxh) - xh * mean(dxh*xh)),  dxh = dy*gamma,
+// xh = (h-mean)*rstd.  Partials per block: [3][H] fp32 = {dgamma, dbeta, db}.
+extern "C" __global__ void k_addln_bwd(
+    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ h,
+    const unsigned short* __restrict__ gamma,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    unsigned short* __restrict__ dh_out,
+    float* __restrict__ partials,  // [gridDim.x][3][H]
+    int R, int H) {
+  const int epl = H >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wlocal = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  const int wid = blockIdx.x * wpb + wlocal;
+  const int nwaves = gridDim.x * wpb;
+  const int c0 = lane * epl;
+
+  float g[MAX_EPL], acc_dg[MAX_EPL], acc_db[MAX_EPL], acc_dbias[MAX_EPL];
+  for (int c = 0; c < epl; c += 4) {
+    ushort4 gv = *(const ushort4*)(gamma + c0 + c);
+    g[c] = bf2f(gv.x); g[c + 1] = bf2f(gv.y); g[c + 2] = bf2f(gv.z); g[c + 3] = bf2f(gv.w);
+    acc_dg[c] = acc_dg[c + 1] = acc_dg[c + 2] = acc_dg[c + 3] = 0.f;
+    acc_db[c] = acc_db[c + 1] = acc_db[c + 2] = acc_db[c + 3] = 0.f;
+    acc_dbias[c] = acc_dbias[c + 1] = acc_dbias[c + 2] = acc_dbias[c + 3] = 0.f;
+  }
+
+  float dyv[MAX_EPL], xh[MAX_EPL];
+  for (int row = wid; row < R; row += nwaves) {
+    const float mean = mean_in[row], rstd = rstd_in[row];
+    const unsigned short* dyr = dy + (long long)row * H + c0;
+    const unsigned short* hr = h + (long long)row * H + c0;
+    float s1 = 0.f, s2 = 0.f;
+    for (int c = 0; c < epl; c += 4) {
+      ushort4 dv = *(const ushort4*)(dyr + c);
+      ushort4 hv = *(const ushort4*)(hr + c);
+      dyv[c] = bf2f(dv.x); dyv[c + 1] = bf2f(dv.y);
+      dyv[c + 2] = bf2f(dv.z); dyv[c + 3] = bf2f(dv.w);
+      xh[c] = (bf2f(hv.x) - mean) * rstd; xh[c + 1] = (bf2f(hv.y) - mean) * rstd;
+      xh[c + 2] = (bf2f(hv.z) - mean) * rstd; xh[c + 3] = (bf2f(hv.w) - mean) * rstd;
+      for (int k = 0; k < 4; ++k) {
+        const float dxh = dyv[c + k] * g[c + k];
+        s1 += dxh;
+        s2 = fmaf(dxh, xh[c + k], s2);
+      }
+    }
+    s1 = wave_sum(s1) / H;
+    s2 = wave_sum(s2) / H;
+    unsigned short* dhr = dh_out + (long long)row * H + c0;
+    for (int c = 0; c < epl; c += 4) {
+      ushort4 o;
+      float dh[4];
+      for (int k = 0; k < 4; ++k) {
+        const float dxh = dyv[c + k] * g[c + k];
+        dh[k] = rstd * (dxh - s1 - xh[c + k] * s2);
+        acc_dg[c + k] = fmaf(dyv[c + k], xh[c + k], acc_dg[c + k]);
+        acc_db[c + k] += dyv[c + k];
+        acc_dbias[c + k] += dh[k];
+      }
+      o.x = f2bf(dh[0]); o.y = f2bf(dh[1]); o.z = f2bf(dh[2]); o.w = f2bf(dh[3]);
+      *(ushort4*)(dhr + c) = o;
+    }
+  }
+
+  // combine the block's waves in LDS, then one fp32 partial slab per block
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][3][H]
+  float* my = lds + ((size_t)wlocal * 3 * H);
+  for (int c = 0; c < epl; ++c) {
+    my[c0 + c] = acc_dg[c];
+    my[H + c0 + c] = acc_db[c];
+    my[2 * H + c0 + c] = acc_dbias[c];
+  }
+  __syncthreads();
+  float* out = partials + (size_t)blockIdx.x * 3 * H;
+  for (int i = threadIdx.x; i < 3 * H; i += blockDim.x) {
+    float s = 0.f;
+    for (int w = 0; w < wpb; ++w) s += lds[(size_t)w * 3 * H + i];
+    out[i] = s;
+  }
+}
+
+// ---------------- bias + GELU (tanh approx, matches torch) ----------------
+#define GELU_C0 0.7978845608028654f
+#define GELU_C1 0.044715f
+
+static inline __device__ float gelu_fwd1(float h) {
+  const float u = GELU_C0 * fmaf(GELU_C1 * h * h, h, h);
+  return 0.5f * h * (1.f + tanhf(u));
+}
+
+static inline __device__ float gelu_bwd1(float h, float dy) {
+  const float u = GELU_C0 * fmaf(GELU_C1 * h * h, h, h);
+  const float t = tanhf(u);
+  const float du = GELU_C0 * fmaf(3.f * GELU_C1 * h, h, 1.f);
+  return dy * (0.5f * (1.f + t) + 0.5f * h * (1.f - t * t) * du);
+}
+
+extern "C" __global__ void k_biasgelu_fwd(
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ y, long long total, int H) {
+  long long stride = (long long)gridDim.x * blockDim.x * 4;
+  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < total;
+       i += stride) {
+    ushort4 xv = *(const ushort4*)(x + i);
+    const int c = (int)(i % H);
+    ushort4 bv = *(const ushort4*)(bias + c);
+    ushort4 o;
+    o.x = f2bf(gelu_fwd1(bf2f(xv.x) + bf2f(bv.x)));
+    o.y = f2bf(gelu_fwd1(bf2f(xv.y) + bf2f(bv.y)));
+    o.z = f2bf(gelu_fwd1(bf2f(xv.z) + bf2f(bv.z)));
+    o.w = f2bf(gelu_fwd1(bf2f(xv.w) + bf2f(bv.w)));
+    *(ushort4*)(y + i) = o;
+  }
+}
+
+#define MAX_EPL_G 64  // H <= 4096 for the gelu bias reduction
+
+extern "C" __global__ void k_biasgelu_bwd(
+    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ dx_out,
+    float* __restrict__ partials,  // [gridDim.x][H]
+    int R, int H) {
+  const int epl = H >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wlocal = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  const int wid = blockIdx.x * wpb + wlocal;
+  const int nwaves = gridDim.x * wpb;
+  const int c0 = lane * epl;
+
+  float acc[MAX_EPL_G];
+  for (int c = 0; c < epl; ++c) acc[c] = 0.f;
+
+  for (int row = wid; row < R; row += nwaves) {
+    const unsigned short* dyr = dy + (long long)row * H + c0;
+    const unsigned short* xr = x + (long long)row * H + c0;
+    unsigned short* dxr = dx_out + (long long)row * H + c0;
+    for (int c = 0; c < epl; c += 4) {
+      ushort4 dv = *(const ushort4*)(dyr + c);
+      ushort4 xv = *(const ushort4*)(xr + c);
+      ushort4 bv = *(const ushort4*)(bias + c0 + c);
+      float d0 = gelu_bwd1(bf2f(xv.x) + bf2f(bv.x), bf2f(dv.x));
+      float d1 = gelu_bwd1(bf2f(xv.y) + bf2f(bv.y), bf2f(dv.y));
+      float d2 = gelu_bwd1(bf2f(xv.z) + bf2f(bv.z), bf2f(dv.z));
+      float d3 = gelu_bwd1(bf2f(xv.w) + bf2f(bv.w), bf2f(dv.w));
+      acc[c] += d0; acc[c + 1] += d1; acc[c + 2] += d2; acc[c + 3] += d3;
+      ushort4 o;
+      o.x = f2bf(d0); o.y = f2bf(d1); o.z = f2bf(d2); o.w = f2bf(d3);
+      *(ushort4*)(dxr + c) = o;
+    }
+  }
+
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][H]
+  float* my = lds + (size_t)wlocal * H;
+  for (int c = 0; c < epl; ++c) my[c0 + c] = acc[c];
+  __syncthreads();
+  float* out = partials + (size_t)blockIdx.x * H;
+  for (int i = threadIdx.x; i < H; i += blockDim.x) {
+    float s = 0.f;
+    for (int w = 0; w < wpb; ++w) s += lds[(size_t)w * H + i];
+    out[i] = s;
+  }
+}
+
+// ---------------- column-partials -> flat fp32 accum slices ----------------
+// Reduces partials [NB][C] over NB and ADDS result into up to 3 destination
+// fp32 slices of the engine's flat accumulation buffer:
+//   dest0 gets cols [0, n0), dest1 [n0, n0+n1), dest2 [n0+n1, C).
+extern "C" __global__ void k_colreduce_acc(
+    const float* __restrict__ partials, int NB, int C,
+    float* __restrict__ dest0, int n0,
+    float* __restrict__ dest1, int n1,
+    float* __restrict__ dest2) {
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < C;
+       i += gridDim.x * blockDim.x) {
+    float s = 0.f;
+    for (int b = 0; b < NB; ++b) s += partials[(size_t)b * C + i];
+    if (i < n0) dest0[i] += s;
+    else if (i < n0 + n1) dest1[i - n0] += s;
+    else dest2[i - n0 - n1] += s;
+  }
+}

@@ -20,6 +20,20 @@ extern "C" __global__ void k_apply_bf16(float4*, float4*, float4*, float4*, usho
                                         const float*, const float*,
                                         long long, long long,
                                         float, float, float, float, float, float);
+extern "C" __global__ void k_addln_fwd(const unsigned short*, const unsigned short*,
+                                       const unsigned short*, const unsigned short*,
+                                       const unsigned short*, unsigned short*,
+                                       unsigned short*, float*, float*, int, int, float);
+extern "C" __global__ void k_addln_bwd(const unsigned short*, const unsigned short*,
+                                       const unsigned short*, const float*, const float*,
+                                       unsigned short*, float*, int, int);
+extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned short*,
+                                          unsigned short*, long long, int);
+extern "C" __global__ void k_biasgelu_bwd(const unsigned short*, const unsigned short*,
+                                          const unsigned short*, unsigned short*,
+                                          float*, int, int);
+extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
+                                           float*, int, float*);
 
 namespace {
 
@@ -114,6 +128,108 @@ void fused_apply(at::Tensor accum, at::Tensor m, at::Tensor v, at::Tensor master
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_apply launch failed");
 }
 
+// ---------------- fused LN / GELU ----------------
+
+inline const unsigned short* bfp(const at::Tensor& t) {
+  return (const unsigned short*)t.data_ptr();
+}
+inline unsigned short* bfp_mut(at::Tensor& t) { return (unsigned short*)t.data_ptr(); }
+
+inline void check_bf16_2d(const at::Tensor& t, const char* name, int H) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() && t.scalar_type() == at::kBFloat16,
+              name, " must be contiguous bf16 on GPU");
+  TORCH_CHECK(t.numel() % H == 0, name, " numel not divisible by H");
+}
+
+std::vector<at::Tensor> addln_fwd(at::Tensor x, c10::optional<at::Tensor> res,
+                                  c10::optional<at::Tensor> bias,
+                                  at::Tensor gamma, at::Tensor beta, double eps) {
+  const int H = (int)gamma.numel();
+  TORCH_CHECK(H % 256 == 0 && H <= 1024, "LN hidden must be %256==0 and <=1024");
+  check_bf16_2d(x, "x", H);
+  const int R = (int)(x.numel() / H);
+  auto y = at::empty_like(x);
+  auto h = at::empty_like(x);
+  auto mean = at::empty({R}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int blocks = std::min((R + 3) / 4, 2048);
+  hipLaunchKernelGGL(k_addln_fwd, dim3(blocks), dim3(256), 0, stream,
+                     bfp(x), res ? bfp(*res) : nullptr, bias ? bfp(*bias) : nullptr,
+                     bfp(gamma), bfp(beta), bfp_mut(y), bfp_mut(h),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), R, H, (float)eps);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_addln_fwd launch failed");
+  return {y, h, mean, rstd};
+}
+
+std::vector<at::Tensor> addln_bwd(at::Tensor dy, at::Tensor h, at::Tensor gamma,
+                                  at::Tensor mean, at::Tensor rstd) {
+  const int H = (int)gamma.numel();
+  check_bf16_2d(dy, "dy", H);
+  const int R = (int)(dy.numel() / H);
+  auto dh = at::empty_like(dy);
+  int blocks = std::max(1, std::min((R + 3) / 4, 128));
+  auto partials = at::empty({blocks, 3, H}, dy.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  size_t lds = (size_t)4 * 3 * H * sizeof(float);
+  hipLaunchKernelGGL(k_addln_bwd, dim3(blocks), dim3(256), lds, stream,
+                     bfp(dy), bfp(h), bfp(gamma), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), bfp_mut(dh),
+                     partials.data_ptr<float>(), R, H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_addln_bwd launch failed");
+  return {dh, partials};
+}
+
+at::Tensor biasgelu_fwd(at::Tensor x, at::Tensor bias) {
+  const int H = (int)bias.numel();
+  TORCH_CHECK(H % 256 == 0 && H <= 4096, "gelu hidden must be %256==0 and <=4096");
+  check_bf16_2d(x, "x", H);
+  auto y = at::empty_like(x);
+  const long long total = x.numel();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+  hipLaunchKernelGGL(k_biasgelu_fwd, dim3(blocks), dim3(256), 0, stream,
+                     bfp(x), bfp(bias), bfp_mut(y), total, H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_fwd launch failed");
+  return y;
+}
+
+std::vector<at::Tensor> biasgelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bias) {
+  const int H = (int)bias.numel();
+  check_bf16_2d(dy, "dy", H);
+  const int R = (int)(dy.numel() / H);
+  auto dx = at::empty_like(dy);
+  int blocks = std::max(1, std::min((R + 3) / 4, 128));
+  auto partials = at::empty({blocks, H}, dy.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  size_t lds = (size_t)4 * H * sizeof(float);
+  hipLaunchKernelGGL(k_biasgelu_bwd, dim3(blocks), dim3(256), lds, stream,
+                     bfp(dy), bfp(x), bfp(bias), bfp_mut(dx),
+                     partials.data_ptr<float>(), R, H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_bwd launch failed");
+  return {dx, partials};
+}
+
+void colreduce_acc(at::Tensor partials, at::Tensor dest0,
+                   c10::optional<at::Tensor> dest1, c10::optional<at::Tensor> dest2) {
+  TORCH_CHECK(partials.dim() >= 2 && partials.scalar_type() == at::kFloat &&
+              partials.is_contiguous(), "partials must be contiguous fp32 [NB,...]");
+  const int NB = (int)partials.size(0);
+  const int C = (int)(partials.numel() / NB);
+  const int n0 = (int)dest0.numel();
+  const int n1 = dest1 ? (int)dest1->numel() : 0;
+  const int n2 = dest2 ? (int)dest2->numel() : 0;
+  TORCH_CHECK(n0 + n1 + n2 == C, "dest sizes must sum to partial columns");
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int blocks = std::min((C + 255) / 256, 256);
+  hipLaunchKernelGGL(k_colreduce_acc, dim3(blocks), dim3(256), 0, stream,
+                     partials.data_ptr<float>(), NB, C,
+                     dest0.data_ptr<float>(), n0,
+                     dest1 ? dest1->data_ptr<float>() : nullptr, n1,
+                     dest2 ? dest2->data_ptr<float>() : nullptr);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_colreduce_acc launch failed");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
@@ -121,4 +237,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("sqnorm", &sqnorm, "out[0] = sum(accum^2)");
   mod.def("fused_apply", &fused_apply,
           "normalize + clip + AdamWeightDecay + bf16 write-back + zero accum");
+  mod.def("addln_fwd", &addln_fwd, "fused residual+bias+LayerNorm forward");
+  mod.def("addln_bwd", &addln_bwd, "fused LayerNorm backward -> dh + fp32 partials");
+  mod.def("biasgelu_fwd", &biasgelu_fwd, "y = gelu_tanh(x + bias)");
+  mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
+  mod.def("colreduce_acc", &colreduce_acc,
+          "reduce partials over blocks, ADD into flat fp32 accum slices");
 }

@@ -1,0 +1,168 @@
+"""Fused residual+bias+LayerNorm and bias+GELU modules (bf16, gfx950).
+
+The backward of these modules can write parameter gradients DIRECTLY into
+the engine's flat fp32 accumulation buffer (``bind_direct_grad``): the
+reference's per-variable ``accum_grad.assign_add(grad)`` (optimization.py:81,
+93) happens inside the op's backward kernel chain, in fp32, with no .grad
+round-trip -- eliminating one AccumulateGrad add + one bf16 grad buffer per
+parameter per micro-step (see profiles/r01_bench_bert_small_kernel_trace.md
+for why that matters at micro-batch 8).
+
+Unbound (standalone) they fall back to returning ordinary grads; on CPU (or
+non-bf16 GPU tensors) they fall back to plain PyTorch math with identical
+semantics.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import require_hip
+
+
+def _use_hip(x: torch.Tensor) -> bool:
+    return x.is_cuda and x.dtype == torch.bfloat16
+
+
+class _AddLayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, proj_bias, eps, module):
+        hip = require_hip()
+        x = x.contiguous()
+        res = residual.contiguous() if residual is not None else None
+        pb = proj_bias.contiguous() if proj_bias is not None else None
+        y, h, mean, rstd = hip.addln_fwd(x, res, pb, gamma, beta, eps)
+        ctx.save_for_backward(h, gamma, mean, rstd)
+        ctx.module = module
+        ctx.has_res = residual is not None
+        ctx.has_pb = proj_bias is not None
+        return y.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        h, gamma, mean, rstd = ctx.saved_tensors
+        dh, partials = hip.addln_bwd(dy.contiguous(), h, gamma, mean, rstd)
+        mod = ctx.module
+        dgamma = dbeta = dpb = None
+        views = getattr(mod, "_accum_views", None)
+        if views is not None:
+            vg, vb, vpb = views
+            hip.colreduce_acc(partials, vg, vb, vpb if vpb is not None else mod._sink())
+        else:
+            s = partials.sum(0)
+            dgamma = s[0].to(dy.dtype)
+            dbeta = s[1].to(dy.dtype)
+            if ctx.has_pb:
+                dpb = s[2].to(dy.dtype)
+        dh = dh.view_as(dy)
+        dres = dh if ctx.has_res else None
+        return dh, dres, dgamma, dbeta, dpb, None, None
+
+
+class FusedAddLayerNorm(nn.Module):
+    """y = LayerNorm(x + residual + proj_bias) * weight + bias.
+
+    ``proj_bias`` (optional) is the bias of the Linear that produced ``x`` --
+    folding it here removes the per-step bias-grad reduce_kernel from that
+    Linear's backward. Named *LayerNorm* so the engine's weight-decay regex
+    exclusion (optimization.py:65,179-187) treats weight/bias/proj_bias like
+    the reference's LayerNorm/bias variables.
+    """
+
+    def __init__(self, hidden: int, eps: float = 1e-12, proj_bias: bool = False):
+        super().__init__()
+        self.hidden = hidden
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(hidden))
+        self.bias = nn.Parameter(torch.zeros(hidden))
+        if proj_bias:
+            self.proj_bias = nn.Parameter(torch.zeros(hidden))
+        else:
+            self.register_parameter("proj_bias", None)
+        self._accum_views = None
+        self._sink_buf = None
+
+    def _sink(self):
+        if self._sink_buf is None or self._sink_buf.device != self.weight.device:
+            self._sink_buf = torch.zeros(
+                self.hidden, dtype=torch.float32, device=self.weight.device
+            )
+        return self._sink_buf
+
+    def forward(self, x, residual=None):
+        if _use_hip(x):
+            return _AddLayerNormFn.apply(
+                x, residual, self.weight, self.bias, self.proj_bias, self.eps, self
+            )
+        h = x
+        if residual is not None:
+            h = h + residual
+        if self.proj_bias is not None:
+            h = h + self.proj_bias
+        return F.layer_norm(h, (self.hidden,), self.weight, self.bias, self.eps)
+
+
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias, module):
+        hip = require_hip()
+        x = x.contiguous()
+        y = hip.biasgelu_fwd(x, bias)
+        ctx.save_for_backward(x, bias)
+        ctx.module = module
+        return y.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        x, bias = ctx.saved_tensors
+        dx, partials = hip.biasgelu_bwd(dy.contiguous(), x, bias)
+        dbias = None
+        view = getattr(ctx.module, "_accum_view", None)
+        if view is not None:
+            hip.colreduce_acc(partials, view, None, None)
+        else:
+            dbias = partials.sum(0).to(dy.dtype)
+        return dx.view_as(dy), dbias, None
+
+
+class FusedBiasGelu(nn.Module):
+    """y = gelu_tanh(x + bias); dbias computed inside the fused backward."""
+
+    def __init__(self, hidden: int):
+        super().__init__()
+        self.hidden = hidden
+        self.bias = nn.Parameter(torch.zeros(hidden))
+        self._accum_view = None
+
+    def forward(self, x):
+        if _use_hip(x):
+            return _BiasGeluFn.apply(x, self.bias, self)
+        return F.gelu(x + self.bias, approximate="tanh")
+
+
+def bind_direct_grad(model: nn.Module, engine) -> int:
+    """Wire every fused module's backward to the engine's flat fp32 accum
+    slices. Returns the number of modules bound. No-op for non-hip engines."""
+    if engine.backend != "hip":
+        return 0
+    n = 0
+    for mod in model.modules():
+        if isinstance(mod, FusedAddLayerNorm):
+            mod._accum_views = (
+                engine.state.accum_view(mod.weight),
+                engine.state.accum_view(mod.bias),
+                engine.state.accum_view(mod.proj_bias)
+                if mod.proj_bias is not None
+                else None,
+            )
+            n += 1
+        elif isinstance(mod, FusedBiasGelu):
+            mod._accum_view = engine.state.accum_view(mod.bias)
+            n += 1
+    return n

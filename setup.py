@@ -25,6 +25,7 @@ setup(
             name="gradient_accumulation_tf_estimator_amd.ops._ga_hip",
             sources=[
                 os.path.join(CSRC, "ga_kernels.hip"),
+                os.path.join(CSRC, "fused_ln_gelu.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],
             extra_compile_args={
