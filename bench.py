@@ -40,6 +40,8 @@ def parse_args():
     p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
                    help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
+    p.add_argument("--fused", default="on", choices=["on", "off"],
+                   help="fused LN/GELU HIP modules (A/B switch)")
     p.add_argument("--sdpa", default="auto",
                    choices=["auto", "flash", "efficient", "math"],
                    help="force a scaled_dot_product_attention backend")
@@ -78,6 +80,7 @@ def main():
 
     torch.manual_seed(1234 + rank)
     cfg = CONFIGS[args.model]()
+    cfg.fused = args.fused == "on"
     dtype = torch.bfloat16 if (args.dtype == "bf16" and use_cuda) else torch.float32
     model = BertForSequenceClassification(cfg).to(device=device, dtype=dtype)
     model.train()

@@ -41,6 +41,7 @@ class BertConfig:
     dropout: float = 0.0  # 0 keeps the step hipGraph-capturable & deterministic
     num_labels: int = 2
     initializer_range: float = 0.02
+    fused: bool = True  # use the fused LN/GELU HIP modules (A/B switch)
 
 
 # BASELINE.json model configs
@@ -65,7 +66,8 @@ class BertEmbeddings(nn.Module):
         self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.position_embeddings = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
         self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
-        self.LayerNorm = FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.LayerNorm = (FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+                          if cfg.fused else nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps))
         self.dropout = nn.Dropout(cfg.dropout)
         self.register_buffer(
             "position_ids", torch.arange(cfg.max_position_embeddings).unsqueeze(0), persistent=False
@@ -119,22 +121,31 @@ class BertLayer(nn.Module):
 
     def __init__(self, cfg: BertConfig):
         super().__init__()
+        self.fused = cfg.fused
         self.attention = BertSelfAttention(cfg)
-        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size, bias=False)
-        self.attention_LayerNorm = FusedAddLayerNorm(
-            cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
-        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
-        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
-        self.output_LayerNorm = FusedAddLayerNorm(
-            cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
+        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size, bias=not cfg.fused)
+        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=not cfg.fused)
+        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=not cfg.fused)
         self.dropout = nn.Dropout(cfg.dropout)
+        if cfg.fused:
+            self.attention_LayerNorm = FusedAddLayerNorm(
+                cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
+            self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
+            self.output_LayerNorm = FusedAddLayerNorm(
+                cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
+        else:
+            self.attention_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+            self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
 
     def forward(self, x, attn_mask=None):
         a = self.attention(x, attn_mask)
-        x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
-        h = self.output(self.intermediate_act(self.intermediate(x)))
-        return self.output_LayerNorm(self.dropout(h), residual=x)
+        if self.fused:
+            x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
+            h = self.output(self.intermediate_act(self.intermediate(x)))
+            return self.output_LayerNorm(self.dropout(h), residual=x)
+        x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
+        h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))
+        return self.output_LayerNorm(x + self.dropout(h))
 
 
 class BertModel(nn.Module):
