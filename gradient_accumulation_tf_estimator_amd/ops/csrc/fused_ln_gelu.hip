@@ -44,27 +44,30 @@ static inline __device__ float wave_sum(float s) {
   return s;
 }
 
-#define MAX_EPL 16  // H <= 1024 for LN
-
 // ---------------- fused add + bias + LayerNorm forward ----------------
-extern "C" __global__ void k_addln_fwd(
+// EPL = H/64 is a template parameter: per-lane arrays must be indexed by
+// compile-time constants or they land in scratch (cdna_hip_programming.md
+// rule 20) -- the first version of these kernels paid 3-6x for that.
+template <int EPL>
+__device__ __forceinline__ void addln_fwd_body(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ res,
     const unsigned short* __restrict__ bias,  // [H] or null
     const unsigned short* __restrict__ gamma, const unsigned short* __restrict__ beta,
     unsigned short* __restrict__ y, unsigned short* __restrict__ h_out,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     int R, int H, float eps) {
-  const int epl = H >> 6;
+  constexpr int epl = EPL;
   const int lane = threadIdx.x & 63;
   const int wid = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   const int nwaves = gridDim.x * (blockDim.x >> 6);
   const int c0 = lane * epl;
 
-  float hv[MAX_EPL];
+  float hv[EPL];
   for (int row = wid; row < R; row += nwaves) {
     const unsigned short* xr = x + (long long)row * H + c0;
     const unsigned short* rr = res ? res + (long long)row * H + c0 : nullptr;
     float s = 0.f;
+#pragma unroll
     for (int c = 0; c < epl; c += 4) {
       ushort4 xv = *(const ushort4*)(xr + c);
       hv[c] = bf2f(xv.x); hv[c + 1] = bf2f(xv.y);
@@ -83,6 +86,7 @@ extern "C" __global__ void k_addln_fwd(
     }
     const float mean = wave_sum(s) / H;
     float sq = 0.f;
+#pragma unroll
     for (int c = 0; c < epl; ++c) {
       const float d = hv[c] - mean;
       sq = fmaf(d, d, sq);
@@ -90,6 +94,7 @@ extern "C" __global__ void k_addln_fwd(
     const float rstd = rsqrtf(wave_sum(sq) / H + eps);
     unsigned short* yr = y + (long long)row * H + c0;
     unsigned short* hr = h_out + (long long)row * H + c0;
+#pragma unroll
     for (int c = 0; c < epl; c += 4) {
       ushort4 gv = *(const ushort4*)(gamma + c0 + c);
       ushort4 bv = *(const ushort4*)(beta + c0 + c);
@@ -110,14 +115,15 @@ extern "C" __global__ void k_addln_fwd(
 // ---------------- fused LayerNorm backward ----------------
 // dh = rstd * (dxh - mean(dxh) - xh * mean(dxh*xh)),  dxh = dy*gamma,
 // xh = (h-mean)*rstd.  Partials per block: [3][H] fp32 = {dgamma, dbeta, db}.
-extern "C" __global__ void k_addln_bwd(
+template <int EPL>
+__device__ __forceinline__ void addln_bwd_body(
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ h,
     const unsigned short* __restrict__ gamma,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
     unsigned short* __restrict__ dh_out,
     float* __restrict__ partials,  // [gridDim.x][3][H]
     int R, int H) {
-  const int epl = H >> 6;
+  constexpr int epl = EPL;
   const int lane = threadIdx.x & 63;
   const int wlocal = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
@@ -125,7 +131,8 @@ extern "C" __global__ void k_addln_bwd(
   const int nwaves = gridDim.x * wpb;
   const int c0 = lane * epl;
 
-  float g[MAX_EPL], acc_dg[MAX_EPL], acc_db[MAX_EPL], acc_dbias[MAX_EPL];
+  float g[EPL], acc_dg[EPL], acc_db[EPL], acc_dbias[EPL];
+#pragma unroll
   for (int c = 0; c < epl; c += 4) {
     ushort4 gv = *(const ushort4*)(gamma + c0 + c);
     g[c] = bf2f(gv.x); g[c + 1] = bf2f(gv.y); g[c + 2] = bf2f(gv.z); g[c + 3] = bf2f(gv.w);
@@ -134,12 +141,13 @@ extern "C" __global__ void k_addln_bwd(
     acc_dbias[c] = acc_dbias[c + 1] = acc_dbias[c + 2] = acc_dbias[c + 3] = 0.f;
   }
 
-  float dyv[MAX_EPL], xh[MAX_EPL];
+  float dyv[EPL], xh[EPL];
   for (int row = wid; row < R; row += nwaves) {
     const float mean = mean_in[row], rstd = rstd_in[row];
     const unsigned short* dyr = dy + (long long)row * H + c0;
     const unsigned short* hr = h + (long long)row * H + c0;
     float s1 = 0.f, s2 = 0.f;
+#pragma unroll
     for (int c = 0; c < epl; c += 4) {
       ushort4 dv = *(const ushort4*)(dyr + c);
       ushort4 hv = *(const ushort4*)(hr + c);
@@ -156,6 +164,7 @@ extern "C" __global__ void k_addln_bwd(
     s1 = wave_sum(s1) / H;
     s2 = wave_sum(s2) / H;
     unsigned short* dhr = dh_out + (long long)row * H + c0;
+#pragma unroll
     for (int c = 0; c < epl; c += 4) {
       ushort4 o;
       float dh[4];
@@ -174,6 +183,7 @@ extern "C" __global__ void k_addln_bwd(
   // combine the block's waves in LDS, then one fp32 partial slab per block
   extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][3][H]
   float* my = lds + ((size_t)wlocal * 3 * H);
+#pragma unroll
   for (int c = 0; c < epl; ++c) {
     my[c0 + c] = acc_dg[c];
     my[H + c0 + c] = acc_db[c];
@@ -222,15 +232,14 @@ extern "C" __global__ void k_biasgelu_fwd(
   }
 }
 
-#define MAX_EPL_G 64  // H <= 4096 for the gelu bias reduction
-
-extern "C" __global__ void k_biasgelu_bwd(
+template <int EPL>
+__device__ __forceinline__ void biasgelu_bwd_body(
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ bias,
     unsigned short* __restrict__ dx_out,
     float* __restrict__ partials,  // [gridDim.x][H]
     int R, int H) {
-  const int epl = H >> 6;
+  constexpr int epl = EPL;
   const int lane = threadIdx.x & 63;
   const int wlocal = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
@@ -238,13 +247,15 @@ extern "C" __global__ void k_biasgelu_bwd(
   const int nwaves = gridDim.x * wpb;
   const int c0 = lane * epl;
 
-  float acc[MAX_EPL_G];
+  float acc[EPL];
+#pragma unroll
   for (int c = 0; c < epl; ++c) acc[c] = 0.f;
 
   for (int row = wid; row < R; row += nwaves) {
     const unsigned short* dyr = dy + (long long)row * H + c0;
     const unsigned short* xr = x + (long long)row * H + c0;
     unsigned short* dxr = dx_out + (long long)row * H + c0;
+#pragma unroll
     for (int c = 0; c < epl; c += 4) {
       ushort4 dv = *(const ushort4*)(dyr + c);
       ushort4 xv = *(const ushort4*)(xr + c);
@@ -262,6 +273,7 @@ extern "C" __global__ void k_biasgelu_bwd(
 
   extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][H]
   float* my = lds + (size_t)wlocal * H;
+#pragma unroll
   for (int c = 0; c < epl; ++c) my[c0 + c] = acc[c];
   __syncthreads();
   float* out = partials + (size_t)blockIdx.x * H;
@@ -276,17 +288,65 @@ extern "C" __global__ void k_biasgelu_bwd(
 // Reduces partials [NB][C] over NB and ADDS result into up to 3 destination
 // fp32 slices of the engine's flat accumulation buffer:
 //   dest0 gets cols [0, n0), dest1 [n0, n0+n1), dest2 [n0+n1, C).
+#define CR_CHUNK 8
+
 extern "C" __global__ void k_colreduce_acc(
     const float* __restrict__ partials, int NB, int C,
     float* __restrict__ dest0, int n0,
     float* __restrict__ dest1, int n1,
     float* __restrict__ dest2) {
-  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < C;
-       i += gridDim.x * blockDim.x) {
+  const int nch = (NB + CR_CHUNK - 1) / CR_CHUNK;
+  const long long total = (long long)C * nch;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long long)gridDim.x * blockDim.x) {
+    const int i = (int)(t % C);
+    const int b0 = (int)(t / C) * CR_CHUNK;
+    const int b1 = min(b0 + CR_CHUNK, NB);
     float s = 0.f;
-    for (int b = 0; b < NB; ++b) s += partials[(size_t)b * C + i];
-    if (i < n0) dest0[i] += s;
-    else if (i < n0 + n1) dest1[i - n0] += s;
-    else dest2[i - n0 - n1] += s;
+#pragma unroll CR_CHUNK
+    for (int b = b0; b < b1; ++b) s += partials[(size_t)b * C + i];
+    float* d = (i < n0) ? dest0 + i
+               : (i < n0 + n1) ? dest1 + (i - n0)
+                               : dest2 + (i - n0 - n1);
+    if (nch == 1) *d += s;
+    else atomicAdd(d, s);
   }
 }
+
+
+// ---------------- explicit instantiations (H = 64*EPL) ----------------
+#define GA_LN_INST(EPL)                                                        \
+  extern "C" __global__ void k_addln_fwd_##EPL(                                \
+      const unsigned short* x, const unsigned short* res,                      \
+      const unsigned short* bias, const unsigned short* gamma,                 \
+      const unsigned short* beta, unsigned short* y, unsigned short* h_out,    \
+      float* mean_out, float* rstd_out, int R, int H, float eps) {             \
+    addln_fwd_body<EPL>(x, res, bias, gamma, beta, y, h_out, mean_out,         \
+                        rstd_out, R, H, eps);                                  \
+  }                                                                            \
+  extern "C" __global__ void k_addln_bwd_##EPL(                                \
+      const unsigned short* dy, const unsigned short* h,                       \
+      const unsigned short* gamma, const float* mean_in, const float* rstd_in, \
+      unsigned short* dh_out, float* partials, int R, int H) {                 \
+    addln_bwd_body<EPL>(dy, h, gamma, mean_in, rstd_in, dh_out, partials, R,   \
+                        H);                                                    \
+  }
+
+GA_LN_INST(4)
+GA_LN_INST(8)
+GA_LN_INST(12)
+GA_LN_INST(16)
+
+#define GA_GELU_INST(EPL)                                                      \
+  extern "C" __global__ void k_biasgelu_bwd_##EPL(                             \
+      const unsigned short* dy, const unsigned short* x,                       \
+      const unsigned short* bias, unsigned short* dx_out, float* partials,     \
+      int R, int H) {                                                          \
+    biasgelu_bwd_body<EPL>(dy, x, bias, dx_out, partials, R, H);               \
+  }
+
+GA_GELU_INST(16)
+GA_GELU_INST(24)
+GA_GELU_INST(32)
+GA_GELU_INST(48)
+GA_GELU_INST(64)

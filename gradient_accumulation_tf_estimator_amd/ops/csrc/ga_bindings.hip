@@ -20,18 +20,22 @@ extern "C" __global__ void k_apply_bf16(float4*, float4*, float4*, float4*, usho
                                         const float*, const float*,
                                         long long, long long,
                                         float, float, float, float, float, float);
-extern "C" __global__ void k_addln_fwd(const unsigned short*, const unsigned short*,
-                                       const unsigned short*, const unsigned short*,
-                                       const unsigned short*, unsigned short*,
-                                       unsigned short*, float*, float*, int, int, float);
-extern "C" __global__ void k_addln_bwd(const unsigned short*, const unsigned short*,
-                                       const unsigned short*, const float*, const float*,
-                                       unsigned short*, float*, int, int);
+#define GA_DECL_LN(EPL)                                                                  \
+  extern "C" __global__ void k_addln_fwd_##EPL(                                          \
+      const unsigned short*, const unsigned short*, const unsigned short*,               \
+      const unsigned short*, const unsigned short*, unsigned short*, unsigned short*,    \
+      float*, float*, int, int, float);                                                  \
+  extern "C" __global__ void k_addln_bwd_##EPL(                                          \
+      const unsigned short*, const unsigned short*, const unsigned short*, const float*, \
+      const float*, unsigned short*, float*, int, int);
+GA_DECL_LN(4) GA_DECL_LN(8) GA_DECL_LN(12) GA_DECL_LN(16)
+#define GA_DECL_GELU(EPL)                                                 \
+  extern "C" __global__ void k_biasgelu_bwd_##EPL(                        \
+      const unsigned short*, const unsigned short*, const unsigned short*, \
+      unsigned short*, float*, int, int);
+GA_DECL_GELU(16) GA_DECL_GELU(24) GA_DECL_GELU(32) GA_DECL_GELU(48) GA_DECL_GELU(64)
 extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned short*,
                                           unsigned short*, long long, int);
-extern "C" __global__ void k_biasgelu_bwd(const unsigned short*, const unsigned short*,
-                                          const unsigned short*, unsigned short*,
-                                          float*, int, int);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
 
@@ -154,7 +158,12 @@ std::vector<at::Tensor> addln_fwd(at::Tensor x, c10::optional<at::Tensor> res,
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   int blocks = std::min((R + 3) / 4, 2048);
-  hipLaunchKernelGGL(k_addln_fwd, dim3(blocks), dim3(256), 0, stream,
+  void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
+               const unsigned short*, const unsigned short*, unsigned short*,
+               unsigned short*, float*, float*, int, int, float) =
+      H == 256 ? k_addln_fwd_4 : H == 512 ? k_addln_fwd_8
+      : H == 768 ? k_addln_fwd_12 : k_addln_fwd_16;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, stream,
                      bfp(x), res ? bfp(*res) : nullptr, bias ? bfp(*bias) : nullptr,
                      bfp(gamma), bfp(beta), bfp_mut(y), bfp_mut(h),
                      mean.data_ptr<float>(), rstd.data_ptr<float>(), R, H, (float)eps);
@@ -168,11 +177,15 @@ std::vector<at::Tensor> addln_bwd(at::Tensor dy, at::Tensor h, at::Tensor gamma,
   check_bf16_2d(dy, "dy", H);
   const int R = (int)(dy.numel() / H);
   auto dh = at::empty_like(dy);
-  int blocks = std::max(1, std::min((R + 3) / 4, 128));
+  int blocks = std::max(1, std::min((R + 3) / 4, 256));
   auto partials = at::empty({blocks, 3, H}, dy.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   size_t lds = (size_t)4 * 3 * H * sizeof(float);
-  hipLaunchKernelGGL(k_addln_bwd, dim3(blocks), dim3(256), lds, stream,
+  void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
+               const float*, const float*, unsigned short*, float*, int, int) =
+      H == 256 ? k_addln_bwd_4 : H == 512 ? k_addln_bwd_8
+      : H == 768 ? k_addln_bwd_12 : k_addln_bwd_16;
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), lds, stream,
                      bfp(dy), bfp(h), bfp(gamma), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(), bfp_mut(dh),
                      partials.data_ptr<float>(), R, H);
@@ -199,11 +212,18 @@ std::vector<at::Tensor> biasgelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bia
   check_bf16_2d(dy, "dy", H);
   const int R = (int)(dy.numel() / H);
   auto dx = at::empty_like(dy);
-  int blocks = std::max(1, std::min((R + 3) / 4, 128));
+  int blocks = std::max(1, std::min((R + 3) / 4, 256));
   auto partials = at::empty({blocks, H}, dy.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   size_t lds = (size_t)4 * H * sizeof(float);
-  hipLaunchKernelGGL(k_biasgelu_bwd, dim3(blocks), dim3(256), lds, stream,
+  void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
+               unsigned short*, float*, int, int) =
+      H == 1024 ? k_biasgelu_bwd_16 : H == 1536 ? k_biasgelu_bwd_24
+      : H == 2048 ? k_biasgelu_bwd_32 : H == 3072 ? k_biasgelu_bwd_48
+                                                  : k_biasgelu_bwd_64;
+  TORCH_CHECK(H == 1024 || H == 1536 || H == 2048 || H == 3072 || H == 4096,
+              "gelu hidden must be one of 1024/1536/2048/3072/4096");
+  hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), lds, stream,
                      bfp(dy), bfp(x), bfp(bias), bfp_mut(dx),
                      partials.data_ptr<float>(), R, H);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_bwd launch failed");
@@ -221,7 +241,8 @@ void colreduce_acc(at::Tensor partials, at::Tensor dest0,
   const int n2 = dest2 ? (int)dest2->numel() : 0;
   TORCH_CHECK(n0 + n1 + n2 == C, "dest sizes must sum to partial columns");
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = std::min((C + 255) / 256, 256);
+  const int nch = (NB + 7) / 8;
+  int blocks = (int)std::min<long long>(((long long)C * nch + 255) / 256, 2048);
   hipLaunchKernelGGL(k_colreduce_acc, dim3(blocks), dim3(256), 0, stream,
                      partials.data_ptr<float>(), NB, C,
                      dest0.data_ptr<float>(), n0,
