@@ -57,10 +57,12 @@ class AccumEngine:
         process_group=None,
         allreduce_bucket_mb: int = 64,
         backend: str = "auto",
+        direct_names=(),
     ):
         if gradient_accumulation_multiplier < 1:
             raise ValueError("gradient_accumulation_multiplier must be >= 1")
-        self.state = FlatState(list(named_params), exclude_from_weight_decay)
+        self.state = FlatState(list(named_params), exclude_from_weight_decay,
+                               direct_names)
         self.K = int(gradient_accumulation_multiplier)
         self.init_lr = float(init_lr)
         self.num_train_steps = int(num_train_steps)
@@ -115,10 +117,19 @@ class AccumEngine:
     # ---- the per-micro-step body (the thing bench.py hipGraph-captures) ----
     def accumulate(self) -> None:
         st = self.state
-        if self._hip is not None:
-            self._hip.accumulate(st.accum, st.grads)
+        lo, hi = st.layout.grad_lo, st.layout.grad_hi
+        if hi <= lo:
+            return  # every param accumulates directly (ops/fused.py)
+        if lo == 0 and hi == st.layout.total:
+            accum, grads = st.accum, st.grads
         else:
-            eager_ops.accumulate(st.accum, st.grads)
+            # direct-accum params (fused modules) bypass .grad entirely --
+            # K1 only touches the contiguous grad-path region
+            accum, grads = st.accum[lo:hi], st.grads[lo:hi]
+        if self._hip is not None:
+            self._hip.accumulate(accum, grads)
+        else:
+            eager_ops.accumulate(accum, grads)
 
     def _allreduce_accum(self) -> None:
         import torch.distributed as dist

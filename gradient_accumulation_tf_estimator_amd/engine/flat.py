@@ -68,6 +68,8 @@ class FlatLayout:
     slices: List[ParamSlice]
     total: int  # total padded elements
     decay_boundary: int  # elements < boundary get weight decay
+    grad_lo: int = 0  # contiguous region whose grads flow through .grad
+    grad_hi: int = 0
 
     @property
     def names(self) -> List[str]:
@@ -77,23 +79,45 @@ class FlatLayout:
 def build_layout(
     named_params: Sequence[Tuple[str, torch.Tensor]],
     exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+    direct_names: Sequence[str] = (),
 ) -> FlatLayout:
-    """Order params [decay..., no-decay...] and assign aligned flat offsets."""
+    """Order params and assign aligned flat offsets.
+
+    Grouping: [decay-direct | decay-grad | nodecay-grad | nodecay-direct].
+    ``direct_names`` are params whose grads bypass ``.grad`` (fused modules
+    write them straight into the accum buffer, ops/fused.py), so the K1
+    accumulate kernel only has to touch the contiguous middle
+    ``[grad_lo, grad_hi)`` region; the decay boundary stays a single index.
+    """
+    direct = set(direct_names)
     decay_group = [(n, p) for n, p in named_params if use_weight_decay(n, exclude_from_weight_decay)]
     nodecay_group = [(n, p) for n, p in named_params if not use_weight_decay(n, exclude_from_weight_decay)]
+    ordered = (
+        [(n, p, True) for n, p in decay_group if n in direct]
+        + [(n, p, True) for n, p in decay_group if n not in direct]
+        + [(n, p, False) for n, p in nodecay_group if n not in direct]
+        + [(n, p, False) for n, p in nodecay_group if n in direct]
+    )
     slices: List[ParamSlice] = []
     off = 0
-    for group, decay in ((decay_group, True), (nodecay_group, False)):
-        for name, p in group:
-            n = p.numel()
-            padded = _round_up(n)
-            slices.append(ParamSlice(name, p.shape, n, off, padded, decay))
-            off += padded
+    decay_boundary = 0
+    grad_lo = grad_hi = None
+    for name, p, decay in ordered:
+        n = p.numel()
+        padded = _round_up(n)
+        slices.append(ParamSlice(name, p.shape, n, off, padded, decay))
+        if name not in direct:
+            if grad_lo is None:
+                grad_lo = off
+            grad_hi = off + padded
+        off += padded
         if decay:
             decay_boundary = off
-    if not decay_group:
-        decay_boundary = 0
-    return FlatLayout(slices=slices, total=off, decay_boundary=decay_boundary)
+    if grad_lo is None:
+        grad_lo = grad_hi = 0
+    lay = FlatLayout(slices=slices, total=off, decay_boundary=decay_boundary)
+    lay.grad_lo, lay.grad_hi = grad_lo, grad_hi
+    return lay
 
 
 class FlatState:
@@ -117,6 +141,7 @@ class FlatState:
         self,
         named_params: Sequence[Tuple[str, torch.Tensor]],
         exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+        direct_names: Sequence[str] = (),
     ):
         named_params = [(n, p) for n, p in named_params if p.requires_grad]
         if not named_params:
@@ -132,7 +157,8 @@ class FlatState:
         if self.dtype not in (torch.float32, torch.bfloat16):
             raise ValueError(f"unsupported param dtype {self.dtype}")
 
-        self.layout = build_layout(named_params, exclude_from_weight_decay)
+        self.layout = build_layout(named_params, exclude_from_weight_decay,
+                                   direct_names)
         N = self.layout.total
         dev = self.device
 

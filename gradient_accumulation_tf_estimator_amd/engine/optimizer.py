@@ -80,6 +80,20 @@ def create_optimizer(
         named = list(model.named_parameters())
     else:
         named = list(model_or_params)
+    # fused modules (ops/fused.py) accumulate their param grads directly into
+    # the flat fp32 accum buffer on the GPU path; K1 then skips their region
+    direct = ()
+    will_bind = False
+    if model is not None and backend != "eager":
+        try:
+            p0 = next(model.parameters())
+            will_bind = p0.is_cuda and p0.dtype == torch.bfloat16
+        except StopIteration:
+            pass
+    if will_bind:
+        from ..ops.fused import direct_param_names
+
+        direct = direct_param_names(model)
     engine = AccumEngine(
         named,
         init_lr=init_lr,
@@ -95,10 +109,9 @@ def create_optimizer(
         strict_reference_semantics=strict_reference_semantics,
         process_group=process_group,
         backend=backend,
+        direct_names=direct,
     )
-    if model is not None and engine.backend == "hip":
-        # fused modules (ops/fused.py) accumulate their param grads directly
-        # into the flat fp32 accum buffer, bypassing .grad
+    if will_bind and engine.backend == "hip":
         from ..ops.fused import bind_direct_grad
 
         bind_direct_grad(model, engine)

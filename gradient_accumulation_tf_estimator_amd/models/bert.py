@@ -25,7 +25,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.fused import FusedAddLayerNorm, FusedBiasGelu
+from ..ops.fused import DirectEmbedding, DirectLinear, FusedAddLayerNorm, FusedBiasGelu
 
 
 @dataclass
@@ -63,9 +63,10 @@ CONFIGS = {"bert-small": bert_small, "bert-base": bert_base, "bert-large": bert_
 class BertEmbeddings(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
-        self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
-        self.position_embeddings = nn.Embedding(cfg.max_position_embeddings, cfg.hidden_size)
-        self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size, cfg.hidden_size)
+        Emb = DirectEmbedding if cfg.fused else nn.Embedding
+        self.word_embeddings = Emb(cfg.vocab_size, cfg.hidden_size)
+        self.position_embeddings = Emb(cfg.max_position_embeddings, cfg.hidden_size)
+        self.token_type_embeddings = Emb(cfg.type_vocab_size, cfg.hidden_size)
         self.LayerNorm = (FusedAddLayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
                           if cfg.fused else nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps))
         self.dropout = nn.Dropout(cfg.dropout)
@@ -94,7 +95,8 @@ class BertSelfAttention(nn.Module):
         super().__init__()
         self.num_heads = cfg.num_heads
         self.head_dim = cfg.hidden_size // cfg.num_heads
-        self.qkv = nn.Linear(cfg.hidden_size, 3 * cfg.hidden_size)
+        Lin = DirectLinear if cfg.fused else nn.Linear
+        self.qkv = Lin(cfg.hidden_size, 3 * cfg.hidden_size)
         self.dropout_p = cfg.dropout
 
     def forward(self, x, attn_mask=None):
@@ -122,10 +124,11 @@ class BertLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.fused = cfg.fused
+        Lin = DirectLinear if cfg.fused else nn.Linear
         self.attention = BertSelfAttention(cfg)
-        self.attention_output = nn.Linear(cfg.hidden_size, cfg.hidden_size, bias=not cfg.fused)
-        self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=not cfg.fused)
-        self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=not cfg.fused)
+        self.attention_output = Lin(cfg.hidden_size, cfg.hidden_size, bias=not cfg.fused)
+        self.intermediate = Lin(cfg.hidden_size, cfg.intermediate_size, bias=not cfg.fused)
+        self.output = Lin(cfg.intermediate_size, cfg.hidden_size, bias=not cfg.fused)
         self.dropout = nn.Dropout(cfg.dropout)
         if cfg.fused:
             self.attention_LayerNorm = FusedAddLayerNorm(
@@ -154,13 +157,14 @@ class BertModel(nn.Module):
         self.cfg = cfg
         self.embeddings = BertEmbeddings(cfg)
         self.encoder = nn.ModuleList(BertLayer(cfg) for _ in range(cfg.num_layers))
-        self.pooler = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.pooler = (DirectLinear if cfg.fused else nn.Linear)(
+            cfg.hidden_size, cfg.hidden_size)
         self.apply(self._init)
 
     def _init(self, m):
-        if isinstance(m, (nn.Linear, nn.Embedding)):
+        if isinstance(m, (nn.Linear, nn.Embedding, DirectLinear, DirectEmbedding)):
             nn.init.normal_(m.weight, std=self.cfg.initializer_range)
-            if isinstance(m, nn.Linear) and m.bias is not None:
+            if getattr(m, "bias", None) is not None and not isinstance(m, nn.LayerNorm):
                 nn.init.zeros_(m.bias)
         elif isinstance(m, nn.LayerNorm):
             nn.init.ones_(m.weight)
@@ -185,6 +189,8 @@ class BertForSequenceClassification(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.bert = BertModel(cfg)
+        # classifier output dim (num_labels=2) is far below the GEMM kernels'
+        # alignment; plain nn.Linear keeps it on the stock torch path
         self.classifier = nn.Linear(cfg.hidden_size, cfg.num_labels)
         nn.init.normal_(self.classifier.weight, std=cfg.initializer_range)
         nn.init.zeros_(self.classifier.bias)

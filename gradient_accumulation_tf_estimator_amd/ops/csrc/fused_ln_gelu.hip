@@ -202,14 +202,23 @@ __device__ __forceinline__ void addln_bwd_body(
 #define GELU_C0 0.7978845608028654f
 #define GELU_C1 0.044715f
 
+// tanh via the native v_exp_f32 path (__expf): ocml tanhf is a branchy
+// precise routine that dominated the gelu kernels' time. |u| clamped to 9
+// so exp never overflows; bf16 output cannot see the ~2-ulp fast-exp error.
+static inline __device__ float fast_tanh(float u) {
+  u = fminf(fmaxf(u, -9.f), 9.f);
+  const float e = __expf(2.f * u);
+  return (e - 1.f) / (e + 1.f);
+}
+
 static inline __device__ float gelu_fwd1(float h) {
   const float u = GELU_C0 * fmaf(GELU_C1 * h * h, h, h);
-  return 0.5f * h * (1.f + tanhf(u));
+  return 0.5f * h * (1.f + fast_tanh(u));
 }
 
 static inline __device__ float gelu_bwd1(float h, float dy) {
   const float u = GELU_C0 * fmaf(GELU_C1 * h * h, h, h);
-  const float t = tanhf(u);
+  const float t = fast_tanh(u);
   const float du = GELU_C0 * fmaf(3.f * GELU_C1 * h, h, 1.f);
   return dy * (0.5f * (1.f + t) + 0.5f * h * (1.f - t * t) * du);
 }
@@ -350,3 +359,27 @@ GA_GELU_INST(24)
 GA_GELU_INST(32)
 GA_GELU_INST(48)
 GA_GELU_INST(64)
+
+
+// ---------------- embedding backward -> flat fp32 accum ----------------
+// Scatter-add bf16 dy rows into the embedding weight's accum slice by token
+// id. Replaces the dense zero-init + scatter + AccumulateGrad add + K1
+// coverage of a [vocab,H] gradient (the vocab table dominates BERT-Small's
+// flat buffer) with R*H fp32 atomics (R = tokens in the micro-batch).
+extern "C" __global__ void k_embgrad_acc(
+    const unsigned short* __restrict__ dy, const long long* __restrict__ ids,
+    float* __restrict__ accum, long long R, int H) {
+  const long long total = R * (H >> 2);
+  const int h4 = H >> 2;
+  for (long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long long)gridDim.x * blockDim.x) {
+    const long long row = t / h4;
+    const int c = (int)(t % h4) << 2;
+    const ushort4 v = *(const ushort4*)(dy + row * H + c);
+    float* dst = accum + ids[row] * (long long)H + c;
+    atomicAdd(dst + 0, bf2f(v.x));
+    atomicAdd(dst + 1, bf2f(v.y));
+    atomicAdd(dst + 2, bf2f(v.z));
+    atomicAdd(dst + 3, bf2f(v.w));
+  }
+}

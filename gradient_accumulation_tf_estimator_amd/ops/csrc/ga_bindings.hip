@@ -38,6 +38,8 @@ extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned 
                                           unsigned short*, long long, int);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
+extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*,
+                                         float*, long long, int);
 
 namespace {
 
@@ -251,9 +253,30 @@ void colreduce_acc(at::Tensor partials, at::Tensor dest0,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_colreduce_acc launch failed");
 }
 
+void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t H) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
+              "dy must be contiguous bf16");
+  TORCH_CHECK(ids.is_cuda() && ids.is_contiguous() && ids.scalar_type() == at::kLong,
+              "ids must be contiguous int64");
+  TORCH_CHECK(accum_slice.scalar_type() == at::kFloat && accum_slice.is_contiguous());
+  TORCH_CHECK(H % 4 == 0 && dy.numel() == ids.numel() * H, "shape mismatch");
+  const long long R = ids.numel();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  long long total = R * (H / 4);
+  int blocks = (int)std::min<long long>((total + 255) / 256, 2048);
+  hipLaunchKernelGGL(k_embgrad_acc, dim3(blocks), dim3(256), 0, stream,
+                     (const unsigned short*)dy.data_ptr(),
+                     (const long long*)ids.data_ptr<int64_t>(),
+                     accum_slice.data_ptr<float>(), R, (int)H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_embgrad_acc launch failed");
+}
+
 }  // namespace
 
+void register_blas_acc(pybind11::module_& mod);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  register_blas_acc(mod);
   mod.def("accumulate", &accumulate, "accum += grad (fp32 upcast); grad = 0");
   mod.def("sqnorm", &sqnorm, "out[0] = sum(accum^2)");
   mod.def("fused_apply", &fused_apply,
@@ -264,4 +287,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
   mod.def("colreduce_acc", &colreduce_acc,
           "reduce partials over blocks, ADD into flat fp32 accum slices");
+  mod.def("embgrad_acc", &embgrad_acc,
+          "scatter-add embedding grads into the flat fp32 accum slice");
 }

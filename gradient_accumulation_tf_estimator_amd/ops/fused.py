@@ -99,6 +99,11 @@ class FusedAddLayerNorm(nn.Module):
             return _AddLayerNormFn.apply(
                 x, residual, self.weight, self.bias, self.proj_bias, self.eps, self
             )
+        if self._accum_views is not None:
+            raise RuntimeError(
+                "FusedAddLayerNorm is bound to an engine (direct-accum grads) "
+                "but received a non-bf16/non-GPU input -- the torch fallback "
+                "would silently drop its parameter gradients")
         h = x
         if residual is not None:
             h = h + residual
@@ -143,7 +148,117 @@ class FusedBiasGelu(nn.Module):
     def forward(self, x):
         if _use_hip(x):
             return _BiasGeluFn.apply(x, self.bias, self)
+        if self._accum_view is not None:
+            raise RuntimeError("bound FusedBiasGelu got non-bf16/non-GPU input")
         return F.gelu(x + self.bias, approximate="tanh")
+
+
+class _DirectLinearFn(torch.autograd.Function):
+    """Linear whose weight gradient is a hipBLASLt bf16x bf16 -> fp32 GEMM
+    with beta=1 straight into the flat accum slice (no .grad, no K1 pass)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, module):
+        ctx.save_for_backward(x, weight)
+        ctx.module = module
+        ctx.has_bias = bias is not None
+        return F.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dy.matmul(weight)
+        x2d = x.reshape(-1, x.shape[-1])
+        dy2d = dy.reshape(-1, dy.shape[-1])
+        hip.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
+        db = dy2d.sum(0) if ctx.has_bias else None
+        return dx, None, db, None
+
+
+class DirectLinear(nn.Module):
+    """nn.Linear drop-in; when bound to an engine on GPU, the weight grad
+    accumulates directly into the flat fp32 buffer (bias grad, if any, stays
+    on the .grad path -- it is tiny)."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = True):
+        super().__init__()
+        self.in_features, self.out_features = in_features, out_features
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        if bias:
+            self.bias = nn.Parameter(torch.zeros(out_features))
+        else:
+            self.register_parameter("bias", None)
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+        self._accum_view_w = None
+
+    def forward(self, x):
+        if self._accum_view_w is not None:
+            if not _use_hip(x):
+                raise RuntimeError("bound DirectLinear got non-bf16/non-GPU input")
+            return _DirectLinearFn.apply(x, self.weight, self.bias, self)
+        return F.linear(x, self.weight, self.bias)
+
+
+class _DirectEmbeddingFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, weight, module):
+        ctx.save_for_backward(ids)
+        ctx.module = module
+        ctx.H = weight.shape[1]
+        return F.embedding(ids, weight)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        (ids,) = ctx.saved_tensors
+        hip.embgrad_acc(dy.contiguous().reshape(-1, ctx.H), ids.reshape(-1),
+                        ctx.module._accum_view_w, ctx.H)
+        return None, None, None
+
+
+class DirectEmbedding(nn.Module):
+    """nn.Embedding drop-in; bound backward scatter-adds bf16 rows into the
+    flat fp32 accum slice by token id -- no dense [vocab,H] grad buffer, no
+    fill, no AccumulateGrad add, no K1 coverage of the vocab table."""
+
+    def __init__(self, num_embeddings: int, embedding_dim: int):
+        super().__init__()
+        self.num_embeddings, self.embedding_dim = num_embeddings, embedding_dim
+        self.weight = nn.Parameter(torch.empty(num_embeddings, embedding_dim))
+        nn.init.normal_(self.weight)
+        self._accum_view_w = None
+
+    def forward(self, ids):
+        if self._accum_view_w is not None:
+            if not (self.weight.is_cuda and self.weight.dtype == torch.bfloat16):
+                raise RuntimeError("bound DirectEmbedding got non-bf16 weights")
+            return _DirectEmbeddingFn.apply(ids, self.weight, self)
+        return F.embedding(ids, self.weight)
+
+
+def direct_param_names(model: nn.Module):
+    """Names of params whose grads will bypass .grad when bound on GPU --
+    passed to FlatState so K1 can skip their (contiguous) flat region."""
+    names = []
+    by_mod = {id(m): n for n, m in model.named_modules()}
+
+    def pname(mod, attr):
+        prefix = by_mod[id(mod)]
+        return f"{prefix}.{attr}" if prefix else attr
+
+    for mod in model.modules():
+        if isinstance(mod, FusedAddLayerNorm):
+            names.append(pname(mod, "weight"))
+            names.append(pname(mod, "bias"))
+            if mod.proj_bias is not None:
+                names.append(pname(mod, "proj_bias"))
+        elif isinstance(mod, FusedBiasGelu):
+            names.append(pname(mod, "bias"))
+        elif isinstance(mod, (DirectLinear, DirectEmbedding)):
+            names.append(pname(mod, "weight"))
+    return names
 
 
 def bind_direct_grad(model: nn.Module, engine) -> int:
@@ -164,5 +279,13 @@ def bind_direct_grad(model: nn.Module, engine) -> int:
             n += 1
         elif isinstance(mod, FusedBiasGelu):
             mod._accum_view = engine.state.accum_view(mod.bias)
+            n += 1
+        elif isinstance(mod, DirectLinear):
+            mod._accum_view_w = engine.state.accum_view(mod.weight).view(
+                mod.out_features, mod.in_features)
+            n += 1
+        elif isinstance(mod, DirectEmbedding):
+            mod._accum_view_w = engine.state.accum_view(mod.weight).view(
+                mod.num_embeddings, mod.embedding_dim)
             n += 1
     return n

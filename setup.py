@@ -26,8 +26,10 @@ setup(
             sources=[
                 os.path.join(CSRC, "ga_kernels.hip"),
                 os.path.join(CSRC, "fused_ln_gelu.hip"),
+                os.path.join(CSRC, "blas_acc.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],

@@ -86,10 +86,19 @@ def test_fused_biasgelu_forward_backward(H):
 
 
 def test_direct_accum_equals_grad_path():
-    """Bound modules write grads into engine accum == what .grad would get."""
+    """Bound modules write grads into engine accum == what .grad would get.
+
+    Covers FusedAddLayerNorm (colreduce path), DirectLinear (hipBLASLt
+    beta=1 wgrad), and DirectEmbedding (scatter-add) against the unbound
+    .grad + K1 path on an identical twin network."""
     import torch.nn as nn
     from gradient_accumulation_tf_estimator_amd.engine.accum import AccumEngine
-    from gradient_accumulation_tf_estimator_amd.ops.fused import bind_direct_grad
+    from gradient_accumulation_tf_estimator_amd.ops.fused import (
+        DirectEmbedding,
+        DirectLinear,
+        bind_direct_grad,
+        direct_param_names,
+    )
 
     H = 512
     torch.manual_seed(0)
@@ -97,10 +106,12 @@ def test_direct_accum_equals_grad_path():
     class Net(nn.Module):
         def __init__(self):
             super().__init__()
-            self.lin = nn.Linear(H, H, bias=False)
+            self.emb = DirectEmbedding(64, H)
+            self.lin = DirectLinear(H, H, bias=True)
             self.LayerNorm = FusedAddLayerNorm(H, proj_bias=True)
 
-        def forward(self, x):
+        def forward(self, ids):
+            x = self.emb(ids)
             return self.LayerNorm(self.lin(x), residual=x)
 
     def make():
@@ -110,23 +121,31 @@ def test_direct_accum_equals_grad_path():
     netA, netB = make(), make()
     kw = dict(init_lr=0.0, num_train_steps=100, num_warmup_steps=0,
               gradient_accumulation_multiplier=4, clip_norm=None)
-    engA = AccumEngine(list(netA.named_parameters()), backend="hip", **kw)
+    engA = AccumEngine(list(netA.named_parameters()), backend="hip",
+                       direct_names=direct_param_names(netA), **kw)
     engB = AccumEngine(list(netB.named_parameters()), backend="hip", **kw)
-    assert bind_direct_grad(netA, engA) == 1  # A: direct-accum path
+    assert bind_direct_grad(netA, engA) == 3  # A: direct-accum path
     # B: unbound -> grads go through .grad + K1
 
     for i in range(2):
         torch.manual_seed(10 + i)
-        x = torch.randn(32, H, device="cuda").bfloat16()
+        ids = torch.randint(0, 64, (4, 32), device="cuda")
         for net, eng in ((netA, engA), (netB, engB)):
-            loss = (net(x).float() ** 2).mean()
+            loss = (net(ids).float() ** 2).mean()
             loss.backward()
             eng.accumulate()
     torch.cuda.synchronize()
-    a = engA.state.accum.cpu().numpy()
-    b = engB.state.accum.cpu().numpy()
-    # direct path accumulates in fp32 (more precise than B's bf16 .grad hop):
-    # compare at bf16-quantization tolerance
-    np.testing.assert_allclose(a, b, rtol=2e-2, atol=2e-3)
-    # and it must actually have written something
-    assert np.abs(a).sum() > 0
+    # compare per-parameter via each engine's own layout (layouts differ:
+    # direct params are grouped to the edges of the flat buffer)
+    stA = engA.state
+    stB = engB.state
+    by_name_B = {s.name: s for s in stB.layout.slices}
+    for sA in stA.layout.slices:
+        sB = by_name_B[sA.name]
+        a = stA.accum[sA.offset : sA.offset + sA.numel].cpu().numpy()
+        b = stB.accum[sB.offset : sB.offset + sB.numel].cpu().numpy()
+        # direct path accumulates in fp32 (more precise than B's bf16 .grad
+        # hop): compare at bf16-quantization tolerance
+        np.testing.assert_allclose(a, b, rtol=2e-2, atol=3e-3,
+                                   err_msg=f"accum mismatch for {sA.name}")
+        assert (abs(a).sum() > 0) or "bias" in sA.name
