@@ -170,7 +170,8 @@ class _DirectLinearFn(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
         dx = dy.matmul(weight)
-        x2d = x.reshape(-1, x.shape[-1])
+        # e.g. the pooler's CLS slice is a non-contiguous view
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
         dy2d = dy.reshape(-1, dy.shape[-1])
         hip.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
         db = dy2d.sum(0) if ctx.has_bias else None
