@@ -354,6 +354,8 @@ GA_LN_INST(16)
     biasgelu_bwd_body<EPL>(dy, x, bias, dx_out, partials, R, H);               \
   }
 
+GA_GELU_INST(8)
+GA_GELU_INST(12)
 GA_GELU_INST(16)
 GA_GELU_INST(24)
 GA_GELU_INST(32)

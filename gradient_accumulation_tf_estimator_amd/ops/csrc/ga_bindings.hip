@@ -33,7 +33,7 @@ GA_DECL_LN(4) GA_DECL_LN(8) GA_DECL_LN(12) GA_DECL_LN(16)
   extern "C" __global__ void k_biasgelu_bwd_##EPL(                        \
       const unsigned short*, const unsigned short*, const unsigned short*, \
       unsigned short*, float*, int, int);
-GA_DECL_GELU(16) GA_DECL_GELU(24) GA_DECL_GELU(32) GA_DECL_GELU(48) GA_DECL_GELU(64)
+GA_DECL_GELU(8) GA_DECL_GELU(12) GA_DECL_GELU(16) GA_DECL_GELU(24) GA_DECL_GELU(32) GA_DECL_GELU(48) GA_DECL_GELU(64)
 extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned short*,
                                           unsigned short*, long long, int);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
@@ -220,11 +220,13 @@ std::vector<at::Tensor> biasgelu_bwd(at::Tensor dy, at::Tensor x, at::Tensor bia
   size_t lds = (size_t)4 * H * sizeof(float);
   void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
                unsigned short*, float*, int, int) =
-      H == 1024 ? k_biasgelu_bwd_16 : H == 1536 ? k_biasgelu_bwd_24
+      H == 512 ? k_biasgelu_bwd_8 : H == 768 ? k_biasgelu_bwd_12
+      : H == 1024 ? k_biasgelu_bwd_16 : H == 1536 ? k_biasgelu_bwd_24
       : H == 2048 ? k_biasgelu_bwd_32 : H == 3072 ? k_biasgelu_bwd_48
                                                   : k_biasgelu_bwd_64;
-  TORCH_CHECK(H == 1024 || H == 1536 || H == 2048 || H == 3072 || H == 4096,
-              "gelu hidden must be one of 1024/1536/2048/3072/4096");
+  TORCH_CHECK(H == 512 || H == 768 || H == 1024 || H == 1536 || H == 2048 ||
+                  H == 3072 || H == 4096,
+              "gelu hidden must be one of 512/768/1024/1536/2048/3072/4096");
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), lds, stream,
                      bfp(dy), bfp(x), bfp(bias), bfp_mut(dx),
                      partials.data_ptr<float>(), R, H);
