@@ -17,6 +17,7 @@
 #include <hipblaslt/hipblaslt.h>
 
 #include <map>
+#include <vector>
 #include <mutex>
 #include <tuple>
 
@@ -27,34 +28,34 @@
                 " at " #expr);                                           \
   } while (0)
 
-namespace {
-
-struct LtPlan {
-  hipblasLtMatmulDesc_t desc;
-  hipblasLtMatrixLayout_t la, lb, lc;
-  hipblasLtMatmulAlgo_t algo;
-  bool has_algo;
-};
-
-hipblasLtHandle_t lt_handle() {
+hipblasLtHandle_t ga_lt_handle() {
   static hipblasLtHandle_t h = [] {
     hipblasLtHandle_t hh;
-    HIPBLASLT_CHECK(hipblasLtCreate(&hh));
+    hipblasStatus_t st = hipblasLtCreate(&hh);
+    TORCH_CHECK(st == HIPBLAS_STATUS_SUCCESS, "hipblasLtCreate failed");
     return hh;
   }();
   return h;
 }
 
-constexpr size_t kWorkspaceBytes = 64ull << 20;
-
-void* lt_workspace() {
+void* ga_lt_workspace() {
   static void* ws = [] {
     void* p = nullptr;
-    (void)hipMalloc(&p, kWorkspaceBytes);
+    (void)hipMalloc(&p, 64ull << 20);
     return p;
   }();
   return ws;
 }
+
+namespace {
+
+struct LtPlan {
+  hipblasLtMatmulDesc_t desc;
+  hipblasLtMatrixLayout_t la, lb, lc;
+  std::vector<hipblasLtMatmulHeuristicResult_t> algos;
+};
+
+constexpr size_t kWorkspaceBytes = 64ull << 20;
 
 LtPlan& plan_for(int64_t K, int64_t N, int64_t R) {
   static std::map<std::tuple<int64_t, int64_t, int64_t>, LtPlan> cache;
@@ -83,18 +84,23 @@ LtPlan& plan_for(int64_t K, int64_t N, int64_t R) {
   size_t ws = kWorkspaceBytes;
   HIPBLASLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
       pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws)));
-  hipblasLtMatmulHeuristicResult_t heur[4];
+  hipblasLtMatmulHeuristicResult_t heur[16];
   int found = 0;
   hipblasStatus_t st = hipblasLtMatmulAlgoGetHeuristic(
-      lt_handle(), p.desc, p.la, p.lb, p.lc, p.lc, pref, 4, heur, &found);
+      ga_lt_handle(), p.desc, p.la, p.lb, p.lc, p.lc, pref, 16, heur, &found);
   hipblasLtMatmulPreferenceDestroy(pref);
-  p.has_algo = (st == HIPBLAS_STATUS_SUCCESS && found > 0);
-  if (p.has_algo) p.algo = heur[0].algo;
+  TORCH_CHECK(st == HIPBLAS_STATUS_SUCCESS && found > 0, "no wgrad algo");
+  p.algos.assign(heur, heur + found);
   return cache.emplace(key, p).first->second;
 }
 
+int64_t wgrad_algo_count(int64_t K, int64_t N, int64_t R) {
+  return (int64_t)plan_for(K, N, R).algos.size();
+}
+
 // dW accum_slice[N*K fp32] += dy[R,N]^T @ x[R,K]
-void wgrad_acc(at::Tensor x, at::Tensor dy, at::Tensor accum_slice) {
+void wgrad_acc(at::Tensor x, at::Tensor dy, at::Tensor accum_slice,
+               int64_t algo_idx) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.scalar_type() == at::kBFloat16,
               "x must be contiguous bf16");
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
@@ -108,18 +114,22 @@ void wgrad_acc(at::Tensor x, at::Tensor dy, at::Tensor accum_slice) {
   TORCH_CHECK(accum_slice.numel() == N * K, "accum slice size mismatch");
 
   auto& p = plan_for(K, N, R);
-  TORCH_CHECK(p.has_algo, "no hipblaslt algo for wgrad shape");
   const float alpha = 1.f, beta = 1.f;
   auto stream = c10::hip::getCurrentHIPStream().stream();
+  int idx = (int)std::min<int64_t>(std::max<int64_t>(algo_idx, 0),
+                                   (int64_t)p.algos.size() - 1);
   HIPBLASLT_CHECK(hipblasLtMatmul(
-      lt_handle(), p.desc, &alpha, x.data_ptr(), p.la, dy.data_ptr(), p.lb,
+      ga_lt_handle(), p.desc, &alpha, x.data_ptr(), p.la, dy.data_ptr(), p.lb,
       &beta, accum_slice.data_ptr(), p.lc, accum_slice.data_ptr(), p.lc,
-      p.has_algo ? &p.algo : nullptr, lt_workspace(), kWorkspaceBytes, stream));
+      &p.algos[idx].algo, ga_lt_workspace(), kWorkspaceBytes, stream));
 }
 
 }  // namespace
 
 void register_blas_acc(pybind11::module_& mod) {
   mod.def("wgrad_acc", &wgrad_acc,
-          "accum_slice[N,K] += dy[R,N]^T @ x[R,K] (bf16 in, fp32 accumulate)");
+          "accum_slice[N,K] += dy[R,N]^T @ x[R,K] (bf16 in, fp32 accumulate)",
+          pybind11::arg("x"), pybind11::arg("dy"), pybind11::arg("accum_slice"),
+          pybind11::arg("algo_idx") = 0);
+  mod.def("wgrad_algo_count", &wgrad_algo_count);
 }

@@ -159,21 +159,27 @@ class _DirectLinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, module):
+        from . import gemm
+
+        # e.g. the pooler's CLS slice is a non-contiguous view
+        x = x.contiguous()
         ctx.save_for_backward(x, weight)
         ctx.module = module
         ctx.has_bias = bias is not None
-        return F.linear(x, weight, bias)
+        x2d = x.reshape(-1, x.shape[-1])
+        y = gemm.linear_fwd(x2d, weight, bias)
+        return y.reshape(*x.shape[:-1], weight.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        hip = require_hip()
+        from . import gemm
+
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
-        dx = dy.matmul(weight)
-        # e.g. the pooler's CLS slice is a non-contiguous view
-        x2d = x.reshape(-1, x.shape[-1]).contiguous()
         dy2d = dy.reshape(-1, dy.shape[-1])
-        hip.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
+        dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
+        x2d = x.reshape(-1, x.shape[-1])
+        gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
         db = dy2d.sum(0) if ctx.has_bias else None
         return dx, None, db, None
 

@@ -149,3 +149,34 @@ def test_direct_accum_equals_grad_path():
         np.testing.assert_allclose(a, b, rtol=2e-2, atol=3e-3,
                                    err_msg=f"accum mismatch for {sA.name}")
         assert (abs(a).sum() > 0) or "bias" in sA.name
+
+
+def test_lt_gemm_matches_torch():
+    """Autotuned hipBLASLt fwd/dgrad/wgrad vs torch matmul references."""
+    from gradient_accumulation_tf_estimator_amd.ops import gemm
+
+    torch.manual_seed(0)
+    R, N, K = 1024, 1536, 512
+    x = torch.randn(R, K, device="cuda").bfloat16()
+    w = (torch.randn(N, K, device="cuda") * 0.05).bfloat16()
+    b = torch.randn(N, device="cuda").bfloat16()
+    dy = torch.randn(R, N, device="cuda").bfloat16()
+
+    y = gemm.linear_fwd(x, w, b)
+    yref = F.linear(x, w, b)
+    np.testing.assert_allclose(y.float().cpu(), yref.float().cpu(),
+                               rtol=2e-2, atol=2e-1)
+
+    dx = gemm.dgrad(dy, w)
+    dxref = dy.matmul(w)
+    np.testing.assert_allclose(dx.float().cpu(), dxref.float().cpu(),
+                               rtol=2e-2, atol=2e-1)
+
+    accum = torch.randn(N * K, device="cuda")
+    a0 = accum.clone()
+    gemm.wgrad_acc(x, dy, accum.view(N, K))
+    ref = a0.view(N, K) + dy.t().float().matmul(x.float())
+    torch.cuda.synchronize()
+    # bf16 inputs, fp32 accumulate: tolerance covers bf16 product rounding
+    np.testing.assert_allclose(accum.view(N, K).cpu(), ref.cpu(),
+                               rtol=2e-2, atol=5e-1)
