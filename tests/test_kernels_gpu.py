@@ -138,3 +138,41 @@ def test_bf16_model_training_step():
     torch.cuda.synchronize()
     assert not torch.equal(net[0].weight.detach().float(), p0)
     assert torch.isfinite(net[0].weight.detach().float()).all()
+
+
+def test_gpu_checkpoint_resume_bit_exact():
+    """Mid-accumulation-window checkpoint/resume on the HIP engine is
+    bit-exact (reference semantics: accum/m/v/step all saved -- SURVEY.md
+    2.2.8). Uses a .grad-path model (no atomics) so determinism holds."""
+    import torch.nn as nn
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+
+    def make():
+        torch.manual_seed(9)
+        return nn.Sequential(nn.Linear(128, 128), nn.ReLU(),
+                             nn.Linear(128, 10)).cuda()
+
+    xs = [torch.randn(16, 128, device="cuda",
+                      generator=torch.Generator("cuda").manual_seed(50 + i))
+          for i in range(6)]
+
+    netA = make()
+    opA = create_optimizer(netA, 1e-3, 1000, 10,
+                           gradient_accumulation_multiplier=4, clip_norm=1.0)
+    for x in xs[:2]:
+        opA.step((netA(x) ** 2).mean())
+    sd = {k: (v.clone() if torch.is_tensor(v) else v)
+          for k, v in opA.state_dict().items()}
+    for x in xs[2:]:
+        opA.step((netA(x) ** 2).mean())
+    ref = netA[0].weight.detach().clone()
+
+    netB = make()
+    opB = create_optimizer(netB, 1e-3, 1000, 10,
+                           gradient_accumulation_multiplier=4, clip_norm=1.0)
+    opB.load_state_dict(sd)
+    assert opB.global_step == 2
+    for x in xs[2:]:
+        opB.step((netB(x) ** 2).mean())
+    torch.cuda.synchronize()
+    assert torch.equal(netB[0].weight.detach(), ref)
