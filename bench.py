@@ -40,6 +40,8 @@ def parse_args():
     p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
                    help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
+    p.add_argument("--wgrad-overlap", default="on", choices=["on", "off"],
+                   help="run weight-grad GEMMs on a side HIP stream")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")
     p.add_argument("--sdpa", default="auto",
@@ -77,6 +79,11 @@ def main():
 
     from gradient_accumulation_tf_estimator_amd import create_optimizer
     from gradient_accumulation_tf_estimator_amd.models.bert import CONFIGS, BertForSequenceClassification
+
+    if use_cuda and args.wgrad_overlap == "on":
+        from gradient_accumulation_tf_estimator_amd.ops import fused as fused_ops
+
+        fused_ops.set_wgrad_overlap(True)
 
     torch.manual_seed(1234 + rank)
     cfg = CONFIGS[args.model]()

@@ -156,11 +156,22 @@ class AccumEngine:
         self.last_lr = lr
         self._lr_dev.fill_(lr)
 
+    def _join_wgrad_stream(self) -> None:
+        """Fence the optional wgrad side stream before anything reads the
+        accum buffer (all-reduce, global norm, apply)."""
+        if self._hip is None:
+            return
+        from ..ops import fused as fused_ops
+
+        if fused_ops.wgrad_overlap_enabled():
+            torch.cuda.current_stream().wait_stream(fused_ops.wgrad_stream())
+
     def apply_from_device(self) -> None:
         """Launch the apply kernels reading lr from the device scalar; no
         host-side lr computation. Used inside hipGraph capture (bench.py)."""
         if self._hip is None:
             raise RuntimeError("apply_from_device requires the HIP backend")
+        self._join_wgrad_stream()
         st = self.state
         model = None if st.master is st.model else st.model
         self._hip.fused_apply(
@@ -182,6 +193,7 @@ class AccumEngine:
         )
 
     def apply(self, lr: Optional[float] = None) -> None:
+        self._join_wgrad_stream()
         st = self.state
         if lr is None:
             lr = self.lr_at(self.global_step)
@@ -234,6 +246,7 @@ class AccumEngine:
         self.accumulate()
         applied = self.is_apply_step()
         if applied:
+            self._join_wgrad_stream()
             self._allreduce_accum()
             self.apply()
         self.global_step += 1
@@ -241,6 +254,7 @@ class AccumEngine:
 
     # ---- checkpoint (accum + m/v + step all saved) ----
     def state_dict(self) -> Dict:
+        self._join_wgrad_stream()
         d = self.state.state_dict()
         d["global_step"] = self.global_step
         d["K"] = self.K

@@ -28,6 +28,31 @@ def _use_hip(x: torch.Tensor) -> bool:
     return x.is_cuda and x.dtype == torch.bfloat16
 
 
+# ---- optional weight-gradient stream overlap -------------------------------
+# wgrad GEMMs accumulate into disjoint slices of the flat fp32 buffer and
+# nothing reads them until the apply boundary, so they can run on a side HIP
+# stream concurrently with the backward dgrad chain. The engine joins the
+# stream before all-reduce/apply (AccumEngine.apply*).
+_WGRAD_OVERLAP = False
+_wgrad_stream = None
+
+
+def set_wgrad_overlap(enabled: bool) -> None:
+    global _WGRAD_OVERLAP
+    _WGRAD_OVERLAP = bool(enabled)
+
+
+def wgrad_overlap_enabled() -> bool:
+    return _WGRAD_OVERLAP
+
+
+def wgrad_stream():
+    global _wgrad_stream
+    if _wgrad_stream is None:
+        _wgrad_stream = torch.cuda.Stream()
+    return _wgrad_stream
+
+
 class _AddLayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, residual, gamma, beta, proj_bias, eps, module):
@@ -179,7 +204,15 @@ class _DirectLinearFn(torch.autograd.Function):
         dy2d = dy.reshape(-1, dy.shape[-1])
         dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
         x2d = x.reshape(-1, x.shape[-1])
-        gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
+        if _WGRAD_OVERLAP:
+            s = wgrad_stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
+            x2d.record_stream(s)
+            dy2d.record_stream(s)
+        else:
+            gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
         db = dy2d.sum(0) if ctx.has_bias else None
         return dx, None, db, None
 
