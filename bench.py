@@ -216,6 +216,9 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
         loss = model.loss(static_ids, static_lab)
         loss.backward()
         engine.accumulate()
+        # capture must end with every forked stream joined; within-step
+        # wgrad overlap is preserved, only the step boundary serializes
+        engine._join_wgrad_stream()
         return loss
 
     # warmup on a side stream (torch.cuda.graphs requirement)
