@@ -40,8 +40,10 @@ def parse_args():
     p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
                    help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
-    p.add_argument("--wgrad-overlap", default="on", choices=["on", "off"],
-                   help="run weight-grad GEMMs on a side HIP stream")
+    p.add_argument("--wgrad-overlap", default="off", choices=["on", "off"],
+                   help="EXPERIMENTAL: wgrad GEMMs on a side HIP stream "
+                        "(measured slower under hipGraphs; also needs a "
+                        "per-stream hipBLASLt workspace before production)")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")
     p.add_argument("--sdpa", default="auto",
