@@ -44,6 +44,8 @@ def parse_args():
                    help="EXPERIMENTAL: wgrad GEMMs on a side HIP stream "
                         "(measured slower under hipGraphs; also needs a "
                         "per-stream hipBLASLt workspace before production)")
+    p.add_argument("--grouped-wgrad", default="off", choices=["on", "off"],
+                   help="issue the micro-step wgrads as one grouped hipBLASLt launch")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")
     p.add_argument("--sdpa", default="auto",
@@ -86,6 +88,10 @@ def main():
         from gradient_accumulation_tf_estimator_amd.ops import fused as fused_ops
 
         fused_ops.set_wgrad_overlap(True)
+    if use_cuda and args.grouped_wgrad == "on":
+        from gradient_accumulation_tf_estimator_amd.ops import fused as fused_ops
+
+        fused_ops.set_grouped_wgrad(True)
 
     torch.manual_seed(1234 + rank)
     cfg = CONFIGS[args.model]()

@@ -116,6 +116,10 @@ class AccumEngine:
 
     # ---- the per-micro-step body (the thing bench.py hipGraph-captures) ----
     def accumulate(self) -> None:
+        if self._hip is not None:
+            from ..ops import fused as fused_ops
+
+            fused_ops.flush_pending_wgrads()
         st = self.state
         lo, hi = st.layout.grad_lo, st.layout.grad_hi
         if hi <= lo:

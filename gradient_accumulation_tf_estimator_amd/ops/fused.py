@@ -36,6 +36,27 @@ def _use_hip(x: torch.Tensor) -> bool:
 _WGRAD_OVERLAP = False
 _wgrad_stream = None
 
+# ---- grouped wgrad -----------------------------------------------------
+# Collect the micro-step's wgrads during backward and issue them as ONE
+# hipBLASLt grouped-gemm launch at flush (engine.accumulate calls it).
+_GROUPED_WGRAD = False
+_pending_wgrads = []
+
+
+def set_grouped_wgrad(enabled: bool) -> None:
+    global _GROUPED_WGRAD
+    _GROUPED_WGRAD = bool(enabled)
+
+
+def flush_pending_wgrads() -> None:
+    global _pending_wgrads
+    if not _pending_wgrads:
+        return
+    hip = require_hip()
+    xs, dys, accs = zip(*_pending_wgrads)
+    _pending_wgrads = []
+    hip.grouped_wgrad_acc(list(xs), list(dys), list(accs))
+
 
 def set_wgrad_overlap(enabled: bool) -> None:
     global _WGRAD_OVERLAP
@@ -204,7 +225,9 @@ class _DirectLinearFn(torch.autograd.Function):
         dy2d = dy.reshape(-1, dy.shape[-1])
         dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
         x2d = x.reshape(-1, x.shape[-1])
-        if _WGRAD_OVERLAP:
+        if _GROUPED_WGRAD:
+            _pending_wgrads.append((x2d, dy2d, ctx.module._accum_view_w))
+        elif _WGRAD_OVERLAP:
             s = wgrad_stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):

@@ -28,6 +28,7 @@ setup(
                 os.path.join(CSRC, "fused_ln_gelu.hip"),
                 os.path.join(CSRC, "blas_acc.hip"),
                 os.path.join(CSRC, "lt_gemm.hip"),
+                os.path.join(CSRC, "grouped_wgrad.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],
             libraries=["hipblaslt"],
