@@ -101,7 +101,7 @@ void grouped_wgrad_acc(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
       size_t need = 0;
       if (gg->isAlgoSupported(h.algo, need) == HIPBLAS_STATUS_SUCCESS &&
           need <= kGroupedWorkspace) {
-        TORCH_CHECK(gg->initialize(h.algo, grouped_workspace(), false, stream) ==
+        TORCH_CHECK(gg->initialize(h.algo, grouped_workspace(), true, stream) ==
                         HIPBLAS_STATUS_SUCCESS,
                     "grouped wgrad initialize failed");
         ok = true;
