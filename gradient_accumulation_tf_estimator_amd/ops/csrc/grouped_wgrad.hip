@@ -96,19 +96,12 @@ void grouped_wgrad_acc(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     TORCH_CHECK(gg->algoGetHeuristic(8, pref, heur) == HIPBLAS_STATUS_SUCCESS &&
                     !heur.empty(),
                 "grouped wgrad: no heuristic");
-    bool ok = false;
-    for (auto& h : heur) {
-      size_t need = 0;
-      if (gg->isAlgoSupported(h.algo, need) == HIPBLAS_STATUS_SUCCESS &&
-          need <= kGroupedWorkspace) {
-        TORCH_CHECK(gg->initialize(h.algo, grouped_workspace(), true, stream) ==
-                        HIPBLAS_STATUS_SUCCESS,
-                    "grouped wgrad initialize failed");
-        ok = true;
-        break;
-      }
-    }
-    TORCH_CHECK(ok, "grouped wgrad: no supported algo");
+    // isAlgoSupported asserts "hardware != nullptr" inside this hipblaslt
+    // build for grouped problems; the heuristic list already respected the
+    // workspace preference, so initialize straight from the first result.
+    TORCH_CHECK(gg->initialize(heur[0].algo, grouped_workspace(), true, stream) ==
+                    HIPBLAS_STATUS_SUCCESS,
+                "grouped wgrad initialize failed");
     it = cache.emplace(std::move(key), std::move(gg)).first;
     // bound the cache (eager fallback churns pointer sets)
     if (cache.size() > 32) cache.erase(cache.begin());
