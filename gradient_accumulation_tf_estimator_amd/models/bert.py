@@ -25,7 +25,14 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.fused import DirectEmbedding, DirectLinear, FusedAddLayerNorm, FusedBiasGelu
+from ..ops.fused import (
+    DirectEmbedding,
+    DirectLinear,
+    FusedAddLayerNorm,
+    FusedBiasGelu,
+    fused_attention,
+    fused_attention_supported,
+)
 
 
 @dataclass
@@ -101,13 +108,17 @@ class BertSelfAttention(nn.Module):
 
     def forward(self, x, attn_mask=None):
         B, S, H = x.shape
-        qkv = self.qkv(x).view(B, S, 3, self.num_heads, self.head_dim)
+        qkv = self.qkv(x)  # [B, S, 3H]
+        dp = self.dropout_p if self.training else 0.0
+        if (x.is_cuda and x.dtype == torch.bfloat16 and attn_mask is None
+                and dp == 0.0 and self.head_dim == 64
+                and fused_attention_supported(S, self.head_dim, False)):
+            # hand-written MFMA attention straight over the packed projection
+            return fused_attention(qkv, self.num_heads)
+        qkv = qkv.view(B, S, 3, self.num_heads, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)  # [3, B, heads, S, head_dim]
         q, k, v = qkv[0], qkv[1], qkv[2]
-        o = F.scaled_dot_product_attention(
-            q, k, v, attn_mask=attn_mask,
-            dropout_p=self.dropout_p if self.training else 0.0,
-        )
+        o = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_mask, dropout_p=dp)
         return o.transpose(1, 2).reshape(B, S, H)
 
 

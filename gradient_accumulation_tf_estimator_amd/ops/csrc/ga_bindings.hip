@@ -40,6 +40,11 @@ extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
 extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*,
                                          float*, long long, int);
+extern "C" __global__ void k_attn_fwd(const unsigned short*, unsigned short*,
+                                      float*, int, int, int);
+extern "C" __global__ void k_attn_bwd(const unsigned short*, const unsigned short*,
+                                      const unsigned short*, const float*,
+                                      unsigned short*, int, int, int);
 
 namespace {
 
@@ -255,6 +260,42 @@ void colreduce_acc(at::Tensor partials, at::Tensor dest0,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_colreduce_acc launch failed");
 }
 
+// qkv [B,S,3,H] bf16 -> (out [B,S,H], lse [B,nh,S] fp32). S<=128, S%32==0,
+// head_dim 64 (all reference BERT configs).
+std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() &&
+              qkv.scalar_type() == at::kBFloat16, "qkv must be contiguous bf16");
+  TORCH_CHECK(qkv.dim() == 4 && qkv.size(2) == 3, "qkv must be [B,S,3,H]");
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1), H = (int)qkv.size(3);
+  TORCH_CHECK(H == nh * 64, "head_dim must be 64");
+  TORCH_CHECK(S <= 128 && S % 32 == 0, "attn kernel needs S<=128, S%32==0");
+  auto out = at::empty({B, S, H}, qkv.options());
+  auto lse = at::empty({B, nh, S}, qkv.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_attn_fwd, dim3(B * (int)nh), dim3(256), 32768, stream,
+                     (const unsigned short*)qkv.data_ptr(),
+                     (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                     B, S, (int)nh);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd launch failed");
+  return {out, lse};
+}
+
+at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
+                    at::Tensor lse, int64_t nh) {
+  const int B = (int)qkv.size(0), S = (int)qkv.size(1), H = (int)qkv.size(3);
+  TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == at::kBFloat16);
+  auto dqkv = at::empty_like(qkv);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_attn_bwd, dim3(B * (int)nh), dim3(256), 114688, stream,
+                     (const unsigned short*)qkv.data_ptr(),
+                     (const unsigned short*)out.data_ptr(),
+                     (const unsigned short*)dout.data_ptr(),
+                     lse.data_ptr<float>(), (unsigned short*)dqkv.data_ptr(),
+                     B, S, (int)nh);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd launch failed");
+  return dqkv;
+}
+
 void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t H) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
               "dy must be contiguous bf16");
@@ -295,4 +336,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "reduce partials over blocks, ADD into flat fp32 accum slices");
   mod.def("embgrad_acc", &embgrad_acc,
           "scatter-add embedding grads into the flat fp32 accum slice");
+  mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");
+  mod.def("attn_bwd", &attn_bwd, "fused MFMA attention bwd -> packed dqkv");
 }

@@ -29,6 +29,7 @@ setup(
                 os.path.join(CSRC, "blas_acc.hip"),
                 os.path.join(CSRC, "lt_gemm.hip"),
                 os.path.join(CSRC, "grouped_wgrad.hip"),
+                os.path.join(CSRC, "attn.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],
             libraries=["hipblaslt"],

@@ -180,3 +180,31 @@ def test_lt_gemm_matches_torch():
     # bf16 inputs, fp32 accumulate: tolerance covers bf16 product rounding
     np.testing.assert_allclose(accum.view(N, K).cpu(), ref.cpu(),
                                rtol=2e-2, atol=5e-1)
+
+
+@pytest.mark.parametrize("S", [32, 64, 128])
+@pytest.mark.parametrize("nh", [2, 8])
+def test_fused_attention_matches_sdpa(S, nh):
+    """Hand-written MFMA attention (fwd+bwd) vs fp32 math reference."""
+    from gradient_accumulation_tf_estimator_amd.ops.fused import fused_attention
+
+    torch.manual_seed(S * 10 + nh)
+    B, dh = 3, 64
+    H = nh * dh
+    qkv = (torch.randn(B, S, 3 * H, device="cuda") * 0.5).bfloat16().requires_grad_()
+
+    o = fused_attention(qkv, nh)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    # fp32 reference via math SDPA
+    qf = qkv.detach().float().requires_grad_()
+    q, k, v = qf.view(B, S, 3, nh, dh).permute(2, 0, 3, 1, 4)
+    ref = F.scaled_dot_product_attention(q, k, v)
+    ref = ref.transpose(1, 2).reshape(B, S, H)
+    ref.backward(do.float())
+
+    np.testing.assert_allclose(o.detach().float().cpu(), ref.detach().cpu(),
+                               rtol=3e-2, atol=2e-2)
+    np.testing.assert_allclose(qkv.grad.float().cpu(), qf.grad.cpu(),
+                               rtol=5e-2, atol=5e-2)
