@@ -332,8 +332,12 @@ class _AttentionFn(torch.autograd.Function):
 
 
 def fused_attention_supported(S: int, head_dim: int, training_extras: bool) -> bool:
+    import os
+
     from . import hip_available
 
+    if os.environ.get("GA_FUSED_ATTN", "1") == "0":  # A/B switch
+        return False
     return (head_dim == 64 and S <= 128 and S % 32 == 0
             and not training_extras and hip_available())
 
