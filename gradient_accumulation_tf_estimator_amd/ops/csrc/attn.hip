@@ -7,130 +7,146 @@
 //
 // Why hand-written: at micro-batch 8 the torch flash path costs far more in
 // layout copies (packed-QKV permutes), dq/dk/dv zero-fills and kernel count
-// than in math. This kernel reads the fused-QKV projection's packed
+// than in math. These kernels read the fused-QKV projection's packed
 // [B,S,3,H] output DIRECTLY and the backward writes the packed gradient
 // buffer completely (no fills, no permutes, no .contiguous()).
 //
-// Structure (forward), one 4-wave block per (batch, head):
-//   * K and V^T staged in LDS (XOR-swizzled rows for conflict-free
-//     ds_read_b128 fragment reads; V transposed at stage time so the PV
-//     B-operand reads are row-contiguous).
-//   * "Swapped" QK^T: mfma(A=K, B=Q) gives S^T[k][q] so each lane holds one
-//     q-column -> row softmax needs only one __shfl_xor(32) lane-pair
-//     combine (cdna_hip_programming.md App. B attention recipe).
-//   * P^T fp32 -> PV A-fragments fully in registers via v_cvt_pk_bf16_f32
-//     pairs + __builtin_amdgcn_permlane32_swap (T12/T21 primitives).
-//   * lse (base-2) saved for the backward's P recompute.
+// Geometry: ONE wave64 per block, one block per (batch*head, 32-row tile) --
+// grid B*nh*(S/32) (256 workgroups at the bench shape) so the 256-CU chip
+// fills; operand panels are re-staged/re-read per block and ride L2. The
+// backward trades recompute for parallelism: a q-tile kernel (dQ) and a
+// k-tile kernel (dK,dV) each recompute P in the orientation whose lane
+// layout matches their output fragments, instead of exchanging P via memory.
 //
 // MFMA fragment maps used throughout (v_mfma_f32_32x32x16_bf16):
 //   A[i][k]: lane l holds i = l&31, k = (l>>5)*8 + c, c = 0..7
 //   B[k][j]: lane l holds j = l&31, k = (l>>5)*8 + c
 //   C/D   : lane l holds col j = l&31, row i = (r&3) + 8*(r>>2) + 4*(l>>5)
+// "cvt+swap": v_cvt_pk_bf16_f32 pairs + permlane32_swap turn a C/D-layout
+// fp32 tile (col = lane) into A-fragments whose i is that col dimension,
+// fully in registers (T12/T21 primitives, cdna_hip_programming.md).
 
 #include <hip/hip_runtime.h>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
-typedef __attribute__((ext_vector_type(4))) unsigned short ush4;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 #define ATTN_D 64
 #define LOG2E 1.44269504088896340736f
 
-static inline __device__ float attn_bf2f(unsigned short u) {
-  union { unsigned int i; float f; } c;
-  c.i = (unsigned int)u << 16;
-  return c.f;
+static inline __device__ unsigned short f2bf_rne(float f) {
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  return (unsigned short)((c.i + (((c.i >> 16) & 1u) + 0x7fffu)) >> 16);
 }
 
-// XOR-swizzled byte offset inside a [rows][64] bf16 tile (128 B rows):
-// spread each 16-lane ds_read_b128 group over slots (G4 recipe).
+// XOR-swizzled byte offsets: [rows][64] bf16 (128 B rows) and [rows][128]
+// bf16 (256 B rows); the mask hits bits 4-6 so 16 B chunks stay intact (G4).
 static inline __device__ int swz(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ ((row & 7) << 4));
 }
-
-// [rows][128] bf16 tiles (256 B rows), same idea.
 static inline __device__ int swz256(int row, int byte_in_row) {
   return row * 256 + (byte_in_row ^ ((row & 7) << 4));
 }
 
-// Stage a [S][64] bf16 matrix from global (row stride row_stride_elems) into
-// LDS with swz layout. 256 threads; each handles S*64/ (256*8) rows of 8.
 static __device__ void stage_64(const unsigned short* g, int row_stride,
                                 unsigned short* lds, int S) {
-  const int chunks = S * 8;  // 16 B chunks (8 bf16)
+  const int chunks = S * 8;  // 16 B chunks
   for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
     const int row = c >> 3, off = (c & 7) << 4;
     const unsigned short* src = g + (long long)row * row_stride + ((c & 7) << 3);
-    // XOR mask touches bits 4-6 only, so the 16 B chunk stays one block
     *(bf16x8*)((char*)lds + swz(row, off)) = *(const bf16x8*)src;
   }
 }
 
-// Stage V transposed: LDS V^T[64][S] (S*2-byte rows padded to 256 B) with
-// swz256 on rows of 128 bf16. Narrow writes, amortized once per block.
+// transpose-stage a [S][64] panel into LDS [64][S<=128] (swz256 rows)
 static __device__ void stage_64_T(const unsigned short* g, int row_stride,
                                   unsigned short* lds, int S) {
-  // element (s, d) -> lds row d, col s
   const int total = S * ATTN_D;
   for (int e = threadIdx.x * 4; e < total; e += blockDim.x * 4) {
     const int s = e / ATTN_D, d0 = e % ATTN_D;
     const unsigned short* src = g + (long long)s * row_stride + d0;
-    // 4 consecutive d of one s -> 4 different LDS rows, same col
-    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
       *(unsigned short*)((char*)lds + swz256(d0 + t, s * 2)) = src[t];
-    }
   }
 }
 
-extern "C" __global__ __launch_bounds__(256) void k_attn_fwd(
+// cvt+swap: 8 consecutive C/D regs (rb..rb+7) -> one bf16x8 A-fragment
+// covering 16 contraction rows across the lane halves.
+static inline __device__ bf16x8 cvt_swap(const f32x16& acc, int rb) {
+  unsigned int u0, u1, v0, v1;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+      : "=v"(u0) : "v"(acc[rb + 0]), "v"(acc[rb + 1]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+      : "=v"(u1) : "v"(acc[rb + 2]), "v"(acc[rb + 3]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+      : "=v"(v0) : "v"(acc[rb + 4]), "v"(acc[rb + 5]));
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
+      : "=v"(v1) : "v"(acc[rb + 6]), "v"(acc[rb + 7]));
+  auto r0 = __builtin_amdgcn_permlane32_swap(u0, v0, false, false);
+  auto r1 = __builtin_amdgcn_permlane32_swap(u1, v1, false, false);
+  bf16x8 out;
+  unsigned int* pw = (unsigned int*)&out;
+  pw[0] = r0[0];
+  pw[1] = r1[0];
+  pw[2] = r0[1];
+  pw[3] = r1[1];
+  return out;
+}
+
+static __device__ void write_tile_bf16(unsigned short* base, long long row_stride,
+                                       int hi, const f32x16& acc) {
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    base[row * row_stride] = f2bf_rne(acc[r]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// forward: block = (bh, q-tile), one wave
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
     const unsigned short* __restrict__ qkv,  // [B,S,3,H]
     unsigned short* __restrict__ out,        // [B,S,H]
     float* __restrict__ lse_out,             // [B,nh,S] base-2 lse
     int B, int S, int nh) {
   const int H = nh * ATTN_D;
-  const int b = blockIdx.x / nh, h = blockIdx.x % nh;
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int lo31 = lane & 31;
-  const int hi = lane >> 5;
+  const int NT = S / 32;
+  const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* Klds = (unsigned short*)smem;            // [S][64] swz
-  unsigned short* Vtlds = (unsigned short*)(smem + 16384); // [64][128] swz256
+  unsigned short* Vtlds = (unsigned short*)(smem + 16384); // [64][S] swz256
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
-  stage_64(qkv + base + H, 3 * H, Klds, S);        // K
-  stage_64_T(qkv + base + 2 * H, 3 * H, Vtlds, S); // V -> V^T
+  stage_64(qkv + base + H, 3 * H, Klds, S);
+  stage_64_T(qkv + base + 2 * H, 3 * H, Vtlds, S);
   __syncthreads();
 
-  const int q0 = wave * 32;
-  const int NT = S / 32;  // k-tiles
-  if (q0 >= S) return;    // safe: no further barriers
-
-  // Q B-fragments from global: lane -> q row lo31, d chunk hi*8 + 16*kk
+  const int q0 = qt * 32;
   bf16x8 qf[4];
-  {
-    const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk)
-      qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
-  }
+  for (int kk = 0; kk < 4; ++kk)
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
 
-  // ---- S^T = K Q^T : per k-tile 32x32 acc ----
+  // swapped S^T = K Q^T: acc[t] = S^T[32t + rows][q = q0 + lo31]
   f32x16 acc[4];
 #pragma unroll
   for (int t = 0; t < 4; ++t) acc[t] = (f32x16)(0.f);
-  for (int t = 0; t < NT; ++t) {
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      // A = K[k=32t+lo31][d = hi*8 + 16kk ..]: swizzled b128 read
       bf16x8 a = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
       acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
     }
-  }
 
-  // ---- softmax over k (lane pair l <-> l^32 holds one q column) ----
-  const float scale2 = 0.125f * LOG2E;  // 1/sqrt(64) folded into exp2
+  const float scale2 = 0.125f * LOG2E;
   float m2 = -1e30f;
 #pragma unroll
   for (int t = 0; t < 4; ++t)
@@ -149,305 +165,233 @@ extern "C" __global__ __launch_bounds__(256) void k_attn_fwd(
       }
   sum += __shfl_xor(sum, 32, 64);
   const float inv_sum = 1.f / sum;
-  const float lse2 = m2 + log2f(sum);
-  if (hi == 0 && lo31 < 32) lse_out[((long long)b * nh + h) * S + q0 + lo31] = lse2;
+  if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m2 + log2f(sum);
 
-  // normalize now so O needs no epilogue divide
 #pragma unroll
   for (int t = 0; t < 4; ++t)
     if (t < NT)
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[t][r] *= inv_sum;
 
-  // ---- P^T fp32 -> PV A-fragments in registers (cvt_pk + permlane) ----
-  // For k-tile t: frag(t,0) covers k' 32t+0..15, frag(t,1) k' 32t+16..31.
-  bf16x8 pf[4][2];
-#pragma unroll
-  for (int t = 0; t < 4; ++t) {
-    if (t >= NT) break;
-#pragma unroll
-    for (int halfk = 0; halfk < 2; ++halfk) {
-      const int rb = halfk * 8;
-      unsigned int u0, u1, v0, v1;
-      // s_nop 1 inside the string: asm writes are invisible to the hazard
-      // recognizer and v_permlane reads them within 2 states (T21 hazard)
-      asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-          : "=v"(u0) : "v"(acc[t][rb + 0]), "v"(acc[t][rb + 1]));
-      asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-          : "=v"(u1) : "v"(acc[t][rb + 2]), "v"(acc[t][rb + 3]));
-      asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-          : "=v"(v0) : "v"(acc[t][rb + 4]), "v"(acc[t][rb + 5]));
-      asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-          : "=v"(v1) : "v"(acc[t][rb + 6]), "v"(acc[t][rb + 7]));
-      auto r0 = __builtin_amdgcn_permlane32_swap(u0, v0, false, false);
-      auto r1 = __builtin_amdgcn_permlane32_swap(u1, v1, false, false);
-      unsigned int* pw = (unsigned int*)&pf[t][halfk];
-      pw[0] = r0[0];
-      pw[1] = r1[0];
-      pw[2] = r0[1];
-      pw[3] = r1[1];
-    }
-  }
-
-  // ---- O = P V : two 32-d output tiles ----
+  // O = P V: cvt+swap A-fragments (i = q), V^T row B-reads
 #pragma unroll
   for (int dt = 0; dt < 2; ++dt) {
     f32x16 oc = (f32x16)(0.f);
-    for (int t = 0; t < NT; ++t) {
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int halfk = 0; halfk < 2; ++halfk) {
-        // B = V^T[d = dt*32 + lo31][k' = 32t + 16*halfk + hi*8 ..]
+        bf16x8 pa = cvt_swap(acc[t], halfk * 8);
         bf16x8 bv = *(const bf16x8*)((char*)Vtlds +
                                      swz256(dt * 32 + lo31,
                                             (32 * t + 16 * halfk + hi * 8) * 2));
-        bf16x8 pa = pf[t][halfk];
         oc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, bv, oc, 0, 0, 0);
       }
-    }
-    // D: col d = dt*32 + lo31, row q = (r&3) + 8*(r>>2) + 4*hi
-    unsigned short* obase = out + ((long long)b * S + q0) * H + h * ATTN_D + dt * 32 + lo31;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int q = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      union { float f; unsigned int i; } c;
-      c.f = oc[r];
-      unsigned int x = c.i;
-      unsigned short bf =
-          (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
-      obase[(long long)q * H] = bf;
-    }
+    write_tile_bf16(out + ((long long)b * S + q0) * H + h * ATTN_D + dt * 32 + lo31,
+                    H, hi, oc);
   }
 }
 
 // ---------------------------------------------------------------------------
-// Backward: one 4-wave block per (b,h); wave w owns q-tile w for S^T/dS/dQ
-// and k-tile w for dV/dK. P / dS^T staged in LDS (bf16) for the operand
-// transposes; everything reads/writes the packed [B,S,3,H] buffers directly.
-//
-//   D_q   = rowsum(dO o O)
-//   P^T   = exp2(scale2 * S^T - lse2)          (normalized, recomputed)
-//   dP^T  = V dO^T                              mfma(A=V, B=dO-frags)
-//   dS^T  = scale * P^T o (dP^T - D_q)
-//   dV    = P^T dO                              mfma(A=P-lds, B=dO^T-lds)
-//   dQ    = dS K                                mfma(A=dS-frags, B=K^T-lds)
-//   dK    = dS^T Q                              mfma(A=dS-lds, B=Q^T-lds)
-
-static __device__ void write_tile_bf16(unsigned short* base, long long row_stride,
-                                       int lo31, int hi, const f32x16& acc) {
+// backward part 0: D[b,h,q] = rowsum(dO o O)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void k_attn_bwd_d(
+    const unsigned short* __restrict__ out, const unsigned short* __restrict__ dout,
+    float* __restrict__ Dtab, int B, int S, int nh) {
+  const int H = nh * ATTN_D;
+  const long long total = (long long)B * nh * S;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int q = (int)(i % S);
+    const int h = (int)((i / S) % nh);
+    const long long b = i / ((long long)S * nh);
+    const unsigned short* o = out + ((long long)b * S + q) * H + h * ATTN_D;
+    const unsigned short* g = dout + ((long long)b * S + q) * H + h * ATTN_D;
+    float s = 0.f;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    union { float f; unsigned int i; } c;
-    c.f = acc[r];
-    unsigned int x = c.i;
-    base[row * row_stride] =
-        (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
+    for (int c = 0; c < 8; ++c) {
+      bf16x8 ov = *(const bf16x8*)(o + c * 8);
+      bf16x8 gv = *(const bf16x8*)(g + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) s += (float)ov[e] * (float)gv[e];
+    }
+    Dtab[i] = s;
   }
 }
 
-extern "C" __global__ __launch_bounds__(256) void k_attn_bwd(
-    const unsigned short* __restrict__ qkv,  // [B,S,3,H]
-    const unsigned short* __restrict__ out,  // [B,S,H] (forward O)
-    const unsigned short* __restrict__ dout, // [B,S,H]
-    const float* __restrict__ lse_in,        // [B,nh,S] base-2
-    unsigned short* __restrict__ dqkv,       // [B,S,3,H]
-    int B, int S, int nh) {
+// ---------------------------------------------------------------------------
+// backward part 1: per (bh, q-tile): dQ  (swapped orientation, lanes own q)
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
   const int H = nh * ATTN_D;
-  const int b = blockIdx.x / nh, h = blockIdx.x % nh;
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int lo31 = lane & 31;
-  const int hi = lane >> 5;
   const int NT = S / 32;
+  const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  unsigned short* Klds = (unsigned short*)smem;             // [S][64]   swz
-  unsigned short* Ktlds = (unsigned short*)(smem + 16384);  // [64][128] swz256
-  unsigned short* Vlds = (unsigned short*)(smem + 32768);   // [S][64]   swz
-  unsigned short* dOtlds = (unsigned short*)(smem + 49152); // [64][128] swz256
-  unsigned short* Qtlds = (unsigned short*)(smem + 65536);  // [64][128] swz256
-  unsigned short* Plds = (unsigned short*)(smem + 81920);   // [S][128]  swz256
+  unsigned short* Klds = (unsigned short*)smem;             // [S][64] swz
+  unsigned short* Vlds = (unsigned short*)(smem + 16384);   // [S][64] swz
+  unsigned short* Ktlds = (unsigned short*)(smem + 32768);  // [64][S] swz256
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
   stage_64(qkv + base + H, 3 * H, Klds, S);
-  stage_64_T(qkv + base + H, 3 * H, Ktlds, S);
   stage_64(qkv + base + 2 * H, 3 * H, Vlds, S);
-  stage_64_T(dout + obase, H, dOtlds, S);
-  stage_64_T(qkv + base, 3 * H, Qtlds, S);
+  stage_64_T(qkv + base + H, 3 * H, Ktlds, S);
   __syncthreads();
 
-  const int q0 = wave * 32;
-  const int active = q0 < S;  // inactive waves still hit barriers below
+  const int q0 = qt * 32;
+  const float lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
+  const float D_q = Dtab[((long long)b * nh + h) * S + q0 + lo31];
 
-  // ---- per-q D = rowsum(dO o O), lse ----
-  float D_q = 0.f, lse2 = 0.f;
-  if (active) {
-    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
-    const unsigned short* orow = out + obase + (long long)(q0 + lo31) * H + hi * 32;
-    float s = 0.f;
+  bf16x8 qf[4], dof[4];
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+  const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
-      bf16x8 ov = *(const bf16x8*)(orow + c * 8);
-#pragma unroll
-      for (int e = 0; e < 8; ++e) s += (float)dv[e] * (float)ov[e];
-    }
-    D_q = s + __shfl_xor(s, 32, 64);
-    lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
+  for (int kk = 0; kk < 4; ++kk) {
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+    dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
   }
 
-  // ---- recompute S^T and P^T; dP^T ----
   f32x16 acc[4], dacc[4];
 #pragma unroll
   for (int t = 0; t < 4; ++t) {
     acc[t] = (f32x16)(0.f);
     dacc[t] = (f32x16)(0.f);
   }
-  bf16x8 qf[4], dof[4];
-  if (active) {
-    const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
-    const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
-      dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+      bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+      bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
+      dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
     }
-    for (int t = 0; t < NT; ++t) {
-#pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
-        bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
-        bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
-        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
-        dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
-      }
-    }
-    const float scale2 = 0.125f * LOG2E;
-    const float scale = 0.125f;
-#pragma unroll
-    for (int t = 0; t < 4; ++t)
-      if (t < NT)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
-          acc[t][r] = p;                                  // P^T
-          dacc[t][r] = scale * p * (dacc[t][r] - D_q);    // dS^T (scaled)
-        }
-  }
 
-  // ---- stage P^T -> Plds[k][q] (bf16, narrow writes) ----
-  if (active) {
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      if (t >= NT) break;
+  for (int t = 0; t < 4; ++t)
+    if (t < NT)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int k = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        union { float f; unsigned int i; } c;
-        c.f = acc[t][r];
-        unsigned int x = c.i;
-        *(unsigned short*)((char*)Plds + swz256(k, (q0 + lo31) * 2)) =
-            (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
+        const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+        dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T (col q = lo31)
       }
-    }
-  }
-  __syncthreads();
 
-  // ---- dV tile (k-rows 32*wave..+32): mfma over q ----
-  const int k0 = wave * 32;
-  if (k0 < S) {
+  // dQ = dS K: A-frags (i = q) from dacc via cvt+swap, B = K^T rows
 #pragma unroll
-    for (int dt = 0; dt < 2; ++dt) {
-      f32x16 a = (f32x16)(0.f);
-      for (int qs = 0; qs < S / 16; ++qs) {
-        bf16x8 ap = *(const bf16x8*)((char*)Plds +
-                                     swz256(k0 + lo31, (qs * 16 + hi * 8) * 2));
-        bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
-                                     swz256(dt * 32 + lo31, (qs * 16 + hi * 8) * 2));
-        a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, a, 0, 0, 0);
-      }
-      write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
-                      3 * H, lo31, hi, a);
-    }
-  }
-  __syncthreads();
-
-  // ---- stage dS^T -> Plds[k][q] (overwrite) ----
-  if (active) {
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      if (t >= NT) break;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int k = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        union { float f; unsigned int i; } c;
-        c.f = dacc[t][r];
-        unsigned int x = c.i;
-        *(unsigned short*)((char*)Plds + swz256(k, (q0 + lo31) * 2)) =
-            (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
-      }
-    }
-  }
-
-  // ---- dQ (own q-tile): A = dS frags from regs (cvt_pk + permlane) ----
-  if (active) {
-    bf16x8 sf[4][2];
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      if (t >= NT) break;
+  for (int dt = 0; dt < 2; ++dt) {
+    f32x16 a = (f32x16)(0.f);
+    for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int halfk = 0; halfk < 2; ++halfk) {
-        const int rb = halfk * 8;
-        unsigned int u0, u1, v0, v1;
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-            : "=v"(u0) : "v"(dacc[t][rb + 0]), "v"(dacc[t][rb + 1]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-            : "=v"(u1) : "v"(dacc[t][rb + 2]), "v"(dacc[t][rb + 3]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-            : "=v"(v0) : "v"(dacc[t][rb + 4]), "v"(dacc[t][rb + 5]));
-        asm("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1"
-            : "=v"(v1) : "v"(dacc[t][rb + 6]), "v"(dacc[t][rb + 7]));
-        auto r0 = __builtin_amdgcn_permlane32_swap(u0, v0, false, false);
-        auto r1 = __builtin_amdgcn_permlane32_swap(u1, v1, false, false);
-        unsigned int* pw = (unsigned int*)&sf[t][halfk];
-        pw[0] = r0[0];
-        pw[1] = r1[0];
-        pw[2] = r0[1];
-        pw[3] = r1[1];
+        bf16x8 as = cvt_swap(dacc[t], halfk * 8);
+        bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfk + hi * 8) * 2));
+        a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bk, a, 0, 0, 0);
       }
-    }
-#pragma unroll
-    for (int dt = 0; dt < 2; ++dt) {
-      f32x16 a = (f32x16)(0.f);
-      for (int t = 0; t < NT; ++t) {
-#pragma unroll
-        for (int halfk = 0; halfk < 2; ++halfk) {
-          bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
-                                       swz256(dt * 32 + lo31,
-                                              (32 * t + 16 * halfk + hi * 8) * 2));
-          a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(sf[t][halfk], bk, a, 0, 0, 0);
-        }
-      }
-      write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
-                      3 * H, lo31, hi, a);
-    }
+    write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, a);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward part 2: per (bh, k-tile): dK, dV (lanes own k: S/dP computed in
+// the q-rows orientation -- A = Q/dO row frags, B = this tile's K/V row
+// frags straight from global -- then cvt+swap gives A[i=k][k'=q]).
+// ---------------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int bh = blockIdx.x / NT, kt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dOtlds = (unsigned short*)smem;           // [64][S] swz256
+  unsigned short* Qtlds = (unsigned short*)(smem + 16384);  // [64][S] swz256
+  float* lsetab = (float*)(smem + 32768);                   // [S]
+  float* dtab = (float*)(smem + 32768 + 512);               // [S]
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  stage_64_T(dout + obase, H, dOtlds, S);
+  stage_64_T(qkv + base, 3 * H, Qtlds, S);
+  for (int i = threadIdx.x; i < S; i += blockDim.x) {
+    lsetab[i] = lse_in[((long long)b * nh + h) * S + i];
+    dtab[i] = Dtab[((long long)b * nh + h) * S + i];
   }
   __syncthreads();
 
-  // ---- dK tile (k-rows 32*wave..+32): A = dS-lds, B = Q^T-lds ----
-  if (k0 < S) {
+  const int k0 = kt * 32;
+  // this tile's K and V rows as B-fragments, straight from global
+  bf16x8 kf[4], vf[4];
+  {
+    const unsigned short* krow = qkv + base + H + (long long)(k0 + lo31) * 3 * H;
+    const unsigned short* vrow = qkv + base + 2 * H + (long long)(k0 + lo31) * 3 * H;
 #pragma unroll
-    for (int dt = 0; dt < 2; ++dt) {
-      f32x16 a = (f32x16)(0.f);
-      for (int qs = 0; qs < S / 16; ++qs) {
-        bf16x8 as = *(const bf16x8*)((char*)Plds +
-                                     swz256(k0 + lo31, (qs * 16 + hi * 8) * 2));
-        bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
-                                     swz256(dt * 32 + lo31, (qs * 16 + hi * 8) * 2));
-        a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, a, 0, 0, 0);
-      }
-      write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
-                      3 * H, lo31, hi, a);
+    for (int kk = 0; kk < 4; ++kk) {
+      kf[kk] = *(const bf16x8*)(krow + kk * 16 + hi * 8);
+      vf[kk] = *(const bf16x8*)(vrow + kk * 16 + hi * 8);
     }
+  }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  f32x16 p_qt[4], ds_qt[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    if (t >= NT) break;
+    f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
+    const unsigned short* qrow = qkv + base + (long long)(t * 32 + lo31) * 3 * H;
+    const unsigned short* drow = dout + obase + (long long)(t * 32 + lo31) * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 aq = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+      bf16x8 ad = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[kk], sacc, 0, 0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[kk], dpacc, 0, 0, 0);
+    }
+    // rows are q here: per-reg lse/D from the LDS tables
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const float p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+      p_qt[t][r] = p;
+      ds_qt[t][r] = scale * p * (dpacc[r] - dtab[q]);
+    }
+  }
+
+  // dV = P^T dO (A i=k from cvt+swap of p_qt; B = dO^T rows)
+  // dK = dS^T Q (A i=k from cvt+swap of ds_qt; B = Q^T rows)
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    f32x16 av_ = (f32x16)(0.f), ak_ = (f32x16)(0.f);
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int halfq = 0; halfq < 2; ++halfq) {
+        bf16x8 ap = cvt_swap(p_qt[t], halfq * 8);
+        bf16x8 as = cvt_swap(ds_qt[t], halfq * 8);
+        bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfq + hi * 8) * 2));
+        bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfq + hi * 8) * 2));
+        av_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, av_, 0, 0, 0);
+        ak_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, ak_, 0, 0, 0);
+      }
+    write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, av_);
+    write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, ak_);
   }
 }

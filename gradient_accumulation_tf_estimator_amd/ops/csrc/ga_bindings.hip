@@ -42,9 +42,14 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
                                          float*, long long, int);
 extern "C" __global__ void k_attn_fwd(const unsigned short*, unsigned short*,
                                       float*, int, int, int);
-extern "C" __global__ void k_attn_bwd(const unsigned short*, const unsigned short*,
-                                      const unsigned short*, const float*,
-                                      unsigned short*, int, int, int);
+extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
+                                        float*, int, int, int);
+extern "C" __global__ void k_attn_bwd_q(const unsigned short*, const unsigned short*,
+                                        const float*, const float*, unsigned short*,
+                                        int, int, int);
+extern "C" __global__ void k_attn_bwd_kv(const unsigned short*, const unsigned short*,
+                                         const float*, const float*, unsigned short*,
+                                         int, int, int);
 
 namespace {
 
@@ -272,7 +277,9 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   auto out = at::empty({B, S, H}, qkv.options());
   auto lse = at::empty({B, nh, S}, qkv.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_attn_fwd, dim3(B * (int)nh), dim3(256), 32768, stream,
+  const int NT = S / 32;
+  const size_t lds = (size_t)S * 128 + 64 * 256;  // K + V^T panels
+  hipLaunchKernelGGL(k_attn_fwd, dim3(B * (int)nh * NT), dim3(64), lds, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
                      B, S, (int)nh);
@@ -285,13 +292,27 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   const int B = (int)qkv.size(0), S = (int)qkv.size(1), H = (int)qkv.size(3);
   TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == at::kBFloat16);
   auto dqkv = at::empty_like(qkv);
+  auto Dtab = at::empty({B, (long)nh, S}, lse.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_attn_bwd, dim3(B * (int)nh), dim3(256), 114688, stream,
-                     (const unsigned short*)qkv.data_ptr(),
+  const int NT = S / 32;
+  const long long rows = (long long)B * nh * S;
+  int dblocks = (int)std::min<long long>((rows + 255) / 256, 1024);
+  hipLaunchKernelGGL(k_attn_bwd_d, dim3(dblocks), dim3(256), 0, stream,
                      (const unsigned short*)out.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
-                     lse.data_ptr<float>(), (unsigned short*)dqkv.data_ptr(),
-                     B, S, (int)nh);
+                     Dtab.data_ptr<float>(), B, S, (int)nh);
+  const size_t lds_q = (size_t)S * 128 * 2 + 64 * 256;  // K + V + K^T
+  hipLaunchKernelGGL(k_attn_bwd_q, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
+                     (const unsigned short*)qkv.data_ptr(),
+                     (const unsigned short*)dout.data_ptr(),
+                     lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                     (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+  const size_t lds_kv = (size_t)64 * 256 * 2 + 1024;  // dO^T + Q^T + tables
+  hipLaunchKernelGGL(k_attn_bwd_kv, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
+                     (const unsigned short*)qkv.data_ptr(),
+                     (const unsigned short*)dout.data_ptr(),
+                     lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                     (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd launch failed");
   return dqkv;
 }
