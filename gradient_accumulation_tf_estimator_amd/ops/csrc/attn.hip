@@ -59,16 +59,31 @@ static __device__ void stage_64(const unsigned short* g, int row_stride,
   }
 }
 
-// transpose-stage a [S][64] panel into LDS [64][S<=128] (swz256 rows)
+// transpose-stage a [S][64] panel into LDS [64][S<=128] (swz256 rows).
+// Each thread loads a 4(s) x 8(d) sub-block as four bf16x8 rows and writes
+// eight 8-byte packs (4 s-values of one d) -- compile-time vector extracts,
+// no scalar LDS writes.
+typedef __attribute__((ext_vector_type(4))) unsigned short ush4v;
+
 static __device__ void stage_64_T(const unsigned short* g, int row_stride,
                                   unsigned short* lds, int S) {
-  const int total = S * ATTN_D;
-  for (int e = threadIdx.x * 4; e < total; e += blockDim.x * 4) {
-    const int s = e / ATTN_D, d0 = e % ATTN_D;
-    const unsigned short* src = g + (long long)s * row_stride + d0;
+  const int blocks4 = (S / 4) * (ATTN_D / 8);  // 4s x 8d sub-blocks
+  for (int blk = threadIdx.x; blk < blocks4; blk += blockDim.x) {
+    const int s0 = (blk / (ATTN_D / 8)) * 4;
+    const int d0 = (blk % (ATTN_D / 8)) * 8;
+    bf16x8 r0 = *(const bf16x8*)(g + (long long)(s0 + 0) * row_stride + d0);
+    bf16x8 r1 = *(const bf16x8*)(g + (long long)(s0 + 1) * row_stride + d0);
+    bf16x8 r2 = *(const bf16x8*)(g + (long long)(s0 + 2) * row_stride + d0);
+    bf16x8 r3 = *(const bf16x8*)(g + (long long)(s0 + 3) * row_stride + d0);
+    const unsigned short* u0 = (const unsigned short*)&r0;
+    const unsigned short* u1 = (const unsigned short*)&r1;
+    const unsigned short* u2 = (const unsigned short*)&r2;
+    const unsigned short* u3 = (const unsigned short*)&r3;
 #pragma unroll
-    for (int t = 0; t < 4; ++t)
-      *(unsigned short*)((char*)lds + swz256(d0 + t, s * 2)) = src[t];
+    for (int c = 0; c < 8; ++c) {
+      ush4v pack = {u0[c], u1[c], u2[c], u3[c]};
+      *(ush4v*)((char*)lds + swz256(d0 + c, s0 * 2)) = pack;
+    }
   }
 }
 

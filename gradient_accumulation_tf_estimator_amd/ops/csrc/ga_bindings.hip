@@ -278,7 +278,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   auto lse = at::empty({B, nh, S}, qkv.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
-  const size_t lds = (size_t)S * 128 + 64 * 256;  // K + V^T panels
+  const size_t lds = 16384 + 64 * 256;  // K + V^T panels (fixed offsets)
   hipLaunchKernelGGL(k_attn_fwd, dim3(B * (int)nh * NT), dim3(64), lds, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
@@ -301,7 +301,7 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
                      (const unsigned short*)out.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
                      Dtab.data_ptr<float>(), B, S, (int)nh);
-  const size_t lds_q = (size_t)S * 128 * 2 + 64 * 256;  // K + V + K^T
+  const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
   hipLaunchKernelGGL(k_attn_bwd_q, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
