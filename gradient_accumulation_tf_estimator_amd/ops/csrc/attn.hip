@@ -49,13 +49,22 @@ static inline __device__ int swz256(int row, int byte_in_row) {
   return row * 256 + (byte_in_row ^ ((row & 7) << 4));
 }
 
-static __device__ void stage_64(const unsigned short* g, int row_stride,
-                                unsigned short* lds, int S) {
-  const int chunks = S * 8;  // 16 B chunks
-  for (int c = threadIdx.x; c < chunks; c += blockDim.x) {
-    const int row = c >> 3, off = (c & 7) << 4;
-    const unsigned short* src = g + (long long)row * row_stride + ((c & 7) << 3);
-    *(bf16x8*)((char*)lds + swz(row, off)) = *(const bf16x8*)src;
+template <int S>
+static __device__ __forceinline__ void stage_64(const unsigned short* g,
+                                                int row_stride,
+                                                unsigned short* lds) {
+  // fully unrolled so all loads issue before the first LDS write waits
+  constexpr int ITER = S * 8 / 64;  // 16 B chunks per thread (64 threads)
+  bf16x8 v[ITER];
+#pragma unroll
+  for (int i = 0; i < ITER; ++i) {
+    const int c = threadIdx.x + i * 64;
+    v[i] = *(const bf16x8*)(g + (long long)(c >> 3) * row_stride + ((c & 7) << 3));
+  }
+#pragma unroll
+  for (int i = 0; i < ITER; ++i) {
+    const int c = threadIdx.x + i * 64;
+    *(bf16x8*)((char*)lds + swz(c >> 3, (c & 7) << 4)) = v[i];
   }
 }
 
@@ -65,20 +74,30 @@ static __device__ void stage_64(const unsigned short* g, int row_stride,
 // no scalar LDS writes.
 typedef __attribute__((ext_vector_type(4))) unsigned short ush4v;
 
-static __device__ void stage_64_T(const unsigned short* g, int row_stride,
-                                  unsigned short* lds, int S) {
-  const int blocks4 = (S / 4) * (ATTN_D / 8);  // 4s x 8d sub-blocks
-  for (int blk = threadIdx.x; blk < blocks4; blk += blockDim.x) {
+template <int S>
+static __device__ __forceinline__ void stage_64_T(const unsigned short* g,
+                                                  int row_stride,
+                                                  unsigned short* lds) {
+  constexpr int ITER = (S / 4) * (ATTN_D / 8) / 64;  // 4s x 8d blocks/thread
+  bf16x8 r[ITER][4];
+#pragma unroll
+  for (int i = 0; i < ITER; ++i) {
+    const int blk = threadIdx.x + i * 64;
     const int s0 = (blk / (ATTN_D / 8)) * 4;
     const int d0 = (blk % (ATTN_D / 8)) * 8;
-    bf16x8 r0 = *(const bf16x8*)(g + (long long)(s0 + 0) * row_stride + d0);
-    bf16x8 r1 = *(const bf16x8*)(g + (long long)(s0 + 1) * row_stride + d0);
-    bf16x8 r2 = *(const bf16x8*)(g + (long long)(s0 + 2) * row_stride + d0);
-    bf16x8 r3 = *(const bf16x8*)(g + (long long)(s0 + 3) * row_stride + d0);
-    const unsigned short* u0 = (const unsigned short*)&r0;
-    const unsigned short* u1 = (const unsigned short*)&r1;
-    const unsigned short* u2 = (const unsigned short*)&r2;
-    const unsigned short* u3 = (const unsigned short*)&r3;
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+      r[i][t] = *(const bf16x8*)(g + (long long)(s0 + t) * row_stride + d0);
+  }
+#pragma unroll
+  for (int i = 0; i < ITER; ++i) {
+    const int blk = threadIdx.x + i * 64;
+    const int s0 = (blk / (ATTN_D / 8)) * 4;
+    const int d0 = (blk % (ATTN_D / 8)) * 8;
+    const unsigned short* u0 = (const unsigned short*)&r[i][0];
+    const unsigned short* u1 = (const unsigned short*)&r[i][1];
+    const unsigned short* u2 = (const unsigned short*)&r[i][2];
+    const unsigned short* u3 = (const unsigned short*)&r[i][3];
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       ush4v pack = {u0[c], u1[c], u2[c], u3[c]};
@@ -122,13 +141,14 @@ static __device__ void write_tile_bf16(unsigned short* base, long long row_strid
 // ---------------------------------------------------------------------------
 // forward: block = (bh, q-tile), one wave
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
+template <int S>
+__device__ __forceinline__ void attn_fwd_body(
     const unsigned short* __restrict__ qkv,  // [B,S,3,H]
     unsigned short* __restrict__ out,        // [B,S,H]
     float* __restrict__ lse_out,             // [B,nh,S] base-2 lse
-    int B, int S, int nh) {
+    int B, int nh) {
   const int H = nh * ATTN_D;
-  const int NT = S / 32;
+  constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
   const int b = bh / nh, h = bh % nh;
   const int lo31 = threadIdx.x & 31;
@@ -139,8 +159,8 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
   unsigned short* Vtlds = (unsigned short*)(smem + 16384); // [64][S] swz256
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
-  stage_64(qkv + base + H, 3 * H, Klds, S);
-  stage_64_T(qkv + base + 2 * H, 3 * H, Vtlds, S);
+  stage_64<S>(qkv + base + H, 3 * H, Klds);
+  stage_64_T<S>(qkv + base + 2 * H, 3 * H, Vtlds);
   __syncthreads();
 
   const int q0 = qt * 32;
@@ -151,9 +171,10 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
     qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
 
   // swapped S^T = K Q^T: acc[t] = S^T[32t + rows][q = q0 + lo31]
-  f32x16 acc[4];
+  f32x16 acc[NT];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) acc[t] = (f32x16)(0.f);
+  for (int t = 0; t < NT; ++t) acc[t] = (f32x16)(0.f);
+#pragma unroll
   for (int t = 0; t < NT; ++t)
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
@@ -164,34 +185,32 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
   const float scale2 = 0.125f * LOG2E;
   float m2 = -1e30f;
 #pragma unroll
-  for (int t = 0; t < 4; ++t)
-    if (t < NT)
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) m2 = fmaxf(m2, acc[t][r]);
+    for (int r = 0; r < 16; ++r) m2 = fmaxf(m2, acc[t][r]);
   m2 = fmaxf(m2, __shfl_xor(m2, 32, 64)) * scale2;
   float sum = 0.f;
 #pragma unroll
-  for (int t = 0; t < 4; ++t)
-    if (t < NT)
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        acc[t][r] = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - m2);
-        sum += acc[t][r];
-      }
+    for (int r = 0; r < 16; ++r) {
+      acc[t][r] = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - m2);
+      sum += acc[t][r];
+    }
   sum += __shfl_xor(sum, 32, 64);
   const float inv_sum = 1.f / sum;
   if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m2 + log2f(sum);
 
 #pragma unroll
-  for (int t = 0; t < 4; ++t)
-    if (t < NT)
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) acc[t][r] *= inv_sum;
+    for (int r = 0; r < 16; ++r) acc[t][r] *= inv_sum;
 
   // O = P V: cvt+swap A-fragments (i = q), V^T row B-reads
 #pragma unroll
   for (int dt = 0; dt < 2; ++dt) {
     f32x16 oc = (f32x16)(0.f);
+#pragma unroll
     for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int halfk = 0; halfk < 2; ++halfk) {
@@ -205,6 +224,17 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd(
                     H, hi, oc);
   }
 }
+
+#define GA_ATTN_FWD_INST(S)                                                   \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_##S(            \
+      const unsigned short* qkv, unsigned short* out, float* lse_out, int B,  \
+      int nh) {                                                               \
+    attn_fwd_body<S>(qkv, out, lse_out, B, nh);                               \
+  }
+GA_ATTN_FWD_INST(32)
+GA_ATTN_FWD_INST(64)
+GA_ATTN_FWD_INST(96)
+GA_ATTN_FWD_INST(128)
 
 // ---------------------------------------------------------------------------
 // backward part 0: D[b,h,q] = rowsum(dO o O)
@@ -236,12 +266,13 @@ extern "C" __global__ void k_attn_bwd_d(
 // ---------------------------------------------------------------------------
 // backward part 1: per (bh, q-tile): dQ  (swapped orientation, lanes own q)
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
+template <int S>
+__device__ __forceinline__ void attn_bwd_q_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, const float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
+    unsigned short* __restrict__ dqkv, int B, int nh) {
   const int H = nh * ATTN_D;
-  const int NT = S / 32;
+  constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
   const int b = bh / nh, h = bh % nh;
   const int lo31 = threadIdx.x & 31;
@@ -254,9 +285,9 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
-  stage_64(qkv + base + H, 3 * H, Klds, S);
-  stage_64(qkv + base + 2 * H, 3 * H, Vlds, S);
-  stage_64_T(qkv + base + H, 3 * H, Ktlds, S);
+  stage_64<S>(qkv + base + H, 3 * H, Klds);
+  stage_64<S>(qkv + base + 2 * H, 3 * H, Vlds);
+  stage_64_T<S>(qkv + base + H, 3 * H, Ktlds);
   __syncthreads();
 
   const int q0 = qt * 32;
@@ -272,12 +303,13 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
     dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
   }
 
-  f32x16 acc[4], dacc[4];
+  f32x16 acc[NT], dacc[NT];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
+  for (int t = 0; t < NT; ++t) {
     acc[t] = (f32x16)(0.f);
     dacc[t] = (f32x16)(0.f);
   }
+#pragma unroll
   for (int t = 0; t < NT; ++t)
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
@@ -289,18 +321,18 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
 #pragma unroll
-  for (int t = 0; t < 4; ++t)
-    if (t < NT)
+  for (int t = 0; t < NT; ++t)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
-        dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T (col q = lo31)
-      }
+    for (int r = 0; r < 16; ++r) {
+      const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+      dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T (col q = lo31)
+    }
 
   // dQ = dS K: A-frags (i = q) from dacc via cvt+swap, B = K^T rows
 #pragma unroll
   for (int dt = 0; dt < 2; ++dt) {
     f32x16 a = (f32x16)(0.f);
+#pragma unroll
     for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int halfk = 0; halfk < 2; ++halfk) {
@@ -315,17 +347,30 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q(
   }
 }
 
+#define GA_ATTN_BWDQ_INST(S)                                                  \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_##S(          \
+      const unsigned short* qkv, const unsigned short* dout,                  \
+      const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,    \
+      int nh) {                                                               \
+    attn_bwd_q_body<S>(qkv, dout, lse_in, Dtab, dqkv, B, nh);                 \
+  }
+GA_ATTN_BWDQ_INST(32)
+GA_ATTN_BWDQ_INST(64)
+GA_ATTN_BWDQ_INST(96)
+GA_ATTN_BWDQ_INST(128)
+
 // ---------------------------------------------------------------------------
 // backward part 2: per (bh, k-tile): dK, dV (lanes own k: S/dP computed in
 // the q-rows orientation -- A = Q/dO row frags, B = this tile's K/V row
 // frags straight from global -- then cvt+swap gives A[i=k][k'=q]).
 // ---------------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
+template <int S>
+__device__ __forceinline__ void attn_bwd_kv_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, const float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
+    unsigned short* __restrict__ dqkv, int B, int nh) {
   const int H = nh * ATTN_D;
-  const int NT = S / 32;
+  constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, kt = blockIdx.x % NT;
   const int b = bh / nh, h = bh % nh;
   const int lo31 = threadIdx.x & 31;
@@ -339,8 +384,8 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
-  stage_64_T(dout + obase, H, dOtlds, S);
-  stage_64_T(qkv + base, 3 * H, Qtlds, S);
+  stage_64_T<S>(dout + obase, H, dOtlds);
+  stage_64_T<S>(qkv + base, 3 * H, Qtlds);
   for (int i = threadIdx.x; i < S; i += blockDim.x) {
     lsetab[i] = lse_in[((long long)b * nh + h) * S + i];
     dtab[i] = Dtab[((long long)b * nh + h) * S + i];
@@ -361,10 +406,9 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
   }
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
-  f32x16 p_qt[4], ds_qt[4];
+  f32x16 p_qt[NT], ds_qt[NT];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
-    if (t >= NT) break;
+  for (int t = 0; t < NT; ++t) {
     f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
     const unsigned short* qrow = qkv + base + (long long)(t * 32 + lo31) * 3 * H;
     const unsigned short* drow = dout + obase + (long long)(t * 32 + lo31) * H;
@@ -390,6 +434,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
 #pragma unroll
   for (int dt = 0; dt < 2; ++dt) {
     f32x16 av_ = (f32x16)(0.f), ak_ = (f32x16)(0.f);
+#pragma unroll
     for (int t = 0; t < NT; ++t)
 #pragma unroll
       for (int halfq = 0; halfq < 2; ++halfq) {
@@ -410,3 +455,15 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv(
                     3 * H, hi, ak_);
   }
 }
+
+#define GA_ATTN_BWDKV_INST(S)                                                 \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_##S(         \
+      const unsigned short* qkv, const unsigned short* dout,                  \
+      const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,    \
+      int nh) {                                                               \
+    attn_bwd_kv_body<S>(qkv, dout, lse_in, Dtab, dqkv, B, nh);                \
+  }
+GA_ATTN_BWDKV_INST(32)
+GA_ATTN_BWDKV_INST(64)
+GA_ATTN_BWDKV_INST(96)
+GA_ATTN_BWDKV_INST(128)

@@ -40,16 +40,18 @@ extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
 extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*,
                                          float*, long long, int);
-extern "C" __global__ void k_attn_fwd(const unsigned short*, unsigned short*,
-                                      float*, int, int, int);
+#define GA_DECL_ATTN(S)                                                        \
+  extern "C" __global__ void k_attn_fwd_##S(const unsigned short*,             \
+                                            unsigned short*, float*, int, int);\
+  extern "C" __global__ void k_attn_bwd_q_##S(                                 \
+      const unsigned short*, const unsigned short*, const float*,              \
+      const float*, unsigned short*, int, int);                                \
+  extern "C" __global__ void k_attn_bwd_kv_##S(                                \
+      const unsigned short*, const unsigned short*, const float*,              \
+      const float*, unsigned short*, int, int);
+GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
 extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
                                         float*, int, int, int);
-extern "C" __global__ void k_attn_bwd_q(const unsigned short*, const unsigned short*,
-                                        const float*, const float*, unsigned short*,
-                                        int, int, int);
-extern "C" __global__ void k_attn_bwd_kv(const unsigned short*, const unsigned short*,
-                                         const float*, const float*, unsigned short*,
-                                         int, int, int);
 
 namespace {
 
@@ -279,10 +281,13 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
   const size_t lds = 16384 + 64 * 256;  // K + V^T panels (fixed offsets)
-  hipLaunchKernelGGL(k_attn_fwd, dim3(B * (int)nh * NT), dim3(64), lds, stream,
+  void (*fk)(const unsigned short*, unsigned short*, float*, int, int) =
+      S == 32 ? k_attn_fwd_32 : S == 64 ? k_attn_fwd_64
+      : S == 96 ? k_attn_fwd_96 : k_attn_fwd_128;
+  hipLaunchKernelGGL(fk, dim3(B * (int)nh * NT), dim3(64), lds, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
-                     B, S, (int)nh);
+                     B, (int)nh);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd launch failed");
   return {out, lse};
 }
@@ -302,17 +307,25 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
                      (const unsigned short*)dout.data_ptr(),
                      Dtab.data_ptr<float>(), B, S, (int)nh);
   const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
-  hipLaunchKernelGGL(k_attn_bwd_q, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
+  void (*qk)(const unsigned short*, const unsigned short*, const float*,
+             const float*, unsigned short*, int, int) =
+      S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
+      : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
+  hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
                      lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                     (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+                     (unsigned short*)dqkv.data_ptr(), B, (int)nh);
   const size_t lds_kv = (size_t)64 * 256 * 2 + 1024;  // dO^T + Q^T + tables
-  hipLaunchKernelGGL(k_attn_bwd_kv, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
+  void (*kvk)(const unsigned short*, const unsigned short*, const float*,
+              const float*, unsigned short*, int, int) =
+      S == 32 ? k_attn_bwd_kv_32 : S == 64 ? k_attn_bwd_kv_64
+      : S == 96 ? k_attn_bwd_kv_96 : k_attn_bwd_kv_128;
+  hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
                      (const unsigned short*)qkv.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
                      lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                     (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+                     (unsigned short*)dqkv.data_ptr(), B, (int)nh);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd launch failed");
   return dqkv;
 }
