@@ -52,6 +52,11 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
 GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
 extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
                                         float*, int, int, int);
+struct WgTile { int g, tn, tk, pad; };
+extern "C" __global__ void k_wgrad_mfma(const unsigned long long*,
+                                        const unsigned long long*,
+                                        const unsigned long long*, const int*,
+                                        const WgTile*, int);
 
 namespace {
 
@@ -330,6 +335,19 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   return dqkv;
 }
 
+// One launch over a precomputed tile table: see wgrad_mfma.hip.
+void wgrad_mfma(at::Tensor x_ptrs, at::Tensor dy_ptrs, at::Tensor acc_ptrs,
+                at::Tensor dims, at::Tensor tiles, int64_t R, int64_t ntiles) {
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 32768, stream,
+                     (const unsigned long long*)x_ptrs.data_ptr<int64_t>(),
+                     (const unsigned long long*)dy_ptrs.data_ptr<int64_t>(),
+                     (const unsigned long long*)acc_ptrs.data_ptr<int64_t>(),
+                     dims.data_ptr<int>(), (const WgTile*)tiles.data_ptr<int>(),
+                     (int)R);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
+}
+
 void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t H) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
               "dy must be contiguous bf16");
@@ -371,5 +389,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embgrad_acc", &embgrad_acc,
           "scatter-add embedding grads into the flat fp32 accum slice");
   mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");
+  mod.def("wgrad_mfma", &wgrad_mfma,
+          "batched MFMA wgrad: accum_g += dy_g^T @ x_g over a tile table");
   mod.def("attn_bwd", &attn_bwd, "fused MFMA attention bwd -> packed dqkv");
 }

@@ -1,0 +1,127 @@
+// Batched MFMA weight-gradient kernel (gfx950).
+//
+//   for each problem g:  accum_g[N,K] += dy_g[R,N]^T @ x_g[R,K]   (bf16 in,
+//   fp32 read-modify-write accumulate into the flat accum buffer slices)
+//
+// Replaces the 16 per-layer hipBLASLt wgrad launches of a BERT micro-step
+// (~205 us/step at the reference micro-batch: each GEMM is ~1-2 GFLOP and
+// fills a fraction of the chip) with ONE launch over a precomputed tile
+// table: every 128x128 output tile of every problem is one 4-wave
+// workgroup, so the whole set fills 256 CUs at once. Totals ~26 GFLOP per
+// micro-step -> tens of us at moderate MFMA utilization.
+//
+// Both operands are contraction-major in memory (R is the row dim of x and
+// dy), so each 64-deep R-chunk of both panels is staged TRANSPOSED into LDS
+// ([cols][64r], 128 B rows, XOR-swizzled) with the same vectorized
+// 4-row-pack transpose used by the attention kernels; A- and B-fragments
+// then read 16 B rows. MFMA fragment maps as in attn.hip.
+
+#include <hip/hip_runtime.h>
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(4))) unsigned short ush4w;
+
+// XOR swizzle for [rows][64] bf16 tiles (128 B rows): bits 4-6.
+static inline __device__ int wswz(int row, int byte_in_row) {
+  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
+}
+
+// Transpose-stage a [64 r][C cols] global panel (row stride = ld elems,
+// start col c0) into LDS [C][64 r] (128 B rows). 256 threads, C in {128}.
+template <int C>
+static __device__ __forceinline__ void stage_T64(const unsigned short* g,
+                                                 long long ld, int c0,
+                                                 unsigned short* lds) {
+  constexpr int ITER = (64 / 4) * (C / 8) / 256;  // 4r x 8c sub-blocks
+#pragma unroll
+  for (int i = 0; i < ITER; ++i) {
+    const int blk = threadIdx.x + i * 256;
+    const int r0 = (blk / (C / 8)) * 4;
+    const int cc = (blk % (C / 8)) * 8;
+    bf16x8 r0v = *(const bf16x8*)(g + (long long)(r0 + 0) * ld + c0 + cc);
+    bf16x8 r1v = *(const bf16x8*)(g + (long long)(r0 + 1) * ld + c0 + cc);
+    bf16x8 r2v = *(const bf16x8*)(g + (long long)(r0 + 2) * ld + c0 + cc);
+    bf16x8 r3v = *(const bf16x8*)(g + (long long)(r0 + 3) * ld + c0 + cc);
+    const unsigned short* u0 = (const unsigned short*)&r0v;
+    const unsigned short* u1 = (const unsigned short*)&r1v;
+    const unsigned short* u2 = (const unsigned short*)&r2v;
+    const unsigned short* u3 = (const unsigned short*)&r3v;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      ush4w pack = {u0[c], u1[c], u2[c], u3[c]};
+      *(ush4w*)((char*)lds + wswz(cc + c, r0 * 2)) = pack;
+    }
+  }
+}
+
+// tile table entry: problem id + tile coords
+struct WgTile {
+  int g;
+  int tn;
+  int tk;
+  int pad;
+};
+
+extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
+    const unsigned long long* __restrict__ x_ptrs,
+    const unsigned long long* __restrict__ dy_ptrs,
+    const unsigned long long* __restrict__ acc_ptrs,
+    const int* __restrict__ dims,  // [G][2] = (N, K)
+    const WgTile* __restrict__ tiles, int R) {
+  const WgTile t = tiles[blockIdx.x];
+  const int N = dims[t.g * 2 + 0], K = dims[t.g * 2 + 1];
+  const unsigned short* x = (const unsigned short*)x_ptrs[t.g];
+  const unsigned short* dy = (const unsigned short*)dy_ptrs[t.g];
+  float* acc_out = (float*)acc_ptrs[t.g];
+  const int n0 = t.tn * 128, k0 = t.tk * 128;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int wn = wave >> 1, wk = wave & 1;  // 2x2 waves -> 64x64 quadrants
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dyT = (unsigned short*)smem;            // [128 n][64 r]
+  unsigned short* xT = (unsigned short*)(smem + 16384);   // [128 k][64 r]
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.f);
+
+  for (int r0 = 0; r0 < R; r0 += 64) {
+    stage_T64<128>(dy + (long long)r0 * N, N, n0, dyT);
+    stage_T64<128>(x + (long long)r0 * K, K, k0, xT);
+    __syncthreads();
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {  // 16-deep r sub-steps
+      // A: dy^T rows (n), B: x^T rows (k); both 16 B swizzled reads
+      bf16x8 a0 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 a1 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b0 = *(const bf16x8*)((char*)xT + wswz(wk * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 b1 = *(const bf16x8*)((char*)xT + wswz(wk * 64 + 32 + lo31, s * 32 + hi * 16));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: fp32 accumulate into acc_out[n][k] (D: col=k, rows n)
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      float* base = acc_out + (long long)(n0 + wn * 64 + i * 32) * K + k0 +
+                    wk * 64 + j * 32 + lo31;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        base[(long long)row * K] += acc[i][j][r];
+      }
+    }
+}

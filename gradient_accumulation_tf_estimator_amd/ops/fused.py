@@ -48,23 +48,58 @@ def set_grouped_wgrad(enabled: bool) -> None:
     _GROUPED_WGRAD = bool(enabled)
 
 
+_wgrad_table_cache = {}
+
+
 def flush_pending_wgrads() -> None:
     global _pending_wgrads
     if not _pending_wgrads:
         return
     hip = require_hip()
-    xs, dys, accs = zip(*_pending_wgrads)
+    all_pending = _pending_wgrads
     _pending_wgrads = []
-    try:
-        hip.grouped_wgrad_acc(list(xs), list(dys), list(accs))
-    except RuntimeError:
-        # this hipblaslt build asserts on grouped bf16->fp32 problems;
-        # fall back to per-problem wgrads and stop grouping
-        set_grouped_wgrad(False)
-        from . import gemm
 
-        for x, dy, acc in zip(xs, dys, accs):
+    # the batched MFMA kernel covers N,K % 128 == 0 with a common R % 64 == 0
+    # (every encoder-layer wgrad); odd shapes (e.g. the pooler's R=batch) go
+    # through per-problem hipBLASLt
+    from . import gemm
+
+    pending = []
+    for x, dy, acc in all_pending:
+        R_i = x.numel() // x.shape[-1]
+        if (x.shape[-1] % 128 == 0 and dy.shape[-1] % 128 == 0 and R_i % 64 == 0
+                and (not pending or R_i == pending[0][0].numel() // pending[0][0].shape[-1])):
+            pending.append((x, dy, acc))
+        else:
             gemm.wgrad_acc(x, dy, acc.view(dy.shape[-1], x.shape[-1]))
+    if not pending:
+        return
+    xs, dys, accs = zip(*pending)
+    R = xs[0].numel() // xs[0].shape[-1]
+
+    key = tuple((x.data_ptr(), dy.data_ptr(), a.data_ptr(),
+                 x.shape[-1], dy.shape[-1]) for x, dy, a in pending) + (R,)
+    entry = _wgrad_table_cache.get(key)
+    if entry is None:
+        dev = xs[0].device
+        x_ptrs = torch.tensor([x.data_ptr() for x in xs], dtype=torch.int64, device=dev)
+        dy_ptrs = torch.tensor([d.data_ptr() for d in dys], dtype=torch.int64, device=dev)
+        acc_ptrs = torch.tensor([a.data_ptr() for a in accs], dtype=torch.int64, device=dev)
+        dims = torch.tensor([[d.shape[-1], x.shape[-1]] for x, d in zip(xs, dys)],
+                            dtype=torch.int32, device=dev)
+        tiles = []
+        for g, (x, dy) in enumerate(zip(xs, dys)):
+            N, K = dy.shape[-1], x.shape[-1]
+            for tn in range(N // 128):
+                for tk in range(K // 128):
+                    tiles.append((g, tn, tk, 0))
+        tiles_t = torch.tensor(tiles, dtype=torch.int32, device=dev)
+        entry = (x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, len(tiles))
+        if len(_wgrad_table_cache) > 32:
+            _wgrad_table_cache.clear()
+        _wgrad_table_cache[key] = entry
+    x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, ntiles = entry
+    hip.wgrad_mfma(x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, R, ntiles)
 
 
 def set_wgrad_overlap(enabled: bool) -> None:
