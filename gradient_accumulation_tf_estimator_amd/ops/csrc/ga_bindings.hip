@@ -52,11 +52,15 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
 GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
 extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
                                         float*, int, int, int);
-struct WgTile { int g, tn, tk, pad; };
-extern "C" __global__ void k_wgrad_mfma(const unsigned long long*,
-                                        const unsigned long long*,
-                                        const unsigned long long*, const int*,
-                                        const WgTile*, int);
+#define WG_MAX_G 24
+struct WgArgs {
+  unsigned long long x[WG_MAX_G];
+  unsigned long long dy[WG_MAX_G];
+  unsigned long long acc[WG_MAX_G];
+  int nk[WG_MAX_G * 2];
+  int G;
+};
+extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
 
 namespace {
 
@@ -335,16 +339,30 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   return dqkv;
 }
 
-// One launch over a precomputed tile table: see wgrad_mfma.hip.
-void wgrad_mfma(at::Tensor x_ptrs, at::Tensor dy_ptrs, at::Tensor acc_ptrs,
-                at::Tensor dims, at::Tensor tiles, int64_t R, int64_t ntiles) {
+// One launch over every pending wgrad; metadata passed by value (see
+// wgrad_mfma.hip) so it is hipGraph-capture-safe.
+void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
+                std::vector<at::Tensor> accs, int64_t R) {
+  const int G = (int)xs.size();
+  TORCH_CHECK(G > 0 && G <= WG_MAX_G, "wgrad_mfma: 1..24 problems");
+  WgArgs args{};
+  args.G = G;
+  long long ntiles = 0;
+  for (int g = 0; g < G; ++g) {
+    const int K = (int)xs[g].size(-1), N = (int)dys[g].size(-1);
+    TORCH_CHECK(N % 128 == 0 && K % 128 == 0, "wgrad_mfma: N,K % 128");
+    TORCH_CHECK(xs[g].numel() / K == R && dys[g].numel() / N == R, "R mismatch");
+    TORCH_CHECK(accs[g].numel() == (long long)N * K, "acc slice size");
+    args.x[g] = (unsigned long long)xs[g].data_ptr();
+    args.dy[g] = (unsigned long long)dys[g].data_ptr();
+    args.acc[g] = (unsigned long long)accs[g].data_ptr();
+    args.nk[g * 2 + 0] = N;
+    args.nk[g * 2 + 1] = K;
+    ntiles += (long long)(N / 128) * (K / 128);
+  }
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 32768, stream,
-                     (const unsigned long long*)x_ptrs.data_ptr<int64_t>(),
-                     (const unsigned long long*)dy_ptrs.data_ptr<int64_t>(),
-                     (const unsigned long long*)acc_ptrs.data_ptr<int64_t>(),
-                     dims.data_ptr<int>(), (const WgTile*)tiles.data_ptr<int>(),
-                     (int)R);
+  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 32768,
+                     stream, args, (int)R);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
 }
 

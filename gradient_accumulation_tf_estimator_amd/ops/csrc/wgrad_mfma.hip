@@ -55,26 +55,34 @@ static __device__ __forceinline__ void stage_T64(const unsigned short* g,
   }
 }
 
-// tile table entry: problem id + tile coords
-struct WgTile {
-  int g;
-  int tn;
-  int tk;
-  int pad;
+// All problem metadata travels BY VALUE in the kernel arguments and the
+// tile -> (problem, tn, tk) map is computed in-kernel (G <= 24 scalar loop),
+// so the launch is hipGraph-capture-safe with zero device-side metadata.
+#define WG_MAX_G 24
+struct WgArgs {
+  unsigned long long x[WG_MAX_G];
+  unsigned long long dy[WG_MAX_G];
+  unsigned long long acc[WG_MAX_G];
+  int nk[WG_MAX_G * 2];  // (N, K) pairs
+  int G;
 };
 
 extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
-    const unsigned long long* __restrict__ x_ptrs,
-    const unsigned long long* __restrict__ dy_ptrs,
-    const unsigned long long* __restrict__ acc_ptrs,
-    const int* __restrict__ dims,  // [G][2] = (N, K)
-    const WgTile* __restrict__ tiles, int R) {
-  const WgTile t = tiles[blockIdx.x];
-  const int N = dims[t.g * 2 + 0], K = dims[t.g * 2 + 1];
-  const unsigned short* x = (const unsigned short*)x_ptrs[t.g];
-  const unsigned short* dy = (const unsigned short*)dy_ptrs[t.g];
-  float* acc_out = (float*)acc_ptrs[t.g];
-  const int n0 = t.tn * 128, k0 = t.tk * 128;
+    WgArgs args, int R) {
+  int id = blockIdx.x, g = 0;
+  int N = 0, K = 0;
+  for (; g < args.G; ++g) {
+    N = args.nk[g * 2 + 0];
+    K = args.nk[g * 2 + 1];
+    const int tg = (N / 128) * (K / 128);
+    if (id < tg) break;
+    id -= tg;
+  }
+  const int tn = id / (K / 128), tk = id % (K / 128);
+  const unsigned short* x = (const unsigned short*)args.x[g];
+  const unsigned short* dy = (const unsigned short*)args.dy[g];
+  float* acc_out = (float*)args.acc[g];
+  const int n0 = tn * 128, k0 = tk * 128;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;

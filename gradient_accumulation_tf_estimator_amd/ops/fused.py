@@ -77,29 +77,8 @@ def flush_pending_wgrads() -> None:
     xs, dys, accs = zip(*pending)
     R = xs[0].numel() // xs[0].shape[-1]
 
-    key = tuple((x.data_ptr(), dy.data_ptr(), a.data_ptr(),
-                 x.shape[-1], dy.shape[-1]) for x, dy, a in pending) + (R,)
-    entry = _wgrad_table_cache.get(key)
-    if entry is None:
-        dev = xs[0].device
-        x_ptrs = torch.tensor([x.data_ptr() for x in xs], dtype=torch.int64, device=dev)
-        dy_ptrs = torch.tensor([d.data_ptr() for d in dys], dtype=torch.int64, device=dev)
-        acc_ptrs = torch.tensor([a.data_ptr() for a in accs], dtype=torch.int64, device=dev)
-        dims = torch.tensor([[d.shape[-1], x.shape[-1]] for x, d in zip(xs, dys)],
-                            dtype=torch.int32, device=dev)
-        tiles = []
-        for g, (x, dy) in enumerate(zip(xs, dys)):
-            N, K = dy.shape[-1], x.shape[-1]
-            for tn in range(N // 128):
-                for tk in range(K // 128):
-                    tiles.append((g, tn, tk, 0))
-        tiles_t = torch.tensor(tiles, dtype=torch.int32, device=dev)
-        entry = (x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, len(tiles))
-        if len(_wgrad_table_cache) > 32:
-            _wgrad_table_cache.clear()
-        _wgrad_table_cache[key] = entry
-    x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, ntiles = entry
-    hip.wgrad_mfma(x_ptrs, dy_ptrs, acc_ptrs, dims, tiles_t, R, ntiles)
+    hip.wgrad_mfma(list(xs), list(dys),
+                   [a.reshape(-1) for a in accs], R)
 
 
 def set_wgrad_overlap(enabled: bool) -> None:
