@@ -361,7 +361,7 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     ntiles += (long long)(N / 128) * (K / 128);
   }
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 32768,
+  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
                      stream, args, (int)R);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
 }

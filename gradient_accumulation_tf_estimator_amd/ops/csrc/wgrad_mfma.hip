@@ -29,24 +29,41 @@ static inline __device__ int wswz(int row, int byte_in_row) {
 
 // Transpose-stage a [64 r][C cols] global panel (row stride = ld elems,
 // start col c0) into LDS [C][64 r] (128 B rows). 256 threads, C in {128}.
+// split into issue (global->regs) and write (regs->LDS transposed) so the
+// load latency hides under the previous chunk's MFMAs (T14 async-STAGE).
 template <int C>
-static __device__ __forceinline__ void stage_T64(const unsigned short* g,
-                                                 long long ld, int c0,
-                                                 unsigned short* lds) {
-  constexpr int ITER = (64 / 4) * (C / 8) / 256;  // 4r x 8c sub-blocks
+struct StageRegs {
+  static constexpr int ITER = (64 / 4) * (C / 8) / 256;
+  bf16x8 r[ITER][4];
+};
+
+template <int C>
+static __device__ __forceinline__ void stage_issue(const unsigned short* g,
+                                                   long long ld, int c0,
+                                                   StageRegs<C>& sr) {
 #pragma unroll
-  for (int i = 0; i < ITER; ++i) {
+  for (int i = 0; i < StageRegs<C>::ITER; ++i) {
     const int blk = threadIdx.x + i * 256;
     const int r0 = (blk / (C / 8)) * 4;
     const int cc = (blk % (C / 8)) * 8;
-    bf16x8 r0v = *(const bf16x8*)(g + (long long)(r0 + 0) * ld + c0 + cc);
-    bf16x8 r1v = *(const bf16x8*)(g + (long long)(r0 + 1) * ld + c0 + cc);
-    bf16x8 r2v = *(const bf16x8*)(g + (long long)(r0 + 2) * ld + c0 + cc);
-    bf16x8 r3v = *(const bf16x8*)(g + (long long)(r0 + 3) * ld + c0 + cc);
-    const unsigned short* u0 = (const unsigned short*)&r0v;
-    const unsigned short* u1 = (const unsigned short*)&r1v;
-    const unsigned short* u2 = (const unsigned short*)&r2v;
-    const unsigned short* u3 = (const unsigned short*)&r3v;
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+      sr.r[i][t] = *(const bf16x8*)(g + (long long)(r0 + t) * ld + c0 + cc);
+  }
+}
+
+template <int C>
+static __device__ __forceinline__ void stage_write(unsigned short* lds,
+                                                   const StageRegs<C>& sr) {
+#pragma unroll
+  for (int i = 0; i < StageRegs<C>::ITER; ++i) {
+    const int blk = threadIdx.x + i * 256;
+    const int r0 = (blk / (C / 8)) * 4;
+    const int cc = (blk % (C / 8)) * 8;
+    const unsigned short* u0 = (const unsigned short*)&sr.r[i][0];
+    const unsigned short* u1 = (const unsigned short*)&sr.r[i][1];
+    const unsigned short* u2 = (const unsigned short*)&sr.r[i][2];
+    const unsigned short* u3 = (const unsigned short*)&sr.r[i][3];
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       ush4w pack = {u0[c], u1[c], u2[c], u3[c]};
@@ -91,8 +108,13 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
   const int wn = wave >> 1, wk = wave & 1;  // 2x2 waves -> 64x64 quadrants
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  unsigned short* dyT = (unsigned short*)smem;            // [128 n][64 r]
-  unsigned short* xT = (unsigned short*)(smem + 16384);   // [128 k][64 r]
+  // double-buffered [128][64 r] panels: dyT/xT x {0,1}
+  unsigned short* dyT[2];
+  unsigned short* xT[2];
+  dyT[0] = (unsigned short*)smem;
+  xT[0] = (unsigned short*)(smem + 16384);
+  dyT[1] = (unsigned short*)(smem + 32768);
+  xT[1] = (unsigned short*)(smem + 49152);
 
   f32x16 acc[2][2];
 #pragma unroll
@@ -100,23 +122,38 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.f);
 
+  // 2-phase pipeline (cdna_hip_programming.md T3 minimum form + T14 split):
+  // issue chunk t+1's loads before chunk t's MFMAs; write them to the other
+  // LDS buffer after the barrier.
+  StageRegs<128> sdy, sx;
+  stage_issue<128>(dy, N, n0, sdy);
+  stage_issue<128>(x, K, k0, sx);
+  stage_write<128>(dyT[0], sdy);
+  stage_write<128>(xT[0], sx);
+  int cur = 0;
   for (int r0 = 0; r0 < R; r0 += 64) {
-    stage_T64<128>(dy + (long long)r0 * N, N, n0, dyT);
-    stage_T64<128>(x + (long long)r0 * K, K, k0, xT);
+    if (r0 + 64 < R) {
+      stage_issue<128>(dy + (long long)(r0 + 64) * N, N, n0, sdy);
+      stage_issue<128>(x + (long long)(r0 + 64) * K, K, k0, sx);
+    }
     __syncthreads();
 #pragma unroll
     for (int s = 0; s < 4; ++s) {  // 16-deep r sub-steps
-      // A: dy^T rows (n), B: x^T rows (k); both 16 B swizzled reads
-      bf16x8 a0 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + lo31, s * 32 + hi * 16));
-      bf16x8 a1 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
-      bf16x8 b0 = *(const bf16x8*)((char*)xT + wswz(wk * 64 + lo31, s * 32 + hi * 16));
-      bf16x8 b1 = *(const bf16x8*)((char*)xT + wswz(wk * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 a0 = *(const bf16x8*)((char*)dyT[cur] + wswz(wn * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 a1 = *(const bf16x8*)((char*)dyT[cur] + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b0 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 b1 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 64 + 32 + lo31, s * 32 + hi * 16));
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
     }
+    if (r0 + 64 < R) {
+      stage_write<128>(dyT[cur ^ 1], sdy);
+      stage_write<128>(xT[cur ^ 1], sx);
+    }
     __syncthreads();
+    cur ^= 1;
   }
 
   // epilogue: fp32 accumulate into acc_out[n][k] (D: col=k, rows n)
