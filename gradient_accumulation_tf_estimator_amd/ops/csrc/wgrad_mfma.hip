@@ -156,17 +156,46 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
     cur ^= 1;
   }
 
-  // epilogue: fp32 accumulate into acc_out[n][k] (D: col=k, rows n)
+  // epilogue: bounce the 128x128 fp32 tile through LDS (reusing the staging
+  // buffers) so the global accumulate is coalesced full rows -- per-lane
+  // scattered 4 B RMWs amplify ~16x through 64 B cache lines.
+  float* ftile = (float*)smem;  // [128 k][128 n] fp32, 512 B rows, swizzled
+  __syncthreads();
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
-      float* base = acc_out + (long long)(n0 + wn * 64 + i * 32) * K + k0 +
-                    wk * 64 + j * 32 + lo31;
+      const int k = k0 - k0 + wk * 64 + j * 32 + lo31;  // tile-local k (col)
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        base[(long long)row * K] += acc[i][j][r];
+      for (int rq = 0; rq < 4; ++rq) {  // pack 4 consecutive n per write
+        const int n_base = wn * 64 + i * 32 + 8 * rq + 4 * hi;
+        float4 pk = make_float4(acc[i][j][rq * 4 + 0], acc[i][j][rq * 4 + 1],
+                                acc[i][j][rq * 4 + 2], acc[i][j][rq * 4 + 3]);
+        *(float4*)((char*)ftile + (long long)k * 512 +
+                   ((n_base * 4) ^ ((k & 7) << 4))) = pk;
       }
     }
+  __syncthreads();
+  // coalesced rows: thread -> (n row, 64-wide k half), float4 global RMW
+  {
+    const int n = threadIdx.x >> 1;
+    const int kh = (threadIdx.x & 1) * 64;
+    float* grow = acc_out + (long long)(n0 + n) * K + k0 + kh;
+#pragma unroll
+    for (int c = 0; c < 16; ++c) {
+      const int k4 = kh + c * 4;
+      float4 v;
+      v.x = *(const float*)((char*)ftile + (long long)(k4 + 0) * 512 +
+                            ((n * 4) ^ (((k4 + 0) & 7) << 4)));
+      v.y = *(const float*)((char*)ftile + (long long)(k4 + 1) * 512 +
+                            ((n * 4) ^ (((k4 + 1) & 7) << 4)));
+      v.z = *(const float*)((char*)ftile + (long long)(k4 + 2) * 512 +
+                            ((n * 4) ^ (((k4 + 2) & 7) << 4)));
+      v.w = *(const float*)((char*)ftile + (long long)(k4 + 3) * 512 +
+                            ((n * 4) ^ (((k4 + 3) & 7) << 4)));
+      float4 old = *(const float4*)(grow + c * 4);
+      old.x += v.x; old.y += v.y; old.z += v.z; old.w += v.w;
+      *(float4*)(grow + c * 4) = old;
+    }
+  }
 }
