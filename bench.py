@@ -44,7 +44,7 @@ def parse_args():
                    help="EXPERIMENTAL: wgrad GEMMs on a side HIP stream "
                         "(measured slower under hipGraphs; also needs a "
                         "per-stream hipBLASLt workspace before production)")
-    p.add_argument("--grouped-wgrad", default="off", choices=["on", "off"],
+    p.add_argument("--grouped-wgrad", default="on", choices=["on", "off"],
                    help="issue the micro-step wgrads as one grouped hipBLASLt launch")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")

@@ -86,7 +86,15 @@ struct WgArgs {
 
 extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
     WgArgs args, int R) {
-  int id = blockIdx.x, g = 0;
+  // XCD-aware bijective remap (T1): consecutive tile ids share operand
+  // panels; keep runs on one XCD so re-reads hit that XCD's L2.
+  int id;
+  {
+    const int nwg = gridDim.x, orig = blockIdx.x;
+    const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  }
+  int g = 0;
   int N = 0, K = 0;
   for (; g < args.G; ++g) {
     N = args.nk[g * 2 + 0];
