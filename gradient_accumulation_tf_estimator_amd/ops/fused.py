@@ -77,8 +77,10 @@ def flush_pending_wgrads() -> None:
     xs, dys, accs = zip(*pending)
     R = xs[0].numel() // xs[0].shape[-1]
 
-    hip.wgrad_mfma(list(xs), list(dys),
-                   [a.reshape(-1) for a in accs], R)
+    # the kernel takes metadata by value: at most 24 problems per launch
+    for lo in range(0, len(xs), 24):
+        hip.wgrad_mfma(list(xs[lo : lo + 24]), list(dys[lo : lo + 24]),
+                       [a.reshape(-1) for a in accs[lo : lo + 24]], R)
 
 
 def set_wgrad_overlap(enabled: bool) -> None:
