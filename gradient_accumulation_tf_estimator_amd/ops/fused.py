@@ -67,7 +67,10 @@ def flush_pending_wgrads() -> None:
     pending = []
     for x, dy, acc in all_pending:
         R_i = x.numel() // x.shape[-1]
+        # the single-launch MFMA path wins at the bench micro-batch; at large
+        # R (seq512 configs) hipBLASLt's split-K wgrads are faster
         if (x.shape[-1] % 128 == 0 and dy.shape[-1] % 128 == 0 and R_i % 64 == 0
+                and R_i <= 2048
                 and (not pending or R_i == pending[0][0].numel() // pending[0][0].shape[-1])):
             pending.append((x, dy, acc))
         else:
