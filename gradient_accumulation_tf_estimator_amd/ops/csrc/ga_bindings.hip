@@ -57,6 +57,7 @@ struct WgArgs {
   unsigned long long x[WG_MAX_G];
   unsigned long long dy[WG_MAX_G];
   unsigned long long acc[WG_MAX_G];
+  unsigned long long dbias[WG_MAX_G];
   int nk[WG_MAX_G * 2];
   int G;
 };
@@ -342,9 +343,11 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
 // One launch over every pending wgrad; metadata passed by value (see
 // wgrad_mfma.hip) so it is hipGraph-capture-safe.
 void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
-                std::vector<at::Tensor> accs, int64_t R) {
+                std::vector<at::Tensor> accs, std::vector<at::Tensor> dbias,
+                int64_t R) {
   const int G = (int)xs.size();
   TORCH_CHECK(G > 0 && G <= WG_MAX_G, "wgrad_mfma: 1..24 problems");
+  TORCH_CHECK(dbias.empty() || dbias.size() == xs.size(), "dbias size");
   WgArgs args{};
   args.G = G;
   long long ntiles = 0;
@@ -356,6 +359,11 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     args.x[g] = (unsigned long long)xs[g].data_ptr();
     args.dy[g] = (unsigned long long)dys[g].data_ptr();
     args.acc[g] = (unsigned long long)accs[g].data_ptr();
+    if (!dbias.empty() && dbias[g].numel() > 0) {
+      TORCH_CHECK(dbias[g].numel() == N && dbias[g].scalar_type() == at::kFloat,
+                  "dbias slice must be fp32 [N]");
+      args.dbias[g] = (unsigned long long)dbias[g].data_ptr();
+    }
     args.nk[g * 2 + 0] = N;
     args.nk[g * 2 + 1] = K;
     ntiles += (long long)(N / 128) * (K / 128);

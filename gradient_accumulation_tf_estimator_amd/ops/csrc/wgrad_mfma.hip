@@ -80,6 +80,7 @@ struct WgArgs {
   unsigned long long x[WG_MAX_G];
   unsigned long long dy[WG_MAX_G];
   unsigned long long acc[WG_MAX_G];
+  unsigned long long dbias[WG_MAX_G];  // fp32 accum slice for colsum(dy); 0 = skip
   int nk[WG_MAX_G * 2];  // (N, K) pairs
   int G;
 };
@@ -130,6 +131,13 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = (f32x16)(0.f);
 
+  // tk==0 blocks also fold the Linear bias gradient (colsum over r of dy)
+  // out of the already-staged dy panels: thread -> (n = tid/2, r-half)
+  const bool do_bias = (tk == 0) && args.dbias[g] != 0ull;
+  const int bn = threadIdx.x >> 1;            // n within tile
+  const int brh = (threadIdx.x & 1) * 32;     // r half
+  float bias_acc = 0.f;
+
   // 2-phase pipeline (cdna_hip_programming.md T3 minimum form + T14 split):
   // issue chunk t+1's loads before chunk t's MFMAs; write them to the other
   // LDS buffer after the barrier.
@@ -145,6 +153,14 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
       stage_issue<128>(x + (long long)(r0 + 64) * K, K, k0, sx);
     }
     __syncthreads();
+    if (do_bias) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8 v = *(const bf16x8*)((char*)dyT[cur] + wswz(bn, (brh + c * 8) * 2));
+#pragma unroll
+        for (int e = 0; e < 8; ++e) bias_acc += (float)v[e];
+      }
+    }
 #pragma unroll
     for (int s = 0; s < 4; ++s) {  // 16-deep r sub-steps
       bf16x8 a0 = *(const bf16x8*)((char*)dyT[cur] + wswz(wn * 64 + lo31, s * 32 + hi * 16));
@@ -162,6 +178,13 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
     }
     __syncthreads();
     cur ^= 1;
+  }
+
+  if (do_bias) {
+    // combine the (n, r-half) pairs: adjacent lanes share n
+    bias_acc += __shfl_xor(bias_acc, 1, 64);
+    if ((threadIdx.x & 1) == 0)
+      ((float*)args.dbias[g])[n0 + bn] += bias_acc;
   }
 
   // epilogue: bounce the 128x128 fp32 tile through LDS (reusing the staging

@@ -65,25 +65,32 @@ def flush_pending_wgrads() -> None:
     from . import gemm
 
     pending = []
-    for x, dy, acc in all_pending:
+    for x, dy, acc, vb in all_pending:
         R_i = x.numel() // x.shape[-1]
         # the single-launch MFMA path wins at the bench micro-batch; at large
         # R (seq512 configs) hipBLASLt's split-K wgrads are faster
         if (x.shape[-1] % 128 == 0 and dy.shape[-1] % 128 == 0 and R_i % 64 == 0
                 and R_i <= 2048
                 and (not pending or R_i == pending[0][0].numel() // pending[0][0].shape[-1])):
-            pending.append((x, dy, acc))
+            pending.append((x, dy, acc, vb))
         else:
             gemm.wgrad_acc(x, dy, acc.view(dy.shape[-1], x.shape[-1]))
+            if vb is not None:
+                vb.add_(dy.sum(0, dtype=torch.float32))
     if not pending:
         return
-    xs, dys, accs = zip(*pending)
+    xs, dys, accs, vbs = zip(*pending)
     R = xs[0].numel() // xs[0].shape[-1]
+    empty = None
 
     # the kernel takes metadata by value: at most 24 problems per launch
     for lo in range(0, len(xs), 24):
+        chunk_vbs = vbs[lo : lo + 24]
+        if empty is None and any(v is None for v in chunk_vbs):
+            empty = torch.empty(0, dtype=torch.float32, device=xs[0].device)
         hip.wgrad_mfma(list(xs[lo : lo + 24]), list(dys[lo : lo + 24]),
-                       [a.reshape(-1) for a in accs[lo : lo + 24]], R)
+                       [a.reshape(-1) for a in accs[lo : lo + 24]],
+                       [v if v is not None else empty for v in chunk_vbs], R)
 
 
 def set_wgrad_overlap(enabled: bool) -> None:
@@ -253,8 +260,9 @@ class _DirectLinearFn(torch.autograd.Function):
         dy2d = dy.reshape(-1, dy.shape[-1])
         dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
         x2d = x.reshape(-1, x.shape[-1])
+        vb = ctx.module._accum_view_b
         if _GROUPED_WGRAD:
-            _pending_wgrads.append((x2d, dy2d, ctx.module._accum_view_w))
+            _pending_wgrads.append((x2d, dy2d, ctx.module._accum_view_w, vb))
         elif _WGRAD_OVERLAP:
             s = wgrad_stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -262,9 +270,13 @@ class _DirectLinearFn(torch.autograd.Function):
                 gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
             x2d.record_stream(s)
             dy2d.record_stream(s)
+            if vb is not None:
+                vb.add_(dy2d.sum(0, dtype=torch.float32))
         else:
             gemm.wgrad_acc(x2d, dy2d, ctx.module._accum_view_w)
-        db = dy2d.sum(0) if ctx.has_bias else None
+            if vb is not None:
+                vb.add_(dy2d.sum(0, dtype=torch.float32))
+        db = dy2d.sum(0) if (ctx.has_bias and vb is None) else None
         return dx, None, db, None
 
 
@@ -283,6 +295,7 @@ class DirectLinear(nn.Module):
             self.register_parameter("bias", None)
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
         self._accum_view_w = None
+        self._accum_view_b = None
 
     def forward(self, x):
         if self._accum_view_w is not None:
@@ -387,6 +400,8 @@ def direct_param_names(model: nn.Module):
             names.append(pname(mod, "bias"))
         elif isinstance(mod, (DirectLinear, DirectEmbedding)):
             names.append(pname(mod, "weight"))
+            if isinstance(mod, DirectLinear) and mod.bias is not None:
+                names.append(pname(mod, "bias"))
     return names
 
 
@@ -412,6 +427,8 @@ def bind_direct_grad(model: nn.Module, engine) -> int:
         elif isinstance(mod, DirectLinear):
             mod._accum_view_w = engine.state.accum_view(mod.weight).view(
                 mod.out_features, mod.in_features)
+            if mod.bias is not None:
+                mod._accum_view_b = engine.state.accum_view(mod.bias)
             n += 1
         elif isinstance(mod, DirectEmbedding):
             mod._accum_view_w = engine.state.accum_view(mod.weight).view(
