@@ -219,23 +219,33 @@ def test_wgrad_mfma_batched_matches_reference():
     torch.manual_seed(0)
     R = 256
     shapes = [(512, 1536), (512, 512), (2048, 512), (512, 2048)]  # (K, N)
-    xs, dys, accs, refs = [], [], [], []
-    for K, N in shapes:
+    xs, dys, accs, vbs, refs, vrefs = [], [], [], [], [], []
+    for idx, (K, N) in enumerate(shapes):
         x = (torch.randn(R, K, device="cuda") * 0.3).bfloat16()
         dy = (torch.randn(R, N, device="cuda") * 0.3).bfloat16()
         a = torch.randn(N * K, device="cuda")
         refs.append(a.view(N, K) + dy.t().float() @ x.float())
+        vb = torch.randn(N, device="cuda") if idx % 2 == 0 else None
+        vbs.append(vb)
+        vrefs.append(vb + dy.float().sum(0) if vb is not None else None)
+        if vb is not None:
+            vb = vb.clone()
+            vbs[-1] = vb
         xs.append(x); dys.append(dy); accs.append(a)
 
     fops.set_grouped_wgrad(True)
     try:
-        for x, dy, a in zip(xs, dys, accs):
-            fops._pending_wgrads.append((x, dy, a.view(dy.shape[-1], x.shape[-1])))
+        for x, dy, a, vb in zip(xs, dys, accs, vbs):
+            fops._pending_wgrads.append(
+                (x, dy, a.view(dy.shape[-1], x.shape[-1]), vb))
         fops.flush_pending_wgrads()
     finally:
         fops.set_grouped_wgrad(False)
     torch.cuda.synchronize()
-    for (K, N), a, ref in zip(shapes, accs, refs):
+    for (K, N), a, ref, vb, vref in zip(shapes, accs, refs, vbs, vrefs):
         np.testing.assert_allclose(a.view(N, K).cpu(), ref.cpu(),
                                    rtol=2e-2, atol=3e-1,
                                    err_msg=f"wgrad mismatch for K={K},N={N}")
+        if vb is not None:
+            np.testing.assert_allclose(vb.cpu(), vref.cpu(), rtol=2e-2, atol=3e-1,
+                                       err_msg=f"dbias mismatch for N={N}")
