@@ -62,6 +62,7 @@ struct WgArgs {
   int G;
 };
 extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
+extern "C" __global__ void k_wgrad_mfma256(WgArgs, int);
 
 namespace {
 
@@ -368,9 +369,20 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     args.nk[g * 2 + 1] = K;
     ntiles += (long long)(N / 128) * (K / 128);
   }
+  bool all256 = true;
+  for (int g = 0; g < G; ++g)
+    all256 &= (args.nk[g * 2] % 256 == 0) && (args.nk[g * 2 + 1] % 256 == 0);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
-                     stream, args, (int)R);
+  if (all256) {
+    long long nt256 = 0;
+    for (int g = 0; g < G; ++g)
+      nt256 += (long long)(args.nk[g * 2] / 256) * (args.nk[g * 2 + 1] / 256);
+    hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)nt256), dim3(512), 65536,
+                       stream, args, (int)R);
+  } else {
+    hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
+                       stream, args, (int)R);
+  }
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
 }
 

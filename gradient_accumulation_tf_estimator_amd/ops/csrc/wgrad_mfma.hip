@@ -230,3 +230,177 @@ extern "C" __global__ __launch_bounds__(256) void k_wgrad_mfma(
     }
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// 256x256-tile variant (8 waves / 512 threads): halves operand panel
+// re-reads vs the 128 kernel. Requires every N,K % 256 == 0 (all BERT
+// encoder shapes); single-buffered 64 KB staging.
+// ---------------------------------------------------------------------------
+struct StageRegs512 {
+  bf16x8 r[4];
+};
+
+// [64 r][256 c] global panel chunk -> regs (4r x 8c sub-block per thread)
+static __device__ __forceinline__ void stage512_issue(const unsigned short* g,
+                                                      long long ld, int c0,
+                                                      StageRegs512& sr) {
+  const int blk = threadIdx.x;  // 512 sub-blocks exactly
+  const int r0 = (blk / 32) * 4;
+  const int cc = (blk % 32) * 8;
+#pragma unroll
+  for (int t = 0; t < 4; ++t)
+    sr.r[t] = *(const bf16x8*)(g + (long long)(r0 + t) * ld + c0 + cc);
+}
+
+static __device__ __forceinline__ void stage512_write(unsigned short* lds,
+                                                      const StageRegs512& sr) {
+  const int blk = threadIdx.x;
+  const int r0 = (blk / 32) * 4;
+  const int cc = (blk % 32) * 8;
+  const unsigned short* u0 = (const unsigned short*)&sr.r[0];
+  const unsigned short* u1 = (const unsigned short*)&sr.r[1];
+  const unsigned short* u2 = (const unsigned short*)&sr.r[2];
+  const unsigned short* u3 = (const unsigned short*)&sr.r[3];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    ush4w pack = {u0[c], u1[c], u2[c], u3[c]};
+    *(ush4w*)((char*)lds + wswz(cc + c, r0 * 2)) = pack;
+  }
+}
+
+extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
+    WgArgs args, int R) {
+  int id;
+  {
+    const int nwg = gridDim.x, orig = blockIdx.x;
+    const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
+    id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
+  }
+  int g = 0, N = 0, K = 0;
+  for (; g < args.G; ++g) {
+    N = args.nk[g * 2 + 0];
+    K = args.nk[g * 2 + 1];
+    const int tg = (N / 256) * (K / 256);
+    if (id < tg) break;
+    id -= tg;
+  }
+  const int tn = id / (K / 256), tk = id % (K / 256);
+  const unsigned short* x = (const unsigned short*)args.x[g];
+  const unsigned short* dy = (const unsigned short*)args.dy[g];
+  float* acc_out = (float*)args.acc[g];
+  const int n0 = tn * 256, k0 = tk * 256;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int wn = wave >> 1, wk = wave & 1;  // 4(n: 64 rows) x 2(k: 128 cols)
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dyT = (unsigned short*)smem;           // [256 n][64 r]
+  unsigned short* xT = (unsigned short*)(smem + 32768);  // [256 k][64 r]
+
+  f32x16 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x16)(0.f);
+
+  const bool do_bias = (tk == 0) && args.dbias[g] != 0ull;
+  const int bn = threadIdx.x >> 1;         // n within tile (0..255)
+  const int brh = (threadIdx.x & 1) * 32;  // r half
+  float bias_acc = 0.f;
+
+  StageRegs512 sdy, sx;
+  stage512_issue(dy, N, n0, sdy);
+  stage512_issue(x, K, k0, sx);
+  for (int r0 = 0; r0 < R; r0 += 64) {
+    __syncthreads();  // previous chunk's readers done
+    stage512_write(dyT, sdy);
+    stage512_write(xT, sx);
+    if (r0 + 64 < R) {
+      stage512_issue(dy + (long long)(r0 + 64) * N, N, n0, sdy);
+      stage512_issue(x + (long long)(r0 + 64) * K, K, k0, sx);
+    }
+    __syncthreads();
+    if (do_bias) {
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8 v = *(const bf16x8*)((char*)dyT + wswz(bn, (brh + c * 8) * 2));
+#pragma unroll
+        for (int e = 0; e < 8; ++e) bias_acc += (float)v[e];
+      }
+    }
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8 a0 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 a1 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b0 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + lo31, s * 32 + hi * 16));
+      bf16x8 b1 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b2 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 64 + lo31, s * 32 + hi * 16));
+      bf16x8 b3 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 96 + lo31, s * 32 + hi * 16));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[0][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b2, acc[0][2], 0, 0, 0);
+      acc[0][3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b3, acc[0][3], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[1][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b2, acc[1][2], 0, 0, 0);
+      acc[1][3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b3, acc[1][3], 0, 0, 0);
+    }
+  }
+
+  if (do_bias) {
+    bias_acc += __shfl_xor(bias_acc, 1, 64);
+    if ((threadIdx.x & 1) == 0)
+      ((float*)args.dbias[g])[n0 + bn] += bias_acc;
+  }
+
+  // epilogue in 4 passes of 64 k-rows, bounced through a [64 k][256 n]
+  // fp32 LDS tile (64 KB, reusing the staging buffers)
+  float* ftile = (float*)smem;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    __syncthreads();
+    if (wk == (p >> 1)) {
+#pragma unroll
+      for (int jj = 0; jj < 2; ++jj) {
+        const int j = (p & 1) * 2 + jj;
+        const int kl = j * 32 + lo31 - (p & 1) * 64;  // k within this pass
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int rq = 0; rq < 4; ++rq) {
+            const int nb = wn * 64 + i * 32 + 8 * rq + 4 * hi;
+            float4 pk = make_float4(acc[i][j][rq * 4 + 0], acc[i][j][rq * 4 + 1],
+                                    acc[i][j][rq * 4 + 2], acc[i][j][rq * 4 + 3]);
+            *(float4*)((char*)ftile + (long long)kl * 1024 +
+                       ((nb * 4) ^ ((kl & 7) << 4))) = pk;
+          }
+      }
+    }
+    __syncthreads();
+    {
+      const int n = threadIdx.x >> 1;
+      const int kh = (threadIdx.x & 1) * 32;
+      float* grow = acc_out + (long long)(n0 + n) * K + k0 + p * 64 + kh;
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        const int k4 = kh + c * 4;
+        float4 v;
+        v.x = *(const float*)((char*)ftile + (long long)(k4 + 0) * 1024 +
+                              ((n * 4) ^ (((k4 + 0) & 7) << 4)));
+        v.y = *(const float*)((char*)ftile + (long long)(k4 + 1) * 1024 +
+                              ((n * 4) ^ (((k4 + 1) & 7) << 4)));
+        v.z = *(const float*)((char*)ftile + (long long)(k4 + 2) * 1024 +
+                              ((n * 4) ^ (((k4 + 2) & 7) << 4)));
+        v.w = *(const float*)((char*)ftile + (long long)(k4 + 3) * 1024 +
+                              ((n * 4) ^ (((k4 + 3) & 7) << 4)));
+        float4 old = *(const float4*)(grow + c * 4);
+        old.x += v.x; old.y += v.y; old.z += v.z; old.w += v.w;
+        *(float4*)(grow + c * 4) = old;
+      }
+    }
+  }
+}
