@@ -26,6 +26,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.fused import (
+    CEClassifier,
     DirectEmbedding,
     DirectLinear,
     FusedAddLayerNorm,
@@ -189,8 +190,8 @@ class BertModel(nn.Module):
         x = self.embeddings(input_ids, token_type_ids)
         for layer in self.encoder:
             x = layer(x, mask)
-        pooled = torch.tanh(self.pooler(x[:, 0]))
-        return x, pooled
+        # pre-tanh pooler output; the tanh lives in the head (fused CE path)
+        return x, self.pooler(x[:, 0])
 
 
 class BertForSequenceClassification(nn.Module):
@@ -200,19 +201,15 @@ class BertForSequenceClassification(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.bert = BertModel(cfg)
-        # classifier output dim (num_labels=2) is far below the GEMM kernels'
-        # alignment; plain nn.Linear keeps it on the stock torch path
-        self.classifier = nn.Linear(cfg.hidden_size, cfg.num_labels)
-        nn.init.normal_(self.classifier.weight, std=cfg.initializer_range)
-        nn.init.zeros_(self.classifier.bias)
+        self.classifier = CEClassifier(cfg.hidden_size, cfg.num_labels)
 
     def forward(self, input_ids, token_type_ids=None, attention_mask=None):
-        _, pooled = self.bert(input_ids, token_type_ids, attention_mask)
-        return self.classifier(pooled)
+        _, pre = self.bert(input_ids, token_type_ids, attention_mask)
+        return self.classifier.logits(pre)
 
     def loss(self, input_ids, labels, token_type_ids=None, attention_mask=None):
-        logits = self.forward(input_ids, token_type_ids, attention_mask)
-        return F.cross_entropy(logits.float(), labels)
+        _, pre = self.bert(input_ids, token_type_ids, attention_mask)
+        return self.classifier.loss(pre, labels)
 
 
 def build_model(name: str, **overrides) -> BertForSequenceClassification:

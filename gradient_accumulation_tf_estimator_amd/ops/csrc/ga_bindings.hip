@@ -63,6 +63,13 @@ struct WgArgs {
 };
 extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
 extern "C" __global__ void k_wgrad_mfma256(WgArgs, int);
+extern "C" __global__ void k_cls_head_fwd(const unsigned short*, const unsigned short*,
+                                          const unsigned short*, const long long*,
+                                          unsigned short*, float*, float*, int, int, int);
+extern "C" __global__ void k_cls_head_bwd(const float*, const unsigned short*,
+                                          const float*, const long long*,
+                                          const unsigned short*, unsigned short*,
+                                          float*, float*, int, int, int);
 
 namespace {
 
@@ -386,6 +393,46 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
 }
 
+std::vector<at::Tensor> cls_head_fwd(at::Tensor pre, at::Tensor Wc, at::Tensor bc,
+                                     at::Tensor labels) {
+  const int B = (int)pre.size(0), H = (int)pre.size(1), C = (int)Wc.size(0);
+  TORCH_CHECK(B <= 64 && C <= 8 && H % 512 == 0 && H <= 1024,
+              "cls head: B<=64, C<=8, H in {512, 1024}");
+  TORCH_CHECK(pre.is_contiguous() && pre.scalar_type() == at::kBFloat16);
+  auto t = at::empty_like(pre);
+  auto probs = at::empty({B, C}, pre.options().dtype(at::kFloat));
+  auto loss = at::empty({}, pre.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_cls_head_fwd, dim3(1), dim3(256), 0, stream,
+                     (const unsigned short*)pre.data_ptr(),
+                     (const unsigned short*)Wc.data_ptr(),
+                     (const unsigned short*)bc.data_ptr(),
+                     (const long long*)labels.data_ptr<int64_t>(),
+                     (unsigned short*)t.data_ptr(), probs.data_ptr<float>(),
+                     loss.data_ptr<float>(), B, H, C);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_cls_head_fwd launch failed");
+  return {loss, t, probs};
+}
+
+at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
+                        at::Tensor labels, at::Tensor Wc,
+                        c10::optional<at::Tensor> acc_w,
+                        c10::optional<at::Tensor> acc_b) {
+  const int B = (int)t.size(0), H = (int)t.size(1), C = (int)Wc.size(0);
+  auto dpre = at::empty_like(t);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_cls_head_bwd, dim3(1), dim3(256), 0, stream,
+                     dloss.data_ptr<float>(), (const unsigned short*)t.data_ptr(),
+                     probs.data_ptr<float>(),
+                     (const long long*)labels.data_ptr<int64_t>(),
+                     (const unsigned short*)Wc.data_ptr(),
+                     (unsigned short*)dpre.data_ptr(),
+                     acc_w ? acc_w->data_ptr<float>() : nullptr,
+                     acc_b ? acc_b->data_ptr<float>() : nullptr, B, H, C);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_cls_head_bwd launch failed");
+  return dpre;
+}
+
 void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t H) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
               "dy must be contiguous bf16");
@@ -427,6 +474,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("embgrad_acc", &embgrad_acc,
           "scatter-add embedding grads into the flat fp32 accum slice");
   mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");
+  mod.def("cls_head_fwd", &cls_head_fwd, "tanh+classifier+CE forward");
+  mod.def("cls_head_bwd", &cls_head_bwd, "fused head backward -> d(pre-tanh)");
   mod.def("wgrad_mfma", &wgrad_mfma,
           "batched MFMA wgrad: accum_g += dy_g^T @ x_g over a tile table");
   mod.def("attn_bwd", &attn_bwd, "fused MFMA attention bwd -> packed dqkv");

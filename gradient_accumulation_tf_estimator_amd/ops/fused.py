@@ -380,6 +380,56 @@ def fused_attention(qkv: torch.Tensor, nh: int) -> torch.Tensor:
     return _AttentionFn.apply(qkv.view(B, S, 3, H3 // 3), nh)
 
 
+class _ClsHeadFn(torch.autograd.Function):
+    """tanh(pooler-out) -> classifier -> mean softmax-CE, one kernel each
+    way (ops/csrc/cls_head.hip); classifier grads go straight into accum."""
+
+    @staticmethod
+    def forward(ctx, pre, weight, bias, labels, module):
+        hip = require_hip()
+        loss, t, probs = hip.cls_head_fwd(pre.contiguous(), weight, bias, labels)
+        ctx.save_for_backward(t, probs, labels, weight)
+        ctx.module = module
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        hip = require_hip()
+        t, probs, labels, weight = ctx.saved_tensors
+        mod = ctx.module
+        dpre = hip.cls_head_bwd(dloss.contiguous().float().reshape(1), t, probs,
+                                labels, weight, mod._accum_view_w, mod._accum_view_b)
+        return dpre, None, None, None, None
+
+
+class CEClassifier(nn.Module):
+    """Classifier head with a fused tanh+GEMV+cross-entropy loss path.
+
+    ``forward`` (predict/eval) runs plain torch ops; ``loss`` uses the fused
+    kernels when bound to an engine on GPU bf16 (grads accumulate directly).
+    """
+
+    def __init__(self, hidden: int, num_labels: int):
+        super().__init__()
+        self.hidden, self.num_labels = hidden, num_labels
+        self.weight = nn.Parameter(torch.empty(num_labels, hidden))
+        self.bias = nn.Parameter(torch.zeros(num_labels))
+        nn.init.normal_(self.weight, std=0.02)
+        self._accum_view_w = None
+        self._accum_view_b = None
+        self.fusable = num_labels <= 8 and hidden in (512, 1024)
+
+    def logits(self, pre):
+        return F.linear(torch.tanh(pre), self.weight, self.bias)
+
+    def loss(self, pre, labels):
+        if self._accum_view_w is not None:
+            if not (_use_hip(pre) and pre.shape[0] <= 64):
+                raise RuntimeError("bound CEClassifier needs bf16 GPU input, B<=64")
+            return _ClsHeadFn.apply(pre, self.weight, self.bias, labels, self)
+        return F.cross_entropy(self.logits(pre).float(), labels)
+
+
 def direct_param_names(model: nn.Module):
     """Names of params whose grads will bypass .grad when bound on GPU --
     passed to FlatState so K1 can skip their (contiguous) flat region."""
@@ -402,6 +452,9 @@ def direct_param_names(model: nn.Module):
             names.append(pname(mod, "weight"))
             if isinstance(mod, DirectLinear) and mod.bias is not None:
                 names.append(pname(mod, "bias"))
+        elif isinstance(mod, CEClassifier) and mod.fusable:
+            names.append(pname(mod, "weight"))
+            names.append(pname(mod, "bias"))
     return names
 
 
@@ -433,5 +486,9 @@ def bind_direct_grad(model: nn.Module, engine) -> int:
         elif isinstance(mod, DirectEmbedding):
             mod._accum_view_w = engine.state.accum_view(mod.weight).view(
                 mod.num_embeddings, mod.embedding_dim)
+            n += 1
+        elif isinstance(mod, CEClassifier) and mod.fusable:
+            mod._accum_view_w = engine.state.accum_view(mod.weight)
+            mod._accum_view_b = engine.state.accum_view(mod.bias)
             n += 1
     return n
