@@ -126,7 +126,7 @@ def main():
         engine.micro_step()
         return loss
 
-    use_graphs = use_cuda and (args.graphs == "on" or (args.graphs == "auto" and world == 1))
+    use_graphs = use_cuda and args.graphs != "off"
     graphed = None
     if use_graphs:
         try:
@@ -213,15 +213,20 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
     becomes a host-side choice between two captured graphs.
     lr reaches the apply kernel through a device scalar so the schedule
     updates without re-capture.
+
+    Data-parallel (world > 1): only the accumulate micro-step is captured;
+    the apply boundary replays it and then runs the RCCL all-reduce + fused
+    apply eagerly -- K-1 of K steps run at full graph speed with no
+    collective inside any capture.
     """
-    if world > 1:
-        raise RuntimeError("graph capture currently enabled for 1 GPU only")
     assert engine.backend == "hip"
     static_ids = pool_ids[0].clone()
     static_lab = pool_lab[0].clone()
 
     def fwd_bwd_accum():
         loss = model.loss(static_ids, static_lab)
+        if world > 1:
+            loss = loss * (1.0 / world)
         loss.backward()
         engine.accumulate()
         # capture must end with every forked stream joined; within-step
@@ -245,13 +250,16 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
     g_accum = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g_accum):
         loss_accum = fwd_bwd_accum()
-    g_apply = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(g_apply, pool=g_accum.pool()):
-        loss_apply = fwd_bwd_accum()
-        engine.apply_from_device()
-    torch.cuda.synchronize()
+    if world == 1:
+        g_apply = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_apply, pool=g_accum.pool()):
+            loss_apply = fwd_bwd_accum()
+            engine.apply_from_device()
+        torch.cuda.synchronize()
+    else:
+        g_apply = loss_apply = None
+        torch.cuda.synchronize()
 
-    K = engine.K
     POOL = pool_ids.shape[0]
 
     def run(i):
@@ -259,9 +267,17 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
         static_lab.copy_(pool_lab[i % POOL])
         if engine.is_apply_step():
             engine.set_lr(engine.lr_at(engine.global_step))
-            g_apply.replay()
+            if g_apply is not None:
+                g_apply.replay()
+            else:
+                # DP: replay the captured micro-step, then all-reduce the
+                # flat accum buffer over RCCL and apply -- eager, outside
+                # any capture
+                g_accum.replay()
+                engine._allreduce_accum()
+                engine.apply_from_device()
             engine.global_step += 1
-            return loss_apply
+            return loss_apply if loss_apply is not None else loss_accum
         g_accum.replay()
         engine.global_step += 1
         return loss_accum
