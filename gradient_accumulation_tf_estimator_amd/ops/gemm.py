@@ -57,10 +57,19 @@ def _algo_for(kind: int, R: int, N: int, K: int, count_fn, run) -> int:
     return _tune(key, count_fn(), run)
 
 
+# the custom MFMA Linear kernels win in the one-block-wave regime (token
+# rows ~= 1024); bigger shapes go to autotuned hipBLASLt
+mfma_linear_enabled = True
+_MFMA_MAX_R = 2048
+
+
 def linear_fwd(x2d: torch.Tensor, w: torch.Tensor, bias) -> torch.Tensor:
     hip = require_hip()
     N, K = w.shape
     R = x2d.numel() // K
+    if (mfma_linear_enabled and R % 128 == 0 and R <= _MFMA_MAX_R
+            and N % 128 == 0 and K % 64 == 0):
+        return hip.linear_fwd_mfma(x2d, w, bias)
     kind = 1 if bias is not None else 0
     idx = _algo_for(kind, R, N, K,
                     lambda: hip.lt_algo_count(kind, R, N, K),
@@ -72,6 +81,9 @@ def dgrad(dy2d: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     hip = require_hip()
     N, K = w.shape
     R = dy2d.numel() // N
+    if (mfma_linear_enabled and R % 128 == 0 and R <= _MFMA_MAX_R
+            and K % 128 == 0 and N % 64 == 0):
+        return hip.linear_dgrad_mfma(dy2d, w)
     idx = _algo_for(2, R, N, K,
                     lambda: hip.lt_algo_count(2, R, N, K),
                     lambda i: hip.lt_dgrad(dy2d, w, i))

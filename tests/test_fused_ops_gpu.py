@@ -307,3 +307,29 @@ def test_cls_head_matches_torch():
     got = st.accum[sl.offset : sl.offset + sl.numel].cpu().numpy()
     refg = (pre.grad.t().float() @ xf).reshape(-1)
     np.testing.assert_allclose(got, refg.cpu().numpy(), rtol=5e-2, atol=5e-3)
+
+
+@pytest.mark.parametrize("N,K", [(512, 512), (1536, 512), (512, 2048), (2048, 512)])
+def test_linear_mfma_matches_torch(N, K):
+    """Custom MFMA Linear fwd/dgrad vs torch references."""
+    from gradient_accumulation_tf_estimator_amd import ops
+
+    hip = ops.require_hip()
+    torch.manual_seed(N + K)
+    R = 1024
+    x = (torch.randn(R, K, device="cuda") * 0.4).bfloat16()
+    w = (torch.randn(N, K, device="cuda") * 0.05).bfloat16()
+    b = torch.randn(N, device="cuda").bfloat16()
+    dy = (torch.randn(R, N, device="cuda") * 0.4).bfloat16()
+
+    y = hip.linear_fwd_mfma(x, w, b)
+    yref = F.linear(x.float(), w.float(), b.float())
+    np.testing.assert_allclose(y.float().cpu(), yref.cpu(), rtol=3e-2, atol=2e-1)
+
+    y2 = hip.linear_fwd_mfma(x, w, None)
+    np.testing.assert_allclose(y2.float().cpu(), (yref - b.float()).cpu(),
+                               rtol=3e-2, atol=2e-1)
+
+    dx = hip.linear_dgrad_mfma(dy, w)
+    dxref = dy.float() @ w.float()
+    np.testing.assert_allclose(dx.float().cpu(), dxref.cpu(), rtol=3e-2, atol=2e-1)
