@@ -57,9 +57,12 @@ def _algo_for(kind: int, R: int, N: int, K: int, count_fn, run) -> int:
     return _tune(key, count_fn(), run)
 
 
-# the custom MFMA Linear kernels win in the one-block-wave regime (token
-# rows ~= 1024); bigger shapes go to autotuned hipBLASLt
-mfma_linear_enabled = True
+# EXPERIMENTAL: the custom MFMA Linear kernels (ops/csrc/linear_mfma.hip)
+# lose to hipBLASLt at the bench shapes (one serial K-loop per block, no
+# cross-problem TLP -- measured fwd 24 us / dgrad 57 us vs ~10 us); routing
+# stays off until they get the deep-pipeline treatment. The batched wgrad
+# kernel wins because it launches every problem's tiles at once.
+mfma_linear_enabled = False
 _MFMA_MAX_R = 2048
 
 
