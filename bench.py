@@ -207,80 +207,20 @@ def main():
 
 
 def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
-    """hipGraph-capture the accumulate micro-step and the apply micro-step.
+    """Wrap the framework's hipGraph-captured micro-batch loop
+    (engine/graphs.py) with the bench's static input buffers."""
+    from gradient_accumulation_tf_estimator_amd.engine.graphs import GraphedTrainLoop
 
-    The tf.cond(apply, accumulate) of the reference (optimization.py:91-94)
-    becomes a host-side choice between two captured graphs.
-    lr reaches the apply kernel through a device scalar so the schedule
-    updates without re-capture.
-
-    Data-parallel (world > 1): only the accumulate micro-step is captured;
-    the apply boundary replays it and then runs the RCCL all-reduce + fused
-    apply eagerly -- K-1 of K steps run at full graph speed with no
-    collective inside any capture.
-    """
-    assert engine.backend == "hip"
     static_ids = pool_ids[0].clone()
     static_lab = pool_lab[0].clone()
-
-    def fwd_bwd_accum():
-        loss = model.loss(static_ids, static_lab)
-        if world > 1:
-            loss = loss * (1.0 / world)
-        loss.backward()
-        engine.accumulate()
-        # capture must end with every forked stream joined; within-step
-        # wgrad overlap is preserved, only the step boundary serializes
-        engine._join_wgrad_stream()
-        return loss
-
-    # warmup on a side stream (torch.cuda.graphs requirement)
-    s = torch.cuda.Stream()
-    s.wait_stream(torch.cuda.current_stream())
-    with torch.cuda.stream(s):
-        for _ in range(3):
-            fwd_bwd_accum()
-        engine.set_lr(engine.lr_at(0))
-        engine.apply_from_device()
-    torch.cuda.current_stream().wait_stream(s)
-    torch.cuda.synchronize()
-    # reset optimizer progress perturbed by warmup
-    engine.state.accum.zero_()
-
-    g_accum = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(g_accum):
-        loss_accum = fwd_bwd_accum()
-    if world == 1:
-        g_apply = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g_apply, pool=g_accum.pool()):
-            loss_apply = fwd_bwd_accum()
-            engine.apply_from_device()
-        torch.cuda.synchronize()
-    else:
-        g_apply = loss_apply = None
-        torch.cuda.synchronize()
-
+    loop = GraphedTrainLoop(engine, lambda: model.loss(static_ids, static_lab),
+                            world=world)
     POOL = pool_ids.shape[0]
 
     def run(i):
         static_ids.copy_(pool_ids[i % POOL])
         static_lab.copy_(pool_lab[i % POOL])
-        if engine.is_apply_step():
-            engine.set_lr(engine.lr_at(engine.global_step))
-            if g_apply is not None:
-                g_apply.replay()
-            else:
-                # DP: replay the captured micro-step, then all-reduce the
-                # flat accum buffer over RCCL and apply -- eager, outside
-                # any capture
-                g_accum.replay()
-                engine._allreduce_accum()
-                engine.apply_from_device()
-            engine.global_step += 1
-            return loss_apply if loss_apply is not None else loss_accum
-        g_accum.replay()
-        engine.global_step += 1
-        return loss_accum
+        return loop.step()
 
     return run
 

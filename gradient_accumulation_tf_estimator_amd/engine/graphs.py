@@ -1,0 +1,94 @@
+"""hipGraph-captured micro-batch loop (the tf.cond replacement).
+
+The reference's ``tf.cond(step % K == 0, apply, accumulate)`` train_op
+(optimization.py:91-94) becomes a host-side choice between two captured
+hipGraphs: an accumulate graph (fwd + bwd + K1) replayed K-1 times per
+window and an apply graph (same + global-norm + fused AdamW) replayed once
+(SURVEY.md section 2.3). The schedule's lr reaches the apply kernel through
+a device scalar so no re-capture is ever needed.
+
+Data-parallel (world > 1): only the accumulate micro-step is captured; the
+apply boundary replays it and then runs the RCCL all-reduce + fused apply
+eagerly -- K-1 of K steps run at full graph speed with no collective inside
+any capture.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Optional
+
+import torch
+
+from .accum import AccumEngine
+
+
+class GraphedTrainLoop:
+    """Captures ``loss_fn() -> loss; backward; accumulate`` into hipGraphs.
+
+    ``loss_fn`` must read its inputs from static (caller-owned) device
+    buffers; update those buffers between ``step()`` calls. All optimizer
+    state lives in the engine's flat buffers, so capture needs no special
+    handling beyond the device-scalar lr.
+    """
+
+    def __init__(self, engine: AccumEngine, loss_fn: Callable[[], torch.Tensor],
+                 *, world: int = 1, warmup_iters: int = 3):
+        if engine.backend != "hip":
+            raise RuntimeError("GraphedTrainLoop requires the HIP engine backend")
+        self.engine = engine
+        self.world = world
+        inv_world = 1.0 / world
+
+        def fwd_bwd_accum():
+            loss = loss_fn()
+            if world > 1:
+                loss = loss * inv_world
+            loss.backward()
+            engine.accumulate()
+            engine._join_wgrad_stream()
+            return loss
+
+        # torch.cuda.graphs warmup protocol: a few eager iterations on a side
+        # stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup_iters):
+                fwd_bwd_accum()
+            engine.set_lr(engine.lr_at(0))
+            engine.apply_from_device()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        engine.state.accum.zero_()  # discard warmup's accumulation
+
+        self.g_accum = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_accum):
+            self.loss_accum = fwd_bwd_accum()
+        if world == 1:
+            self.g_apply = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_apply, pool=self.g_accum.pool()):
+                self.loss_apply = fwd_bwd_accum()
+                engine.apply_from_device()
+        else:
+            self.g_apply = None
+            self.loss_apply = None
+        torch.cuda.synchronize()
+
+    def step(self) -> torch.Tensor:
+        """One reference micro-step (static inputs must already be set).
+        Returns the static loss tensor of the replayed graph."""
+        engine = self.engine
+        if engine.is_apply_step():
+            engine.set_lr(engine.lr_at(engine.global_step))
+            if self.g_apply is not None:
+                self.g_apply.replay()
+                engine.global_step += 1
+                return self.loss_apply
+            self.g_accum.replay()
+            engine._allreduce_accum()
+            engine.apply_from_device()
+            engine.global_step += 1
+            return self.loss_accum
+        self.g_accum.replay()
+        engine.global_step += 1
+        return self.loss_accum

@@ -176,3 +176,45 @@ def test_gpu_checkpoint_resume_bit_exact():
         opB.step((netB(x) ** 2).mean())
     torch.cuda.synchronize()
     assert torch.equal(netB[0].weight.detach(), ref)
+
+
+def test_graphed_train_loop():
+    """Framework-level hipGraph capture: GraphedTrainLoop matches the eager
+    engine step-for-step on the same inputs."""
+    import torch.nn as nn
+    from gradient_accumulation_tf_estimator_amd.engine.accum import AccumEngine
+    from gradient_accumulation_tf_estimator_amd.engine.graphs import GraphedTrainLoop
+
+    def make():
+        torch.manual_seed(21)
+        return nn.Sequential(nn.Linear(128, 128), nn.ReLU(),
+                             nn.Linear(128, 10)).cuda()
+
+    xs = torch.randn(12, 16, 128, device="cuda",
+                     generator=torch.Generator("cuda").manual_seed(3))
+
+    netA = make()
+    engA = AccumEngine(list(netA.named_parameters()), init_lr=1e-3,
+                       num_train_steps=1000, num_warmup_steps=0,
+                       gradient_accumulation_multiplier=4, clip_norm=1.0,
+                       backend="hip")
+    static = xs[0].clone()
+    loop = GraphedTrainLoop(engA, lambda: (netA(static) ** 2).mean())
+    for i in range(8):
+        static.copy_(xs[i])
+        loop.step()
+    torch.cuda.synchronize()
+
+    netB = make()
+    engB = AccumEngine(list(netB.named_parameters()), init_lr=1e-3,
+                       num_train_steps=1000, num_warmup_steps=0,
+                       gradient_accumulation_multiplier=4, clip_norm=1.0,
+                       backend="hip")
+    for i in range(8):
+        loss = (netB(xs[i]) ** 2).mean()
+        loss.backward()
+        engB.micro_step()
+    torch.cuda.synchronize()
+    np.testing.assert_allclose(engA.state.master.cpu(), engB.state.master.cpu(),
+                               rtol=1e-5, atol=1e-6)
+    assert engA.global_step == engB.global_step == 8
