@@ -49,7 +49,16 @@ class GraphedTrainLoop:
             return loss
 
         # torch.cuda.graphs warmup protocol: a few eager iterations on a side
-        # stream, then capture
+        # stream, then capture. The warmup runs a REAL apply, so snapshot the
+        # optimizer state and restore it afterwards -- capture must be
+        # state-neutral.
+        st = engine.state
+        snap = {
+            "master": st.master.clone(),
+            "m": st.m.clone(),
+            "v": st.v.clone(),
+            "model": None if st.model is st.master else st.model.clone(),
+        }
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
@@ -59,7 +68,6 @@ class GraphedTrainLoop:
             engine.apply_from_device()
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
-        engine.state.accum.zero_()  # discard warmup's accumulation
 
         self.g_accum = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_accum):
@@ -72,6 +80,18 @@ class GraphedTrainLoop:
         else:
             self.g_apply = None
             self.loss_apply = None
+        torch.cuda.synchronize()
+
+        # capture itself executed the kernels: restore the optimizer state
+        # (the captured graphs reference the buffers, values are free to set)
+        st.master.copy_(snap["master"])
+        st.m.copy_(snap["m"])
+        st.v.copy_(snap["v"])
+        if snap["model"] is not None:
+            st.model.copy_(snap["model"])
+        st.accum.zero_()
+        st.grads.zero_()
+        del snap
         torch.cuda.synchronize()
 
     def step(self) -> torch.Tensor:
