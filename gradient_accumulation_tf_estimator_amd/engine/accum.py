@@ -120,6 +120,7 @@ class AccumEngine:
             from ..ops import fused as fused_ops
 
             fused_ops.flush_pending_wgrads()
+            fused_ops.flush_pending_colreduce()
         st = self.state
         lo, hi = st.layout.grad_lo, st.layout.grad_hi
         if hi <= lo:

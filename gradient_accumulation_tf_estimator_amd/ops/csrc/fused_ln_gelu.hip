@@ -385,3 +385,54 @@ extern "C" __global__ void k_embgrad_acc(
     atomicAdd(dst + 3, bf2f(v.w));
   }
 }
+
+
+// ---------------------------------------------------------------------------
+// Batched column-reduce flush: every fused-LN/GELU backward's fp32 partial
+// slab of a micro-step reduced into its accum slices in ONE launch (problem
+// metadata by value -> hipGraph-capture-safe; same trick as wgrad_mfma).
+// ---------------------------------------------------------------------------
+#define CRB_MAX_G 16
+struct CrbArgs {
+  unsigned long long part[CRB_MAX_G];   // [NB][C] fp32
+  unsigned long long d0[CRB_MAX_G];     // dest slices (d1/d2 may be 0)
+  unsigned long long d1[CRB_MAX_G];
+  unsigned long long d2[CRB_MAX_G];
+  int nb[CRB_MAX_G];
+  int c[CRB_MAX_G];
+  int n0[CRB_MAX_G];
+  int n1[CRB_MAX_G];
+  int G;
+};
+
+extern "C" __global__ void k_colreduce_batch(CrbArgs args) {
+  // block -> (problem, chunk) via per-problem block counts
+  int b = blockIdx.x, g = 0;
+  int nblk = 0;
+  for (; g < args.G; ++g) {
+    const int nch = (args.nb[g] + CR_CHUNK - 1) / CR_CHUNK;
+    nblk = (args.c[g] * nch + 255) / 256;
+    if (b < nblk) break;
+    b -= nblk;
+  }
+  const int NB = args.nb[g], C = args.c[g];
+  const int nch = (NB + CR_CHUNK - 1) / CR_CHUNK;
+  const float* partials = (const float*)args.part[g];
+  float* dest0 = (float*)args.d0[g];
+  float* dest1 = (float*)args.d1[g];
+  float* dest2 = (float*)args.d2[g];
+  const int n0 = args.n0[g], n1 = args.n1[g];
+  const long long t = (long long)b * 256 + threadIdx.x;
+  if (t >= (long long)C * nch) return;
+  const int i = (int)(t % C);
+  const int b0 = (int)(t / C) * CR_CHUNK;
+  const int b1 = min(b0 + CR_CHUNK, NB);
+  float s = 0.f;
+#pragma unroll CR_CHUNK
+  for (int bb = b0; bb < b1; ++bb) s += partials[(size_t)bb * C + i];
+  float* d = (i < n0) ? dest0 + i
+             : (i < n0 + n1) ? dest1 + (i - n0)
+                             : dest2 + (i - n0 - n1);
+  if (nch == 1) *d += s;
+  else atomicAdd(d, s);
+}

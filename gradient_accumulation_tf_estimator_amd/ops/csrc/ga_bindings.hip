@@ -38,6 +38,19 @@ extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned 
                                           unsigned short*, long long, int);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
+#define CRB_MAX_G 16
+struct CrbArgs {
+  unsigned long long part[CRB_MAX_G];
+  unsigned long long d0[CRB_MAX_G];
+  unsigned long long d1[CRB_MAX_G];
+  unsigned long long d2[CRB_MAX_G];
+  int nb[CRB_MAX_G];
+  int c[CRB_MAX_G];
+  int n0[CRB_MAX_G];
+  int n1[CRB_MAX_G];
+  int G;
+};
+extern "C" __global__ void k_colreduce_batch(CrbArgs);
 extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*,
                                          float*, long long, int);
 #define GA_DECL_ATTN(S)                                                        \
@@ -471,6 +484,40 @@ at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
   return dpre;
 }
 
+// problems: list of (partials [NB,...,C], dest0, dest1?, dest2?)
+void colreduce_batch(std::vector<at::Tensor> parts,
+                     std::vector<at::Tensor> d0s,
+                     std::vector<c10::optional<at::Tensor>> d1s,
+                     std::vector<c10::optional<at::Tensor>> d2s) {
+  const int G = (int)parts.size();
+  TORCH_CHECK(G > 0 && G <= CRB_MAX_G, "colreduce_batch: 1..16 problems");
+  CrbArgs args{};
+  args.G = G;
+  long long blocks = 0;
+  for (int g = 0; g < G; ++g) {
+    const int NB = (int)parts[g].size(0);
+    const int C = (int)(parts[g].numel() / NB);
+    const int na = (int)d0s[g].numel();
+    const int nb_ = d1s[g] ? (int)d1s[g]->numel() : 0;
+    const int nc = d2s[g] ? (int)d2s[g]->numel() : 0;
+    TORCH_CHECK(na + nb_ + nc == C, "colreduce_batch dest sizes");
+    args.part[g] = (unsigned long long)parts[g].data_ptr();
+    args.d0[g] = (unsigned long long)d0s[g].data_ptr();
+    args.d1[g] = d1s[g] ? (unsigned long long)d1s[g]->data_ptr() : 0ull;
+    args.d2[g] = d2s[g] ? (unsigned long long)d2s[g]->data_ptr() : 0ull;
+    args.nb[g] = NB;
+    args.c[g] = C;
+    args.n0[g] = na;
+    args.n1[g] = nb_;
+    const int nch = (NB + 7) / 8;
+    blocks += ((long long)C * nch + 255) / 256;
+  }
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(k_colreduce_batch, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, args);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_colreduce_batch launch failed");
+}
+
 void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t H) {
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.scalar_type() == at::kBFloat16,
               "dy must be contiguous bf16");
@@ -509,6 +556,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
   mod.def("colreduce_acc", &colreduce_acc,
           "reduce partials over blocks, ADD into flat fp32 accum slices");
+  mod.def("colreduce_batch", &colreduce_batch,
+          "one launch reducing every pending LN/GELU partial slab into accum");
   mod.def("embgrad_acc", &embgrad_acc,
           "scatter-add embedding grads into the flat fp32 accum slice");
   mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");

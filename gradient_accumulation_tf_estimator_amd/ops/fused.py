@@ -50,6 +50,28 @@ def set_grouped_wgrad(enabled: bool) -> None:
 
 _wgrad_table_cache = {}
 
+# deferred LN/GELU partial-slab reductions, flushed as ONE batched launch
+_pending_colreduce = []
+
+
+def _queue_colreduce(partials, d0, d1, d2):
+    _pending_colreduce.append((partials, d0, d1, d2))
+
+
+def flush_pending_colreduce() -> None:
+    global _pending_colreduce
+    if not _pending_colreduce:
+        return
+    hip = require_hip()
+    pend = _pending_colreduce
+    _pending_colreduce = []
+    for lo in range(0, len(pend), 16):
+        chunk = pend[lo : lo + 16]
+        hip.colreduce_batch([p for p, _, _, _ in chunk],
+                            [d0 for _, d0, _, _ in chunk],
+                            [d1 for _, _, d1, _ in chunk],
+                            [d2 for _, _, _, d2 in chunk])
+
 
 def flush_pending_wgrads() -> None:
     global _pending_wgrads
@@ -133,7 +155,8 @@ class _AddLayerNormFn(torch.autograd.Function):
         views = getattr(mod, "_accum_views", None)
         if views is not None:
             vg, vb, vpb = views
-            hip.colreduce_acc(partials, vg, vb, vpb if vpb is not None else mod._sink())
+            _queue_colreduce(partials, vg, vb,
+                             vpb if vpb is not None else mod._sink())
         else:
             s = partials.sum(0)
             dgamma = s[0].to(dy.dtype)
@@ -211,7 +234,7 @@ class _BiasGeluFn(torch.autograd.Function):
         dbias = None
         view = getattr(ctx.module, "_accum_view", None)
         if view is not None:
-            hip.colreduce_acc(partials, view, None, None)
+            _queue_colreduce(partials, view, None, None)
         else:
             dbias = partials.sum(0).to(dy.dtype)
         return dx.view_as(dy), dbias, None
