@@ -268,8 +268,9 @@ extern "C" __global__ void k_attn_bwd_d(
 // ---------------------------------------------------------------------------
 template <int S>
 __device__ __forceinline__ void attn_bwd_q_body(
-    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
-    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, float* __restrict__ Dtab,
     unsigned short* __restrict__ dqkv, int B, int nh) {
   const int H = nh * ATTN_D;
   constexpr int NT = S / 32;
@@ -292,7 +293,24 @@ __device__ __forceinline__ void attn_bwd_q_body(
 
   const int q0 = qt * 32;
   const float lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
-  const float D_q = Dtab[((long long)b * nh + h) * S + q0 + lo31];
+
+  // D_q = rowsum(dO o O) computed here (lane owns row q) and published to
+  // Dtab for the k-tile kernel -- replaces the separate k_attn_bwd_d launch
+  float D_q;
+  {
+    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
+    const unsigned short* orw = out + obase + (long long)(q0 + lo31) * H + hi * 32;
+    float sd = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
+      bf16x8 ov = *(const bf16x8*)(orw + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
+    }
+    D_q = sd + __shfl_xor(sd, 32, 64);
+    if (hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+  }
 
   bf16x8 qf[4], dof[4];
   const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
@@ -349,10 +367,10 @@ __device__ __forceinline__ void attn_bwd_q_body(
 
 #define GA_ATTN_BWDQ_INST(S)                                                  \
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_##S(          \
-      const unsigned short* qkv, const unsigned short* dout,                  \
-      const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,    \
-      int nh) {                                                               \
-    attn_bwd_q_body<S>(qkv, dout, lse_in, Dtab, dqkv, B, nh);                 \
+      const unsigned short* qkv, const unsigned short* out,                   \
+      const unsigned short* dout, const float* lse_in, float* Dtab,           \
+      unsigned short* dqkv, int B, int nh) {                                  \
+    attn_bwd_q_body<S>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh);            \
   }
 GA_ATTN_BWDQ_INST(32)
 GA_ATTN_BWDQ_INST(64)

@@ -57,8 +57,8 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
   extern "C" __global__ void k_attn_fwd_##S(const unsigned short*,             \
                                             unsigned short*, float*, int, int);\
   extern "C" __global__ void k_attn_bwd_q_##S(                                 \
-      const unsigned short*, const unsigned short*, const float*,              \
-      const float*, unsigned short*, int, int);                                \
+      const unsigned short*, const unsigned short*, const unsigned short*,     \
+      const float*, float*, unsigned short*, int, int);                        \
   extern "C" __global__ void k_attn_bwd_kv_##S(                                \
       const unsigned short*, const unsigned short*, const float*,              \
       const float*, unsigned short*, int, int);
@@ -336,19 +336,15 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   auto Dtab = at::empty({B, (long)nh, S}, lse.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
-  const long long rows = (long long)B * nh * S;
-  int dblocks = (int)std::min<long long>((rows + 255) / 256, 1024);
-  hipLaunchKernelGGL(k_attn_bwd_d, dim3(dblocks), dim3(256), 0, stream,
-                     (const unsigned short*)out.data_ptr(),
-                     (const unsigned short*)dout.data_ptr(),
-                     Dtab.data_ptr<float>(), B, S, (int)nh);
   const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
-  void (*qk)(const unsigned short*, const unsigned short*, const float*,
-             const float*, unsigned short*, int, int) =
+  void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
+             const float*, float*, unsigned short*, int, int) =
       S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
       : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
+  // bwd_q computes + publishes the D table (rowsum(dO o O)) for bwd_kv
   hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
                      (const unsigned short*)qkv.data_ptr(),
+                     (const unsigned short*)out.data_ptr(),
                      (const unsigned short*)dout.data_ptr(),
                      lse.data_ptr<float>(), Dtab.data_ptr<float>(),
                      (unsigned short*)dqkv.data_ptr(), B, (int)nh);
