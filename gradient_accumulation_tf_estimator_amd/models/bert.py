@@ -31,6 +31,7 @@ from ..ops.fused import (
     DirectLinear,
     FusedAddLayerNorm,
     FusedBiasGelu,
+    FusedFFN,
     fused_attention,
     fused_attention_supported,
 )
@@ -139,13 +140,14 @@ class BertLayer(nn.Module):
         Lin = DirectLinear if cfg.fused else nn.Linear
         self.attention = BertSelfAttention(cfg)
         self.attention_output = Lin(cfg.hidden_size, cfg.hidden_size, bias=not cfg.fused)
-        self.intermediate = Lin(cfg.hidden_size, cfg.intermediate_size, bias=not cfg.fused)
-        self.output = Lin(cfg.intermediate_size, cfg.hidden_size, bias=not cfg.fused)
+        if not cfg.fused:
+            self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
+            self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
         self.dropout = nn.Dropout(cfg.dropout)
         if cfg.fused:
             self.attention_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
-            self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
+            self.ffn = FusedFFN(cfg.hidden_size, cfg.intermediate_size)
             self.output_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
         else:
@@ -156,8 +158,7 @@ class BertLayer(nn.Module):
         a = self.attention(x, attn_mask)
         if self.fused:
             x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
-            h = self.output(self.intermediate_act(self.intermediate(x)))
-            return self.output_LayerNorm(self.dropout(h), residual=x)
+            return self.output_LayerNorm(self.dropout(self.ffn(x)), residual=x)
         x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
         h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))
         return self.output_LayerNorm(x + self.dropout(h))

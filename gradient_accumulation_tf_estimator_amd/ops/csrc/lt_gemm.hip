@@ -42,7 +42,8 @@ struct Plan {
   std::vector<hipblasLtMatmulHeuristicResult_t> algos;
 };
 
-// kind: 0 = fwd (no bias), 1 = fwd + bias epilogue, 2 = dgrad
+// kind: 0 = fwd (no bias), 1 = fwd + bias epilogue, 2 = dgrad,
+//       3 = fwd + GELU_AUX_BIAS epilogue, 4 = dgrad + DGELU epilogue
 using Key = std::tuple<int, int64_t, int64_t, int64_t>;
 
 Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
@@ -56,7 +57,7 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
   Plan p{};
   LT_CHECK(hipblasLtMatmulDescCreate(&p.desc, HIPBLAS_COMPUTE_32F, HIP_R_32F));
   hipblasOperation_t opN = HIPBLAS_OP_N, opT = HIPBLAS_OP_T;
-  if (kind == 2) {
+  if (kind == 2 || kind == 4) {
     // dx_cm(K,R) = W_cm(K,N) opN  @  dy_cm(N,R) opN
     LT_CHECK(hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_TRANSA,
                                              &opN, sizeof(opN)));
@@ -65,6 +66,14 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, K, N, K));  // W
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, N, R, N));  // dy
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lc, HIP_R_16BF, K, R, K));  // dx
+    if (kind == 4) {
+      hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_DGELU;
+      LT_CHECK(hipblasLtMatmulDescSetAttribute(
+          p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+      int64_t ld = K;
+      LT_CHECK(hipblasLtMatmulDescSetAttribute(
+          p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld)));
+    }
   } else {
     // y_cm(N,R) = W_cm(K,N) opT  @  x_cm(K,R) opN
     LT_CHECK(hipblasLtMatmulDescSetAttribute(p.desc, HIPBLASLT_MATMUL_DESC_TRANSA,
@@ -74,10 +83,16 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.la, HIP_R_16BF, K, N, K));  // W
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lb, HIP_R_16BF, K, R, K));  // x
     LT_CHECK(hipblasLtMatrixLayoutCreate(&p.lc, HIP_R_16BF, N, R, N));  // y
-    if (kind == 1) {
-      hipblasLtEpilogue_t epi = HIPBLASLT_EPILOGUE_BIAS;
+    if (kind == 1 || kind == 3) {
+      hipblasLtEpilogue_t epi = kind == 1 ? HIPBLASLT_EPILOGUE_BIAS
+                                          : HIPBLASLT_EPILOGUE_GELU_AUX_BIAS;
       LT_CHECK(hipblasLtMatmulDescSetAttribute(
           p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &epi, sizeof(epi)));
+      if (kind == 3) {
+        int64_t ld = N;
+        LT_CHECK(hipblasLtMatmulDescSetAttribute(
+            p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld)));
+      }
     }
   }
 
@@ -98,11 +113,15 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
 }
 
 void run(Plan& p, const void* A, const void* B, void* D, const void* bias,
-         int64_t algo_idx) {
+         int64_t algo_idx, void* aux = nullptr) {
   const float alpha = 1.f, beta = 0.f;
   if (bias) {
     LT_CHECK(hipblasLtMatmulDescSetAttribute(
         p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
+  }
+  if (aux) {
+    LT_CHECK(hipblasLtMatmulDescSetAttribute(
+        p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &aux, sizeof(aux)));
   }
   auto stream = c10::hip::getCurrentHIPStream().stream();
   int idx = (int)std::min<int64_t>(std::max<int64_t>(algo_idx, 0),
@@ -142,9 +161,47 @@ at::Tensor lt_dgrad(at::Tensor dy, at::Tensor w, int64_t algo_idx) {
   return dx;
 }
 
+std::vector<at::Tensor> lt_linear_gelu(at::Tensor x, at::Tensor w, at::Tensor bias,
+                                       int64_t algo_idx) {
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  const int64_t K = w.size(1), N = w.size(0);
+  const int64_t R = x.numel() / K;
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  auto aux = at::empty(sizes, x.options());
+  auto& p = plan_for(3, R, N, K);
+  run(p, w.data_ptr(), x.data_ptr(), y.data_ptr(), bias.data_ptr(), algo_idx,
+      aux.data_ptr());
+  return {y, aux};
+}
+
+at::Tensor lt_dgrad_dgelu(at::Tensor dy, at::Tensor w, at::Tensor aux,
+                          int64_t algo_idx) {
+  TORCH_CHECK(dy.is_contiguous() && w.is_contiguous() && aux.is_contiguous());
+  const int64_t K = w.size(1), N = w.size(0);
+  const int64_t R = dy.numel() / N;
+  auto sizes = dy.sizes().vec();
+  sizes.back() = K;
+  auto dx = at::empty(sizes, dy.options());
+  auto& p = plan_for(4, R, N, K);
+  run(p, w.data_ptr(), dy.data_ptr(), dx.data_ptr(), nullptr, algo_idx,
+      aux.data_ptr());
+  return dx;
+}
+
+int64_t lt_gelu_algo_count(int64_t kind, int64_t R, int64_t N, int64_t K) {
+  return (int64_t)plan_for((int)kind, R, N, K).algos.size();
+}
+
 }  // namespace
 
 void register_lt_gemm(pybind11::module_& mod) {
+  mod.def("lt_linear_gelu", &lt_linear_gelu,
+          "y, aux = gelu(x @ W^T + b) with GELU_AUX_BIAS epilogue");
+  mod.def("lt_dgrad_dgelu", &lt_dgrad_dgelu,
+          "dx = dgelu(aux) o (dy @ W) with DGELU epilogue");
+  mod.def("lt_gelu_algo_count", &lt_gelu_algo_count);
   mod.def("lt_algo_count", &lt_algo_count, "heuristic candidates for a shape");
   mod.def("lt_linear", &lt_linear, "y = x @ W^T (+bias), hipblaslt, algo_idx");
   mod.def("lt_dgrad", &lt_dgrad, "dx = dy @ W, hipblaslt, algo_idx");

@@ -403,6 +403,74 @@ def fused_attention(qkv: torch.Tensor, nh: int) -> torch.Tensor:
     return _AttentionFn.apply(qkv.view(B, S, 3, H3 // 3), nh)
 
 
+class _FFNFn(torch.autograd.Function):
+    """intermediate GEMM + bias + GELU (GELU_AUX_BIAS epilogue) -> output
+    GEMM, with the backward's dgelu fused into the output dgrad (DGELU
+    epilogue) and both wgrads + the intermediate bias grad routed through
+    the direct-accum wgrad path. Replaces the standalone bias+GELU kernels
+    entirely on the bound GPU path."""
+
+    @staticmethod
+    def forward(ctx, x, wi, bi, wo, module):
+        from . import gemm
+
+        x = x.contiguous()
+        x2d = x.reshape(-1, x.shape[-1])
+        h, aux = gemm.linear_gelu(x2d, wi, bi)
+        y = gemm.linear_fwd(h, wo, None)
+        ctx.save_for_backward(x, aux, h, wi, wo)
+        ctx.module = module
+        return y.reshape(*x.shape[:-1], wo.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        from . import gemm
+
+        x, aux, h, wi, wo = ctx.saved_tensors
+        mod = ctx.module
+        dy2d = dy.contiguous().reshape(-1, dy.shape[-1])
+        d_h = gemm.dgrad_dgelu(dy2d, wo, aux)
+        x2d = x.reshape(-1, x.shape[-1])
+        if _GROUPED_WGRAD:
+            _pending_wgrads.append((h, dy2d, mod._accum_view_wo, None))
+            _pending_wgrads.append((x2d, d_h, mod._accum_view_wi, mod._accum_view_bi))
+        else:
+            gemm.wgrad_acc(h, dy2d, mod._accum_view_wo)
+            gemm.wgrad_acc(x2d, d_h, mod._accum_view_wi)
+            mod._accum_view_bi.add_(d_h.sum(0, dtype=torch.float32))
+        dx = gemm.dgrad(d_h, wi).reshape(x.shape)
+        return dx, None, None, None, None
+
+
+class FusedFFN(nn.Module):
+    """The encoder FFN (Linear -> GELU -> Linear) with the activation fused
+    into the GEMM epilogues when bound to an engine on GPU bf16; plain
+    torch ops otherwise. Parameter names keep the reference's weight-decay
+    semantics (weights decay, ``bias_in`` matches the bias exclusion regex).
+    """
+
+    def __init__(self, hidden: int, intermediate: int):
+        super().__init__()
+        self.hidden, self.intermediate = hidden, intermediate
+        self.weight_in = nn.Parameter(torch.empty(intermediate, hidden))
+        self.bias_in = nn.Parameter(torch.zeros(intermediate))
+        self.weight_out = nn.Parameter(torch.empty(hidden, intermediate))
+        nn.init.normal_(self.weight_in, std=0.02)
+        nn.init.normal_(self.weight_out, std=0.02)
+        self._accum_view_wi = None
+        self._accum_view_bi = None
+        self._accum_view_wo = None
+
+    def forward(self, x):
+        if self._accum_view_wi is not None:
+            if not _use_hip(x):
+                raise RuntimeError("bound FusedFFN got non-bf16/non-GPU input")
+            return _FFNFn.apply(x, self.weight_in, self.bias_in, self.weight_out,
+                                self)
+        h = F.gelu(F.linear(x, self.weight_in, self.bias_in), approximate="tanh")
+        return F.linear(h, self.weight_out)
+
+
 class _ClsHeadFn(torch.autograd.Function):
     """tanh(pooler-out) -> classifier -> mean softmax-CE, one kernel each
     way (ops/csrc/cls_head.hip); classifier grads go straight into accum."""
@@ -478,6 +546,10 @@ def direct_param_names(model: nn.Module):
         elif isinstance(mod, CEClassifier) and mod.fusable:
             names.append(pname(mod, "weight"))
             names.append(pname(mod, "bias"))
+        elif isinstance(mod, FusedFFN):
+            names.append(pname(mod, "weight_in"))
+            names.append(pname(mod, "bias_in"))
+            names.append(pname(mod, "weight_out"))
     return names
 
 
@@ -513,5 +585,12 @@ def bind_direct_grad(model: nn.Module, engine) -> int:
         elif isinstance(mod, CEClassifier) and mod.fusable:
             mod._accum_view_w = engine.state.accum_view(mod.weight)
             mod._accum_view_b = engine.state.accum_view(mod.bias)
+            n += 1
+        elif isinstance(mod, FusedFFN):
+            mod._accum_view_wi = engine.state.accum_view(mod.weight_in).view(
+                mod.intermediate, mod.hidden)
+            mod._accum_view_bi = engine.state.accum_view(mod.bias_in)
+            mod._accum_view_wo = engine.state.accum_view(mod.weight_out).view(
+                mod.hidden, mod.intermediate)
             n += 1
     return n

@@ -111,3 +111,25 @@ def wgrad_acc(x2d: torch.Tensor, dy2d: torch.Tensor, accum_2d: torch.Tensor) -> 
                         lambda i: hip.wgrad_acc(x2d, dy2d, scratch, i))
             del scratch
     hip.wgrad_acc(x2d, dy2d, accum_2d, idx)
+
+
+def linear_gelu(x2d: torch.Tensor, w: torch.Tensor, bias: torch.Tensor):
+    """y, aux(pre-gelu) = gelu(x @ W^T + b) via the GELU_AUX_BIAS epilogue."""
+    hip = require_hip()
+    N, K = w.shape
+    R = x2d.numel() // K
+    idx = _algo_for(3, R, N, K,
+                    lambda: hip.lt_gelu_algo_count(3, R, N, K),
+                    lambda i: hip.lt_linear_gelu(x2d, w, bias, i))
+    return hip.lt_linear_gelu(x2d, w, bias, idx)
+
+
+def dgrad_dgelu(dy2d: torch.Tensor, w: torch.Tensor, aux: torch.Tensor):
+    """dx = dgelu(aux) o (dy @ W) via the DGELU epilogue."""
+    hip = require_hip()
+    N, K = w.shape
+    R = dy2d.numel() // N
+    idx = _algo_for(4, R, N, K,
+                    lambda: hip.lt_gelu_algo_count(4, R, N, K),
+                    lambda i: hip.lt_dgrad_dgelu(dy2d, w, aux, i))
+    return hip.lt_dgrad_dgelu(dy2d, w, aux, idx)

@@ -335,3 +335,59 @@ def test_linear_mfma_matches_torch(N, K):
     dx = hip.linear_dgrad_mfma(dy, w)
     dxref = dy.float() @ w.float()
     np.testing.assert_allclose(dx.float().cpu(), dxref.cpu(), rtol=3e-2, atol=2e-1)
+
+
+def test_fused_ffn_matches_torch():
+    """FusedFFN (GELU epilogues + direct-accum wgrads) vs fp32 reference."""
+    import torch.nn as nn
+    from gradient_accumulation_tf_estimator_amd.engine.accum import AccumEngine
+    from gradient_accumulation_tf_estimator_amd.ops.fused import (
+        FusedFFN, bind_direct_grad, direct_param_names)
+
+    H, I, R = 512, 2048, 1024
+    torch.manual_seed(2)
+
+    class Net(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.ffn = FusedFFN(H, I)
+
+        def forward(self, x):
+            return self.ffn(x)
+
+    torch.manual_seed(5)
+    net = Net().cuda().bfloat16()
+    eng = AccumEngine(list(net.named_parameters()), init_lr=0.0,
+                      num_train_steps=10, gradient_accumulation_multiplier=4,
+                      clip_norm=None, backend="hip",
+                      direct_names=direct_param_names(net))
+    bind_direct_grad(net, eng)
+
+    x = (torch.randn(R, H, device="cuda") * 0.5).bfloat16().requires_grad_()
+    y = net(x)
+    dy = (torch.randn_like(y) * 0.3)
+    y.backward(dy)
+    eng.accumulate()
+    torch.cuda.synchronize()
+
+    xf = x.detach().float().requires_grad_()
+    wi = net.ffn.weight_in.detach().float().requires_grad_()
+    bi = net.ffn.bias_in.detach().float().requires_grad_()
+    wo = net.ffn.weight_out.detach().float().requires_grad_()
+    href = F.gelu(F.linear(xf, wi, bi), approximate="tanh")
+    yref = F.linear(href, wo)
+    yref.backward(dy.float())
+
+    # hipblaslt's gelu flavor may be erf vs our tanh approx: loose tolerances
+    np.testing.assert_allclose(y.detach().float().cpu(), yref.detach().cpu(),
+                               rtol=5e-2, atol=1e-1)
+    np.testing.assert_allclose(x.grad.float().cpu(), xf.grad.cpu(),
+                               rtol=5e-2, atol=1e-1)
+    st = eng.state
+    for name, ref in [("ffn.weight_in", wi.grad.reshape(-1)),
+                      ("ffn.bias_in", bi.grad),
+                      ("ffn.weight_out", wo.grad.reshape(-1))]:
+        sl = [s for s in st.layout.slices if s.name == name][0]
+        got = st.accum[sl.offset : sl.offset + sl.numel].cpu().numpy()
+        np.testing.assert_allclose(got, ref.detach().cpu().numpy(), rtol=5e-2,
+                                   atol=8e-1, err_msg=name)
