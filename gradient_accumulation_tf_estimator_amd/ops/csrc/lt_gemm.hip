@@ -73,6 +73,10 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
       int64_t ld = K;
       LT_CHECK(hipblasLtMatmulDescSetAttribute(
           p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld)));
+      hipDataType auxt = HIP_R_16BF;
+      LT_CHECK(hipblasLtMatmulDescSetAttribute(
+          p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &auxt,
+          sizeof(auxt)));
     }
   } else {
     // y_cm(N,R) = W_cm(K,N) opT  @  x_cm(K,R) opN
@@ -92,6 +96,10 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
         int64_t ld = N;
         LT_CHECK(hipblasLtMatmulDescSetAttribute(
             p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld)));
+        hipDataType auxt = HIP_R_16BF;
+        LT_CHECK(hipblasLtMatmulDescSetAttribute(
+            p.desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &auxt,
+            sizeof(auxt)));
       }
     }
   }
