@@ -147,7 +147,14 @@ class BertLayer(nn.Module):
         if cfg.fused:
             self.attention_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
-            self.ffn = FusedFFN(cfg.hidden_size, cfg.intermediate_size)
+            # NOTE: FusedFFN (GEMM-fused GELU epilogues) stays experimental:
+            # this hipblaslt build returns no GELU_AUX solutions, so the FFN
+            # uses DirectLinear + the fused bias+GELU kernels instead
+            self.intermediate = DirectLinear(cfg.hidden_size,
+                                             cfg.intermediate_size, bias=False)
+            self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
+            self.output = DirectLinear(cfg.intermediate_size, cfg.hidden_size,
+                                       bias=False)
             self.output_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
         else:
@@ -158,7 +165,8 @@ class BertLayer(nn.Module):
         a = self.attention(x, attn_mask)
         if self.fused:
             x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
-            return self.output_LayerNorm(self.dropout(self.ffn(x)), residual=x)
+            h = self.output(self.intermediate_act(self.intermediate(x)))
+            return self.output_LayerNorm(self.dropout(h), residual=x)
         x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
         h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))
         return self.output_LayerNorm(x + self.dropout(h))
