@@ -28,12 +28,13 @@ from gradient_accumulation_tf_estimator_amd.models.bert import (
 
 def batch(n, seq, vocab, gen, device):
     ids = torch.randint(4, vocab, (n, seq), generator=gen)
-    # learnable rule: label = parity of the count of tokens below vocab/2
-    labels = ((ids < vocab // 2).sum(dim=1) & 1).long()
+    # learnable rule: label = whether the first token is in the lower half of
+    # the vocab (readable from one embedding -> converges from random init)
+    labels = (ids[:, 0] < vocab // 2).long()
     return ids.to(device), labels.to(device)
 
 
-def run_arm(name, micro_batch, K, updates, device, dtype, lr=5e-5):
+def run_arm(name, micro_batch, K, updates, device, dtype, lr=1e-4):
     torch.manual_seed(19830610)
     cfg = CONFIGS["bert-small"]()
     model = BertForSequenceClassification(cfg).to(device, dtype)
