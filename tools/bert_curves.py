@@ -27,10 +27,12 @@ from gradient_accumulation_tf_estimator_amd.models.bert import (
 
 
 def batch(n, seq, vocab, gen, device):
-    ids = torch.randint(4, vocab, (n, seq), generator=gen)
-    # learnable rule: label = whether the first token is in the lower half of
-    # the vocab (readable from one embedding -> converges from random init)
-    labels = (ids[:, 0] < vocab // 2).long()
+    # strongly learnable: every token is drawn from a class-dependent vocab
+    # half (bag-of-words separable) -> loss drops fast from random init
+    labels = torch.randint(0, 2, (n,), generator=gen)
+    lo_ids = torch.randint(4, vocab // 2, (n, seq), generator=gen)
+    hi_ids = torch.randint(vocab // 2, vocab, (n, seq), generator=gen)
+    ids = torch.where(labels[:, None] == 0, lo_ids, hi_ids)
     return ids.to(device), labels.to(device)
 
 
