@@ -161,6 +161,14 @@ class BertLayer(nn.Module):
             self.attention_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
             self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
 
+    def _bind_direct_extras(self, engine):
+        # the intermediate Linear's wgrad colsum over d(pre-gelu) IS the
+        # gelu bias gradient: delegate it so gelu backward is elementwise
+        if self.fused:
+            self.intermediate._accum_view_b = engine.state.accum_view(
+                self.intermediate_act.bias)
+            self.intermediate_act._bias_delegated = True
+
     def forward(self, x, attn_mask=None):
         a = self.attention(x, attn_mask)
         if self.fused:

@@ -387,6 +387,29 @@ extern "C" __global__ void k_embgrad_acc(
 }
 
 
+// Pure elementwise gelu backward (no bias-grad reduction): used when the
+// bias gradient is delegated to the downstream Linear's wgrad colsum
+// (colsum(d_pre) == dbias exactly).
+extern "C" __global__ void k_biasgelu_bwd_ew(
+    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ dx_out, long long total, int H) {
+  long long stride = (long long)gridDim.x * blockDim.x * 4;
+  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i < total; i += stride) {
+    ushort4 dv = *(const ushort4*)((const unsigned short*)dy + i);
+    ushort4 xv = *(const ushort4*)((const unsigned short*)x + i);
+    const int c = (int)(i % H);
+    ushort4 bv = *(const ushort4*)(bias + c);
+    ushort4 o;
+    o.x = f2bf(gelu_bwd1(bf2f(xv.x) + bf2f(bv.x), bf2f(dv.x)));
+    o.y = f2bf(gelu_bwd1(bf2f(xv.y) + bf2f(bv.y), bf2f(dv.y)));
+    o.z = f2bf(gelu_bwd1(bf2f(xv.z) + bf2f(bv.z), bf2f(dv.z)));
+    o.w = f2bf(gelu_bwd1(bf2f(xv.w) + bf2f(bv.w), bf2f(dv.w)));
+    *(ushort4*)(dx_out + i) = o;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Batched column-reduce flush: every fused-LN/GELU backward's fp32 partial
 // slab of a micro-step reduced into its accum slices in ONE launch (problem

@@ -36,6 +36,9 @@ GA_DECL_LN(4) GA_DECL_LN(8) GA_DECL_LN(12) GA_DECL_LN(16)
 GA_DECL_GELU(8) GA_DECL_GELU(12) GA_DECL_GELU(16) GA_DECL_GELU(24) GA_DECL_GELU(32) GA_DECL_GELU(48) GA_DECL_GELU(64)
 extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned short*,
                                           unsigned short*, long long, int);
+extern "C" __global__ void k_biasgelu_bwd_ew(const unsigned short*, const unsigned short*,
+                                             const unsigned short*, unsigned short*,
+                                             long long, int);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
 #define CRB_MAX_G 16
@@ -481,6 +484,22 @@ at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
 }
 
 // problems: list of (partials [NB,...,C], dest0, dest1?, dest2?)
+at::Tensor biasgelu_bwd_ew(at::Tensor dy, at::Tensor x, at::Tensor bias) {
+  const int H = (int)bias.numel();
+  check_bf16_2d(dy, "dy", H);
+  auto dx = at::empty_like(dy);
+  const long long total = dy.numel();
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+  hipLaunchKernelGGL(k_biasgelu_bwd_ew, dim3(blocks), dim3(256), 0, stream,
+                     (const unsigned short*)dy.data_ptr(),
+                     (const unsigned short*)x.data_ptr(),
+                     (const unsigned short*)bias.data_ptr(),
+                     (unsigned short*)dx.data_ptr(), total, H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_bwd_ew launch failed");
+  return dx;
+}
+
 void colreduce_batch(std::vector<at::Tensor> parts,
                      std::vector<at::Tensor> d0s,
                      std::vector<c10::optional<at::Tensor>> d1s,
@@ -552,6 +571,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
   mod.def("colreduce_acc", &colreduce_acc,
           "reduce partials over blocks, ADD into flat fp32 accum slices");
+  mod.def("biasgelu_bwd_ew", &biasgelu_bwd_ew,
+          "elementwise gelu backward (bias grad delegated to wgrad colsum)");
   mod.def("colreduce_batch", &colreduce_batch,
           "one launch reducing every pending LN/GELU partial slab into accum");
   mod.def("embgrad_acc", &embgrad_acc,

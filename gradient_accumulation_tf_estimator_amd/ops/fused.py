@@ -240,6 +240,27 @@ class _BiasGeluFn(torch.autograd.Function):
         return dx.view_as(dy), dbias, None
 
 
+class _BiasGeluEwFn(torch.autograd.Function):
+    """Gelu backward with the bias gradient DELEGATED: the downstream
+    Linear's wgrad colsum over d_pre equals dbias exactly, so this backward
+    is a single elementwise kernel."""
+
+    @staticmethod
+    def forward(ctx, x, bias, module):
+        hip = require_hip()
+        x = x.contiguous()
+        y = hip.biasgelu_fwd(x, bias)
+        ctx.save_for_backward(x, bias)
+        return y.view_as(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        x, bias = ctx.saved_tensors
+        dx = hip.biasgelu_bwd_ew(dy.contiguous(), x, bias)
+        return dx.view_as(dy), None, None
+
+
 class FusedBiasGelu(nn.Module):
     """y = gelu_tanh(x + bias); dbias computed inside the fused backward."""
 
@@ -248,9 +269,12 @@ class FusedBiasGelu(nn.Module):
         self.hidden = hidden
         self.bias = nn.Parameter(torch.zeros(hidden))
         self._accum_view = None
+        self._bias_delegated = False  # set when a downstream wgrad owns dbias
 
     def forward(self, x):
         if _use_hip(x):
+            if self._bias_delegated:
+                return _BiasGeluEwFn.apply(x, self.bias, self)
             return _BiasGeluFn.apply(x, self.bias, self)
         if self._accum_view is not None:
             raise RuntimeError("bound FusedBiasGelu got non-bf16/non-GPU input")
@@ -593,4 +617,7 @@ def bind_direct_grad(model: nn.Module, engine) -> int:
             mod._accum_view_wo = engine.state.accum_view(mod.weight_out).view(
                 mod.hidden, mod.intermediate)
             n += 1
+    for mod in model.modules():
+        if hasattr(mod, "_bind_direct_extras"):
+            mod._bind_direct_extras(engine)
     return n

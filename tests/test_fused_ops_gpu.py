@@ -151,6 +151,52 @@ def test_direct_accum_equals_grad_path():
         assert (abs(a).sum() > 0) or "bias" in sA.name
 
 
+def test_bertlayer_delegated_gelu_bias():
+    """A bound fused BertLayer delegates the gelu bias gradient to the
+    intermediate Linear's wgrad colsum (models/bert.py _bind_direct_extras);
+    the accumulated bias grad must match an unbound twin's .grad path."""
+    from gradient_accumulation_tf_estimator_amd.engine.accum import AccumEngine
+    from gradient_accumulation_tf_estimator_amd.models.bert import BertConfig, BertLayer
+    from gradient_accumulation_tf_estimator_amd.ops.fused import (
+        bind_direct_grad, direct_param_names)
+
+    cfg = BertConfig(vocab_size=64, hidden_size=512, num_layers=1,
+                     num_heads=8, intermediate_size=2048,
+                     max_position_embeddings=128, fused=True)
+
+    def make():
+        torch.manual_seed(5)
+        return BertLayer(cfg).cuda().bfloat16()
+
+    layerA, layerB = make(), make()
+    kw = dict(init_lr=0.0, num_train_steps=100, num_warmup_steps=0,
+              gradient_accumulation_multiplier=4, clip_norm=None)
+    engA = AccumEngine(list(layerA.named_parameters()), backend="hip",
+                       direct_names=direct_param_names(layerA), **kw)
+    engB = AccumEngine(list(layerB.named_parameters()), backend="hip", **kw)
+    bind_direct_grad(layerA, engA)
+    assert layerA.intermediate_act._bias_delegated
+    assert layerA.intermediate._accum_view_b is not None
+
+    for i in range(2):
+        torch.manual_seed(20 + i)
+        x = (torch.randn(4, 128, 512, device="cuda") * 0.5).bfloat16()
+        for layer, eng in ((layerA, engA), (layerB, engB)):
+            loss = (layer(x.clone()).float() ** 2).mean()
+            loss.backward()
+            eng.accumulate()
+    torch.cuda.synchronize()
+    stA, stB = engA.state, engB.state
+    by_name_B = {s.name: s for s in stB.layout.slices}
+    for sA in stA.layout.slices:
+        sB = by_name_B[sA.name]
+        a = stA.accum[sA.offset : sA.offset + sA.numel].cpu().numpy()
+        b = stB.accum[sB.offset : sB.offset + sB.numel].cpu().numpy()
+        np.testing.assert_allclose(a, b, rtol=2e-2, atol=4e-3,
+                                   err_msg=f"accum mismatch for {sA.name}")
+        assert abs(a).sum() > 0, sA.name
+
+
 def test_lt_gemm_matches_torch():
     """Autotuned hipBLASLt fwd/dgrad/wgrad vs torch matmul references."""
     from gradient_accumulation_tf_estimator_amd.ops import gemm
