@@ -15,7 +15,9 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 #include <hipblaslt/hipblaslt.h>
+#include <hipblaslt/hipblaslt-ext.hpp>
 
+#include <cstdlib>
 #include <map>
 #include <mutex>
 #include <tuple>
@@ -34,7 +36,7 @@ constexpr size_t kLtWorkspaceBytes = 64ull << 20;
 
 namespace {
 
-constexpr int kMaxAlgos = 16;
+constexpr int kMaxAlgos = 96;
 
 struct Plan {
   hipblasLtMatmulDesc_t desc;
@@ -117,6 +119,33 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
   TORCH_CHECK(st == HIPBLAS_STATUS_SUCCESS && found > 0,
               "no hipblaslt algo for shape R=", R, " N=", N, " K=", K);
   p.algos.assign(heur, heur + found);
+
+  // GA_LT_ALLALGOS=1: widen the tuning pool beyond the heuristic list with
+  // the full solution catalogue for this (transA, transB, dtype) class,
+  // filtered by matmulIsAlgoSupported (epilogue + shape + workspace). The
+  // heuristic picks stay at the front so algo_idx from older tuning runs
+  // keeps meaning.
+  if (kind <= 2 && std::getenv("GA_LT_ALLALGOS")) {
+    std::vector<hipblasLtMatmulHeuristicResult_t> all;
+    hipblasOperation_t eopA = (kind == 2) ? HIPBLAS_OP_N : HIPBLAS_OP_T;
+    if (hipblaslt_ext::getAllAlgos(ga_lt_handle(), hipblaslt_ext::GemmType::HIPBLASLT_GEMM,
+                                   eopA, HIPBLAS_OP_N, HIP_R_16BF, HIP_R_16BF,
+                                   HIP_R_16BF, HIP_R_16BF, HIPBLAS_COMPUTE_32F,
+                                   all) == HIPBLAS_STATUS_SUCCESS) {
+      const float alpha = 1.f, beta = 0.f;
+      for (auto& r : all) {
+        size_t ws_need = 0;
+        if (hipblaslt_ext::matmulIsAlgoSupported(ga_lt_handle(), p.desc, &alpha,
+                                                 p.la, p.lb, &beta, p.lc, p.lc,
+                                                 r.algo, ws_need)
+                == HIPBLAS_STATUS_SUCCESS
+            && ws_need <= kLtWorkspaceBytes) {
+          p.algos.push_back(r);
+          if ((int)p.algos.size() >= 512) break;
+        }
+      }
+    }
+  }
   return cache.emplace(key, p).first->second;
 }
 
