@@ -39,6 +39,39 @@ static inline __device__ unsigned short f2bf(float f) {
   return (unsigned short)((x + (((x >> 16) & 1u) + 0x7fffu)) >> 16);
 }
 
+// 16-byte vectorized bf16 load/store (8 x bf16): the dword4 path is what
+// reaches HBM3E peak -- ushort4 (8 B) access leaves ~2x bandwidth on the
+// table (measured on the engine kernels, which use float4).
+template <int V>
+static inline __device__ void ldv(const unsigned short* p, float* f);
+template <>
+__device__ void ldv<8>(const unsigned short* p, float* f) {
+  union { uint4 v; unsigned short u[8]; } t;
+  t.v = *(const uint4*)p;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) f[k] = bf2f(t.u[k]);
+}
+template <>
+__device__ void ldv<4>(const unsigned short* p, float* f) {
+  ushort4 t = *(const ushort4*)p;
+  f[0] = bf2f(t.x); f[1] = bf2f(t.y); f[2] = bf2f(t.z); f[3] = bf2f(t.w);
+}
+template <int V>
+static inline __device__ void stv(unsigned short* p, const float* f);
+template <>
+__device__ void stv<8>(unsigned short* p, const float* f) {
+  union { uint4 v; unsigned short u[8]; } t;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) t.u[k] = f2bf(f[k]);
+  *(uint4*)p = t.v;
+}
+template <>
+__device__ void stv<4>(unsigned short* p, const float* f) {
+  ushort4 t;
+  t.x = f2bf(f[0]); t.y = f2bf(f[1]); t.z = f2bf(f[2]); t.w = f2bf(f[3]);
+  *(ushort4*)p = t;
+}
+
 static inline __device__ float wave_sum(float s) {
   for (int off = 32; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
   return s;
@@ -62,27 +95,29 @@ __device__ __forceinline__ void addln_fwd_body(
   const int nwaves = gridDim.x * (blockDim.x >> 6);
   const int c0 = lane * epl;
 
+  constexpr int V = (EPL % 8 == 0) ? 8 : 4;
   float hv[EPL];
   for (int row = wid; row < R; row += nwaves) {
     const unsigned short* xr = x + (long long)row * H + c0;
     const unsigned short* rr = res ? res + (long long)row * H + c0 : nullptr;
     float s = 0.f;
 #pragma unroll
-    for (int c = 0; c < epl; c += 4) {
-      ushort4 xv = *(const ushort4*)(xr + c);
-      hv[c] = bf2f(xv.x); hv[c + 1] = bf2f(xv.y);
-      hv[c + 2] = bf2f(xv.z); hv[c + 3] = bf2f(xv.w);
+    for (int c = 0; c < epl; c += V) {
+      ldv<V>(xr + c, hv + c);
       if (rr) {
-        ushort4 rv = *(const ushort4*)(rr + c);
-        hv[c] += bf2f(rv.x); hv[c + 1] += bf2f(rv.y);
-        hv[c + 2] += bf2f(rv.z); hv[c + 3] += bf2f(rv.w);
+        float rv[V];
+        ldv<V>(rr + c, rv);
+#pragma unroll
+        for (int k = 0; k < V; ++k) hv[c + k] += rv[k];
       }
       if (bias) {
-        ushort4 bv = *(const ushort4*)(bias + c0 + c);
-        hv[c] += bf2f(bv.x); hv[c + 1] += bf2f(bv.y);
-        hv[c + 2] += bf2f(bv.z); hv[c + 3] += bf2f(bv.w);
+        float bv[V];
+        ldv<V>(bias + c0 + c, bv);
+#pragma unroll
+        for (int k = 0; k < V; ++k) hv[c + k] += bv[k];
       }
-      s += hv[c] + hv[c + 1] + hv[c + 2] + hv[c + 3];
+#pragma unroll
+      for (int k = 0; k < V; ++k) s += hv[c + k];
     }
     const float mean = wave_sum(s) / H;
     float sq = 0.f;
@@ -95,18 +130,15 @@ __device__ __forceinline__ void addln_fwd_body(
     unsigned short* yr = y + (long long)row * H + c0;
     unsigned short* hr = h_out + (long long)row * H + c0;
 #pragma unroll
-    for (int c = 0; c < epl; c += 4) {
-      ushort4 gv = *(const ushort4*)(gamma + c0 + c);
-      ushort4 bv = *(const ushort4*)(beta + c0 + c);
-      ushort4 ho, yo;
-      ho.x = f2bf(hv[c]);     ho.y = f2bf(hv[c + 1]);
-      ho.z = f2bf(hv[c + 2]); ho.w = f2bf(hv[c + 3]);
-      yo.x = f2bf(fmaf((hv[c] - mean) * rstd, bf2f(gv.x), bf2f(bv.x)));
-      yo.y = f2bf(fmaf((hv[c + 1] - mean) * rstd, bf2f(gv.y), bf2f(bv.y)));
-      yo.z = f2bf(fmaf((hv[c + 2] - mean) * rstd, bf2f(gv.z), bf2f(bv.z)));
-      yo.w = f2bf(fmaf((hv[c + 3] - mean) * rstd, bf2f(gv.w), bf2f(bv.w)));
-      *(ushort4*)(hr + c) = ho;
-      *(ushort4*)(yr + c) = yo;
+    for (int c = 0; c < epl; c += V) {
+      float gv[V], bv[V], yo[V];
+      ldv<V>(gamma + c0 + c, gv);
+      ldv<V>(beta + c0 + c, bv);
+#pragma unroll
+      for (int k = 0; k < V; ++k)
+        yo[k] = fmaf((hv[c + k] - mean) * rstd, gv[k], bv[k]);
+      stv<V>(hr + c, hv + c);
+      stv<V>(yr + c, yo);
     }
     if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
   }
@@ -131,14 +163,13 @@ __device__ __forceinline__ void addln_bwd_body(
   const int nwaves = gridDim.x * wpb;
   const int c0 = lane * epl;
 
+  constexpr int V = (EPL % 8 == 0) ? 8 : 4;
   float g[EPL], acc_dg[EPL], acc_db[EPL], acc_dbias[EPL];
 #pragma unroll
-  for (int c = 0; c < epl; c += 4) {
-    ushort4 gv = *(const ushort4*)(gamma + c0 + c);
-    g[c] = bf2f(gv.x); g[c + 1] = bf2f(gv.y); g[c + 2] = bf2f(gv.z); g[c + 3] = bf2f(gv.w);
-    acc_dg[c] = acc_dg[c + 1] = acc_dg[c + 2] = acc_dg[c + 3] = 0.f;
-    acc_db[c] = acc_db[c + 1] = acc_db[c + 2] = acc_db[c + 3] = 0.f;
-    acc_dbias[c] = acc_dbias[c + 1] = acc_dbias[c + 2] = acc_dbias[c + 3] = 0.f;
+  for (int c = 0; c < epl; c += V) ldv<V>(gamma + c0 + c, g + c);
+#pragma unroll
+  for (int c = 0; c < epl; ++c) {
+    acc_dg[c] = 0.f; acc_db[c] = 0.f; acc_dbias[c] = 0.f;
   }
 
   float dyv[EPL], xh[EPL];
@@ -148,14 +179,13 @@ __device__ __forceinline__ void addln_bwd_body(
     const unsigned short* hr = h + (long long)row * H + c0;
     float s1 = 0.f, s2 = 0.f;
 #pragma unroll
-    for (int c = 0; c < epl; c += 4) {
-      ushort4 dv = *(const ushort4*)(dyr + c);
-      ushort4 hv = *(const ushort4*)(hr + c);
-      dyv[c] = bf2f(dv.x); dyv[c + 1] = bf2f(dv.y);
-      dyv[c + 2] = bf2f(dv.z); dyv[c + 3] = bf2f(dv.w);
-      xh[c] = (bf2f(hv.x) - mean) * rstd; xh[c + 1] = (bf2f(hv.y) - mean) * rstd;
-      xh[c + 2] = (bf2f(hv.z) - mean) * rstd; xh[c + 3] = (bf2f(hv.w) - mean) * rstd;
-      for (int k = 0; k < 4; ++k) {
+    for (int c = 0; c < epl; c += V) {
+      float hv[V];
+      ldv<V>(dyr + c, dyv + c);
+      ldv<V>(hr + c, hv);
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        xh[c + k] = (hv[k] - mean) * rstd;
         const float dxh = dyv[c + k] * g[c + k];
         s1 += dxh;
         s2 = fmaf(dxh, xh[c + k], s2);
@@ -165,18 +195,17 @@ __device__ __forceinline__ void addln_bwd_body(
     s2 = wave_sum(s2) / H;
     unsigned short* dhr = dh_out + (long long)row * H + c0;
 #pragma unroll
-    for (int c = 0; c < epl; c += 4) {
-      ushort4 o;
-      float dh[4];
-      for (int k = 0; k < 4; ++k) {
+    for (int c = 0; c < epl; c += V) {
+      float dh[V];
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
         const float dxh = dyv[c + k] * g[c + k];
         dh[k] = rstd * (dxh - s1 - xh[c + k] * s2);
         acc_dg[c + k] = fmaf(dyv[c + k], xh[c + k], acc_dg[c + k]);
         acc_db[c + k] += dyv[c + k];
         acc_dbias[c + k] += dh[k];
       }
-      o.x = f2bf(dh[0]); o.y = f2bf(dh[1]); o.z = f2bf(dh[2]); o.w = f2bf(dh[3]);
-      *(ushort4*)(dhr + c) = o;
+      stv<V>(dhr + c, dh);
     }
   }
 
@@ -226,18 +255,15 @@ static inline __device__ float gelu_bwd1(float h, float dy) {
 extern "C" __global__ void k_biasgelu_fwd(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ bias,
     unsigned short* __restrict__ y, long long total, int H) {
-  long long stride = (long long)gridDim.x * blockDim.x * 4;
-  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < total;
+  long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
        i += stride) {
-    ushort4 xv = *(const ushort4*)(x + i);
-    const int c = (int)(i % H);
-    ushort4 bv = *(const ushort4*)(bias + c);
-    ushort4 o;
-    o.x = f2bf(gelu_fwd1(bf2f(xv.x) + bf2f(bv.x)));
-    o.y = f2bf(gelu_fwd1(bf2f(xv.y) + bf2f(bv.y)));
-    o.z = f2bf(gelu_fwd1(bf2f(xv.z) + bf2f(bv.z)));
-    o.w = f2bf(gelu_fwd1(bf2f(xv.w) + bf2f(bv.w)));
-    *(ushort4*)(y + i) = o;
+    float xv[8], bv[8], o[8];
+    ldv<8>(x + i, xv);
+    ldv<8>(bias + (int)(i % H), bv);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) o[k] = gelu_fwd1(xv[k] + bv[k]);
+    stv<8>(y + i, o);
   }
 }
 
@@ -256,6 +282,7 @@ __device__ __forceinline__ void biasgelu_bwd_body(
   const int nwaves = gridDim.x * wpb;
   const int c0 = lane * epl;
 
+  constexpr int V = (EPL % 8 == 0) ? 8 : 4;
   float acc[EPL];
 #pragma unroll
   for (int c = 0; c < epl; ++c) acc[c] = 0.f;
@@ -265,18 +292,17 @@ __device__ __forceinline__ void biasgelu_bwd_body(
     const unsigned short* xr = x + (long long)row * H + c0;
     unsigned short* dxr = dx_out + (long long)row * H + c0;
 #pragma unroll
-    for (int c = 0; c < epl; c += 4) {
-      ushort4 dv = *(const ushort4*)(dyr + c);
-      ushort4 xv = *(const ushort4*)(xr + c);
-      ushort4 bv = *(const ushort4*)(bias + c0 + c);
-      float d0 = gelu_bwd1(bf2f(xv.x) + bf2f(bv.x), bf2f(dv.x));
-      float d1 = gelu_bwd1(bf2f(xv.y) + bf2f(bv.y), bf2f(dv.y));
-      float d2 = gelu_bwd1(bf2f(xv.z) + bf2f(bv.z), bf2f(dv.z));
-      float d3 = gelu_bwd1(bf2f(xv.w) + bf2f(bv.w), bf2f(dv.w));
-      acc[c] += d0; acc[c + 1] += d1; acc[c + 2] += d2; acc[c + 3] += d3;
-      ushort4 o;
-      o.x = f2bf(d0); o.y = f2bf(d1); o.z = f2bf(d2); o.w = f2bf(d3);
-      *(ushort4*)(dxr + c) = o;
+    for (int c = 0; c < epl; c += V) {
+      float dv[V], xv[V], bv[V], d[V];
+      ldv<V>(dyr + c, dv);
+      ldv<V>(xr + c, xv);
+      ldv<V>(bias + c0 + c, bv);
+#pragma unroll
+      for (int k = 0; k < V; ++k) {
+        d[k] = gelu_bwd1(xv[k] + bv[k], dv[k]);
+        acc[c + k] += d[k];
+      }
+      stv<V>(dxr + c, d);
     }
   }
 
@@ -394,19 +420,17 @@ extern "C" __global__ void k_biasgelu_bwd_ew(
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ bias,
     unsigned short* __restrict__ dx_out, long long total, int H) {
-  long long stride = (long long)gridDim.x * blockDim.x * 4;
-  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  float bv8[8];
+  long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
        i < total; i += stride) {
-    ushort4 dv = *(const ushort4*)((const unsigned short*)dy + i);
-    ushort4 xv = *(const ushort4*)((const unsigned short*)x + i);
-    const int c = (int)(i % H);
-    ushort4 bv = *(const ushort4*)(bias + c);
-    ushort4 o;
-    o.x = f2bf(gelu_bwd1(bf2f(xv.x) + bf2f(bv.x), bf2f(dv.x)));
-    o.y = f2bf(gelu_bwd1(bf2f(xv.y) + bf2f(bv.y), bf2f(dv.y)));
-    o.z = f2bf(gelu_bwd1(bf2f(xv.z) + bf2f(bv.z), bf2f(dv.z)));
-    o.w = f2bf(gelu_bwd1(bf2f(xv.w) + bf2f(bv.w), bf2f(dv.w)));
-    *(ushort4*)(dx_out + i) = o;
+    float dv[8], xv[8], o[8];
+    ldv<8>(dy + i, dv);
+    ldv<8>(x + i, xv);
+    ldv<8>(bias + (int)(i % H), bv8);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) o[k] = gelu_bwd1(xv[k] + bv8[k], dv[k]);
+    stv<8>(dx_out + i, o);
   }
 }
 

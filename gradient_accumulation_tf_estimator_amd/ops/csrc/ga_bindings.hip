@@ -253,7 +253,7 @@ at::Tensor biasgelu_fwd(at::Tensor x, at::Tensor bias) {
   auto y = at::empty_like(x);
   const long long total = x.numel();
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+  int blocks = (int)std::min<long long>((total / 8 + 255) / 256, 2048);
   hipLaunchKernelGGL(k_biasgelu_fwd, dim3(blocks), dim3(256), 0, stream,
                      bfp(x), bfp(bias), bfp_mut(y), total, H);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_fwd launch failed");
@@ -490,7 +490,7 @@ at::Tensor biasgelu_bwd_ew(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   auto dx = at::empty_like(dy);
   const long long total = dy.numel();
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+  int blocks = (int)std::min<long long>((total / 8 + 255) / 256, 2048);
   hipLaunchKernelGGL(k_biasgelu_bwd_ew, dim3(blocks), dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
