@@ -39,6 +39,17 @@ extern "C" __global__ void k_biasgelu_fwd(const unsigned short*, const unsigned 
 extern "C" __global__ void k_biasgelu_bwd_ew(const unsigned short*, const unsigned short*,
                                              const unsigned short*, unsigned short*,
                                              long long, int);
+extern "C" __global__ void k_lin_fwd_bf16(const unsigned short*, const unsigned short*,
+                                          const unsigned short*, unsigned short*,
+                                          int, int, int);
+extern "C" __global__ void k_lin_fwd_f32(const unsigned short*, const unsigned short*,
+                                         float*, int, int, int, int);
+extern "C" __global__ void k_lin_dgrad_bf16(const unsigned short*, const unsigned short*,
+                                            unsigned short*, int, int, int);
+extern "C" __global__ void k_lin_dgrad_f32(const unsigned short*, const unsigned short*,
+                                           float*, int, int, int, int);
+extern "C" __global__ void k_splitk_combine(const float*, int, long long, int,
+                                            const unsigned short*, unsigned short*);
 extern "C" __global__ void k_colreduce_acc(const float*, int, int, float*, int,
                                            float*, int, float*);
 #define CRB_MAX_G 16
@@ -484,6 +495,74 @@ at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
 }
 
 // problems: list of (partials [NB,...,C], dest0, dest1?, dest2?)
+// ---- small-GEMM MFMA Linear path (linear_small.hip) ----
+at::Tensor lin_fwd_small(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias) {
+  const int N = (int)w.size(0), K = (int)w.size(1);
+  const int R = (int)(x.numel() / K);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous() && x.scalar_type() == at::kBFloat16,
+              "lin_fwd_small: contiguous bf16 required");
+  TORCH_CHECK(R % 64 == 0 && N % 64 == 0 && K % 64 == 0, "lin_fwd_small shape");
+  auto sizes = x.sizes().vec();
+  sizes.back() = N;
+  auto y = at::empty(sizes, x.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int tiles = (R / 64) * (N / 64);
+  const int S = (K >= 1024) ? 4 : 1;
+  const unsigned short* bp = bias ? (const unsigned short*)bias->data_ptr() : nullptr;
+  if (S == 1) {
+    hipLaunchKernelGGL(k_lin_fwd_bf16, dim3(tiles), dim3(64), 0, stream,
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(), bp,
+                       (unsigned short*)y.data_ptr(), R, N, K);
+  } else {
+    auto part = at::empty({S, (long)R, (long)N}, x.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(k_lin_fwd_f32, dim3(tiles, S), dim3(64), 0, stream,
+                       (const unsigned short*)x.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       part.data_ptr<float>(), R, N, K, S);
+    const long long total = (long long)R * N;
+    int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+    hipLaunchKernelGGL(k_splitk_combine, dim3(blocks), dim3(256), 0, stream,
+                       part.data_ptr<float>(), S, total, N, bp,
+                       (unsigned short*)y.data_ptr());
+  }
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "lin_fwd_small launch failed");
+  return y;
+}
+
+at::Tensor lin_dgrad_small(at::Tensor dy, at::Tensor w) {
+  const int N = (int)w.size(0), K = (int)w.size(1);
+  const int R = (int)(dy.numel() / N);
+  TORCH_CHECK(dy.is_contiguous() && w.is_contiguous() && dy.scalar_type() == at::kBFloat16,
+              "lin_dgrad_small: contiguous bf16 required");
+  TORCH_CHECK(R % 64 == 0 && N % 64 == 0 && K % 64 == 0, "lin_dgrad_small shape");
+  auto sizes = dy.sizes().vec();
+  sizes.back() = K;
+  auto dx = at::empty(sizes, dy.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int tiles = (R / 64) * (K / 64);
+  const int S = (N >= 1024) ? 4 : 1;
+  if (S == 1) {
+    hipLaunchKernelGGL(k_lin_dgrad_bf16, dim3(tiles), dim3(64), 0, stream,
+                       (const unsigned short*)dy.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (unsigned short*)dx.data_ptr(), R, N, K);
+  } else {
+    auto part = at::empty({S, (long)R, (long)K}, dy.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(k_lin_dgrad_f32, dim3(tiles, S), dim3(64), 0, stream,
+                       (const unsigned short*)dy.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       part.data_ptr<float>(), R, N, K, S);
+    const long long total = (long long)R * K;
+    int blocks = (int)std::min<long long>((total / 4 + 255) / 256, 2048);
+    hipLaunchKernelGGL(k_splitk_combine, dim3(blocks), dim3(256), 0, stream,
+                       part.data_ptr<float>(), S, total, K, nullptr,
+                       (unsigned short*)dx.data_ptr());
+  }
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "lin_dgrad_small launch failed");
+  return dx;
+}
+
 at::Tensor biasgelu_bwd_ew(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   const int H = (int)bias.numel();
   check_bf16_2d(dy, "dy", H);
@@ -571,6 +650,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
   mod.def("colreduce_acc", &colreduce_acc,
           "reduce partials over blocks, ADD into flat fp32 accum slices");
+  mod.def("lin_fwd_small", &lin_fwd_small,
+          "small-GEMM MFMA forward (64x64 tiles, split-K >= 1024)");
+  mod.def("lin_dgrad_small", &lin_dgrad_small,
+          "small-GEMM MFMA dgrad (W^T LDS transpose staging)");
   mod.def("biasgelu_bwd_ew", &biasgelu_bwd_ew,
           "elementwise gelu backward (bias grad delegated to wgrad colsum)");
   mod.def("colreduce_batch", &colreduce_batch,
