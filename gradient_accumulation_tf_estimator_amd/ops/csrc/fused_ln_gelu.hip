@@ -81,7 +81,7 @@ static inline __device__ float wave_sum(float s) {
 // EPL = H/64 is a template parameter: per-lane arrays must be indexed by
 // compile-time constants or they land in scratch (cdna_hip_programming.md
 // rule 20) -- the first version of these kernels paid 3-6x for that.
-template <int EPL>
+template <int EPL, int SPLIT>
 __device__ __forceinline__ void addln_fwd_body(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ res,
     const unsigned short* __restrict__ bias,  // [H] or null
@@ -89,65 +89,102 @@ __device__ __forceinline__ void addln_fwd_body(
     unsigned short* __restrict__ y, unsigned short* __restrict__ h_out,
     float* __restrict__ mean_out, float* __restrict__ rstd_out,
     int R, int H, float eps) {
-  constexpr int epl = EPL;
+  // SPLIT waves cooperate on one row (SPLIT=2 doubles the wave count at the
+  // bench row counts -- one wave/SIMD has zero latency hiding otherwise);
+  // partial sums cross the wave pair through a tiny static-LDS exchange.
+  constexpr int epl = EPL;  // elements per lane = H / (64*SPLIT)
   const int lane = threadIdx.x & 63;
-  const int wid = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
-  const int nwaves = gridDim.x * (blockDim.x >> 6);
-  const int c0 = lane * epl;
-
+  const int wlocal = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  const int wid = blockIdx.x * wpb + wlocal;
+  const int nwaves = gridDim.x * wpb;
+  const long long units = (long long)R * SPLIT;
+  const int half = (SPLIT == 1) ? 0 : (wid & 1);  // stride nwaves is even
+  const int c0 = half * (H / SPLIT) + lane * epl;
   constexpr int V = (EPL % 8 == 0) ? 8 : 4;
+
+  __shared__ float exs[2][8];  // [phase][wlocal] cross-wave partials
+
   float hv[EPL];
-  for (int row = wid; row < R; row += nwaves) {
-    const unsigned short* xr = x + (long long)row * H + c0;
-    const unsigned short* rr = res ? res + (long long)row * H + c0 : nullptr;
+  const int iters = (int)((units + nwaves - 1) / nwaves);
+  for (int it = 0; it < iters; ++it) {
+    const long long unit = wid + (long long)it * nwaves;
+    const bool active = unit < units;
+    const int row = (int)(unit / SPLIT);
     float s = 0.f;
+    if (active) {
+      const unsigned short* xr = x + (long long)row * H + c0;
+      const unsigned short* rr = res ? res + (long long)row * H + c0 : nullptr;
 #pragma unroll
-    for (int c = 0; c < epl; c += V) {
-      ldv<V>(xr + c, hv + c);
-      if (rr) {
-        float rv[V];
-        ldv<V>(rr + c, rv);
+      for (int c = 0; c < epl; c += V) {
+        ldv<V>(xr + c, hv + c);
+        if (rr) {
+          float rv[V];
+          ldv<V>(rr + c, rv);
 #pragma unroll
-        for (int k = 0; k < V; ++k) hv[c + k] += rv[k];
+          for (int k = 0; k < V; ++k) hv[c + k] += rv[k];
+        }
+        if (bias) {
+          float bv[V];
+          ldv<V>(bias + c0 + c, bv);
+#pragma unroll
+          for (int k = 0; k < V; ++k) hv[c + k] += bv[k];
+        }
+#pragma unroll
+        for (int k = 0; k < V; ++k) s += hv[c + k];
       }
-      if (bias) {
-        float bv[V];
-        ldv<V>(bias + c0 + c, bv);
+    } else {
 #pragma unroll
-        for (int k = 0; k < V; ++k) hv[c + k] += bv[k];
-      }
-#pragma unroll
-      for (int k = 0; k < V; ++k) s += hv[c + k];
+      for (int c = 0; c < epl; ++c) hv[c] = 0.f;
     }
-    const float mean = wave_sum(s) / H;
+    float mean;
+    if (SPLIT == 1) {
+      mean = wave_sum(s) / H;
+    } else {
+      s = wave_sum(s);
+      __syncthreads();  // previous iteration's exchange reads done
+      if (lane == 0) exs[0][wlocal] = s;
+      __syncthreads();
+      mean = (exs[0][wlocal] + exs[0][wlocal ^ 1]) / H;
+    }
     float sq = 0.f;
 #pragma unroll
     for (int c = 0; c < epl; ++c) {
       const float d = hv[c] - mean;
       sq = fmaf(d, d, sq);
     }
-    const float rstd = rsqrtf(wave_sum(sq) / H + eps);
-    unsigned short* yr = y + (long long)row * H + c0;
-    unsigned short* hr = h_out + (long long)row * H + c0;
-#pragma unroll
-    for (int c = 0; c < epl; c += V) {
-      float gv[V], bv[V], yo[V];
-      ldv<V>(gamma + c0 + c, gv);
-      ldv<V>(beta + c0 + c, bv);
-#pragma unroll
-      for (int k = 0; k < V; ++k)
-        yo[k] = fmaf((hv[c + k] - mean) * rstd, gv[k], bv[k]);
-      stv<V>(hr + c, hv + c);
-      stv<V>(yr + c, yo);
+    float rstd;
+    if (SPLIT == 1) {
+      rstd = rsqrtf(wave_sum(sq) / H + eps);
+    } else {
+      sq = wave_sum(sq);
+      if (lane == 0) exs[1][wlocal] = sq;
+      __syncthreads();
+      rstd = rsqrtf((exs[1][wlocal] + exs[1][wlocal ^ 1]) / H + eps);
     }
-    if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+    if (active) {
+      unsigned short* yr = y + (long long)row * H + c0;
+      unsigned short* hr = h_out + (long long)row * H + c0;
+#pragma unroll
+      for (int c = 0; c < epl; c += V) {
+        float gv[V], bv[V], yo[V];
+        ldv<V>(gamma + c0 + c, gv);
+        ldv<V>(beta + c0 + c, bv);
+#pragma unroll
+        for (int k = 0; k < V; ++k)
+          yo[k] = fmaf((hv[c + k] - mean) * rstd, gv[k], bv[k]);
+        stv<V>(hr + c, hv + c);
+        stv<V>(yr + c, yo);
+      }
+      if (lane == 0 && half == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
+    }
   }
 }
 
 // ---------------- fused LayerNorm backward ----------------
 // dh = rstd * (dxh - mean(dxh) - xh * mean(dxh*xh)),  dxh = dy*gamma,
 // xh = (h-mean)*rstd.  Partials per block: [3][H] fp32 = {dgamma, dbeta, db}.
-template <int EPL>
+template <int EPL, int SPLIT>
 __device__ __forceinline__ void addln_bwd_body(
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ h,
     const unsigned short* __restrict__ gamma,
@@ -155,15 +192,20 @@ __device__ __forceinline__ void addln_bwd_body(
     unsigned short* __restrict__ dh_out,
     float* __restrict__ partials,  // [gridDim.x][3][H]
     int R, int H) {
-  constexpr int epl = EPL;
+  constexpr int epl = EPL;  // elements per lane = H / (64*SPLIT)
   const int lane = threadIdx.x & 63;
   const int wlocal = threadIdx.x >> 6;
   const int wpb = blockDim.x >> 6;
   const int wid = blockIdx.x * wpb + wlocal;
   const int nwaves = gridDim.x * wpb;
-  const int c0 = lane * epl;
-
+  const long long units = (long long)R * SPLIT;
+  const int half = (SPLIT == 1) ? 0 : (wid & 1);
+  const int halfH = H / SPLIT;
+  const int c0 = half * halfH + lane * epl;
   constexpr int V = (EPL % 8 == 0) ? 8 : 4;
+
+  __shared__ float exs[2][8];
+
   float g[EPL], acc_dg[EPL], acc_db[EPL], acc_dbias[EPL];
 #pragma unroll
   for (int c = 0; c < epl; c += V) ldv<V>(gamma + c0 + c, g + c);
@@ -173,57 +215,83 @@ __device__ __forceinline__ void addln_bwd_body(
   }
 
   float dyv[EPL], xh[EPL];
-  for (int row = wid; row < R; row += nwaves) {
-    const float mean = mean_in[row], rstd = rstd_in[row];
-    const unsigned short* dyr = dy + (long long)row * H + c0;
-    const unsigned short* hr = h + (long long)row * H + c0;
+  const int iters = (int)((units + nwaves - 1) / nwaves);
+  for (int it = 0; it < iters; ++it) {
+    const long long unit = wid + (long long)it * nwaves;
+    const bool active = unit < units;
+    const int row = (int)(unit / SPLIT);
     float s1 = 0.f, s2 = 0.f;
+    if (active) {
+      const float mean = mean_in[row], rstd = rstd_in[row];
+      const unsigned short* dyr = dy + (long long)row * H + c0;
+      const unsigned short* hr = h + (long long)row * H + c0;
 #pragma unroll
-    for (int c = 0; c < epl; c += V) {
-      float hv[V];
-      ldv<V>(dyr + c, dyv + c);
-      ldv<V>(hr + c, hv);
+      for (int c = 0; c < epl; c += V) {
+        float hvv[V];
+        ldv<V>(dyr + c, dyv + c);
+        ldv<V>(hr + c, hvv);
 #pragma unroll
-      for (int k = 0; k < V; ++k) {
-        xh[c + k] = (hv[k] - mean) * rstd;
-        const float dxh = dyv[c + k] * g[c + k];
-        s1 += dxh;
-        s2 = fmaf(dxh, xh[c + k], s2);
+        for (int k = 0; k < V; ++k) {
+          xh[c + k] = (hvv[k] - mean) * rstd;
+          const float dxh = dyv[c + k] * g[c + k];
+          s1 += dxh;
+          s2 = fmaf(dxh, xh[c + k], s2);
+        }
       }
     }
-    s1 = wave_sum(s1) / H;
-    s2 = wave_sum(s2) / H;
-    unsigned short* dhr = dh_out + (long long)row * H + c0;
+    if (SPLIT == 1) {
+      s1 = wave_sum(s1) / H;
+      s2 = wave_sum(s2) / H;
+    } else {
+      s1 = wave_sum(s1);
+      s2 = wave_sum(s2);
+      __syncthreads();
+      if (lane == 0) { exs[0][wlocal] = s1; exs[1][wlocal] = s2; }
+      __syncthreads();
+      s1 = (exs[0][wlocal] + exs[0][wlocal ^ 1]) / H;
+      s2 = (exs[1][wlocal] + exs[1][wlocal ^ 1]) / H;
+    }
+    if (active) {
+      const float rstd = rstd_in[row];
+      unsigned short* dhr = dh_out + (long long)row * H + c0;
 #pragma unroll
-    for (int c = 0; c < epl; c += V) {
-      float dh[V];
+      for (int c = 0; c < epl; c += V) {
+        float dh[V];
 #pragma unroll
-      for (int k = 0; k < V; ++k) {
-        const float dxh = dyv[c + k] * g[c + k];
-        dh[k] = rstd * (dxh - s1 - xh[c + k] * s2);
-        acc_dg[c + k] = fmaf(dyv[c + k], xh[c + k], acc_dg[c + k]);
-        acc_db[c + k] += dyv[c + k];
-        acc_dbias[c + k] += dh[k];
+        for (int k = 0; k < V; ++k) {
+          const float dxh = dyv[c + k] * g[c + k];
+          dh[k] = rstd * (dxh - s1 - xh[c + k] * s2);
+          acc_dg[c + k] = fmaf(dyv[c + k], xh[c + k], acc_dg[c + k]);
+          acc_db[c + k] += dyv[c + k];
+          acc_dbias[c + k] += dh[k];
+        }
+        stv<V>(dhr + c, dh);
       }
-      stv<V>(dhr + c, dh);
     }
   }
 
-  // combine the block's waves in LDS, then one fp32 partial slab per block
-  extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][3][H]
-  float* my = lds + ((size_t)wlocal * 3 * H);
+  // combine the block's waves in LDS, then one fp32 partial slab per block.
+  // SPLIT=2: each wave owns a half-row slab [wlocal][3][H/SPLIT]; the
+  // columns of the output slab pull from the (row-local, half) wave pair.
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [wpb][3][halfH]
+  float* my = lds + ((size_t)wlocal * 3 * halfH);
+  __syncthreads();  // exs exchange reads done before LDS reuse... (distinct mem, but order stores)
 #pragma unroll
   for (int c = 0; c < epl; ++c) {
-    my[c0 + c] = acc_dg[c];
-    my[H + c0 + c] = acc_db[c];
-    my[2 * H + c0 + c] = acc_dbias[c];
+    const int cl = c0 - half * halfH + c;  // column within the half slab
+    my[cl] = acc_dg[c];
+    my[halfH + cl] = acc_db[c];
+    my[2 * halfH + cl] = acc_dbias[c];
   }
   __syncthreads();
   float* out = partials + (size_t)blockIdx.x * 3 * H;
   for (int i = threadIdx.x; i < 3 * H; i += blockDim.x) {
-    float s = 0.f;
-    for (int w = 0; w < wpb; ++w) s += lds[(size_t)w * 3 * H + i];
-    out[i] = s;
+    const int p = i / H, col = i % H;
+    const int hf = col / halfH, cl = col % halfH;
+    float sum = 0.f;
+    for (int rl = 0; rl < wpb / SPLIT; ++rl)
+      sum += lds[(size_t)(rl * SPLIT + hf) * 3 * halfH + p * halfH + cl];
+    out[i] = sum;
   }
 }
 
@@ -260,7 +328,8 @@ extern "C" __global__ void k_biasgelu_fwd(
        i += stride) {
     float xv[8], bv[8], o[8];
     ldv<8>(x + i, xv);
-    ldv<8>(bias + (int)(i % H), bv);
+    const int c = ((H & (H - 1)) == 0) ? (int)(i & (H - 1)) : (int)(i % H);
+    ldv<8>(bias + c, bv);
 #pragma unroll
     for (int k = 0; k < 8; ++k) o[k] = gelu_fwd1(xv[k] + bv[k]);
     stv<8>(y + i, o);
@@ -350,27 +419,30 @@ extern "C" __global__ void k_colreduce_acc(
 
 
 // ---------------- explicit instantiations (H = 64*EPL) ----------------
-#define GA_LN_INST(EPL)                                                        \
-  extern "C" __global__ void k_addln_fwd_##EPL(                                \
+#define GA_LN_INST(NAME, EPL, SPLIT)                                           \
+  extern "C" __global__ void k_addln_fwd_##NAME(                               \
       const unsigned short* x, const unsigned short* res,                      \
       const unsigned short* bias, const unsigned short* gamma,                 \
       const unsigned short* beta, unsigned short* y, unsigned short* h_out,    \
       float* mean_out, float* rstd_out, int R, int H, float eps) {             \
-    addln_fwd_body<EPL>(x, res, bias, gamma, beta, y, h_out, mean_out,         \
-                        rstd_out, R, H, eps);                                  \
+    addln_fwd_body<EPL, SPLIT>(x, res, bias, gamma, beta, y, h_out, mean_out,  \
+                               rstd_out, R, H, eps);                           \
   }                                                                            \
-  extern "C" __global__ void k_addln_bwd_##EPL(                                \
+  extern "C" __global__ void k_addln_bwd_##NAME(                               \
       const unsigned short* dy, const unsigned short* h,                       \
       const unsigned short* gamma, const float* mean_in, const float* rstd_in, \
       unsigned short* dh_out, float* partials, int R, int H) {                 \
-    addln_bwd_body<EPL>(dy, h, gamma, mean_in, rstd_in, dh_out, partials, R,   \
-                        H);                                                    \
+    addln_bwd_body<EPL, SPLIT>(dy, h, gamma, mean_in, rstd_in, dh_out,         \
+                               partials, R, H);                                \
   }
 
-GA_LN_INST(4)
-GA_LN_INST(8)
-GA_LN_INST(12)
-GA_LN_INST(16)
+GA_LN_INST(4, 4, 1)
+GA_LN_INST(8, 8, 1)
+GA_LN_INST(12, 12, 1)
+GA_LN_INST(16, 16, 1)
+// split-row pairs: H = 64*EPL*2 (H=512 -> s2_4, H=1024 -> s2_8)
+GA_LN_INST(s2_4, 4, 2)
+GA_LN_INST(s2_8, 8, 2)
 
 #define GA_GELU_INST(EPL)                                                      \
   extern "C" __global__ void k_biasgelu_bwd_##EPL(                             \
@@ -427,7 +499,8 @@ extern "C" __global__ void k_biasgelu_bwd_ew(
     float dv[8], xv[8], o[8];
     ldv<8>(dy + i, dv);
     ldv<8>(x + i, xv);
-    ldv<8>(bias + (int)(i % H), bv8);
+    const int c = ((H & (H - 1)) == 0) ? (int)(i & (H - 1)) : (int)(i % H);
+    ldv<8>(bias + c, bv8);
 #pragma unroll
     for (int k = 0; k < 8; ++k) o[k] = gelu_bwd1(xv[k] + bv8[k], dv[k]);
     stv<8>(dx_out + i, o);

@@ -2,6 +2,7 @@
 // Native HIP path only (no CUDA shim, no hipify): launches on the current
 // c10 HIP stream so torch.cuda.graphs capture works.
 
+#include <cstdlib>
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
@@ -29,6 +30,7 @@ extern "C" __global__ void k_apply_bf16(float4*, float4*, float4*, float4*, usho
       const unsigned short*, const unsigned short*, const unsigned short*, const float*, \
       const float*, unsigned short*, float*, int, int);
 GA_DECL_LN(4) GA_DECL_LN(8) GA_DECL_LN(12) GA_DECL_LN(16)
+GA_DECL_LN(s2_4) GA_DECL_LN(s2_8)
 #define GA_DECL_GELU(EPL)                                                 \
   extern "C" __global__ void k_biasgelu_bwd_##EPL(                        \
       const unsigned short*, const unsigned short*, const unsigned short*, \
@@ -221,11 +223,19 @@ std::vector<at::Tensor> addln_fwd(at::Tensor x, c10::optional<at::Tensor> res,
   auto mean = at::empty({R}, x.options().dtype(at::kFloat));
   auto rstd = at::empty({R}, x.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = std::min((R + 3) / 4, 2048);
+  // Split-row (2 waves/row) was measured SLOWER end-to-end on MI355X
+  // (9752 vs 9965 samples/s): the per-row LDS exchanges + barriers cost
+  // more than the extra occupancy buys. GA_LN_SPLIT2=1 re-enables it for
+  // future re-measurement; kernels kept (fused_ln_gelu.hip SPLIT param).
+  static const bool want_split = std::getenv("GA_LN_SPLIT2") != nullptr;
+  const bool split2 = want_split && (H == 512 || H == 1024) && R >= 256;
+  const int units = split2 ? R * 2 : R;
+  int blocks = std::min((units + 3) / 4, 2048);
   void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
                const unsigned short*, const unsigned short*, unsigned short*,
                unsigned short*, float*, float*, int, int, float) =
-      H == 256 ? k_addln_fwd_4 : H == 512 ? k_addln_fwd_8
+      split2 ? (H == 512 ? k_addln_fwd_s2_4 : k_addln_fwd_s2_8)
+      : H == 256 ? k_addln_fwd_4 : H == 512 ? k_addln_fwd_8
       : H == 768 ? k_addln_fwd_12 : k_addln_fwd_16;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, stream,
                      bfp(x), res ? bfp(*res) : nullptr, bias ? bfp(*bias) : nullptr,
@@ -241,13 +251,17 @@ std::vector<at::Tensor> addln_bwd(at::Tensor dy, at::Tensor h, at::Tensor gamma,
   check_bf16_2d(dy, "dy", H);
   const int R = (int)(dy.numel() / H);
   auto dh = at::empty_like(dy);
-  int blocks = std::max(1, std::min((R + 3) / 4, 256));
+  static const bool want_split = std::getenv("GA_LN_SPLIT2") != nullptr;
+  const bool split2 = want_split && (H == 512 || H == 1024) && R >= 256;
+  const int units = split2 ? R * 2 : R;
+  int blocks = std::max(1, std::min((units + 3) / 4, 256));
   auto partials = at::empty({blocks, 3, H}, dy.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  size_t lds = (size_t)4 * 3 * H * sizeof(float);
+  size_t lds = (size_t)4 * 3 * (split2 ? H / 2 : H) * sizeof(float);
   void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
                const float*, const float*, unsigned short*, float*, int, int) =
-      H == 256 ? k_addln_bwd_4 : H == 512 ? k_addln_bwd_8
+      split2 ? (H == 512 ? k_addln_bwd_s2_4 : k_addln_bwd_s2_8)
+      : H == 256 ? k_addln_bwd_4 : H == 512 ? k_addln_bwd_8
       : H == 768 ? k_addln_bwd_12 : k_addln_bwd_16;
   hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), lds, stream,
                      bfp(dy), bfp(h), bfp(gamma), mean.data_ptr<float>(),

@@ -17,9 +17,9 @@ from gradient_accumulation_tf_estimator_amd.ops.fused import (
 
 @pytest.mark.parametrize("H", [512, 768, 1024])
 @pytest.mark.parametrize("with_res,with_pb", [(False, False), (True, True)])
-def test_fused_addln_forward_backward(H, with_res, with_pb):
+@pytest.mark.parametrize("R", [136, 1024])  # small + bench-scale row counts
+def test_fused_addln_forward_backward(H, with_res, with_pb, R):
     torch.manual_seed(H)
-    R = 136
     mod = FusedAddLayerNorm(H, eps=1e-12, proj_bias=with_pb).cuda().bfloat16()
     with torch.no_grad():
         mod.weight.copy_(torch.randn(H) * 0.2 + 1)

@@ -3,11 +3,17 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."
 import torch, time
 from gradient_accumulation_tf_estimator_amd.ops import require_hip
 hip = require_hip()
-def t(fn, n=200):
-    for _ in range(20): fn()
-    torch.cuda.synchronize(); t0=time.perf_counter()
-    for _ in range(n): fn()
-    torch.cuda.synchronize(); return (time.perf_counter()-t0)/n*1e6
+def t(fn, n=50, reps=20):
+    for _ in range(10): fn()
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        for _ in range(n): fn()
+    g.replay(); torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(reps): g.replay()
+    torch.cuda.synchronize()
+    return (time.perf_counter()-t0)/(n*reps)*1e6
 R,H = 1024,512
 x = torch.randn(R,H,device="cuda").bfloat16(); res = torch.randn_like(x)
 g = torch.randn(H,device="cuda").bfloat16(); b = torch.randn_like(g)
