@@ -168,6 +168,14 @@ class BertLayer(nn.Module):
             self.intermediate._accum_view_b = engine.state.accum_view(
                 self.intermediate_act.bias)
             self.intermediate_act._bias_delegated = True
+            # residual-branch grad adds (x feeds attention AND its LN;
+            # x2 feeds the FFN AND its LN) fold into the branch Linear's
+            # dgrad epilogue: attention_LayerNorm's dres rides qkv's dgrad,
+            # output_LayerNorm's dres rides intermediate's dgrad. Ordering:
+            # the LN backward runs strictly before that dgrad, and the
+            # summed grad's consumers run strictly after it.
+            self.attention_LayerNorm._defer_residual_to = self.attention.qkv
+            self.output_LayerNorm._defer_residual_to = self.intermediate
 
     def forward(self, x, attn_mask=None):
         a = self.attention(x, attn_mask)

@@ -150,8 +150,9 @@ Plan& plan_for(int kind, int64_t R, int64_t N, int64_t K) {
 }
 
 void run(Plan& p, const void* A, const void* B, void* D, const void* bias,
-         int64_t algo_idx, void* aux = nullptr) {
-  const float alpha = 1.f, beta = 0.f;
+         int64_t algo_idx, void* aux = nullptr, const void* C = nullptr,
+         float beta = 0.f) {
+  const float alpha = 1.f;
   if (bias) {
     LT_CHECK(hipblasLtMatmulDescSetAttribute(
         p.desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &bias, sizeof(bias)));
@@ -164,7 +165,7 @@ void run(Plan& p, const void* A, const void* B, void* D, const void* bias,
   int idx = (int)std::min<int64_t>(std::max<int64_t>(algo_idx, 0),
                                    (int64_t)p.algos.size() - 1);
   LT_CHECK(hipblasLtMatmul(ga_lt_handle(), p.desc, &alpha, A, p.la, B, p.lb,
-                           &beta, D, p.lc, D, p.lc, &p.algos[idx].algo,
+                           &beta, C ? C : D, p.lc, D, p.lc, &p.algos[idx].algo,
                            ga_lt_workspace(), kLtWorkspaceBytes, stream));
 }
 
@@ -195,6 +196,25 @@ at::Tensor lt_dgrad(at::Tensor dy, at::Tensor w, int64_t algo_idx) {
   auto dx = at::empty(sizes, dy.options());
   auto& p = plan_for(2, R, N, K);
   run(p, w.data_ptr(), dy.data_ptr(), dx.data_ptr(), nullptr, algo_idx);
+  return dx;
+}
+
+// dx = dy @ W + addend in one GEMM (beta=1, C=addend, D=dx): folds the
+// residual-branch gradient add (autograd's two-consumer sum) into the
+// dgrad's epilogue -- C and D are distinct, so the addend (also the
+// deferred wgrad's dy upstream) is never clobbered.
+at::Tensor lt_dgrad_add(at::Tensor dy, at::Tensor w, at::Tensor addend,
+                        int64_t algo_idx) {
+  TORCH_CHECK(dy.is_contiguous() && w.is_contiguous() && addend.is_contiguous());
+  const int64_t K = w.size(1), N = w.size(0);
+  const int64_t R = dy.numel() / N;
+  TORCH_CHECK(addend.numel() == R * K, "lt_dgrad_add: addend shape");
+  auto sizes = dy.sizes().vec();
+  sizes.back() = K;
+  auto dx = at::empty(sizes, dy.options());
+  auto& p = plan_for(2, R, N, K);
+  run(p, w.data_ptr(), dy.data_ptr(), dx.data_ptr(), nullptr, algo_idx,
+      nullptr, addend.data_ptr(), 1.f);
   return dx;
 }
 
@@ -241,5 +261,6 @@ void register_lt_gemm(pybind11::module_& mod) {
   mod.def("lt_gelu_algo_count", &lt_gelu_algo_count);
   mod.def("lt_algo_count", &lt_algo_count, "heuristic candidates for a shape");
   mod.def("lt_linear", &lt_linear, "y = x @ W^T (+bias), hipblaslt, algo_idx");
+  mod.def("lt_dgrad_add", &lt_dgrad_add, "dgrad + residual-grad add epilogue");
   mod.def("lt_dgrad", &lt_dgrad, "dx = dy @ W, hipblaslt, algo_idx");
 }

@@ -143,6 +143,7 @@ class _AddLayerNormFn(torch.autograd.Function):
         ctx.module = module
         ctx.has_res = residual is not None
         ctx.has_pb = proj_bias is not None
+        ctx.defer_to = getattr(module, "_defer_residual_to", None)
         return y.view_as(x)
 
     @staticmethod
@@ -165,6 +166,16 @@ class _AddLayerNormFn(torch.autograd.Function):
                 dpb = s[2].to(dy.dtype)
         dh = dh.view_as(dy)
         dres = dh if ctx.has_res else None
+        if dres is not None and ctx.defer_to is not None:
+            # residual-branch grad rides the deferred Linear's dgrad epilogue
+            # (dx = dy @ W + dres in ONE GEMM) instead of autograd's add
+            # kernel; see BertLayer._bind_direct_extras for the wiring and
+            # the ordering argument (the deferred dgrad runs strictly after
+            # this backward and before any consumer of the summed grad).
+            assert getattr(ctx.defer_to, "_pending_dres_add", None) is None, \
+                "deferred residual grad was never consumed"
+            ctx.defer_to._pending_dres_add = dh
+            dres = None
         return dh, dres, dgamma, dbeta, dpb, None, None
 
 
@@ -305,7 +316,13 @@ class _DirectLinearFn(torch.autograd.Function):
         x, weight = ctx.saved_tensors
         dy = dy.contiguous()
         dy2d = dy.reshape(-1, dy.shape[-1])
-        dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
+        addend = getattr(ctx.module, "_pending_dres_add", None)
+        if addend is not None:
+            ctx.module._pending_dres_add = None
+            dx = gemm.dgrad_add(dy2d, weight,
+                                addend.reshape(-1, addend.shape[-1])).reshape(x.shape)
+        else:
+            dx = gemm.dgrad(dy2d, weight).reshape(x.shape)
         x2d = x.reshape(-1, x.shape[-1])
         vb = ctx.module._accum_view_b
         if _GROUPED_WGRAD:
@@ -343,8 +360,10 @@ class DirectLinear(nn.Module):
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
         self._accum_view_w = None
         self._accum_view_b = None
+        self._pending_dres_add = None
 
     def forward(self, x):
+        self._pending_dres_add = None  # drop any unconsumed deferred grad
         if self._accum_view_w is not None:
             if not _use_hip(x):
                 raise RuntimeError("bound DirectLinear got non-bf16/non-GPU input")

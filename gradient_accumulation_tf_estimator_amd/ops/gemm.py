@@ -93,6 +93,18 @@ def dgrad(dy2d: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return hip.lt_dgrad(dy2d, w, idx)
 
 
+def dgrad_add(dy2d: torch.Tensor, w: torch.Tensor, addend: torch.Tensor) -> torch.Tensor:
+    """dx = dy @ W + addend (one GEMM, beta=1 with C=addend, D=dx): the
+    residual-branch gradient sum folded into the dgrad epilogue."""
+    hip = require_hip()
+    N, K = w.shape
+    R = dy2d.numel() // N
+    idx = _algo_for(2, R, N, K,
+                    lambda: hip.lt_algo_count(2, R, N, K),
+                    lambda i: hip.lt_dgrad(dy2d, w, i))
+    return hip.lt_dgrad_add(dy2d, w, addend, idx)
+
+
 def wgrad_acc(x2d: torch.Tensor, dy2d: torch.Tensor, accum_2d: torch.Tensor) -> None:
     hip = require_hip()
     K = x2d.shape[-1]

@@ -178,13 +178,19 @@ def test_bertlayer_delegated_gelu_bias():
     assert layerA.intermediate_act._bias_delegated
     assert layerA.intermediate._accum_view_b is not None
 
+    xgrads = {}
     for i in range(2):
         torch.manual_seed(20 + i)
         x = (torch.randn(4, 128, 512, device="cuda") * 0.5).bfloat16()
-        for layer, eng in ((layerA, engA), (layerB, engB)):
-            loss = (layer(x.clone()).float() ** 2).mean()
+        for name, (layer, eng) in (("A", (layerA, engA)), ("B", (layerB, engB))):
+            xi = x.clone().requires_grad_()
+            loss = (layer(xi).float() ** 2).mean()
             loss.backward()
             eng.accumulate()
+            xgrads.setdefault(name, []).append(xi.grad.float().cpu().numpy())
+    # input grads cover the deferred residual-add path (dgrad beta=1 epilogue)
+    for ga, gb in zip(xgrads["A"], xgrads["B"]):
+        np.testing.assert_allclose(ga, gb, rtol=3e-2, atol=4e-3)
     torch.cuda.synchronize()
     stA, stB = engA.state, engB.state
     by_name_B = {s.name: s for s in stB.layout.slices}
