@@ -2,7 +2,10 @@
 
 Reference: /root/reference/another-example.py -- CSV feature columns, MLP
 [16,8,4], regression head, train/eval/predict driver, B=59, K=3
-(SURVEY.md C8). Synthetic housing-shaped data stands in for the CSV."""
+(SURVEY.md C8). Synthetic housing-shaped data stands in by default (no
+network in this environment); pass ``--csv path/to/housing.csv`` to use the
+real CSV pipeline (data/csv.py: numeric z-score + CHAS categorical
+indicator, the reference's get_feature_columns)."""
 
 import os
 import sys
@@ -46,9 +49,36 @@ def model_fn(features, labels, mode, params):
     return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
 
 
+def load_csv_datasets(path):
+    """another-example.py's 13 Boston-housing columns through data/csv.py."""
+    from gradient_accumulation_tf_estimator_amd.data.csv import (
+        CategoricalColumn, NumericColumn, build_features, parse_csv)
+    from gradient_accumulation_tf_estimator_amd.data.input_fn import ArrayDataset
+
+    numeric = ["CRIM", "ZN", "INDUS", "NOX", "RM", "AGE", "DIS", "RAD", "TAX",
+               "PTRATIO", "B", "LSTAT"]
+    cols = [NumericColumn(n) for n in numeric] + [
+        CategoricalColumn("CHAS", vocabulary=["0", "1"])]
+    raw, labels = parse_csv(path, cols, "MEDV")
+    x, stats = build_features(raw, cols)
+    y = torch.tensor(labels)
+    n_train = int(0.8 * len(y))
+    return (ArrayDataset(x[:n_train], y[:n_train]),
+            ArrayDataset(x[n_train:], y[n_train:]))
+
+
 if __name__ == "__main__":
-    train_ds = synthetic.housing(n=472, seed=7)
-    eval_ds = synthetic.housing(n=118, seed=8)
+    import argparse
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--csv", default=None,
+                    help="real housing CSV (columns CRIM..LSTAT, CHAS, MEDV)")
+    cli = ap.parse_args()
+    if cli.csv:
+        train_ds, eval_ds = load_csv_datasets(cli.csv)
+    else:
+        train_ds = synthetic.housing(n=472, seed=7)
+        eval_ds = synthetic.housing(n=118, seed=8)
     est = Estimator(
         model_fn,
         RunConfig(model_dir="/tmp/ga_amd_housing", log_step_count_steps=100,

@@ -65,14 +65,26 @@ def make_model_fn():
     return model_fn
 
 
+def _dataset(split, n, seed):
+    """Real raw-IDX MNIST when MNIST_DATA_DIR points at the gz files
+    (data/mnist_idx.py = the reference's mnist_dataset.py); synthetic
+    MNIST-shaped data otherwise (no network in this environment)."""
+    data_dir = os.environ.get("MNIST_DATA_DIR")
+    if data_dir:
+        from gradient_accumulation_tf_estimator_amd.data import mnist_idx
+
+        return mnist_idx.load(data_dir)[split]
+    return synthetic.mnist(n=n, seed=seed)
+
+
 def train_input_fn(batch_size, input_context: InputContext = None, seed=SEED):
-    ds = synthetic.mnist(n=TRAIN_N, seed=1)
+    ds = _dataset("train", TRAIN_N, 1)
     return input_fn_iterator(ds, batch_size, num_epochs=NUM_EPOCHS, seed=seed,
                              input_context=input_context)
 
 
 def eval_input_fn(batch_size=200):
-    ds = synthetic.mnist(n=EVAL_N, seed=2)
+    ds = _dataset("test", EVAL_N, 2)
     return input_fn_iterator(ds, batch_size, num_epochs=1, shuffle=False)
 
 

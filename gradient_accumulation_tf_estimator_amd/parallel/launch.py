@@ -46,6 +46,30 @@ def init_distributed(backend: str = "auto") -> DistContext:
     return DistContext(rank=rank, world_size=world, local_rank=local_rank, device=device)
 
 
+def init_from_tf_config(backend: str = "auto") -> DistContext:
+    """Bootstrap from a reference-style TF_CONFIG env JSON
+    (03:68-74: {"cluster": {"worker": ["host:port", ...]}, "task":
+    {"type": "worker", "index": i}}): maps the cluster spec onto
+    torch.distributed rendezvous (worker 0's host:port becomes the master)
+    and delegates to init_distributed. torchrun env vars, when present,
+    win -- this exists so reference launch scripts port without edits."""
+    import json
+
+    cfg = os.environ.get("TF_CONFIG")
+    if cfg and "WORLD_SIZE" not in os.environ:
+        spec = json.loads(cfg)
+        workers = spec.get("cluster", {}).get("worker", [])
+        index = int(spec.get("task", {}).get("index", 0))
+        if workers:
+            host, _, port = workers[0].partition(":")
+            os.environ["MASTER_ADDR"] = host or "127.0.0.1"
+            os.environ["MASTER_PORT"] = port or "29517"
+            os.environ["WORLD_SIZE"] = str(len(workers))
+            os.environ["RANK"] = str(index)
+            os.environ.setdefault("LOCAL_RANK", str(index))
+    return init_distributed(backend)
+
+
 def cleanup() -> None:
     if dist.is_available() and dist.is_initialized():
         dist.destroy_process_group()
