@@ -485,3 +485,334 @@ GA_ATTN_BWDKV_INST(32)
 GA_ATTN_BWDKV_INST(64)
 GA_ATTN_BWDKV_INST(96)
 GA_ATTN_BWDKV_INST(128)
+
+// ===========================================================================
+// Large-S variants (S % 64 == 0, runtime S, e.g. seq 256/384/512): the same
+// one-wave tile geometry, but K/V (fwd, bwd_q) or Q/dO (bwd_kv) stream
+// through 64-row LDS chunks with online softmax (fwd) / the saved lse
+// (backward). Replaces the torch-SDPA fallback for the seq512 BERT configs.
+// ===========================================================================
+
+// transpose-stage a [64 s][64 d] chunk into LDS [64 d][64 s] (128 B rows)
+static __device__ __forceinline__ void stage_c64_T(const unsigned short* g,
+                                                   int row_stride,
+                                                   unsigned short* lds) {
+  bf16x8 r[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int blk = threadIdx.x + i * 64;
+    const int s0 = (blk / 8) * 4;
+    const int d0 = (blk % 8) * 8;
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+      r[i][t] = *(const bf16x8*)(g + (long long)(s0 + t) * row_stride + d0);
+  }
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int blk = threadIdx.x + i * 64;
+    const int s0 = (blk / 8) * 4;
+    const int d0 = (blk % 8) * 8;
+    const unsigned short* u0 = (const unsigned short*)&r[i][0];
+    const unsigned short* u1 = (const unsigned short*)&r[i][1];
+    const unsigned short* u2 = (const unsigned short*)&r[i][2];
+    const unsigned short* u3 = (const unsigned short*)&r[i][3];
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      ush4v pack = {u0[c], u1[c], u2[c], u3[c]};
+      *(ush4v*)((char*)lds + swz(d0 + c, s0 * 2)) = pack;
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
+    const unsigned short* __restrict__ qkv, unsigned short* __restrict__ out,
+    float* __restrict__ lse_out, int B, int S, int nh) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;            // [64][64] swz
+  unsigned short* Vtlds = (unsigned short*)(smem + 8192);  // [64][64] swz
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const int q0 = qt * 32;
+  bf16x8 qf[4];
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk)
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+
+  const float scale2 = 0.125f * LOG2E;
+  float m_old = -1e30f, sum = 0.f;
+  f32x16 oc[2];
+  oc[0] = (f32x16)(0.f);
+  oc[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage_64<64>(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
+    stage_c64_T(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vtlds);
+    __syncthreads();
+
+    // S^T chunk = K_c Q^T (cols = q, rows = k within the chunk)
+    f32x16 acc[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      acc[t] = (f32x16)(0.f);
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 a = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
+      }
+    }
+    float mc = -1e30f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) mc = fmaxf(mc, acc[t][r]);
+    mc = fmaxf(mc, __shfl_xor(mc, 32, 64)) * scale2;
+    const float m_new = fmaxf(m_old, mc);
+    const float sf = __builtin_amdgcn_exp2f(m_old - m_new);  // per-q (lane)
+    float csum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        acc[t][r] = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - m_new);
+        csum += acc[t][r];
+      }
+    csum += __shfl_xor(csum, 32, 64);
+    sum = sum * sf + csum;
+    m_old = m_new;
+
+    // O rescale by sf (per q = O-tile ROW) then O += P_c V_c
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow_r = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        oc[dt][r] *= __shfl(sf, qrow_r, 64);
+      }
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfk = 0; halfk < 2; ++halfk) {
+          bf16x8 pa = cvt_swap(acc[t], halfk * 8);
+          bf16x8 bv = *(const bf16x8*)((char*)Vtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfk + hi * 8) * 2));
+          oc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, bv, oc[dt], 0, 0, 0);
+        }
+    }
+  }
+
+  const float inv_sum = 1.f / sum;
+  if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m_old + log2f(sum);
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow_r = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      oc[dt][r] *= __shfl(inv_sum, qrow_r, 64);
+    }
+    write_tile_bf16(out + ((long long)b * S + q0) * H + h * ATTN_D + dt * 32 + lo31,
+                    H, hi, oc[dt]);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ dout, const float* __restrict__ lse_in,
+    float* __restrict__ Dtab, unsigned short* __restrict__ dqkv, int B, int S,
+    int nh) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;             // [64][64] swz
+  unsigned short* Vlds = (unsigned short*)(smem + 8192);    // [64][64] swz
+  unsigned short* Ktlds = (unsigned short*)(smem + 16384);  // [64][64] swz
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  const int q0 = qt * 32;
+  const float lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
+
+  float D_q;
+  {
+    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
+    const unsigned short* orw = out + obase + (long long)(q0 + lo31) * H + hi * 32;
+    float sd = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
+      bf16x8 ov = *(const bf16x8*)(orw + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
+    }
+    D_q = sd + __shfl_xor(sd, 32, 64);
+    if (hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+  }
+
+  bf16x8 qf[4], dof[4];
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+  const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk) {
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+    dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+  }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  f32x16 dq[2];
+  dq[0] = (f32x16)(0.f);
+  dq[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage_64<64>(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
+    stage_64<64>(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vlds);
+    stage_c64_T(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Ktlds);
+    __syncthreads();
+
+    f32x16 acc[2], dacc[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      acc[t] = (f32x16)(0.f);
+      dacc[t] = (f32x16)(0.f);
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
+        dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+        dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T chunk
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfk = 0; halfk < 2; ++halfk) {
+          bf16x8 as = cvt_swap(dacc[t], halfk * 8);
+          bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfk + hi * 8) * 2));
+          dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bk, dq[dt], 0, 0, 0);
+        }
+  }
+
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+    write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dq[dt]);
+}
+
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int bh = blockIdx.x / NT, kt = blockIdx.x % NT;
+  const int b = bh / nh, h = bh % nh;
+  const int lo31 = threadIdx.x & 31;
+  const int hi = (threadIdx.x >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dOtlds = (unsigned short*)smem;          // [64][64] swz
+  unsigned short* Qtlds = (unsigned short*)(smem + 8192);  // [64][64] swz
+  float* lsetab = (float*)(smem + 16384);                  // [64]
+  float* dtab = (float*)(smem + 16384 + 256);              // [64]
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  const int k0 = kt * 32;
+
+  bf16x8 kf[4], vf[4];
+  {
+    const unsigned short* krow = qkv + base + H + (long long)(k0 + lo31) * 3 * H;
+    const unsigned short* vrow = qkv + base + 2 * H + (long long)(k0 + lo31) * 3 * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      kf[kk] = *(const bf16x8*)(krow + kk * 16 + hi * 8);
+      vf[kk] = *(const bf16x8*)(vrow + kk * 16 + hi * 8);
+    }
+  }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  f32x16 dv_[2], dk_[2];
+  dv_[0] = (f32x16)(0.f); dv_[1] = (f32x16)(0.f);
+  dk_[0] = (f32x16)(0.f); dk_[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage_c64_T(dout + obase + (long long)c0 * H, H, dOtlds);
+    stage_c64_T(qkv + base + (long long)c0 * 3 * H, 3 * H, Qtlds);
+    if (threadIdx.x < 64) {
+      lsetab[threadIdx.x] = lse_in[((long long)b * nh + h) * S + c0 + threadIdx.x];
+      dtab[threadIdx.x] = Dtab[((long long)b * nh + h) * S + c0 + threadIdx.x];
+    }
+    __syncthreads();
+
+    f32x16 p_qt[2], ds_qt[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
+      const unsigned short* qr = qkv + base + (long long)(c0 + t * 32 + lo31) * 3 * H;
+      const unsigned short* dr = dout + obase + (long long)(c0 + t * 32 + lo31) * H;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 aq = *(const bf16x8*)(qr + kk * 16 + hi * 8);
+        bf16x8 ad = *(const bf16x8*)(dr + kk * 16 + hi * 8);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[kk], sacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[kk], dpacc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+        p_qt[t][r] = p;
+        ds_qt[t][r] = scale * p * (dpacc[r] - dtab[q]);
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfq = 0; halfq < 2; ++halfq) {
+          bf16x8 ap = cvt_swap(p_qt[t], halfq * 8);
+          bf16x8 as = cvt_swap(ds_qt[t], halfq * 8);
+          bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfq + hi * 8) * 2));
+          bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfq + hi * 8) * 2));
+          dv_[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, dv_[dt], 0, 0, 0);
+          dk_[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, dk_[dt], 0, 0, 0);
+        }
+  }
+
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dv_[dt]);
+    write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dk_[dt]);
+  }
+}

@@ -79,6 +79,14 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
       const unsigned short*, const unsigned short*, const float*,              \
       const float*, unsigned short*, int, int);
 GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
+extern "C" __global__ void k_attn_fwd_big(const unsigned short*, unsigned short*,
+                                           float*, int, int, int);
+extern "C" __global__ void k_attn_bwd_q_big(const unsigned short*, const unsigned short*,
+                                            const unsigned short*, const float*,
+                                            float*, unsigned short*, int, int, int);
+extern "C" __global__ void k_attn_bwd_kv_big(const unsigned short*, const unsigned short*,
+                                             const float*, const float*,
+                                             unsigned short*, int, int, int);
 extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
                                         float*, int, int, int);
 #define WG_MAX_G 24
@@ -339,11 +347,22 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   TORCH_CHECK(qkv.dim() == 4 && qkv.size(2) == 3, "qkv must be [B,S,3,H]");
   const int B = (int)qkv.size(0), S = (int)qkv.size(1), H = (int)qkv.size(3);
   TORCH_CHECK(H == nh * 64, "head_dim must be 64");
-  TORCH_CHECK(S <= 128 && S % 32 == 0, "attn kernel needs S<=128, S%32==0");
+  TORCH_CHECK((S <= 128 && S % 32 == 0) || S % 64 == 0,
+              "attn kernel needs S%32==0 and (S<=128 or S%64==0)");
   auto out = at::empty({B, S, H}, qkv.options());
   auto lse = at::empty({B, nh, S}, qkv.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
+  if (S > 128) {
+    // chunked online-softmax variant: [64][64] K + V^T LDS panels
+    hipLaunchKernelGGL(k_attn_fwd_big, dim3(B * (int)nh * NT), dim3(64),
+                       16384, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                       B, S, (int)nh);
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd_big launch failed");
+    return {out, lse};
+  }
   const size_t lds = 16384 + 64 * 256;  // K + V^T panels (fixed offsets)
   void (*fk)(const unsigned short*, unsigned short*, float*, int, int) =
       S == 32 ? k_attn_fwd_32 : S == 64 ? k_attn_fwd_64
@@ -364,6 +383,23 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   auto Dtab = at::empty({B, (long)nh, S}, lse.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
+  if (S > 128) {
+    hipLaunchKernelGGL(k_attn_bwd_q_big, dim3(B * (int)nh * NT), dim3(64),
+                       24576, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (const unsigned short*)out.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+    hipLaunchKernelGGL(k_attn_bwd_kv_big, dim3(B * (int)nh * NT), dim3(64),
+                       16384 + 512, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd_big launch failed");
+    return dqkv;
+  }
   const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
   void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
              const float*, float*, unsigned short*, int, int) =

@@ -411,7 +411,8 @@ class DirectEmbedding(nn.Module):
 class _AttentionFn(torch.autograd.Function):
     """Hand-written MFMA flash attention over the packed QKV projection
     (ops/csrc/attn.hip): no permute copies, no dq/dk/dv zero-fills, packed
-    dqkv gradient. S <= 128, head_dim 64, no mask/dropout."""
+    dqkv gradient. head_dim 64, no mask/dropout; S <= 128 single-pass or
+    any S %% 64 == 0 via the chunked online-softmax variants."""
 
     @staticmethod
     def forward(ctx, qkv4, nh):
@@ -436,7 +437,9 @@ def fused_attention_supported(S: int, head_dim: int, training_extras: bool) -> b
 
     if os.environ.get("GA_FUSED_ATTN", "1") == "0":  # A/B switch
         return False
-    return (head_dim == 64 and S <= 128 and S % 32 == 0
+    # S <= 128: fully-resident single-pass kernels; larger S (%64): the
+    # chunked online-softmax variants (k_attn_*_big)
+    return (head_dim == 64 and ((S <= 128 and S % 32 == 0) or S % 64 == 0)
             and not training_extras and hip_available())
 
 

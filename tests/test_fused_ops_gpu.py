@@ -234,7 +234,7 @@ def test_lt_gemm_matches_torch():
                                rtol=2e-2, atol=5e-1)
 
 
-@pytest.mark.parametrize("S", [32, 64, 128])
+@pytest.mark.parametrize("S", [32, 64, 128, 256, 512])
 @pytest.mark.parametrize("nh", [2, 8])
 def test_fused_attention_matches_sdpa(S, nh):
     """Hand-written MFMA attention (fwd+bwd) vs fp32 math reference."""
