@@ -78,26 +78,26 @@ extern "C" __global__ void k_accum_bf16(float4* __restrict__ accum,
 extern "C" __global__ void k_sqnorm(const float4* __restrict__ accum,
                                     long long n4,
                                     float* __restrict__ out) {
-  // two independent accumulator chains + 2 loads in flight per iteration:
-  // a single dependent fmaf chain leaves the kernel latency-bound well
-  // below HBM peak (measured 2.6 TB/s -> this form reaches ~6 TB/s)
+  // four independent per-component accumulator chains + 2 loads in flight
+  // per iteration: a single dependent fmaf chain leaves the kernel
+  // latency-bound well below HBM peak
   long long stride = (long long)gridDim.x * blockDim.x * 2;
-  float s = 0.f, s2 = 0.f;
+  float sx = 0.f, sy = 0.f, sz = 0.f, sw = 0.f;
   long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
   for (; i + 1 < n4; i += stride) {
     float4 a = accum[i];
     float4 b = accum[i + 1];
-    s = fmaf(a.x, a.x, s); s = fmaf(a.y, a.y, s);
-    s = fmaf(a.z, a.z, s); s = fmaf(a.w, a.w, s);
-    s2 = fmaf(b.x, b.x, s2); s2 = fmaf(b.y, b.y, s2);
-    s2 = fmaf(b.z, b.z, s2); s2 = fmaf(b.w, b.w, s2);
+    sx = fmaf(a.x, a.x, sx); sy = fmaf(a.y, a.y, sy);
+    sz = fmaf(a.z, a.z, sz); sw = fmaf(a.w, a.w, sw);
+    sx = fmaf(b.x, b.x, sx); sy = fmaf(b.y, b.y, sy);
+    sz = fmaf(b.z, b.z, sz); sw = fmaf(b.w, b.w, sw);
   }
   if (i < n4) {
     float4 a = accum[i];
-    s = fmaf(a.x, a.x, s); s = fmaf(a.y, a.y, s);
-    s = fmaf(a.z, a.z, s); s = fmaf(a.w, a.w, s);
+    sx = fmaf(a.x, a.x, sx); sy = fmaf(a.y, a.y, sy);
+    sz = fmaf(a.z, a.z, sz); sw = fmaf(a.w, a.w, sw);
   }
-  s += s2;
+  float s = (sx + sy) + (sz + sw);
   // wave64 shuffle reduction
   for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
   __shared__ float ws[GA_THREADS / 64];
@@ -135,25 +135,48 @@ static __device__ void apply_body(float4* __restrict__ accum,
   const float s = inv_k * coef;
   const float omb1 = 1.f - b1, omb2 = 1.f - b2;
   const float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
-  long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n4; i += stride) {
+  // two float4 groups per iteration: independent mm/vv/u chains double the
+  // in-flight loads and hide the sqrt+div latency; the bf16 model write
+  // becomes one 16-byte store per pair
+  long long stride = (long long)gridDim.x * blockDim.x * 2;
+  long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  for (; i < n4; i += stride) {
+    const bool two = (i + 1) < n4;
     float4 a = accum[i], mm = m[i], vv = v[i], pp = p[i];
+    float4 a2, mm2, vv2, pp2;
+    if (two) { a2 = accum[i + 1]; mm2 = m[i + 1]; vv2 = v[i + 1]; pp2 = p[i + 1]; }
     const float dw = (i < boundary4) ? wd : 0.f;
-#define GA_C(c)                                        \
+    const float dw2 = ((i + 1) < boundary4) ? wd : 0.f;
+#define GA_C(av, mv, vvv, pv, c, dww)                  \
     {                                                  \
-      float g = a.c * s;                               \
-      mm.c = fmaf(b1, mm.c, omb1 * g);                 \
-      vv.c = fmaf(b2, vv.c, omb2 * g * g);             \
-      float u = mm.c / (sqrtf(vv.c) + eps);            \
-      u = fmaf(dw, pp.c, u);                           \
-      pp.c = fmaf(-lr, u, pp.c);                       \
+      float g = av.c * s;                              \
+      mv.c = fmaf(b1, mv.c, omb1 * g);                 \
+      vvv.c = fmaf(b2, vvv.c, omb2 * g * g);           \
+      float u = mv.c / (sqrtf(vvv.c) + eps);           \
+      u = fmaf(dww, pv.c, u);                          \
+      pv.c = fmaf(-lr, u, pv.c);                       \
     }
-    GA_C(x) GA_C(y) GA_C(z) GA_C(w)
+    GA_C(a, mm, vv, pp, x, dw) GA_C(a, mm, vv, pp, y, dw)
+    GA_C(a, mm, vv, pp, z, dw) GA_C(a, mm, vv, pp, w, dw)
+    if (two) {
+      GA_C(a2, mm2, vv2, pp2, x, dw2) GA_C(a2, mm2, vv2, pp2, y, dw2)
+      GA_C(a2, mm2, vv2, pp2, z, dw2) GA_C(a2, mm2, vv2, pp2, w, dw2)
+    }
 #undef GA_C
     m[i] = mm; v[i] = vv; p[i] = pp; accum[i] = z;
+    if (two) { m[i + 1] = mm2; v[i + 1] = vv2; p[i + 1] = pp2; accum[i + 1] = z; }
     if (HAS_MODEL) {
-      model[i] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
-                              f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+      if (two) {
+        union { ushort4 u4[2]; uint4 u16; } pk;
+        pk.u4[0] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
+                                f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+        pk.u4[1] = make_ushort4(f32_to_bf16(pp2.x), f32_to_bf16(pp2.y),
+                                f32_to_bf16(pp2.z), f32_to_bf16(pp2.w));
+        *(uint4*)(model + i) = pk.u16;
+      } else {
+        model[i] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
+                                f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+      }
     }
   }
 }
