@@ -100,11 +100,6 @@ struct WgArgs {
 };
 extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
 extern "C" __global__ void k_wgrad_mfma256(WgArgs, int);
-extern "C" __global__ void k_linear_fwd(const unsigned short*, const unsigned short*,
-                                        const unsigned short*, unsigned short*,
-                                        int, int, int);
-extern "C" __global__ void k_linear_dgrad(const unsigned short*, const unsigned short*,
-                                          unsigned short*, int, int, int);
 extern "C" __global__ void k_cls_head_fwd(const unsigned short*, const unsigned short*,
                                           const unsigned short*, const long long*,
                                           unsigned short*, float*, float*, int, int, int);
@@ -471,39 +466,6 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_wgrad_mfma launch failed");
 }
 
-at::Tensor linear_fwd_mfma(at::Tensor x, at::Tensor W, c10::optional<at::Tensor> bias) {
-  const int K = (int)W.size(1), N = (int)W.size(0);
-  const int R = (int)(x.numel() / K);
-  TORCH_CHECK(R % 128 == 0 && N % 128 == 0 && K % 64 == 0, "linear_fwd_mfma shape");
-  auto sizes = x.sizes().vec();
-  sizes.back() = N;
-  auto y = at::empty(sizes, x.options());
-  auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_linear_fwd, dim3((R / 128) * (N / 128)), dim3(256), 65536,
-                     stream, (const unsigned short*)x.data_ptr(),
-                     (const unsigned short*)W.data_ptr(),
-                     bias ? (const unsigned short*)bias->data_ptr() : nullptr,
-                     (unsigned short*)y.data_ptr(), R, N, K);
-  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_linear_fwd launch failed");
-  return y;
-}
-
-at::Tensor linear_dgrad_mfma(at::Tensor dy, at::Tensor W) {
-  const int K = (int)W.size(1), N = (int)W.size(0);
-  const int R = (int)(dy.numel() / N);
-  TORCH_CHECK(R % 128 == 0 && K % 128 == 0 && N % 64 == 0, "linear_dgrad_mfma shape");
-  auto sizes = dy.sizes().vec();
-  sizes.back() = K;
-  auto dx = at::empty(sizes, dy.options());
-  auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_linear_dgrad, dim3((R / 128) * (K / 128)), dim3(256), 65536,
-                     stream, (const unsigned short*)dy.data_ptr(),
-                     (const unsigned short*)W.data_ptr(),
-                     (unsigned short*)dx.data_ptr(), R, N, K);
-  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_linear_dgrad launch failed");
-  return dx;
-}
-
 std::vector<at::Tensor> cls_head_fwd(at::Tensor pre, at::Tensor Wc, at::Tensor bc,
                                      at::Tensor labels) {
   const int B = (int)pre.size(0), H = (int)pre.size(1), C = (int)Wc.size(0);
@@ -712,8 +674,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "scatter-add embedding grads into the flat fp32 accum slice");
   mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");
   mod.def("cls_head_fwd", &cls_head_fwd, "tanh+classifier+CE forward");
-  mod.def("linear_fwd_mfma", &linear_fwd_mfma, "custom MFMA y = x @ W^T + b");
-  mod.def("linear_dgrad_mfma", &linear_dgrad_mfma, "custom MFMA dx = dy @ W");
   mod.def("cls_head_bwd", &cls_head_bwd, "fused head backward -> d(pre-tanh)");
   mod.def("wgrad_mfma", &wgrad_mfma,
           "batched MFMA wgrad: accum_g += dy_g^T @ x_g over a tile table");

@@ -57,22 +57,17 @@ def _algo_for(kind: int, R: int, N: int, K: int, count_fn, run) -> int:
     return _tune(key, count_fn(), run)
 
 
-# EXPERIMENTAL: the custom MFMA Linear kernels (ops/csrc/linear_mfma.hip)
-# lose to hipBLASLt at the bench shapes (one serial K-loop per block, no
-# cross-problem TLP -- measured fwd 24 us / dgrad 57 us vs ~10 us); routing
-# stays off until they get the deep-pipeline treatment. The batched wgrad
-# kernel wins because it launches every problem's tiles at once.
-mfma_linear_enabled = False
-_MFMA_MAX_R = 2048
+# NOTE: custom MFMA fwd/dgrad kernels were built twice and measured slower
+# than tuned hipBLASLt at the bench shapes (latest attempt + numbers:
+# ops/csrc/linear_small.hip + tools/lin_small_test.py); hipBLASLt stays the
+# routed path. The batched wgrad kernel (wgrad_mfma.hip) wins because it
+# launches every problem's tiles in one grid.
 
 
 def linear_fwd(x2d: torch.Tensor, w: torch.Tensor, bias) -> torch.Tensor:
     hip = require_hip()
     N, K = w.shape
     R = x2d.numel() // K
-    if (mfma_linear_enabled and R % 128 == 0 and R <= _MFMA_MAX_R
-            and N % 128 == 0 and K % 64 == 0):
-        return hip.linear_fwd_mfma(x2d, w, bias)
     kind = 1 if bias is not None else 0
     idx = _algo_for(kind, R, N, K,
                     lambda: hip.lt_algo_count(kind, R, N, K),
@@ -84,9 +79,6 @@ def dgrad(dy2d: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     hip = require_hip()
     N, K = w.shape
     R = dy2d.numel() // N
-    if (mfma_linear_enabled and R % 128 == 0 and R <= _MFMA_MAX_R
-            and K % 128 == 0 and N % 64 == 0):
-        return hip.linear_dgrad_mfma(dy2d, w)
     idx = _algo_for(2, R, N, K,
                     lambda: hip.lt_algo_count(2, R, N, K),
                     lambda i: hip.lt_dgrad(dy2d, w, i))

@@ -32,7 +32,6 @@ setup(
                 os.path.join(CSRC, "attn.hip"),
                 os.path.join(CSRC, "wgrad_mfma.hip"),
                 os.path.join(CSRC, "cls_head.hip"),
-                os.path.join(CSRC, "linear_mfma.hip"),
                 os.path.join(CSRC, "linear_small.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],

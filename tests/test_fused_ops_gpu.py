@@ -361,9 +361,30 @@ def test_cls_head_matches_torch():
     np.testing.assert_allclose(got, refg.cpu().numpy(), rtol=5e-2, atol=5e-3)
 
 
-@pytest.mark.skip(reason="experimental kernels: routing disabled (lose to "
-                          "hipBLASLt) and numerics not yet settled")
 @pytest.mark.parametrize("N,K", [(512, 512), (1536, 512), (512, 2048), (2048, 512)])
+def test_linear_small_matches_torch(N, K):
+    """Experimental small-GEMM MFMA fwd/dgrad (ops/csrc/linear_small.hip)
+    vs fp32 torch references -- correct but unrouted (hipBLASLt is faster
+    at these shapes; see the file header for the measured comparison)."""
+    from gradient_accumulation_tf_estimator_amd.ops import require_hip
+
+    hip = require_hip()
+    torch.manual_seed(N + K)
+    R = 1024
+    x = (torch.randn(R, K, device="cuda") * 0.5).bfloat16()
+    w = (torch.randn(N, K, device="cuda") * 0.05).bfloat16()
+    b = torch.randn(N, device="cuda").bfloat16()
+    ref = x.float() @ w.float().T + b.float()
+    y = hip.lin_fwd_small(x, w, b)
+    np.testing.assert_allclose(y.float().cpu(), ref.cpu(), rtol=3e-2, atol=3e-2)
+
+    dy = (torch.randn(R, N, device="cuda") * 0.5).bfloat16()
+    dref = dy.float() @ w.float()
+    dx = hip.lin_dgrad_small(dy, w)
+    np.testing.assert_allclose(dx.float().cpu(), dref.cpu(), rtol=3e-2, atol=3e-2)
+
+
+
 def test_linear_mfma_matches_torch(N, K):
     """Custom MFMA Linear fwd/dgrad vs torch references."""
     from gradient_accumulation_tf_estimator_amd import ops
