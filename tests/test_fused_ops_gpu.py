@@ -385,30 +385,6 @@ def test_linear_small_matches_torch(N, K):
 
 
 
-def test_linear_mfma_matches_torch(N, K):
-    """Custom MFMA Linear fwd/dgrad vs torch references."""
-    from gradient_accumulation_tf_estimator_amd import ops
-
-    hip = ops.require_hip()
-    torch.manual_seed(N + K)
-    R = 1024
-    x = (torch.randn(R, K, device="cuda") * 0.4).bfloat16()
-    w = (torch.randn(N, K, device="cuda") * 0.05).bfloat16()
-    b = torch.randn(N, device="cuda").bfloat16()
-    dy = (torch.randn(R, N, device="cuda") * 0.4).bfloat16()
-
-    y = hip.linear_fwd_mfma(x, w, b)
-    yref = F.linear(x.float(), w.float(), b.float())
-    np.testing.assert_allclose(y.float().cpu(), yref.cpu(), rtol=3e-2, atol=2e-1)
-
-    y2 = hip.linear_fwd_mfma(x, w, None)
-    np.testing.assert_allclose(y2.float().cpu(), (yref - b.float()).cpu(),
-                               rtol=3e-2, atol=2e-1)
-
-    dx = hip.linear_dgrad_mfma(dy, w)
-    dxref = dy.float() @ w.float()
-    np.testing.assert_allclose(dx.float().cpu(), dxref.cpu(), rtol=3e-2, atol=2e-1)
-
 
 @pytest.mark.skip(reason="this hipblaslt build has no GELU_AUX solutions; "
                           "FusedFFN is experimental and unused")
