@@ -1,0 +1,85 @@
+"""Property-based sweep of the accumulation-engine contract (hypothesis).
+
+Randomizes what the fixed-case oracle tests (test_accum_semantics.py) pin
+down: K, warmup/total step counts, clip on/off, weight decay, strict vs
+corrected predicate, and parameter shapes -- the engine must track the
+NumPy transcription of optimization.py:76-103,150-171 for every draw.
+"""
+
+import math
+
+import numpy as np
+import torch
+import torch.nn as nn
+from hypothesis import given, settings, strategies as st
+
+from gradient_accumulation_tf_estimator_amd import create_optimizer
+from gradient_accumulation_tf_estimator_amd.engine.flat import use_weight_decay
+
+from test_accum_semantics import NumpyAdamWOracle
+
+
+class RandNet(nn.Module):
+    def __init__(self, din, dh):
+        super().__init__()
+        self.fc1 = nn.Linear(din, dh)
+        self.layer_norm = nn.LayerNorm(dh)
+        self.fc2 = nn.Linear(dh, 2)
+
+    def forward(self, x):
+        return self.fc2(self.layer_norm(torch.tanh(self.fc1(x))))
+
+
+@settings(max_examples=12, deadline=None)
+@given(
+    K=st.integers(1, 6),
+    steps=st.integers(2, 14),
+    warmup=st.integers(0, 6),
+    clip=st.sampled_from([None, 0.5, 1.0]),
+    wd=st.sampled_from([0.0, 0.01, 0.1]),
+    strict=st.booleans(),
+    din=st.integers(2, 9),
+    dh=st.integers(2, 8),
+    seed=st.integers(0, 10_000),
+)
+def test_engine_tracks_oracle_for_random_configs(K, steps, warmup, clip, wd,
+                                                 strict, din, dh, seed):
+    torch.manual_seed(seed)
+    net64 = RandNet(din, dh).double()
+    names = [n for n, _ in net64.named_parameters()]
+    shapes = [tuple(p.shape) for _, p in net64.named_parameters()]
+    decay = [use_weight_decay(n, ("LayerNorm", "layer_norm", "bias"))
+             for n in names]
+    nts = max(steps + 2, warmup + 1)
+    oracle = NumpyAdamWOracle(shapes, decay, K, 3e-3, nts, warmup,
+                              clip_norm=clip, wd=wd, strict=strict)
+    oracle.p = [p.detach().numpy().copy() for _, p in net64.named_parameters()]
+
+    net = RandNet(din, dh)
+    net.load_state_dict({k: v.float() for k, v in net64.state_dict().items()})
+    op = create_optimizer(net, 3e-3, nts, warmup,
+                          gradient_accumulation_multiplier=K, clip_norm=clip,
+                          weight_decay=wd, strict_reference_semantics=strict)
+
+    gen = torch.Generator().manual_seed(seed + 1)
+    for i in range(steps):
+        x = torch.randn(3, din, generator=gen)
+        loss = (net(x) ** 2).mean()
+        applied = op.step(loss)
+
+        loss64 = (net64(x.double()) ** 2).mean()
+        net64.zero_grad()
+        loss64.backward()
+        grads = [p.grad.numpy() for _, p in net64.named_parameters()]
+        applied_o = oracle.micro_step(grads)
+        assert applied == applied_o, f"predicate diverged at micro-step {i}"
+        with torch.no_grad():
+            for (n, p), arr in zip(net64.named_parameters(), oracle.p):
+                p.copy_(torch.from_numpy(arr))
+
+    # final params within fp32-vs-fp64 accumulation noise
+    flat_engine = np.concatenate(
+        [p.detach().numpy().ravel() for _, p in net.named_parameters()])
+    flat_oracle = np.concatenate([a.ravel() for a in oracle.p])
+    np.testing.assert_allclose(flat_engine, flat_oracle, rtol=2e-4, atol=2e-5)
+    assert math.isfinite(float(flat_engine.sum()))
