@@ -3,7 +3,11 @@ whole K-window (multiple micro-steps per graph) worth it?
 
 Captures 1-, 2- and 4-micro-step accumulate graphs of the bench model and
 compares per-step replay time. (The multi-step graphs reuse the same static
-inputs -- timing only.)"""
+inputs -- timing only.)
+
+MEASURED (MI355X): 1-step 712.8 us, 2-step 711.1, 4-step 709.9 per
+micro-step -- replay overhead is ~1.5 us/step, so window-sized graphs are
+not worth the input-plumbing complexity. Kept as the measurement record."""
 import os, sys, time
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
 import torch
