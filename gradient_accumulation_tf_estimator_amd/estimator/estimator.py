@@ -31,7 +31,7 @@ import inspect
 import os
 import time
 from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, Iterable, Iterator, Optional, Tuple
+from typing import Any, Callable, Dict, Iterator, Optional
 
 import torch
 

@@ -11,8 +11,6 @@ from __future__ import annotations
 import importlib
 from typing import Optional
 
-import torch
-
 from . import eager
 
 _hip_mod = None
