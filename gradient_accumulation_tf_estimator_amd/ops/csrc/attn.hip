@@ -271,7 +271,7 @@ __device__ __forceinline__ void attn_bwd_q_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
     const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int nh) {
+    unsigned short* __restrict__ dqkv, int B, int nh, int publish_d) {
   const int H = nh * ATTN_D;
   constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -309,7 +309,9 @@ __device__ __forceinline__ void attn_bwd_q_body(
       for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
     }
     D_q = sd + __shfl_xor(sd, 32, 64);
-    if (hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+    // when bwd_kv runs concurrently on a forked stream it reads the table
+    // written by the standalone k_attn_bwd_d instead (publish_d == 0 here)
+    if (publish_d && hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
   }
 
   bf16x8 qf[4], dof[4];
@@ -369,8 +371,8 @@ __device__ __forceinline__ void attn_bwd_q_body(
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_##S(          \
       const unsigned short* qkv, const unsigned short* out,                   \
       const unsigned short* dout, const float* lse_in, float* Dtab,           \
-      unsigned short* dqkv, int B, int nh) {                                  \
-    attn_bwd_q_body<S>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh);            \
+      unsigned short* dqkv, int B, int nh, int publish_d) {                   \
+    attn_bwd_q_body<S>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh, publish_d); \
   }
 GA_ATTN_BWDQ_INST(32)
 GA_ATTN_BWDQ_INST(64)
@@ -628,7 +630,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse_in,
     float* __restrict__ Dtab, unsigned short* __restrict__ dqkv, int B, int S,
-    int nh) {
+    int nh, int publish_d) {
   const int H = nh * ATTN_D;
   const int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -659,7 +661,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
       for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
     }
     D_q = sd + __shfl_xor(sd, 32, 64);
-    if (hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+    if (publish_d && hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
   }
 
   bf16x8 qf[4], dof[4];

@@ -74,7 +74,7 @@ extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*
                                             unsigned short*, float*, int, int);\
   extern "C" __global__ void k_attn_bwd_q_##S(                                 \
       const unsigned short*, const unsigned short*, const unsigned short*,     \
-      const float*, float*, unsigned short*, int, int);                        \
+      const float*, float*, unsigned short*, int, int, int);                   \
   extern "C" __global__ void k_attn_bwd_kv_##S(                                \
       const unsigned short*, const unsigned short*, const float*,              \
       const float*, unsigned short*, int, int);
@@ -83,7 +83,7 @@ extern "C" __global__ void k_attn_fwd_big(const unsigned short*, unsigned short*
                                            float*, int, int, int);
 extern "C" __global__ void k_attn_bwd_q_big(const unsigned short*, const unsigned short*,
                                             const unsigned short*, const float*,
-                                            float*, unsigned short*, int, int, int);
+                                            float*, unsigned short*, int, int, int, int);
 extern "C" __global__ void k_attn_bwd_kv_big(const unsigned short*, const unsigned short*,
                                              const float*, const float*,
                                              unsigned short*, int, int, int);
@@ -336,6 +336,12 @@ void colreduce_acc(at::Tensor partials, at::Tensor dest0,
 
 // qkv [B,S,3,H] bf16 -> (out [B,S,H], lse [B,nh,S] fp32). S<=128, S%32==0,
 // head_dim 64 (all reference BERT configs).
+// NOTE on a measured negative result: forking bwd_kv onto a side stream
+// (bwd_q and bwd_kv are independent once the D table exists from
+// k_attn_bwd_d) regressed the bench 10390 -> 8286 samples/s -- inside the
+// captured graph the extra event edges + the standalone D kernel cost far
+// more than the ~4 us/layer of overlap they buy. The sequential launch
+// below (bwd_q computes and publishes D itself) is the fast path.
 std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() &&
               qkv.scalar_type() == at::kBFloat16, "qkv must be contiguous bf16");
@@ -378,6 +384,7 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   auto Dtab = at::empty({B, (long)nh, S}, lse.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
+
   if (S > 128) {
     hipLaunchKernelGGL(k_attn_bwd_q_big, dim3(B * (int)nh * NT), dim3(64),
                        24576, stream,
@@ -385,38 +392,36 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
                        (const unsigned short*)out.data_ptr(),
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1);
     hipLaunchKernelGGL(k_attn_bwd_kv_big, dim3(B * (int)nh * NT), dim3(64),
                        16384 + 512, stream,
                        (const unsigned short*)qkv.data_ptr(),
                        (const unsigned short*)dout.data_ptr(),
                        lse.data_ptr<float>(), Dtab.data_ptr<float>(),
                        (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
-    TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd_big launch failed");
-    return dqkv;
+  } else {
+    const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
+    void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
+               const float*, float*, unsigned short*, int, int, int) =
+        S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
+        : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
+    hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (const unsigned short*)out.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                       (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1);
+    const size_t lds_kv = (size_t)64 * 256 * 2 + 1024;  // dO^T + Q^T + tables
+    void (*kvk)(const unsigned short*, const unsigned short*, const float*,
+                const float*, unsigned short*, int, int) =
+        S == 32 ? k_attn_bwd_kv_32 : S == 64 ? k_attn_bwd_kv_64
+        : S == 96 ? k_attn_bwd_kv_96 : k_attn_bwd_kv_128;
+    hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (const unsigned short*)dout.data_ptr(),
+                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                       (unsigned short*)dqkv.data_ptr(), B, (int)nh);
   }
-  const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
-  void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
-             const float*, float*, unsigned short*, int, int) =
-      S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
-      : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
-  // bwd_q computes + publishes the D table (rowsum(dO o O)) for bwd_kv
-  hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
-                     (const unsigned short*)qkv.data_ptr(),
-                     (const unsigned short*)out.data_ptr(),
-                     (const unsigned short*)dout.data_ptr(),
-                     lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                     (unsigned short*)dqkv.data_ptr(), B, (int)nh);
-  const size_t lds_kv = (size_t)64 * 256 * 2 + 1024;  // dO^T + Q^T + tables
-  void (*kvk)(const unsigned short*, const unsigned short*, const float*,
-              const float*, unsigned short*, int, int) =
-      S == 32 ? k_attn_bwd_kv_32 : S == 64 ? k_attn_bwd_kv_64
-      : S == 96 ? k_attn_bwd_kv_96 : k_attn_bwd_kv_128;
-  hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
-                     (const unsigned short*)qkv.data_ptr(),
-                     (const unsigned short*)dout.data_ptr(),
-                     lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                     (unsigned short*)dqkv.data_ptr(), B, (int)nh);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd launch failed");
   return dqkv;
 }
