@@ -271,3 +271,7 @@ class AccumEngine:
         self.global_step = int(d["global_step"])
         if int(d.get("K", self.K)) != self.K:
             raise ValueError("checkpoint K does not match engine K")
+        if bool(d.get("strict", self.strict)) != self.strict:
+            # a predicate mismatch silently shifts every later apply boundary
+            raise ValueError(
+                "checkpoint strict_reference_semantics does not match engine")
