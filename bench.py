@@ -37,8 +37,11 @@ def parse_args():
     p.add_argument("--accum", type=int, default=4)
     p.add_argument("--lr", type=float, default=2e-5)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
-    p.add_argument("--graphs", default="auto", choices=["auto", "on", "off"],
-                   help="hipGraph-capture the micro-step (auto: on for 1 GPU)")
+    p.add_argument("--graphs", default="auto",
+                   choices=["auto", "on", "off", "window"],
+                   help="hipGraph-capture the micro-step; 'window' captures "
+                        "the K-1 accumulate steps as one graph (wgrads "
+                        "overlap the next step on a side stream)")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
     p.add_argument("--wgrad-overlap", default="off", choices=["on", "off"],
                    help="EXPERIMENTAL: wgrad GEMMs on a side HIP stream "
@@ -130,7 +133,8 @@ def main():
     graphed = None
     if use_graphs:
         try:
-            graphed = capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world)
+            graphed = capture_graphs(model, engine, pool_ids, pool_lab, inv_world,
+                                     world, window=args.graphs == "window")
         except Exception as e:
             print(f"[bench] hipGraph capture failed, falling back to eager: {e}",
                   file=sys.stderr)
@@ -206,7 +210,8 @@ def main():
         dist.destroy_process_group()
 
 
-def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
+def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world,
+                   window=False):
     """Wrap the framework's hipGraph-captured micro-batch loop
     (engine/graphs.py) with the bench's static input buffers."""
     from gradient_accumulation_tf_estimator_amd.engine.graphs import GraphedTrainLoop
@@ -216,7 +221,29 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world):
     B, S = pool_ids.shape[1], pool_ids.shape[2]
     POOL = pool_ids.shape[0]
     pool_packed = torch.cat([pool_ids, pool_lab[:, :, None]], dim=2).contiguous()
-    static_packed = pool_packed[0].clone()
+    K = engine.K
+    nslots = K if (window and world == 1 and K > 1) else 1
+    slots = [pool_packed[0].clone() for _ in range(nslots)]
+
+    if nslots > 1:
+        loop = GraphedTrainLoop(
+            engine,
+            lambda k: model.loss(slots[k][:, :S], slots[k][:, S]),
+            world=world, window=True)
+
+        def run(i):
+            pos = i % K
+            if pos == 0:
+                # the window replay consumes slots 0..K-2 at once
+                for k in range(K - 1):
+                    slots[k].copy_(pool_packed[(i + k) % POOL])
+            elif pos == K - 1:
+                slots[K - 1].copy_(pool_packed[i % POOL])
+            return loop.step()
+
+        return run
+
+    static_packed = slots[0]
     static_ids = static_packed[:, :S]
     static_lab = static_packed[:, S]
     loop = GraphedTrainLoop(engine, lambda: model.loss(static_ids, static_lab),

@@ -31,21 +31,29 @@ class GraphedTrainLoop:
     handling beyond the device-scalar lr.
     """
 
-    def __init__(self, engine: AccumEngine, loss_fn: Callable[[], torch.Tensor],
-                 *, world: int = 1, warmup_iters: int = 3):
+    def __init__(self, engine: AccumEngine, loss_fn: Callable[..., torch.Tensor],
+                 *, world: int = 1, warmup_iters: int = 3, window: bool = False):
+        """``window=True`` captures the K-1 accumulate micro-steps as ONE
+        graph (``loss_fn`` is then called with a slot index 0..K-1 and must
+        read that slot's static input buffers): inside a multi-step capture
+        the batched wgrad launches can ride the side stream and overlap the
+        NEXT micro-step's forward -- a per-step graph must join the side
+        stream before capture end, pinning the wgrads to the critical path."""
         if engine.backend != "hip":
             raise RuntimeError("GraphedTrainLoop requires the HIP engine backend")
         self.engine = engine
         self.world = world
+        self.window = bool(window) and world == 1 and engine.K > 1
         inv_world = 1.0 / world
 
-        def fwd_bwd_accum():
-            loss = loss_fn()
+        def fwd_bwd_accum(slot=None, join=True):
+            loss = loss_fn(slot) if slot is not None else loss_fn()
             if world > 1:
                 loss = loss * inv_world
             loss.backward()
             engine.accumulate()
-            engine._join_wgrad_stream()
+            if join:
+                engine._join_wgrad_stream()
             return loss
 
         # torch.cuda.graphs warmup protocol: a few eager iterations on a side
@@ -63,23 +71,37 @@ class GraphedTrainLoop:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(warmup_iters):
-                fwd_bwd_accum()
+                fwd_bwd_accum(0 if self.window else None)
             engine.set_lr(engine.lr_at(0))
             engine.apply_from_device()
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
 
-        self.g_accum = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_accum):
-            self.loss_accum = fwd_bwd_accum()
-        if world == 1:
+        if self.window:
+            K = engine.K
+            self.g_accum = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_accum):
+                self.window_losses = [
+                    fwd_bwd_accum(k, join=False) for k in range(K - 1)
+                ]
+                engine._join_wgrad_stream()
             self.g_apply = torch.cuda.CUDAGraph()
             with torch.cuda.graph(self.g_apply, pool=self.g_accum.pool()):
-                self.loss_apply = fwd_bwd_accum()
+                self.loss_apply = fwd_bwd_accum(K - 1)
                 engine.apply_from_device()
+            self.loss_accum = self.window_losses[0]
         else:
-            self.g_apply = None
-            self.loss_apply = None
+            self.g_accum = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_accum):
+                self.loss_accum = fwd_bwd_accum()
+            if world == 1:
+                self.g_apply = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.g_apply, pool=self.g_accum.pool()):
+                    self.loss_apply = fwd_bwd_accum()
+                    engine.apply_from_device()
+            else:
+                self.g_apply = None
+                self.loss_apply = None
         torch.cuda.synchronize()
 
         # capture itself executed the kernels: restore the optimizer state
@@ -96,8 +118,26 @@ class GraphedTrainLoop:
 
     def step(self) -> torch.Tensor:
         """One reference micro-step (static inputs must already be set).
-        Returns the static loss tensor of the replayed graph."""
+        Returns the static loss tensor of the replayed graph.
+
+        Window mode: the caller must have filled ALL K slot inputs before
+        the first micro-step of the window; the K-1 accumulate steps execute
+        in one replay on that first call (later in-window calls return the
+        already-computed losses)."""
         engine = self.engine
+        if self.window:
+            pos = engine.global_step % engine.K
+            if pos == 0:
+                # one replay covers micro-steps 0..K-2 of this window; the
+                # next K-2 calls just hand back the already-computed losses
+                self.g_accum.replay()
+            if pos < engine.K - 1:
+                engine.global_step += 1
+                return self.window_losses[pos]
+            engine.set_lr(engine.lr_at(engine.global_step))
+            self.g_apply.replay()
+            engine.global_step += 1
+            return self.loss_apply
         if engine.is_apply_step():
             engine.set_lr(engine.lr_at(engine.global_step))
             if self.g_apply is not None:

@@ -78,6 +78,12 @@ def flush_pending_wgrads() -> None:
     hip = require_hip()
     all_pending = _pending_wgrads
     _pending_wgrads = []
+    # grouped + overlap: the batched MFMA launch uses no shared workspace,
+    # so it can ride the side stream (joined by the engine before anything
+    # reads accum; successive flushes serialize on the stream itself, so
+    # same-slice fp32 RMWs never race). hipBLASLt fallback problems stay on
+    # the main stream (shared lt workspace).
+    overlap = _WGRAD_OVERLAP
 
     # the batched MFMA kernel covers N,K % 128 == 0 with a common R % 64 == 0
     # (every encoder-layer wgrad); odd shapes (e.g. the pooler's R=batch) go
@@ -103,14 +109,31 @@ def flush_pending_wgrads() -> None:
     R = xs[0].numel() // xs[0].shape[-1]
     empty = None
 
+    if overlap:
+        side = wgrad_stream()
+        side.wait_stream(torch.cuda.current_stream())
+        ctx = torch.cuda.stream(side)
+    else:
+        ctx = None
+
     # the kernel takes metadata by value: at most 24 problems per launch
-    for lo in range(0, len(xs), 24):
-        chunk_vbs = vbs[lo : lo + 24]
-        if empty is None and any(v is None for v in chunk_vbs):
-            empty = torch.empty(0, dtype=torch.float32, device=xs[0].device)
-        hip.wgrad_mfma(list(xs[lo : lo + 24]), list(dys[lo : lo + 24]),
-                       [a.reshape(-1) for a in accs[lo : lo + 24]],
-                       [v if v is not None else empty for v in chunk_vbs], R)
+    def _launch_all():
+        nonlocal empty
+        for lo in range(0, len(xs), 24):
+            chunk_vbs = vbs[lo : lo + 24]
+            if empty is None and any(v is None for v in chunk_vbs):
+                empty = torch.empty(0, dtype=torch.float32, device=xs[0].device)
+            hip.wgrad_mfma(list(xs[lo : lo + 24]), list(dys[lo : lo + 24]),
+                           [a.reshape(-1) for a in accs[lo : lo + 24]],
+                           [v if v is not None else empty for v in chunk_vbs], R)
+
+    if ctx is not None:
+        with ctx:
+            _launch_all()
+        for t in xs + dys:
+            t.record_stream(side)
+    else:
+        _launch_all()
 
 
 def set_wgrad_overlap(enabled: bool) -> None:
