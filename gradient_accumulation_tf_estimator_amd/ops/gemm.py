@@ -28,16 +28,22 @@ def _tune(key, count: int, run: Callable[[int], None]) -> int:
     stream = torch.cuda.current_stream()
     best, best_t = 0, float("inf")
     ev0, ev1 = torch.cuda.Event(enable_timing=True), torch.cuda.Event(enable_timing=True)
+    # min over repeated 3-iter measurements: a single noisy sample otherwise
+    # lets a slow candidate "win" (matters when GA_LT_ALLALGOS widens the
+    # pool to hundreds of candidates)
+    reps = 3 if count > 24 else 2
     for i in range(count):
         try:
             run(i)  # warm
-            torch.cuda.synchronize()
-            ev0.record(stream)
-            for _ in range(3):
-                run(i)
-            ev1.record(stream)
-            torch.cuda.synchronize()
-            t = ev0.elapsed_time(ev1)
+            t = float("inf")
+            for _ in range(reps):
+                torch.cuda.synchronize()
+                ev0.record(stream)
+                for _ in range(3):
+                    run(i)
+                ev1.record(stream)
+                torch.cuda.synchronize()
+                t = min(t, ev0.elapsed_time(ev1))
         except RuntimeError:
             continue
         if t < best_t:
