@@ -48,6 +48,21 @@ def set_grouped_wgrad(enabled: bool) -> None:
 
 _wgrad_table_cache = {}
 
+
+def _wgrad_mfma_max_r() -> int:
+    import os
+
+    global _WGRAD_MFMA_MAX_R, _WGRAD_MFMA_MAX_R_ENV
+    if _WGRAD_MFMA_MAX_R is None:
+        v = os.environ.get("GA_WGRAD_MFMA_MAX_R")
+        _WGRAD_MFMA_MAX_R_ENV = v is not None
+        _WGRAD_MFMA_MAX_R = int(v) if v is not None else 4096
+    return _WGRAD_MFMA_MAX_R
+
+
+_WGRAD_MFMA_MAX_R = None
+_WGRAD_MFMA_MAX_R_ENV = False
+
 # deferred LN/GELU partial-slab reductions, flushed as ONE batched launch
 _pending_colreduce = []
 
@@ -93,10 +108,17 @@ def flush_pending_wgrads() -> None:
     pending = []
     for x, dy, acc, vb in all_pending:
         R_i = x.numel() // x.shape[-1]
-        # the single-launch MFMA path wins at the bench micro-batch; at large
-        # R (seq512 configs) hipBLASLt's split-K wgrads are faster
+        # measured routing (MI355X): the single-launch MFMA path wins at
+        # R <= 2048 (bench micro-batch) and, since the 256-tile kernel, also
+        # at R = 4096 for bert-base-width shapes (seq512: 1046 -> 1127
+        # samples/s); at bert-large widths (N or K = 4096) hipBLASLt's
+        # split-K wgrads stay faster (442 vs 413). GA_WGRAD_MFMA_MAX_R
+        # overrides the R ceiling for re-measurement.
+        wide = max(x.shape[-1], dy.shape[-1]) > 3072
+        cap_env = _wgrad_mfma_max_r()  # also resolves _WGRAD_MFMA_MAX_R_ENV
+        r_cap = cap_env if _WGRAD_MFMA_MAX_R_ENV else (2048 if wide else 4096)
         if (x.shape[-1] % 128 == 0 and dy.shape[-1] % 128 == 0 and R_i % 64 == 0
-                and R_i <= 2048
+                and R_i <= r_cap
                 and (not pending or R_i == pending[0][0].numel() // pending[0][0].shape[-1])):
             pending.append((x, dy, acc, vb))
         else:
