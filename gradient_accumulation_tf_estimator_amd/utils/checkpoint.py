@@ -71,3 +71,22 @@ def latest(model_dir: str) -> Optional[str]:
 
 def load(path: str, map_location="cpu") -> Dict:
     return torch.load(path, map_location=map_location, weights_only=False)
+
+
+def export_safetensors(path: str, model_state: Dict) -> str:
+    """Export model weights (only) as a .safetensors file -- the
+    interchange format for serving stacks. Engine state (flat accum/m/v,
+    micro-step) stays in the .pt checkpoints: safetensors holds tensors
+    only, and serving needs none of it."""
+    from safetensors.torch import save_file
+
+    flat = {k: v.detach().cpu().contiguous() for k, v in model_state.items()
+            if torch.is_tensor(v)}
+    save_file(flat, path)
+    return path
+
+
+def load_safetensors(path: str) -> Dict:
+    from safetensors.torch import load_file
+
+    return load_file(path)

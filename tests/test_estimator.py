@@ -166,3 +166,21 @@ def test_housing_estimator_end_to_end(tmp_path):
     preds = list(est.predict(lambda mode=None: input_fn_iterator(ds, 59, num_epochs=1,
                                                                  shuffle=False)))
     assert len(preds) == 236
+
+
+def test_safetensors_export_roundtrip(tmp_path):
+    """Serving export: model weights to .safetensors and back, bit-exact."""
+    import torch
+
+    from gradient_accumulation_tf_estimator_amd.models.housing import HousingMLP
+    from gradient_accumulation_tf_estimator_amd.utils import checkpoint as ckpt
+
+    torch.manual_seed(0)
+    m = HousingMLP(hidden=(8, 4))
+    p = str(tmp_path / "m.safetensors")
+    ckpt.export_safetensors(p, m.state_dict())
+    back = ckpt.load_safetensors(p)
+    m2 = HousingMLP(hidden=(8, 4))
+    m2.load_state_dict(back)
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(a, b)
