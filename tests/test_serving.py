@@ -11,7 +11,7 @@ def test_cpu_fallback_matches_model():
 
     torch.manual_seed(0)
     m = HousingMLP(hidden=(8, 4))
-    x = torch.randn(16, 13)
+    x = torch.randn(16, 14)
     pred = GraphedPredictor(m, x)
     with torch.no_grad():
         ref = m(x)
