@@ -320,20 +320,21 @@ static inline __device__ float gelu_bwd1(float h, float dy) {
   return dy * (0.5f * (1.f + t) + 0.5f * h * (1.f - t * t) * du);
 }
 
+// 2D grid: blockIdx.y = row, blockIdx.x = 2048-element column chunk.
+// The column index falls out of the thread id -- no per-thread 64-bit
+// modulo for the bias column (H = 3072 has no pow2 shortcut).
 extern "C" __global__ void k_biasgelu_fwd(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ bias,
     unsigned short* __restrict__ y, long long total, int H) {
-  long long stride = (long long)gridDim.x * blockDim.x * 8;
-  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
-       i += stride) {
-    float xv[8], bv[8], o[8];
-    ldv<8>(x + i, xv);
-    const int c = ((H & (H - 1)) == 0) ? (int)(i & (H - 1)) : (int)(i % H);
-    ldv<8>(bias + c, bv);
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c >= H) return;
+  const long long i = (long long)blockIdx.y * H + c;
+  float xv[8], bv[8], o[8];
+  ldv<8>(x + i, xv);
+  ldv<8>(bias + c, bv);
 #pragma unroll
-    for (int k = 0; k < 8; ++k) o[k] = gelu_fwd1(xv[k] + bv[k]);
-    stv<8>(y + i, o);
-  }
+  for (int k = 0; k < 8; ++k) o[k] = gelu_fwd1(xv[k] + bv[k]);
+  stv<8>(y + i, o);
 }
 
 template <int EPL>
@@ -492,19 +493,16 @@ extern "C" __global__ void k_biasgelu_bwd_ew(
     const unsigned short* __restrict__ dy, const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ bias,
     unsigned short* __restrict__ dx_out, long long total, int H) {
-  float bv8[8];
-  long long stride = (long long)gridDim.x * blockDim.x * 8;
-  for (long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
-       i < total; i += stride) {
-    float dv[8], xv[8], o[8];
-    ldv<8>(dy + i, dv);
-    ldv<8>(x + i, xv);
-    const int c = ((H & (H - 1)) == 0) ? (int)(i & (H - 1)) : (int)(i % H);
-    ldv<8>(bias + c, bv8);
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c >= H) return;
+  const long long i = (long long)blockIdx.y * H + c;
+  float dv[8], xv[8], bv8[8], o[8];
+  ldv<8>(dy + i, dv);
+  ldv<8>(x + i, xv);
+  ldv<8>(bias + c, bv8);
 #pragma unroll
-    for (int k = 0; k < 8; ++k) o[k] = gelu_bwd1(xv[k] + bv8[k], dv[k]);
-    stv<8>(dx_out + i, o);
-  }
+  for (int k = 0; k < 8; ++k) o[k] = gelu_bwd1(xv[k] + bv8[k], dv[k]);
+  stv<8>(dx_out + i, o);
 }
 
 // ---------------------------------------------------------------------------

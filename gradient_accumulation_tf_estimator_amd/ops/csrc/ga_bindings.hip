@@ -280,9 +280,11 @@ at::Tensor biasgelu_fwd(at::Tensor x, at::Tensor bias) {
   check_bf16_2d(x, "x", H);
   auto y = at::empty_like(x);
   const long long total = x.numel();
+  const int R = (int)(total / H);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = (int)std::min<long long>((total / 8 + 255) / 256, 2048);
-  hipLaunchKernelGGL(k_biasgelu_fwd, dim3(blocks), dim3(256), 0, stream,
+  // 2D: (column chunks of 2048 elems) x (rows); no per-thread modulo
+  int cblocks = (H / 8 + 255) / 256;
+  hipLaunchKernelGGL(k_biasgelu_fwd, dim3(cblocks, R), dim3(256), 0, stream,
                      bfp(x), bfp(bias), bfp_mut(y), total, H);
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_biasgelu_fwd launch failed");
   return y;
@@ -585,9 +587,10 @@ at::Tensor biasgelu_bwd_ew(at::Tensor dy, at::Tensor x, at::Tensor bias) {
   check_bf16_2d(dy, "dy", H);
   auto dx = at::empty_like(dy);
   const long long total = dy.numel();
+  const int R2 = (int)(total / H);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  int blocks = (int)std::min<long long>((total / 8 + 255) / 256, 2048);
-  hipLaunchKernelGGL(k_biasgelu_bwd_ew, dim3(blocks), dim3(256), 0, stream,
+  int cblocks = (H / 8 + 255) / 256;
+  hipLaunchKernelGGL(k_biasgelu_bwd_ew, dim3(cblocks, R2), dim3(256), 0, stream,
                      (const unsigned short*)dy.data_ptr(),
                      (const unsigned short*)x.data_ptr(),
                      (const unsigned short*)bias.data_ptr(),
