@@ -85,7 +85,18 @@ class BertEmbeddings(nn.Module):
 
     def forward(self, input_ids, token_type_ids=None):
         S = input_ids.shape[1]
-        x = self.word_embeddings(input_ids)
+        wm = self.word_embeddings
+        if (isinstance(wm, DirectEmbedding) and wm._accum_view_w is not None
+                and input_ids.is_cuda):
+            # bound fast path: one gather-sum kernel for all tables
+            # (ops/fused.py fused_embed3)
+            from ..ops.fused import fused_embed3
+
+            x = fused_embed3(input_ids, self.position_ids[:, :S],
+                             token_type_ids, wm, self.position_embeddings,
+                             self.token_type_embeddings)
+            return self.dropout(self.LayerNorm(x))
+        x = wm(input_ids)
         x = x + self.position_embeddings(self.position_ids[:, :S])
         if token_type_ids is not None:
             x = x + self.token_type_embeddings(token_type_ids)

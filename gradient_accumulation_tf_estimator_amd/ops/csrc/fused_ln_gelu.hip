@@ -462,6 +462,34 @@ GA_GELU_INST(48)
 GA_GELU_INST(64)
 
 
+// ---------------- fused 3-table embedding forward ----------------
+// out[b,s,:] = word[ids[b,s]] + pos[s] (+ tok[tids[b,s]]): one launch
+// replacing two gather kernels + a broadcast add. 2D grid like the gelu
+// kernels: blockIdx.y = token row, blockIdx.x = 2048-element column chunk.
+extern "C" __global__ void k_emb3_fwd(
+    const long long* __restrict__ ids, const long long* __restrict__ tids,
+    const unsigned short* __restrict__ word, const unsigned short* __restrict__ pos,
+    const unsigned short* __restrict__ tok, unsigned short* __restrict__ out,
+    int S, int H) {
+  const int c = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c >= H) return;
+  const int row = blockIdx.y;
+  const int s = row % S;
+  const long long wid = ids[row];
+  float wv[8], pv[8], o[8];
+  ldv<8>(word + wid * H + c, wv);
+  ldv<8>(pos + (long long)s * H + c, pv);
+#pragma unroll
+  for (int k = 0; k < 8; ++k) o[k] = wv[k] + pv[k];
+  if (tids) {
+    float tv[8];
+    ldv<8>(tok + tids[row] * H + c, tv);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) o[k] += tv[k];
+  }
+  stv<8>(out + (long long)row * H + c, o);
+}
+
 // ---------------- embedding backward -> flat fp32 accum ----------------
 // Scatter-add bf16 dy rows into the embedding weight's accum slice by token
 // id. Replaces the dense zero-init + scatter + AccumulateGrad add + K1

@@ -69,6 +69,10 @@ struct CrbArgs {
 extern "C" __global__ void k_colreduce_batch(CrbArgs);
 extern "C" __global__ void k_embgrad_acc(const unsigned short*, const long long*,
                                          float*, long long, int);
+extern "C" __global__ void k_emb3_fwd(const long long*, const long long*,
+                                      const unsigned short*, const unsigned short*,
+                                      const unsigned short*, unsigned short*,
+                                      int, int);
 #define GA_DECL_ATTN(S)                                                        \
   extern "C" __global__ void k_attn_fwd_##S(const unsigned short*,             \
                                             unsigned short*, float*, int, int);\
@@ -515,6 +519,34 @@ at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
 
 // problems: list of (partials [NB,...,C], dest0, dest1?, dest2?)
 // ---- small-GEMM MFMA Linear path (linear_small.hip) ----
+at::Tensor emb3_fwd(at::Tensor ids, c10::optional<at::Tensor> tids,
+                    at::Tensor word, at::Tensor pos,
+                    c10::optional<at::Tensor> tok) {
+  TORCH_CHECK(ids.is_cuda() && ids.scalar_type() == at::kLong && ids.dim() == 2);
+  TORCH_CHECK(word.is_contiguous() && word.scalar_type() == at::kBFloat16);
+  const int B = (int)ids.size(0), S = (int)ids.size(1);
+  const int H = (int)word.size(1);
+  TORCH_CHECK(H % 256 == 0, "emb3_fwd hidden must be %256==0");
+  auto out = at::empty({B, S, H}, word.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  int cblocks = (H / 8 + 255) / 256;
+  auto idsc = ids.contiguous();
+  const long long* tp = nullptr;
+  at::Tensor tidsc;
+  if (tids) {
+    tidsc = tids->contiguous();
+    tp = (const long long*)tidsc.data_ptr<int64_t>();
+  }
+  hipLaunchKernelGGL(k_emb3_fwd, dim3(cblocks, B * S), dim3(256), 0, stream,
+                     (const long long*)idsc.data_ptr<int64_t>(), tp,
+                     (const unsigned short*)word.data_ptr(),
+                     (const unsigned short*)pos.data_ptr(),
+                     tok ? (const unsigned short*)tok->data_ptr() : nullptr,
+                     (unsigned short*)out.data_ptr(), S, H);
+  TORCH_CHECK(hipGetLastError() == hipSuccess, "k_emb3_fwd launch failed");
+  return out;
+}
+
 at::Tensor lin_fwd_small(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias) {
   const int N = (int)w.size(0), K = (int)w.size(1);
   const int R = (int)(x.numel() / K);
@@ -670,6 +702,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("biasgelu_bwd", &biasgelu_bwd, "dx + fp32 dbias partials");
   mod.def("colreduce_acc", &colreduce_acc,
           "reduce partials over blocks, ADD into flat fp32 accum slices");
+  mod.def("emb3_fwd", &emb3_fwd, "fused word+position(+type) embedding gather-sum");
   mod.def("lin_fwd_small", &lin_fwd_small,
           "small-GEMM MFMA forward (64x64 tiles, split-K >= 1024)");
   mod.def("lin_dgrad_small", &lin_dgrad_small,

@@ -431,6 +431,52 @@ class _DirectEmbeddingFn(torch.autograd.Function):
         return None, None, None
 
 
+class _Embed3Fn(torch.autograd.Function):
+    """Fused word+position(+token-type) embedding: one gather-sum kernel
+    forward (replaces two gathers + a broadcast add), backward scatter-adds
+    straight into each table's flat accum slice (BOUND mode only)."""
+
+    @staticmethod
+    def forward(ctx, ids, pos_ids, tok_ids, w_word, w_pos, w_tok,
+                mod_word, mod_pos, mod_tok):
+        hip = require_hip()
+        out = hip.emb3_fwd(ids, tok_ids, w_word, w_pos, w_tok if tok_ids is not None else None)
+        ctx.save_for_backward(ids, pos_ids, tok_ids if tok_ids is not None else ids[:0])
+        ctx.mods = (mod_word, mod_pos, mod_tok)
+        ctx.has_tok = tok_ids is not None
+        ctx.H = w_word.shape[1]
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        hip = require_hip()
+        ids, pos_ids, tok_ids = ctx.saved_tensors
+        mod_word, mod_pos, mod_tok = ctx.mods
+        dy = dy.contiguous()
+        dy2d = dy.reshape(-1, ctx.H)
+        hip.embgrad_acc(dy2d, ids.reshape(-1), mod_word._accum_view_w, ctx.H)
+        # position rows are shared by the whole batch: reduce over B first
+        # (B-way atomic collisions otherwise serialize the scatter)
+        B, S = pos_ids.shape
+        pos_dy = dy.reshape(B, S, ctx.H).sum(0, dtype=torch.float32).bfloat16()
+        hip.embgrad_acc(pos_dy, pos_ids[0].reshape(-1),
+                        mod_pos._accum_view_w, ctx.H)
+        if ctx.has_tok:
+            hip.embgrad_acc(dy2d, tok_ids.reshape(-1), mod_tok._accum_view_w, ctx.H)
+        return (None,) * 9
+
+
+def fused_embed3(ids, pos_ids_row, tok_ids, word_mod, pos_mod, tok_mod):
+    """ids [B,S]; pos_ids_row [1,S] (broadcast over the batch in backward
+    via an expanded scatter); tok_ids [B,S] or None. All modules must be
+    engine-bound DirectEmbeddings on bf16."""
+    B, S = ids.shape
+    pos_exp = pos_ids_row.expand(B, S)
+    return _Embed3Fn.apply(ids, pos_exp, tok_ids, word_mod.weight,
+                           pos_mod.weight, tok_mod.weight,
+                           word_mod, pos_mod, tok_mod)
+
+
 class DirectEmbedding(nn.Module):
     """nn.Embedding drop-in; bound backward scatter-adds bf16 rows into the
     flat fp32 accum slice by token id -- no dense [vocab,H] grad buffer, no
