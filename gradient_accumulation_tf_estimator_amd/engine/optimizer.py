@@ -70,6 +70,7 @@ def create_optimizer(
     beta2: float = 0.999,
     eps: float = 1e-6,
     exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+    use_tpu: bool = False,
     strict_reference_semantics: bool = False,
     process_group=None,
     ddp_scale_loss: bool = True,
@@ -80,6 +81,12 @@ def create_optimizer(
         named = list(model.named_parameters())
     else:
         named = list(model_or_params)
+    if use_tpu:
+        # the reference's vestigial CrossShardOptimizer wrap
+        # (optimization.py:67-68): flag kept for signature parity; there is
+        # no TPU path on MI355X
+        raise ValueError("use_tpu is a vestigial reference flag; this "
+                         "framework targets MI355X (leave it False)")
     # fused modules (ops/fused.py) accumulate their param grads directly into
     # the flat fp32 accum buffer on the GPU path; K1 then skips their region
     direct = ()
