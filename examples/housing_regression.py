@@ -44,7 +44,9 @@ def model_fn(features, labels, mode, params):
     train_op = create_optimizer(
         model, params["learning_rate"], 10**6, 0,
         gradient_accumulation_multiplier=params["gradient_accumulation_multiplier"],
-        clip_norm=None, weight_decay=0.0,  # stock Adam variant (C5)
+        # stock tf.train.AdamOptimizer() variant (C5, another-example.py:139):
+        # bias-corrected, eps=1e-8, no clip, no weight decay
+        optimizer="adam", clip_norm=None,
     )
     return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
 

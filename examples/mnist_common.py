@@ -52,13 +52,14 @@ def make_model_fn():
                 return float((model(f).argmax(-1) == l).float().mean()), l.numel()
             return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
                                  eval_metric_fns={"accuracy": acc})
-        # stock-Adam variant: no clipping, decoupled wd off (02:47-74)
+        # stock-Adam variant (02:41 AdamOptimizer(learning_rate=1e-4)):
+        # bias-corrected, eps=1e-8, no clipping, no weight decay (02:47-74)
         train_op = create_optimizer(
             model, params["learning_rate"], num_train_steps=10**6,
             num_warmup_steps=0,
             gradient_accumulation_multiplier=params.get(
                 "gradient_accumulation_multiplier", 1),
-            clip_norm=None, weight_decay=0.0,
+            optimizer="adam", clip_norm=None,
         )
         return EstimatorSpec(mode, model=model, loss_fn=loss_fn, train_op=train_op)
 

@@ -24,6 +24,14 @@ Semantics knobs:
     micro-step (optimization.py:99-103).
   * Adam without bias correction, eps outside sqrt, decay-by-regex
     (optimization.py:150-187).
+  * ``bias_correction=True`` is the stock ``tf.train.AdamOptimizer`` of the
+    generic/MNIST/distributed variants (another-example.py:139, 02:41):
+    TF folds the correction into a scalar step size
+    ``lr_t = lr * sqrt(1 - beta2^t) / (1 - beta1^t)`` (t = number of
+    ``apply_gradients`` calls so far + 1, NOT micro-steps) and updates
+    ``p -= lr_t * m / (sqrt(v) + eps)`` -- same eps placement as C3, so the
+    fused HIP apply kernel is unchanged and the correction travels through
+    the device-scalar lr.
 """
 
 from __future__ import annotations
@@ -53,6 +61,7 @@ class AccumEngine:
         beta2: float = 0.999,
         eps: float = 1e-6,
         exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
+        bias_correction: bool = False,
         strict_reference_semantics: bool = False,
         process_group=None,
         allreduce_bucket_mb: int = 64,
@@ -70,10 +79,15 @@ class AccumEngine:
         self.clip_norm = clip_norm
         self.weight_decay = float(weight_decay)
         self.beta1, self.beta2, self.eps = float(beta1), float(beta2), float(eps)
+        self.bias_correction = bool(bias_correction)
         self.strict = bool(strict_reference_semantics)
         self.group = process_group
         self.allreduce_bucket_elems = max(1, (allreduce_bucket_mb << 20) // 4)
         self.global_step = 0
+        # number of optimizer updates performed -- TF's beta1_power/beta2_power
+        # advance once per apply_gradients call, which is what bias correction
+        # keys off (tf.train.AdamOptimizer._finish)
+        self.apply_count = 0
         self.last_lr = 0.0
 
         dev = self.state.device
@@ -155,11 +169,21 @@ class AccumEngine:
         for h in handles:
             h.wait()
 
+    def _eff_lr(self, lr: float) -> float:
+        """Step size the update kernel multiplies by: the schedule lr, times
+        TF stock Adam's folded bias correction when enabled (t = the update
+        about to be performed, 1-based)."""
+        if not self.bias_correction:
+            return lr
+        t = self.apply_count + 1
+        return lr * ((1.0 - self.beta2**t) ** 0.5) / (1.0 - self.beta1**t)
+
     def set_lr(self, lr: float) -> None:
         """Write the schedule's lr into the device scalar the apply kernel
-        reads -- the hipGraph-capture-safe path for a changing lr."""
+        reads -- the hipGraph-capture-safe path for a changing lr. The value
+        written is the *effective* step size (bias correction folded in)."""
         self.last_lr = lr
-        self._lr_dev.fill_(lr)
+        self._lr_dev.fill_(self._eff_lr(lr))
 
     def _join_wgrad_stream(self) -> None:
         """Fence the optional wgrad side stream before anything reads the
@@ -203,6 +227,7 @@ class AccumEngine:
         if lr is None:
             lr = self.lr_at(self.global_step)
         self.last_lr = lr
+        lr = self._eff_lr(lr)
         inv_k = 1.0 / self.K
         model = None if st.master is st.model else st.model
         if self._hip is not None:
@@ -241,6 +266,7 @@ class AccumEngine:
                 beta2=self.beta2,
                 eps=self.eps,
             )
+        self.apply_count += 1
 
     def micro_step(self) -> bool:
         """One reference session.run: accumulate, maybe apply, step += 1.
@@ -262,6 +288,7 @@ class AccumEngine:
         self._join_wgrad_stream()
         d = self.state.state_dict()
         d["global_step"] = self.global_step
+        d["apply_count"] = self.apply_count
         d["K"] = self.K
         d["strict"] = self.strict
         return d
@@ -269,6 +296,7 @@ class AccumEngine:
     def load_state_dict(self, d: Dict) -> None:
         self.state.load_state_dict(d)
         self.global_step = int(d["global_step"])
+        self.apply_count = int(d.get("apply_count", 0))
         if int(d.get("K", self.K)) != self.K:
             raise ValueError("checkpoint K does not match engine K")
         if bool(d.get("strict", self.strict)) != self.strict:

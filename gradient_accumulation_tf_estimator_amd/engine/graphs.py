@@ -41,6 +41,14 @@ class GraphedTrainLoop:
         stream before capture end, pinning the wgrads to the critical path."""
         if engine.backend != "hip":
             raise RuntimeError("GraphedTrainLoop requires the HIP engine backend")
+        if window and engine.strict:
+            # window mode bakes the apply into slot K-1 (the corrected
+            # (s+1)%K==0 predicate); a strict engine applies at s%K==0, so
+            # replaying the window graph would silently shift every apply
+            # boundary. Refuse instead.
+            raise ValueError(
+                "window=True is incompatible with strict_reference_semantics: "
+                "the window graph fixes the apply at slot K-1")
         self.engine = engine
         self.world = world
         self.window = bool(window) and world == 1 and engine.K > 1
@@ -137,17 +145,20 @@ class GraphedTrainLoop:
             engine.set_lr(engine.lr_at(engine.global_step))
             self.g_apply.replay()
             engine.global_step += 1
+            engine.apply_count += 1
             return self.loss_apply
         if engine.is_apply_step():
             engine.set_lr(engine.lr_at(engine.global_step))
             if self.g_apply is not None:
                 self.g_apply.replay()
                 engine.global_step += 1
+                engine.apply_count += 1
                 return self.loss_apply
             self.g_accum.replay()
             engine._allreduce_accum()
             engine.apply_from_device()
             engine.global_step += 1
+            engine.apply_count += 1
             return self.loss_accum
         self.g_accum.replay()
         engine.global_step += 1

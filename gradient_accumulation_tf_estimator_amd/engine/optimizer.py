@@ -9,7 +9,8 @@ equivalent: backward -> accumulate-or-apply -> global_step += 1
 (SURVEY.md section 3.3).
 
 The non-clipping generic variant (another-example.py:126-155, 02:47-74,
-04:49-74) is ``clip_norm=None``.
+04:49-74) is ``optimizer="adam"`` (stock ``tf.train.AdamOptimizer``:
+bias-corrected, eps=1e-8, no weight decay) with ``clip_norm=None``.
 """
 
 from __future__ import annotations
@@ -64,11 +65,12 @@ def create_optimizer(
     num_warmup_steps: int = 0,
     *,
     gradient_accumulation_multiplier: int = 1,
+    optimizer: str = "adamw",
     clip_norm: Optional[float] = 1.0,
-    weight_decay: float = 0.01,
+    weight_decay: Optional[float] = None,
     beta1: float = 0.9,
     beta2: float = 0.999,
-    eps: float = 1e-6,
+    eps: Optional[float] = None,
     exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
     use_tpu: bool = False,
     strict_reference_semantics: bool = False,
@@ -76,6 +78,25 @@ def create_optimizer(
     ddp_scale_loss: bool = True,
     backend: str = "auto",
 ) -> TrainOp:
+    """Build the train op.
+
+    ``optimizer``:
+      * ``"adamw"`` -- the reference's AdamWeightDecayOptimizer (C3,
+        optimization.py:107-194): NO bias correction, eps=1e-6 outside the
+        sqrt, decoupled weight decay 0.01 with regex exclusion.
+      * ``"adam"`` -- stock ``tf.train.AdamOptimizer`` (another-example.py:139,
+        02:41): bias correction folded into the step size, eps=1e-8 added to
+        sqrt(v), no weight decay.
+    ``weight_decay``/``eps`` left as None take the chosen optimizer's
+    reference default and may be overridden explicitly.
+    """
+    if optimizer not in ("adamw", "adam"):
+        raise ValueError(f"optimizer must be 'adamw' or 'adam', got {optimizer!r}")
+    bias_correction = optimizer == "adam"
+    if eps is None:
+        eps = 1e-8 if bias_correction else 1e-6
+    if weight_decay is None:
+        weight_decay = 0.0 if bias_correction else 0.01
     model = model_or_params if isinstance(model_or_params, nn.Module) else None
     if model is not None:
         named = list(model.named_parameters())
@@ -113,6 +134,7 @@ def create_optimizer(
         beta2=beta2,
         eps=eps,
         exclude_from_weight_decay=exclude_from_weight_decay,
+        bias_correction=bias_correction,
         strict_reference_semantics=strict_reference_semantics,
         process_group=process_group,
         backend=backend,

@@ -40,7 +40,8 @@ class NumpyAdamWOracle:
     """Straight transcription of optimization.py:150-171 + 76-103 in NumPy."""
 
     def __init__(self, shapes, decay_flags, K, init_lr, num_train_steps, num_warmup_steps,
-                 clip_norm=1.0, wd=0.01, b1=0.9, b2=0.999, eps=1e-6, strict=False):
+                 clip_norm=1.0, wd=0.01, b1=0.9, b2=0.999, eps=1e-6, strict=False,
+                 bias_correction=False):
         self.p = None  # set later
         self.m = [np.zeros(s, np.float64) for s in shapes]
         self.v = [np.zeros(s, np.float64) for s in shapes]
@@ -49,8 +50,12 @@ class NumpyAdamWOracle:
         self.K, self.clip_norm, self.wd = K, clip_norm, wd
         self.b1, self.b2, self.eps = b1, b2, eps
         self.step = 0
+        self.applies = 0
         self.init_lr, self.nts, self.nws = init_lr, num_train_steps, num_warmup_steps
         self.strict = strict
+        # tf.train.AdamOptimizer: lr_t = lr*sqrt(1-b2^t)/(1-b1^t), t counts
+        # apply_gradients calls (another-example.py:139, 02:41)
+        self.bias_correction = bias_correction
 
     def lr(self):
         s = min(self.step, self.nts)
@@ -70,6 +75,9 @@ class NumpyAdamWOracle:
                 coef = self.clip_norm / max(gn, self.clip_norm)
                 norm_g = [g * coef for g in norm_g]
             lr = self.lr()
+            if self.bias_correction:
+                t = self.applies + 1
+                lr = lr * math.sqrt(1 - self.b2**t) / (1 - self.b1**t)
             for i, g in enumerate(norm_g):
                 self.m[i] = self.b1 * self.m[i] + (1 - self.b1) * g
                 self.v[i] = self.b2 * self.v[i] + (1 - self.b2) * g * g
@@ -78,17 +86,22 @@ class NumpyAdamWOracle:
                     u = u + self.wd * self.p[i]
                 self.p[i] = self.p[i] - lr * u
             self.accum = [np.zeros_like(a) for a in self.accum]
+            self.applies += 1
         self.step += 1
         return applied
 
 
-def run_pair(strict, K=3, steps=10, clip_norm=1.0):
+def run_pair(strict, K=3, steps=10, clip_norm=1.0, optimizer="adamw"):
     torch.manual_seed(42)
     net = TinyNet().double()
     names = [n for n, _ in net.named_parameters()]
     shapes = [tuple(p.shape) for _, p in net.named_parameters()]
     decay = [use_weight_decay(n, ("LayerNorm", "layer_norm", "bias")) for n in names]
-    oracle = NumpyAdamWOracle(shapes, decay, K, 1e-2, 100, 5, clip_norm=clip_norm, strict=strict)
+    adam = optimizer == "adam"
+    oracle = NumpyAdamWOracle(shapes, decay, K, 1e-2, 100, 5, clip_norm=clip_norm,
+                              strict=strict, bias_correction=adam,
+                              wd=0.0 if adam else 0.01,
+                              eps=1e-8 if adam else 1e-6)
     oracle.p = [p.detach().numpy().copy() for _, p in net.named_parameters()]
 
     net_f = TinyNet()
@@ -96,6 +109,7 @@ def run_pair(strict, K=3, steps=10, clip_norm=1.0):
     op = create_optimizer(
         net_f, 1e-2, 100, 5,
         gradient_accumulation_multiplier=K,
+        optimizer=optimizer,
         clip_norm=clip_norm,
         strict_reference_semantics=strict,
     )
@@ -138,6 +152,62 @@ def test_strict_step0_applies_immediately():
 def test_corrected_applies_after_full_window():
     _, _, ae, _ = run_pair(strict=False, K=4, steps=9)
     assert ae == [False, False, False, True, False, False, False, True, False]
+
+
+@pytest.mark.parametrize("strict", [False, True])
+def test_stock_adam_matches_numpy_oracle(strict):
+    # the generic/MNIST/distributed variants use bias-corrected
+    # tf.train.AdamOptimizer (another-example.py:139, 02:41)
+    net_f, oracle, ae, ao = run_pair(strict, K=2, steps=8, clip_norm=None,
+                                     optimizer="adam")
+    assert ae == ao
+    for (n, p), arr in zip(net_f.named_parameters(), oracle.p):
+        np.testing.assert_allclose(p.detach().numpy(), arr, rtol=2e-4, atol=2e-5)
+
+
+def test_bias_correction_first_update_magnitude():
+    """Bias-corrected Adam's first update is ~lr*sign(g); the reference's
+    AdamWeightDecay (no correction) gives ~0.316*lr*sign(g) for the same
+    gradient -- the large early-step divergence VERDICT.md item 1 flags."""
+    lr = 1e-3
+    for optimizer in ("adam", "adamw"):
+        p = nn.Parameter(torch.zeros(64))
+        op = create_optimizer([("w", p)], lr, 1000, 0, optimizer=optimizer,
+                              clip_norm=None, weight_decay=0.0)
+        loss = (p * torch.ones(64)).sum()  # g = 1 exactly
+        op.step(loss)
+        # uncorrected: u = 0.1/(sqrt(0.001)+eps) ~ sqrt(1-b2)/... ; corrected ~ 1
+        step_mag = float(p.detach().abs().mean()) / lr
+        if optimizer == "adam":
+            assert abs(step_mag - 1.0) < 1e-3
+        else:
+            # m/(sqrt(v)+eps) = 0.1/(sqrt(0.001)+1e-6) ~ 3.1623
+            assert abs(step_mag - 0.1 / math.sqrt(0.001)) < 1e-3
+
+
+def test_adam_apply_count_checkpointed():
+    torch.manual_seed(3)
+    net = TinyNet()
+    op = create_optimizer(net, 1e-3, 1000, 0, optimizer="adam",
+                          gradient_accumulation_multiplier=2, clip_norm=None)
+    xs = [torch.randn(4, 7) for _ in range(8)]
+    for x in xs[:4]:
+        op.step((net(x) ** 2).mean())
+    assert op.engine.apply_count == 2
+    sd = {k: (v.clone() if torch.is_tensor(v) else v) for k, v in op.state_dict().items()}
+    for x in xs[4:]:
+        op.step((net(x) ** 2).mean())
+    ref = net.fc1.weight.detach().clone()
+
+    torch.manual_seed(3)
+    net2 = TinyNet()
+    op2 = create_optimizer(net2, 1e-3, 1000, 0, optimizer="adam",
+                           gradient_accumulation_multiplier=2, clip_norm=None)
+    op2.load_state_dict(sd)
+    assert op2.engine.apply_count == 2
+    for x in xs[4:]:
+        op2.step((net2(x) ** 2).mean())
+    assert torch.equal(net2.fc1.weight.detach(), ref)
 
 
 def test_no_clip_variant():

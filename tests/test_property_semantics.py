@@ -37,12 +37,13 @@ class RandNet(nn.Module):
     warmup=st.integers(0, 6),
     clip=st.sampled_from([None, 0.5, 1.0]),
     wd=st.sampled_from([0.0, 0.01, 0.1]),
+    opt=st.sampled_from(["adamw", "adam"]),
     strict=st.booleans(),
     din=st.integers(2, 9),
     dh=st.integers(2, 8),
     seed=st.integers(0, 10_000),
 )
-def test_engine_tracks_oracle_for_random_configs(K, steps, warmup, clip, wd,
+def test_engine_tracks_oracle_for_random_configs(K, steps, warmup, clip, wd, opt,
                                                  strict, din, dh, seed):
     torch.manual_seed(seed)
     net64 = RandNet(din, dh).double()
@@ -58,14 +59,15 @@ def test_engine_tracks_oracle_for_random_configs(K, steps, warmup, clip, wd,
     # reference-eps math is pinned by test_accum_semantics.py)
     EPS = 1e-3
     oracle = NumpyAdamWOracle(shapes, decay, K, 3e-3, nts, warmup,
-                              clip_norm=clip, wd=wd, strict=strict, eps=EPS)
+                              clip_norm=clip, wd=wd, strict=strict, eps=EPS,
+                              bias_correction=opt == "adam")
     oracle.p = [p.detach().numpy().copy() for _, p in net64.named_parameters()]
 
     net = RandNet(din, dh)
     net.load_state_dict({k: v.float() for k, v in net64.state_dict().items()})
     op = create_optimizer(net, 3e-3, nts, warmup,
                           gradient_accumulation_multiplier=K, clip_norm=clip,
-                          weight_decay=wd, eps=EPS,
+                          optimizer=opt, weight_decay=wd, eps=EPS,
                           strict_reference_semantics=strict)
 
     gen = torch.Generator().manual_seed(seed + 1)
