@@ -59,6 +59,19 @@ def parse_args():
 
 def main():
     args = parse_args()
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        # driver contract: `python bench.py --gpus N` must itself be the
+        # N-rank launch -- re-exec through torch.distributed.run (one rank
+        # per GPU over RCCL) when not already under a launcher
+        import subprocess
+
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--standalone", "--local-addr", "127.0.0.1",
+            f"--nproc-per-node={args.gpus}",
+            os.path.abspath(__file__),
+        ] + sys.argv[1:]
+        raise SystemExit(subprocess.call(cmd))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))

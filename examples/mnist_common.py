@@ -50,8 +50,14 @@ def make_model_fn():
         if mode == ModeKeys.EVAL:
             def acc(f, l):
                 return float((model(f).argmax(-1) == l).float().mean()), l.numel()
+            # EVAL also carries predictions (01:50-57): logits/classes/probs
+            def preds(f):
+                logits = model(f)
+                return {"logits": logits, "classes": logits.argmax(-1),
+                        "probabilities": logits.softmax(-1)}
             return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
-                                 eval_metric_fns={"accuracy": acc})
+                                 eval_metric_fns={"accuracy": acc},
+                                 predictions_fn=preds)
         # stock-Adam variant (02:41 AdamOptimizer(learning_rate=1e-4)):
         # bias-corrected, eps=1e-8, no clipping, no weight decay (02:47-74)
         train_op = create_optimizer(

@@ -48,6 +48,10 @@ class ModeKeys:
 
 @dataclass
 class EstimatorSpec:
+    """``predictions_fn`` is required for PREDICT and optional for EVAL/TRAIN
+    -- the reference's EVAL spec also carries ``predictions`` (01:50-57);
+    ``evaluate(..., return_predictions=True)`` surfaces them."""
+
     mode: str
     model: Optional[torch.nn.Module] = None
     loss_fn: Optional[Callable] = None
@@ -175,7 +179,8 @@ class Estimator:
         device = self.config.device
         # keep one live iterator per input_fn so chunked train() calls
         # (train_and_evaluate) continue the stream instead of restarting it
-        if self._train_iter is None or self._train_iter_key is not input_fn:
+        fresh_iter = self._train_iter is None or self._train_iter_key is not input_fn
+        if fresh_iter:
             self._train_iter = iter(_call_input_fn(input_fn, ModeKeys.TRAIN))
             self._train_iter_key = input_fn
         it = self._train_iter
@@ -186,9 +191,17 @@ class Estimator:
             return {"global_step": sp.train_op.global_step if sp else 0, "loss": None}
         features, labels = _to_device(first[0], device), _to_device(first[1], device)
 
+        skip = 0
         if self._train_spec is None:
             self._train_spec = self._build_spec(ModeKeys.TRAIN, features, labels)
-            self._restore(self._train_spec, with_engine=True)
+            restored = self._restore(self._train_spec, with_engine=True)
+            if fresh_iter and restored:
+                # deterministic input replay on resume: the checkpointed
+                # engine already consumed `restored` micro-batches of this
+                # (seeded) stream; skip them so the resumed run sees exactly
+                # the batches an uninterrupted run would (makes restart-from-
+                # checkpoint bit-exact, utils/failure.py)
+                skip = int(restored)
         spec = self._train_spec
         op = spec.train_op
 
@@ -196,6 +209,13 @@ class Estimator:
         done_this_call = 0
         t_last, s_last = time.perf_counter(), op.global_step
         pending = (features, labels)
+        if skip:
+            pending = None  # `first` was batch 0 of the replayed stream
+            for _ in range(skip - 1):
+                try:
+                    next(it)
+                except StopIteration:
+                    break
         last_loss = None
         while True:
             step = op.global_step
@@ -232,7 +252,8 @@ class Estimator:
         }
 
     def evaluate(self, input_fn, steps: Optional[int] = None,
-                 checkpoint_path: Optional[str] = None) -> Dict:
+                 checkpoint_path: Optional[str] = None,
+                 return_predictions: bool = False) -> Dict:
         device = self.config.device
         it = iter(_call_input_fn(input_fn, ModeKeys.EVAL))
         try:
@@ -250,6 +271,7 @@ class Estimator:
         loss_m = Mean()
         metric_means = {k: Mean() for k in spec.eval_metric_fns}
         n_batches = 0
+        predictions = [] if (return_predictions and spec.predictions_fn) else None
 
         def run_batch(f, l):
             nonlocal n_batches
@@ -261,6 +283,8 @@ class Estimator:
                     out = fn(f, l)
                     v, cnt = out if isinstance(out, tuple) else (out, n)
                     metric_means[k].update(float(v), cnt)
+                if predictions is not None:
+                    predictions.append(spec.predictions_fn(f))
             n_batches += 1
 
         run_batch(features, labels)
@@ -273,6 +297,8 @@ class Estimator:
                    "global_step": restored_step if restored_step is not None else 0}
         results.update({k: m.result() for k, m in metric_means.items()})
         self._logger.log(eval=True, **{k: v for k, v in results.items()})
+        if predictions is not None:
+            results["predictions"] = predictions
         return results
 
     def predict(self, input_fn, checkpoint_path: Optional[str] = None) -> Iterator:

@@ -466,11 +466,25 @@ class _Embed3Fn(torch.autograd.Function):
         return (None,) * 9
 
 
+_ARANGE_OK = {}
+
+
 def fused_embed3(ids, pos_ids_row, tok_ids, word_mod, pos_mod, tok_mod):
-    """ids [B,S]; pos_ids_row [1,S] (broadcast over the batch in backward
-    via an expanded scatter); tok_ids [B,S] or None. All modules must be
-    engine-bound DirectEmbeddings on bf16."""
+    """ids [B,S]; pos_ids_row [1,S] -- MUST be arange(S): the fused forward
+    kernel derives the position row from the sequence index (k_emb3_fwd) and
+    the backward scatters by pos_ids_row, so any other ordering would give a
+    wrong forward with mismatched gradients. Asserted here (once per
+    distinct tensor) rather than silently trusted. tok_ids [B,S] or None.
+    All modules must be engine-bound DirectEmbeddings on bf16."""
     B, S = ids.shape
+    key = (pos_ids_row.data_ptr(), S)
+    if key not in _ARANGE_OK:
+        expect = torch.arange(S, device=pos_ids_row.device)
+        if not torch.equal(pos_ids_row.reshape(-1), expect):
+            raise ValueError(
+                "fused_embed3 requires pos_ids_row == arange(seq_len); got a "
+                "non-contiguous position ordering (use the unfused path)")
+        _ARANGE_OK[key] = True
     pos_exp = pos_ids_row.expand(B, S)
     return _Embed3Fn.apply(ids, pos_exp, tok_ids, word_mod.weight,
                            pos_mod.weight, tok_mod.weight,

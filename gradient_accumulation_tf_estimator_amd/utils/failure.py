@@ -7,10 +7,14 @@ checkpoints give" (README.md:133). Here the policy is explicit:
   ``is_fatal_comm_error`` classifies it, ``abort_process_group`` tears the
   process group down so peers fail fast too instead of hanging in a
   collective;
-* restart-from-checkpoint: ``train_with_restarts`` re-enters
-  ``Estimator.train`` after a transient failure; the estimator's normal
-  model-dir restore (engine state_dict incl. the mid-window accumulation
-  buffer) makes the resume exact.
+* restart-from-checkpoint: ``train_with_restarts`` discards the estimator's
+  live train spec and input iterator and re-enters ``Estimator.train``; the
+  rebuild restores model + engine state (incl. the mid-window accumulation
+  buffer and step counter) from the latest checkpoint and fast-forwards the
+  seeded input stream to the checkpointed step, so the resumed run is
+  bit-exact vs an uninterrupted one. (Re-entering with the live spec would
+  keep a half-accumulated grad buffer from the failed step -- that is NOT a
+  checkpoint resume.)
 
 No elasticity -- world size is fixed for a job, as in the reference.
 """
@@ -73,4 +77,14 @@ def train_with_restarts(estimator, input_fn, *, max_steps: int,
                 raise
             logger.warning("step failed (%s); restart %d/%d from checkpoint",
                            exc, attempts, max_restarts)
+            _reset_to_checkpoint(estimator)
             time.sleep(backoff_secs)
+
+
+def _reset_to_checkpoint(estimator) -> None:
+    """Drop the live train spec/iterator so the next ``train()`` rebuilds
+    from the latest checkpoint instead of continuing a poisoned in-memory
+    state (stale grads in the flat buffer after a mid-step exception)."""
+    for attr in ("_train_spec", "_train_iter", "_train_iter_key"):
+        if hasattr(estimator, attr):
+            setattr(estimator, attr, None)

@@ -37,7 +37,8 @@ def mnist_model_fn(features, labels, mode, params):
         def acc(f, l):
             return float((model(f).argmax(-1) == l).float().mean()), l.numel()
         return EstimatorSpec(mode, model=model, loss_fn=loss_fn,
-                             eval_metric_fns={"accuracy": acc})
+                             eval_metric_fns={"accuracy": acc},
+                             predictions_fn=lambda f: model(f).argmax(-1))
     train_op = create_optimizer(
         model, params["learning_rate"], 10000, 0,
         gradient_accumulation_multiplier=params["gradient_accumulation_multiplier"],
@@ -75,6 +76,18 @@ def test_train_eval_predict_cycle(tmp_path):
         (f, l) for f, l in eval_input_fn())))
     assert len(preds) == 256
     assert all(0 <= int(p) <= 9 for p in preds)
+
+
+def test_eval_spec_carries_predictions(tmp_path):
+    # the reference EVAL spec also returns predictions (01:50-57)
+    est = make_estimator(tmp_path)
+    est.train(train_input_fn, max_steps=10)
+    ev = est.evaluate(eval_input_fn, return_predictions=True)
+    batches = ev["predictions"]
+    assert sum(p.numel() for p in batches) == 256
+    # default evaluate() keeps the metrics-only result shape
+    ev2 = est.evaluate(eval_input_fn)
+    assert "predictions" not in ev2
 
 
 def test_checkpoint_resume_continues_exactly(tmp_path):
