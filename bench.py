@@ -88,8 +88,10 @@ def main():
         dist = None
 
     if use_cuda:
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # modulo lets a 2-rank shakeout run on a 1-GPU box (RCCL permitting)
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
         if args.sdpa != "auto":
             torch.backends.cuda.enable_flash_sdp(args.sdpa == "flash")
             torch.backends.cuda.enable_mem_efficient_sdp(args.sdpa == "efficient")
