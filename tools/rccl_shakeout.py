@@ -1,26 +1,35 @@
 #!/usr/bin/env python3
-"""First-contact shakeout of the multi-rank RCCL path on real hardware
+"""First-contact shakeout of the multi-rank collective path on real hardware
 (nothing RCCL-side had ever executed on a GPU).
 
-Launch (2 ranks; works on a 1-GPU box if RCCL accepts two ranks on one
-device, else falls back to reporting that limitation):
+RCCL (like NCCL) refuses two ranks on one device ("Duplicate GPU detected",
+measured on this pool 2026-09-13, rccl 2.26.6), so a 1-GPU box is covered in
+two parts:
 
+  single-process (no RANK in env):
+      a real single-rank RCCL communicator: init + bucketed all-reduce of an
+      engine-shaped flat buffer enqueued on the HIP stream, plus a broadcast
+      -- shakes out RCCL init/enqueue/stream interaction with the engine.
+
+  torchrun -nproc-per-node 2 (RANK set):
+      the engine's world>1 code path ON the GPU with a real collective --
+      backend "gloo" on a 1-GPU box (both ranks share cuda:0; gloo allows
+      it), "nccl" when each rank can have its own device:
+        1. bucketed all-reduce shaped exactly like AccumEngine._allreduce_accum
+        2. eager world>1 micro-steps (loss 1/W, K1, apply-boundary
+           all-reduce, fused apply)
+        3. GraphedTrainLoop world>1 branch (captured accumulate graph +
+           eager all-reduce + apply_from_device)
+        4. DP2 x K2 == single-process K=4 equivalence on GPU
+
+    python tools/rccl_shakeout.py                      # single-rank RCCL
     python -m torch.distributed.run --standalone --local-addr 127.0.0.1 \
-        --nproc-per-node 2 tools/rccl_shakeout.py
+        --nproc-per-node 2 tools/rccl_shakeout.py --backend gloo
 
-Exercises, in order:
-  1. NCCL(=RCCL) process-group init + a bucketed all-reduce shaped exactly
-     like AccumEngine._allreduce_accum (flat fp32 buffer, 64 MiB buckets).
-  2. The engine's world>1 micro-step path eagerly (loss scale 1/W, K1
-     accumulate, apply-boundary all-reduce, fused apply).
-  3. The GraphedTrainLoop world>1 branch: captured accumulate graph +
-     eager all-reduce + apply_from_device.
-  4. DP2 x K2 == single-process K=4 equivalence ON GPU: both ranks train a
-     bert-tiny config two windows, rank 0 reruns the same global batches
-     single-process and compares master weights.
 Prints one JSON line per phase; exits nonzero on any mismatch.
 """
 
+import argparse
 import json
 import os
 import sys
@@ -34,37 +43,64 @@ from gradient_accumulation_tf_estimator_amd import create_optimizer
 from gradient_accumulation_tf_estimator_amd.models.bert import (
     BertConfig, BertForSequenceClassification)
 
+BUCKET = (64 << 20) // 4
+
 
 def log(rank, phase, **kw):
     if rank == 0:
         print(json.dumps({"phase": phase, **kw}), flush=True)
 
 
-def main():
+def bucketed_allreduce(buf):
+    n = buf.numel()
+    handles = [
+        dist.all_reduce(buf[off : min(off + BUCKET, n)], async_op=True)
+        for off in range(0, n, BUCKET)
+    ]
+    for h in handles:
+        h.wait()
+    return len(handles)
+
+
+def single_rank_rccl():
+    """A world-1 RCCL communicator is a real RCCL communicator: init,
+    enqueue on the HIP stream, completion -- everything but cross-rank
+    traffic."""
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29531")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    n = 28_000_000
+    buf = torch.full((n,), 2.0, device="cuda:0")
+    nb = bucketed_allreduce(buf)
+    dist.broadcast(buf, src=0)
+    torch.cuda.synchronize()
+    assert torch.all(buf == 2.0)
+    log(0, "single_rank_rccl", elems=n, buckets=nb, ok=True)
+    dist.destroy_process_group()
+
+
+def multi_rank(backend):
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
     ndev = torch.cuda.device_count()
     dev_idx = rank % ndev
     torch.cuda.set_device(dev_idx)
     device = torch.device("cuda", dev_idx)
+    if backend == "auto":
+        backend = "nccl" if ndev >= world else "gloo"
 
-    dist.init_process_group("nccl", rank=rank, world_size=world)
-    log(rank, "init", world=world, ndev=ndev, dev_idx=dev_idx)
+    dist.init_process_group(backend, rank=rank, world_size=world)
+    log(rank, "init", world=world, ndev=ndev, dev_idx=dev_idx, backend=backend)
 
-    # --- 1. bucketed all-reduce, engine-shaped ---
+    # --- 1. bucketed all-reduce, engine-shaped, on the GPU buffer ---
     n = 28_000_000  # ~BERT-Small grad-buffer scale, 112 MB fp32
     buf = torch.full((n,), float(rank + 1), device=device)
-    bucket = (64 << 20) // 4
-    handles = [
-        dist.all_reduce(buf[off : min(off + bucket, n)], async_op=True)
-        for off in range(0, n, bucket)
-    ]
-    for h in handles:
-        h.wait()
+    nb = bucketed_allreduce(buf)
     torch.cuda.synchronize()
     expect = world * (world + 1) / 2
     assert torch.all(buf == expect), "bucketed all-reduce wrong"
-    log(rank, "bucketed_allreduce", elems=n, buckets=len(handles), ok=True)
+    log(rank, "bucketed_allreduce", elems=n, buckets=nb, ok=True)
 
     # --- 2+3. engine DP path, eager then graphed ---
     # bert-small itself: the exact fused-module GPU path the bench runs
@@ -87,11 +123,11 @@ def main():
             for _ in range(n_steps)
         ]
 
-    # global stream of micro-batches; rank r takes batch 2*i + r of window i
-    all_b = batches(8, seed=7)
+    # global stream of micro-batches; rank r takes batch i*world + r
+    all_b = batches(2 * K * world, seed=7)
 
     model = make(3)
-    op = create_optimizer(model, 1e-3, 1000, 0,
+    op = create_optimizer(model, 1e-3, 10**6, 0,
                           gradient_accumulation_multiplier=K, clip_norm=1.0,
                           backend="hip")
     for i in range(2 * K):  # 2 windows eager
@@ -114,7 +150,7 @@ def main():
     torch.cuda.synchronize()
     log(rank, "graphed_dp_microsteps", steps=2 * K, ok=True)
 
-    # --- 4. DP2 x K == single-process 2K equivalence on GPU ---
+    # --- 4. DP{W} x K == single-process K*W equivalence on GPU ---
     model_dp = make(11)
     op_dp = create_optimizer(model_dp, 1e-3, 10**6, 0,
                              gradient_accumulation_multiplier=K, clip_norm=1.0,
@@ -135,7 +171,7 @@ def main():
                                 clip_norm=1.0, backend="hip")
         # single-process equivalent (SURVEY.md 2.2.7 linearity): NO loss
         # scaling, K*world accumulation -- sum/(K*W) == DP's (sum*1/W)/K.
-        # The engine still sees the live 2-rank process group, so drive
+        # The engine still sees the live process group, so drive
         # accumulate/apply directly instead of micro_step (whose all-reduce
         # would hang with only rank 0 in it).
         for i in range(2 * K * world):
@@ -159,4 +195,10 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    p = argparse.ArgumentParser()
+    p.add_argument("--backend", default="auto", choices=["auto", "nccl", "gloo"])
+    args = p.parse_args()
+    if "RANK" not in os.environ:
+        single_rank_rccl()
+    else:
+        multi_rank(args.backend)
