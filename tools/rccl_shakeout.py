@@ -151,10 +151,15 @@ def multi_rank(backend):
     log(rank, "graphed_dp_microsteps", steps=2 * K, ok=True)
 
     # --- 4. DP{W} x K == single-process K*W equivalence on GPU ---
+    # eps=1e-3 bounds Adam's amplification of bf16 rounding noise where
+    # grads are ~0 (u = m/(sqrt(v)+eps) flips to +-3.16 on sign noise with
+    # the reference eps; see tests/test_property_semantics.py) -- measured
+    # 5.3e-3 max diff with eps=1e-6, within 5e-3 with the bounded eps
+    EPS = 1e-3
     model_dp = make(11)
     op_dp = create_optimizer(model_dp, 1e-3, 10**6, 0,
                              gradient_accumulation_multiplier=K, clip_norm=1.0,
-                             backend="hip")
+                             eps=EPS, backend="hip")
     for w in range(2):  # 2 windows
         for k in range(K):
             i = w * K + k
@@ -168,7 +173,7 @@ def multi_rank(backend):
         model_1 = make(11)
         op_1 = create_optimizer(model_1, 1e-3, 10**6, 0,
                                 gradient_accumulation_multiplier=K * world,
-                                clip_norm=1.0, backend="hip")
+                                clip_norm=1.0, eps=EPS, backend="hip")
         # single-process equivalent (SURVEY.md 2.2.7 linearity): NO loss
         # scaling, K*world accumulation -- sum/(K*W) == DP's (sum*1/W)/K.
         # The engine still sees the live process group, so drive
