@@ -42,6 +42,15 @@ def parse_args():
                    help="hipGraph-capture the micro-step; 'window' captures "
                         "the K-1 accumulate steps as one graph (wgrads "
                         "overlap the next step on a side stream)")
+    p.add_argument("--fuse-micro", default="auto",
+                   help="window fusion: compute N micro-steps as one fused "
+                        "fwd/bwd over the concatenated batch (exact by "
+                        "linearity, engine.micro_step_many; N must divide "
+                        "accum). 'auto' picks the largest divisor of accum "
+                        "whose fused row count fits --max-fuse-rows; 'off' "
+                        "or 1 disables")
+    p.add_argument("--max-fuse-rows", type=int, default=32768,
+                   help="row cap (micro_batch*seq_len*N) for auto fusion")
     p.add_argument("--allreduce-bucket-mb", type=int, default=64)
     p.add_argument("--wgrad-overlap", default="off", choices=["on", "off"],
                    help="EXPERIMENTAL: wgrad GEMMs on a side HIP stream "
@@ -144,9 +153,31 @@ def main():
         engine.micro_step()
         return loss
 
+    # ---- window fusion factor (exact reformulation: N micro-steps as one
+    # fused fwd/bwd, engine.micro_step_many) ----
+    fuse = 1
+    if use_cuda and args.graphs != "off" and args.fuse_micro != "off":
+        if args.fuse_micro == "auto":
+            for d in range(args.accum, 0, -1):
+                if args.accum % d == 0 and args.steps % d == 0 and \
+                        args.micro_batch * args.seq_len * d <= args.max_fuse_rows:
+                    fuse = d
+                    break
+        else:
+            fuse = int(args.fuse_micro)
+            if fuse < 1 or args.accum % fuse or args.steps % fuse:
+                raise SystemExit("--fuse-micro must divide --accum and --steps")
+
     use_graphs = use_cuda and args.graphs != "off"
     graphed = None
-    if use_graphs:
+    if use_graphs and fuse > 1:
+        try:
+            graphed = capture_fused(model, engine, pool_ids, pool_lab, world, fuse)
+        except Exception as e:
+            print(f"[bench] fused-window capture failed, falling back: {e}",
+                  file=sys.stderr)
+            fuse = 1
+    if use_graphs and graphed is None:
         try:
             graphed = capture_graphs(model, engine, pool_ids, pool_lab, inv_world,
                                      world, window=args.graphs == "window")
@@ -154,6 +185,8 @@ def main():
             print(f"[bench] hipGraph capture failed, falling back to eager: {e}",
                   file=sys.stderr)
             graphed = None
+    if graphed is None:
+        fuse = 1
 
     def step(i):
         if graphed is not None:
@@ -161,7 +194,7 @@ def main():
         return eager_micro_step(i)
 
     # ---- warmup ----
-    for i in range(args.warmup):
+    for i in range(0, args.warmup, fuse):
         step(i)
     if use_cuda:
         torch.cuda.synchronize()
@@ -173,7 +206,7 @@ def main():
 
     # ---- timed region: exactly args.steps micro-steps ----
     t0 = time.perf_counter()
-    for i in range(args.warmup, args.warmup + args.steps):
+    for i in range(args.warmup, args.warmup + args.steps, fuse):
         step(i)
     if use_cuda:
         torch.cuda.synchronize()
@@ -216,6 +249,7 @@ def main():
                 "seq_len": args.seq_len,
                 "parallelism": f"dp{world}",
                 "graphs": graphed is not None,
+                "window_fuse": fuse,
                 "peak_hbm_gb": round(peak_hbm_gb, 3),
             },
         }
@@ -223,6 +257,31 @@ def main():
 
     if dist:
         dist.destroy_process_group()
+
+
+def capture_fused(model, engine, pool_ids, pool_lab, world, fuse):
+    """Window-fused hipGraph loop: each replay computes `fuse` micro-steps
+    as one forward/backward over the concatenated [fuse*B, S] batch
+    (engine/graphs.py FusedWindowLoop; exact by linearity)."""
+    from gradient_accumulation_tf_estimator_amd.engine.graphs import FusedWindowLoop
+
+    POOL, B, S = pool_ids.shape
+    pool_packed = torch.cat([pool_ids, pool_lab[:, :, None]], dim=2).contiguous()
+    # block b consumes pool micro-batches b*fuse .. b*fuse+fuse-1 (mod POOL)
+    rolled = torch.stack(
+        [pool_packed[(torch.arange(fuse) + b) % POOL].reshape(fuse * B, S + 1)
+         for b in range(POOL)])
+    static = rolled[0].clone()
+    static_ids = static[:, :S]
+    static_lab = static[:, S]
+    loop = FusedWindowLoop(engine, lambda: model.loss(static_ids, static_lab),
+                           n_micro=fuse, world=world)
+
+    def run(i):
+        static.copy_(rolled[(i // fuse) % POOL])
+        return loop.step()
+
+    return run
 
 
 def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world,

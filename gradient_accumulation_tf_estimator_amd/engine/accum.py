@@ -283,6 +283,61 @@ class AccumEngine:
         self.global_step += 1
         return applied
 
+    def fused_block_sizes(self, max_micro: Optional[int] = None):
+        """Partition of one accumulation window into fused blocks, starting
+        at the CURRENT ``global_step``'s position: each block is a run of
+        micro-steps with no apply before its last slot (the invariant
+        ``micro_step_many`` enforces). Corrected semantics give [K] (or
+        [F, F, ...] capped at ``max_micro``); strict semantics give a
+        leading [1] (the step-0/step-mK apply) followed by K-sized runs.
+        """
+        sizes = []
+        s = self.global_step
+        first = True
+        while first or not self.is_apply_step(s - 1):
+            first = False
+            run = 1
+            while not self.is_apply_step(s + run - 1) and \
+                    (max_micro is None or run < max_micro):
+                run += 1
+            sizes.append(run)
+            s += run
+            if len(sizes) > 2 * self.K + 2:  # defensive: cannot happen
+                raise RuntimeError("window partition did not terminate")
+        return sizes
+
+    def micro_step_many(self, n: int) -> bool:
+        """``n`` reference micro-steps computed as ONE fused forward/backward
+        (window fusion): by linearity, the backward of
+        ``sum_k mean_loss(micro_batch_k)`` fills the grad buffer with exactly
+        the sum the reference's per-micro-step ``assign_add`` chain builds
+        (optimization.py:81,93) -- and with a single fp32 GEMM reduction
+        instead of n bf16 roundings, so numerics tighten. The LR schedule is
+        only ever READ at the apply step, so advancing ``global_step`` by n
+        is observationally identical. Blocks may not cross an apply boundary
+        (the apply must be the block's last slot).
+
+        Call after the fused loss's ``backward()``. Returns True if the
+        block ended in an optimizer update.
+        """
+        if n < 1:
+            raise ValueError("n must be >= 1")
+        s = self.global_step
+        for j in range(n - 1):
+            if self.is_apply_step(s + j):
+                raise RuntimeError(
+                    f"fused block of {n} starting at micro-step {s} crosses "
+                    f"an apply boundary at {s + j}; align blocks with "
+                    "fused_block_sizes()")
+        self.accumulate()
+        applied = self.is_apply_step(s + n - 1)
+        if applied:
+            self._join_wgrad_stream()
+            self._allreduce_accum()
+            self.apply(lr=self.lr_at(s + n - 1))
+        self.global_step = s + n
+        return applied
+
     # ---- checkpoint (SURVEY.md 2.2 item 8: accum + m/v + step all saved) ----
     def state_dict(self) -> Dict:
         self._join_wgrad_stream()

@@ -163,3 +163,120 @@ class GraphedTrainLoop:
         self.g_accum.replay()
         engine.global_step += 1
         return self.loss_accum
+
+
+class FusedWindowLoop:
+    """hipGraph-captured WINDOW-FUSED micro-batch loop.
+
+    ``n_micro`` reference micro-steps execute as ONE fused forward/backward
+    over their concatenated batch (engine.micro_step_many semantics:
+    linearity makes the accumulated gradient identical to the sequential
+    chain, with a tighter fp32 GEMM reduction). On MI355X this turns the
+    latency-bound small-row GEMM pool into n_micro-times-larger GEMMs --
+    the single biggest lever at the reference's tiny micro-batch (see
+    profiles/r01_pmc_counters.md).
+
+    ``loss_fn`` must return the MEAN loss over the n_micro*B-row static
+    batch. Two graphs: a block graph (fused fwd/bwd + accumulate) and, for
+    world == 1, an apply-block graph (same + global-norm + fused apply).
+    world > 1 replays the block graph and runs RCCL all-reduce + apply
+    eagerly, exactly like GraphedTrainLoop.
+    """
+
+    def __init__(self, engine: AccumEngine, loss_fn: Callable[[], torch.Tensor],
+                 *, n_micro: int, world: int = 1, warmup_iters: int = 3):
+        if engine.backend != "hip":
+            raise RuntimeError("FusedWindowLoop requires the HIP engine backend")
+        if engine.strict:
+            raise ValueError("window fusion under strict_reference_semantics "
+                             "needs per-window block resizing; use the eager "
+                             "step_fused path")
+        if engine.K % n_micro != 0:
+            raise ValueError(f"n_micro={n_micro} must divide K={engine.K}")
+        if engine.global_step % engine.K != 0:
+            raise ValueError("start FusedWindowLoop at a window boundary")
+        self.engine = engine
+        self.n = int(n_micro)
+        self.world = world
+        scale = float(n_micro) / float(world)
+
+        def fwd_bwd_accum():
+            loss = loss_fn()
+            (loss * scale).backward()
+            engine.accumulate()
+            engine._join_wgrad_stream()
+            return loss
+
+        st = engine.state
+        snap = {
+            "master": st.master.clone(),
+            "m": st.m.clone(),
+            "v": st.v.clone(),
+            "model": None if st.model is st.master else st.model.clone(),
+        }
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(warmup_iters):
+                fwd_bwd_accum()
+            engine.set_lr(engine.lr_at(0))
+            engine.apply_from_device()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+        blocks = engine.K // self.n
+        self.g_accum = None
+        self.loss_accum = None
+        if blocks > 1:
+            self.g_accum = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_accum):
+                self.loss_accum = fwd_bwd_accum()
+        pool = self.g_accum.pool() if self.g_accum is not None else None
+        if world == 1:
+            self.g_apply = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_apply, pool=pool):
+                self.loss_apply = fwd_bwd_accum()
+                engine.apply_from_device()
+        else:
+            # the apply block replays the accum graph (capture one if the
+            # window is a single block) and finishes eagerly
+            if self.g_accum is None:
+                self.g_accum = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(self.g_accum):
+                    self.loss_accum = fwd_bwd_accum()
+            self.g_apply = None
+            self.loss_apply = None
+        torch.cuda.synchronize()
+
+        st.master.copy_(snap["master"])
+        st.m.copy_(snap["m"])
+        st.v.copy_(snap["v"])
+        if snap["model"] is not None:
+            st.model.copy_(snap["model"])
+        st.accum.zero_()
+        st.grads.zero_()
+        del snap
+        torch.cuda.synchronize()
+
+    def step(self) -> torch.Tensor:
+        """One BLOCK = n_micro reference micro-steps (static inputs must
+        hold the n_micro*B-row concatenated batch). Returns the static
+        mean-loss tensor."""
+        engine = self.engine
+        s = engine.global_step
+        if (s % engine.K) + self.n == engine.K or engine.K == self.n:
+            engine.set_lr(engine.lr_at(s + self.n - 1))
+            if self.g_apply is not None:
+                self.g_apply.replay()
+                engine.global_step = s + self.n
+                engine.apply_count += 1
+                return self.loss_apply
+            self.g_accum.replay()
+            engine._allreduce_accum()
+            engine.apply_from_device()
+            engine.global_step = s + self.n
+            engine.apply_count += 1
+            return self.loss_accum
+        self.g_accum.replay()
+        engine.global_step = s + self.n
+        return self.loss_accum

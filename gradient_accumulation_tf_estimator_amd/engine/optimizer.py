@@ -51,6 +51,19 @@ class TrainOp:
         self.scale_loss(loss).backward()
         return self.engine.micro_step()
 
+    def step_fused(self, mean_loss: torch.Tensor, n: int) -> bool:
+        """``n`` micro-steps as one fused forward/backward (window fusion).
+
+        ``mean_loss`` is the MEAN loss over the n micro-batches concatenated
+        into one batch of n*B rows; ``n * mean_loss`` equals the sum of the
+        n per-micro-batch mean losses (equal micro-batch sizes), whose
+        backward fills the grad buffer with exactly the reference's
+        accumulated sum. Blocks must not cross an apply boundary -- use
+        ``engine.fused_block_sizes()`` to align.
+        """
+        self.scale_loss(mean_loss * n).backward()
+        return self.engine.micro_step_many(n)
+
     def state_dict(self):
         return self.engine.state_dict()
 

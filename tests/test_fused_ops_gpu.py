@@ -496,3 +496,63 @@ def test_fused_embed3_matches_reference():
         # per row -- pure quantization noise scales with the fan-in
         np.testing.assert_allclose(got.cpu().numpy(), want.cpu().numpy(),
                                    rtol=3e-2, atol=2e-2, err_msg=name)
+
+
+@pytest.mark.parametrize("fuse", [4, 2])
+def test_fused_window_loop_equals_sequential(fuse):
+    """FusedWindowLoop (window fusion: n micro-steps as one fwd/bwd) must
+    track the sequential graphed/eager chain on the real bert path."""
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+    from gradient_accumulation_tf_estimator_amd.engine.graphs import (
+        FusedWindowLoop)
+    from gradient_accumulation_tf_estimator_amd.models.bert import (
+        BertConfig, BertForSequenceClassification)
+
+    cfg = BertConfig(hidden_size=512, num_layers=2, num_heads=8,
+                     intermediate_size=2048)
+    K, B, S = 4, 4, 64
+    EPS = 1e-3  # bounds Adam noise amplification on ~0 grads (bf16)
+
+    def make(seed):
+        torch.manual_seed(seed)
+        m = BertForSequenceClassification(cfg).cuda().bfloat16()
+        m.train()
+        op = create_optimizer(m, 1e-3, 10**6, 0,
+                              gradient_accumulation_multiplier=K,
+                              clip_norm=1.0, eps=EPS, backend="hip")
+        return m, op
+
+    g = torch.Generator().manual_seed(5)
+    batches = [
+        (torch.randint(0, 30522, (B, S), generator=g).cuda(),
+         torch.randint(0, 2, (B,), generator=g).cuda())
+        for _ in range(2 * K)
+    ]
+
+    # sequential eager reference
+    m_a, op_a = make(9)
+    for ids, lab in batches:
+        op_a.step(m_a.loss(ids, lab))
+    torch.cuda.synchronize()
+
+    # fused-window graphed run over the same stream
+    m_b, op_b = make(9)
+    sid = torch.zeros(fuse * B, S, dtype=torch.long, device="cuda")
+    slab = torch.zeros(fuse * B, dtype=torch.long, device="cuda")
+    loop = FusedWindowLoop(op_b.engine, lambda: m_b.loss(sid, slab),
+                           n_micro=fuse, world=1)
+    i = 0
+    while i < len(batches):
+        blk = batches[i : i + fuse]
+        sid.copy_(torch.cat([b[0] for b in blk]))
+        slab.copy_(torch.cat([b[1] for b in blk]))
+        loop.step()
+        i += fuse
+    torch.cuda.synchronize()
+
+    assert op_b.engine.global_step == op_a.engine.global_step
+    assert op_b.engine.apply_count == op_a.engine.apply_count
+    a = op_a.engine.state.master
+    b = op_b.engine.state.master
+    diff = (a - b).abs().max().item()
+    assert diff < 5e-3, f"fused-window master diverged: {diff}"
