@@ -529,11 +529,17 @@ def test_fused_window_loop_equals_sequential(fuse):
         for _ in range(2 * K)
     ]
 
-    # sequential eager reference
+    # sequential eager reference, run TWICE: the embedding/wgrad scatter
+    # atomics are order-nondeterministic, so even identical runs differ;
+    # the fused path is held to that measured noise floor (x4 margin)
     m_a, op_a = make(9)
     for ids, lab in batches:
         op_a.step(m_a.loss(ids, lab))
+    m_a2, op_a2 = make(9)
+    for ids, lab in batches:
+        op_a2.step(m_a2.loss(ids, lab))
     torch.cuda.synchronize()
+    noise = (op_a.engine.state.master - op_a2.engine.state.master).abs().max().item()
 
     # fused-window graphed run over the same stream
     m_b, op_b = make(9)
@@ -555,4 +561,6 @@ def test_fused_window_loop_equals_sequential(fuse):
     a = op_a.engine.state.master
     b = op_b.engine.state.master
     diff = (a - b).abs().max().item()
-    assert diff < 5e-3, f"fused-window master diverged: {diff}"
+    bound = max(5e-3, 4.0 * noise)
+    assert diff < bound, \
+        f"fused-window master diverged: {diff} (noise floor {noise})"
