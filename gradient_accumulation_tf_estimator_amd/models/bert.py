@@ -119,15 +119,17 @@ class BertSelfAttention(nn.Module):
         self.qkv = Lin(cfg.hidden_size, 3 * cfg.hidden_size)
         self.dropout_p = cfg.dropout
 
-    def forward(self, x, attn_mask=None):
+    def forward(self, x, attn_mask=None, mask8=None, seed=None):
         B, S, H = x.shape
         qkv = self.qkv(x)  # [B, S, 3H]
         dp = self.dropout_p if self.training else 0.0
-        if (x.is_cuda and x.dtype == torch.bfloat16 and attn_mask is None
-                and dp == 0.0 and self.head_dim == 64
+        if (x.is_cuda and x.dtype == torch.bfloat16 and self.head_dim == 64
+                and (dp == 0.0 or seed is not None)
                 and fused_attention_supported(S, self.head_dim, False)):
-            # hand-written MFMA attention straight over the packed projection
-            return fused_attention(qkv, self.num_heads)
+            # hand-written MFMA attention straight over the packed
+            # projection; key-padding mask + prob dropout handled in-kernel
+            return fused_attention(qkv, self.num_heads, mask8=mask8,
+                                   seed=seed, p_drop=dp)
         qkv = qkv.view(B, S, 3, self.num_heads, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)  # [3, B, heads, S, head_dim]
         q, k, v = qkv[0], qkv[1], qkv[2]
@@ -188,8 +190,8 @@ class BertLayer(nn.Module):
             self.attention_LayerNorm._defer_residual_to = self.attention.qkv
             self.output_LayerNorm._defer_residual_to = self.intermediate
 
-    def forward(self, x, attn_mask=None):
-        a = self.attention(x, attn_mask)
+    def forward(self, x, attn_mask=None, mask8=None, seed=None):
+        a = self.attention(x, attn_mask, mask8=mask8, seed=seed)
         if self.fused:
             x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
             h = self.output(self.intermediate_act(self.intermediate(x)))
@@ -207,6 +209,11 @@ class BertModel(nn.Module):
         self.encoder = nn.ModuleList(BertLayer(cfg) for _ in range(cfg.num_layers))
         self.pooler = (DirectLinear if cfg.fused else nn.Linear)(
             cfg.hidden_size, cfg.hidden_size)
+        # per-layer attention-dropout seeds; refreshed each training forward
+        # by a captured RNG op (graph-safe: replays draw fresh philox values)
+        self.register_buffer("_attn_seeds",
+                             torch.zeros(cfg.num_layers, dtype=torch.int64),
+                             persistent=False)
         self.apply(self._init)
 
     def _init(self, m):
@@ -219,13 +226,20 @@ class BertModel(nn.Module):
             nn.init.zeros_(m.bias)
 
     def forward(self, input_ids, token_type_ids=None, attention_mask=None):
-        # attention_mask: [B, S] of 1/0 -> additive float mask for SDPA
-        mask = None
+        # attention_mask: [B, S] of 1/0. Fused path takes it as a u8
+        # key-padding mask; the SDPA fallback as a broadcast bool mask.
+        mask = mask8 = None
         if attention_mask is not None:
             mask = attention_mask[:, None, None, :].to(torch.bool)
+            mask8 = attention_mask.to(torch.uint8).contiguous()
+        seeds = None
+        if self.training and self.cfg.dropout > 0.0 and input_ids.is_cuda:
+            self._attn_seeds.random_()
+            seeds = self._attn_seeds
         x = self.embeddings(input_ids, token_type_ids)
-        for layer in self.encoder:
-            x = layer(x, mask)
+        for i, layer in enumerate(self.encoder):
+            x = layer(x, mask, mask8=mask8,
+                      seed=None if seeds is None else seeds[i])
         # pre-tanh pooler output; the tanh lives in the head (fused CE path)
         return x, self.pooler(x[:, 0])
 

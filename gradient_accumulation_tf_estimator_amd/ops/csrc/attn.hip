@@ -1,9 +1,24 @@
 // Hand-written MFMA attention for the reference bench shapes (gfx950).
 //
-// Scope: seq_len S <= 128 (S % 32 == 0), head_dim 64, no mask, no dropout,
-// bf16 in/out -- the reference's BERT configs all use head_dim 64 and the
-// headline bench is seq128. Larger shapes fall back to torch SDPA
-// (ops/fused.py gating).
+// Scope: seq_len S <= 128 (S % 32 == 0) single-pass or any S % 64 == 0
+// chunked, head_dim 64, bf16 in/out -- the reference's BERT configs all use
+// head_dim 64 and the headline bench is seq128.
+//
+// Key-padding mask: optional [B,S] u8 (1 = attend). Masked keys get score
+// -inf before the online max (fwd; -inf, not a large-negative, so a FULLY
+// masked streaming chunk cannot become its own max and leak exp(0)=1 rows)
+// and p forced to 0 in both backwards -- the exact additive -inf semantics
+// of torch SDPA's bool mask.
+//
+// Dropout (templated DROP so the off path pays nothing): applied to the
+// NORMALIZED probabilities, keep mask M ~ Bernoulli(1-p), P_drop = M*P/(1-p).
+// The mask is never materialized: a counter-based splitmix64 hash of
+// (seed, b, h, q, k) regenerates the identical M in the forward and both
+// backward kernels. The seed lives in DEVICE memory so a hipGraph replay
+// picks up a fresh value the host writes before each replay (same
+// mechanism as the engine's device-scalar lr). Backward identities:
+//   D = rowsum(dO o O) (unchanged: equals rowsum(P_drop o dP_drop)),
+//   dS = scale * P o (M/(1-p) * (dO V^T) - D),  dV = (M*P/(1-p))^T dO.
 //
 // Why hand-written: at micro-batch 8 the torch flash path costs far more in
 // layout copies (packed-QKV permutes), dq/dk/dv zero-fills and kernel count
@@ -38,6 +53,38 @@ static inline __device__ unsigned short f2bf_rne(float f) {
   union { float f; unsigned int i; } c;
   c.f = f;
   return (unsigned short)((c.i + (((c.i >> 16) & 1u) + 0x7fffu)) >> 16);
+}
+
+// counter-based splitmix64 -> uniform [0,1). Deterministic in (seed, idx):
+// the forward and both backward kernels regenerate the same dropout mask.
+static inline __device__ float rng_u01(unsigned long long seed,
+                                       unsigned long long idx) {
+  unsigned long long z = seed + idx * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  z ^= z >> 31;
+  return (float)(unsigned int)(z >> 40) * (1.0f / 16777216.0f);
+}
+
+// dropout multiplier for score element (b,h,q,k): 1/(1-p) kept, 0 dropped
+static inline __device__ float drop_mult(unsigned long long seed,
+                                         long long bh, int S, int q, int k,
+                                         float p_drop, float inv_keep) {
+  const unsigned long long idx =
+      ((unsigned long long)bh * (unsigned)S + (unsigned)q) * (unsigned)S + (unsigned)k;
+  return rng_u01(seed, idx) >= p_drop ? inv_keep : 0.f;
+}
+
+// C/D-fragment register r -> row index within the 32-row tile
+static inline __device__ int crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
+
+// stage a [B,S] u8 mask row as 0/1 floats into LDS (count entries)
+static inline __device__ void stage_mask(const unsigned char* mrow, int count,
+                                         float* lds) {
+  for (int i = threadIdx.x; i < count; i += 64)
+    lds[i] = mrow[i] ? 1.f : 0.f;
 }
 
 // XOR-swizzled byte offsets: [rows][64] bf16 (128 B rows) and [rows][128]
@@ -141,12 +188,14 @@ static __device__ void write_tile_bf16(unsigned short* base, long long row_strid
 // ---------------------------------------------------------------------------
 // forward: block = (bh, q-tile), one wave
 // ---------------------------------------------------------------------------
-template <int S>
+template <int S, bool DROP>
 __device__ __forceinline__ void attn_fwd_body(
     const unsigned short* __restrict__ qkv,  // [B,S,3,H]
     unsigned short* __restrict__ out,        // [B,S,H]
     float* __restrict__ lse_out,             // [B,nh,S] base-2 lse
-    int B, int nh) {
+    int B, int nh,
+    const unsigned char* __restrict__ mask,  // [B,S] 1=attend, or null
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -157,10 +206,12 @@ __device__ __forceinline__ void attn_fwd_body(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* Klds = (unsigned short*)smem;            // [S][64] swz
   unsigned short* Vtlds = (unsigned short*)(smem + 16384); // [64][S] swz256
+  float* maskf = (float*)(smem + 16384 + 16384);           // [S] 0/1
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   stage_64<S>(qkv + base + H, 3 * H, Klds);
   stage_64_T<S>(qkv + base + 2 * H, 3 * H, Vtlds);
+  if (mask) stage_mask(mask + (long long)b * S, S, maskf);
   __syncthreads();
 
   const int q0 = qt * 32;
@@ -182,6 +233,14 @@ __device__ __forceinline__ void attn_fwd_body(
       acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
     }
 
+  if (mask) {
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        if (maskf[32 * t + crow(r, hi)] == 0.f) acc[t][r] = -INFINITY;
+  }
+
   const float scale2 = 0.125f * LOG2E;
   float m2 = -1e30f;
 #pragma unroll
@@ -201,10 +260,21 @@ __device__ __forceinline__ void attn_fwd_body(
   const float inv_sum = 1.f / sum;
   if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m2 + log2f(sum);
 
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
 #pragma unroll
   for (int t = 0; t < NT; ++t)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) acc[t][r] *= inv_sum;
+    for (int r = 0; r < 16; ++r) {
+      acc[t][r] *= inv_sum;
+      if (DROP)
+        acc[t][r] *= drop_mult(seed, bh, S, q0 + lo31, 32 * t + crow(r, hi),
+                               p_drop, inv_keep);
+    }
 
   // O = P V: cvt+swap A-fragments (i = q), V^T row B-reads
 #pragma unroll
@@ -228,8 +298,14 @@ __device__ __forceinline__ void attn_fwd_body(
 #define GA_ATTN_FWD_INST(S)                                                   \
   extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_##S(            \
       const unsigned short* qkv, unsigned short* out, float* lse_out, int B,  \
-      int nh) {                                                               \
-    attn_fwd_body<S>(qkv, out, lse_out, B, nh);                               \
+      int nh, const unsigned char* mask) {                                    \
+    attn_fwd_body<S, false>(qkv, out, lse_out, B, nh, mask, nullptr, 0.f);    \
+  }                                                                           \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_drop_##S(       \
+      const unsigned short* qkv, unsigned short* out, float* lse_out, int B,  \
+      int nh, const unsigned char* mask, const unsigned long long* seed,      \
+      float p_drop) {                                                         \
+    attn_fwd_body<S, true>(qkv, out, lse_out, B, nh, mask, seed, p_drop);     \
   }
 GA_ATTN_FWD_INST(32)
 GA_ATTN_FWD_INST(64)
@@ -266,12 +342,14 @@ extern "C" __global__ void k_attn_bwd_d(
 // ---------------------------------------------------------------------------
 // backward part 1: per (bh, q-tile): dQ  (swapped orientation, lanes own q)
 // ---------------------------------------------------------------------------
-template <int S>
+template <int S, bool DROP>
 __device__ __forceinline__ void attn_bwd_q_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
     const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int nh, int publish_d) {
+    unsigned short* __restrict__ dqkv, int B, int nh, int publish_d,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -283,12 +361,14 @@ __device__ __forceinline__ void attn_bwd_q_body(
   unsigned short* Klds = (unsigned short*)smem;             // [S][64] swz
   unsigned short* Vlds = (unsigned short*)(smem + 16384);   // [S][64] swz
   unsigned short* Ktlds = (unsigned short*)(smem + 32768);  // [64][S] swz256
+  float* maskf = (float*)(smem + 49152);                    // [S] 0/1
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
   stage_64<S>(qkv + base + H, 3 * H, Klds);
   stage_64<S>(qkv + base + 2 * H, 3 * H, Vlds);
   stage_64_T<S>(qkv + base + H, 3 * H, Ktlds);
+  if (mask) stage_mask(mask + (long long)b * S, S, maskf);
   __syncthreads();
 
   const int q0 = qt * 32;
@@ -340,12 +420,23 @@ __device__ __forceinline__ void attn_bwd_q_body(
     }
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
 #pragma unroll
   for (int t = 0; t < NT; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
-      dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T (col q = lo31)
+      const int k = 32 * t + crow(r, hi);
+      float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+      if (mask && maskf[k] == 0.f) p = 0.f;
+      float dp = dacc[t][r];
+      if (DROP)
+        dp *= drop_mult(seed, bh, S, q0 + lo31, k, p_drop, inv_keep);
+      dacc[t][r] = scale * p * (dp - D_q);  // dS^T (col q = lo31)
     }
 
   // dQ = dS K: A-frags (i = q) from dacc via cvt+swap, B = K^T rows
@@ -371,8 +462,19 @@ __device__ __forceinline__ void attn_bwd_q_body(
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_##S(          \
       const unsigned short* qkv, const unsigned short* out,                   \
       const unsigned short* dout, const float* lse_in, float* Dtab,           \
-      unsigned short* dqkv, int B, int nh, int publish_d) {                   \
-    attn_bwd_q_body<S>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh, publish_d); \
+      unsigned short* dqkv, int B, int nh, int publish_d,                     \
+      const unsigned char* mask) {                                            \
+    attn_bwd_q_body<S, false>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh,      \
+                              publish_d, mask, nullptr, 0.f);                 \
+  }                                                                           \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_drop_##S(     \
+      const unsigned short* qkv, const unsigned short* out,                   \
+      const unsigned short* dout, const float* lse_in, float* Dtab,           \
+      unsigned short* dqkv, int B, int nh, int publish_d,                     \
+      const unsigned char* mask, const unsigned long long* seed,              \
+      float p_drop) {                                                         \
+    attn_bwd_q_body<S, true>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh,       \
+                             publish_d, mask, seed, p_drop);                  \
   }
 GA_ATTN_BWDQ_INST(32)
 GA_ATTN_BWDQ_INST(64)
@@ -384,11 +486,13 @@ GA_ATTN_BWDQ_INST(128)
 // the q-rows orientation -- A = Q/dO row frags, B = this tile's K/V row
 // frags straight from global -- then cvt+swap gives A[i=k][k'=q]).
 // ---------------------------------------------------------------------------
-template <int S>
+template <int S, bool DROP>
 __device__ __forceinline__ void attn_bwd_kv_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, const float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int nh) {
+    unsigned short* __restrict__ dqkv, int B, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   constexpr int NT = S / 32;
   const int bh = blockIdx.x / NT, kt = blockIdx.x % NT;
@@ -426,6 +530,16 @@ __device__ __forceinline__ void attn_bwd_kv_body(
   }
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  // this block's k is per-lane: one mask read / dropout column for all q
+  const int k_lane = k0 + lo31;
+  const float kvalid =
+      mask ? (mask[(long long)b * S + k_lane] ? 1.f : 0.f) : 1.f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
   f32x16 p_qt[NT], ds_qt[NT];
 #pragma unroll
   for (int t = 0; t < NT; ++t) {
@@ -443,9 +557,15 @@ __device__ __forceinline__ void attn_bwd_kv_body(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-      const float p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
-      p_qt[t][r] = p;
-      ds_qt[t][r] = scale * p * (dpacc[r] - dtab[q]);
+      const float p =
+          kvalid * __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+      float m = 1.f, dp = dpacc[r];
+      if (DROP) {
+        m = drop_mult(seed, bh, S, q, k_lane, p_drop, inv_keep);
+        dp *= m;
+      }
+      p_qt[t][r] = p * m;  // P_drop rows feed dV
+      ds_qt[t][r] = scale * p * (dp - dtab[q]);
     }
   }
 
@@ -480,8 +600,17 @@ __device__ __forceinline__ void attn_bwd_kv_body(
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_##S(         \
       const unsigned short* qkv, const unsigned short* dout,                  \
       const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,    \
-      int nh) {                                                               \
-    attn_bwd_kv_body<S>(qkv, dout, lse_in, Dtab, dqkv, B, nh);                \
+      int nh, const unsigned char* mask) {                                    \
+    attn_bwd_kv_body<S, false>(qkv, dout, lse_in, Dtab, dqkv, B, nh, mask,    \
+                               nullptr, 0.f);                                 \
+  }                                                                           \
+  extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_drop_##S(    \
+      const unsigned short* qkv, const unsigned short* dout,                  \
+      const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,    \
+      int nh, const unsigned char* mask, const unsigned long long* seed,      \
+      float p_drop) {                                                         \
+    attn_bwd_kv_body<S, true>(qkv, dout, lse_in, Dtab, dqkv, B, nh, mask,     \
+                              seed, p_drop);                                  \
   }
 GA_ATTN_BWDKV_INST(32)
 GA_ATTN_BWDKV_INST(64)
@@ -526,9 +655,12 @@ static __device__ __forceinline__ void stage_c64_T(const unsigned short* g,
   }
 }
 
-extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
+template <bool DROP>
+__device__ __forceinline__ void attn_fwd_big_body(
     const unsigned short* __restrict__ qkv, unsigned short* __restrict__ out,
-    float* __restrict__ lse_out, int B, int S, int nh) {
+    float* __restrict__ lse_out, int B, int S, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   const int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -539,6 +671,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   unsigned short* Klds = (unsigned short*)smem;            // [64][64] swz
   unsigned short* Vtlds = (unsigned short*)(smem + 8192);  // [64][64] swz
+  float* maskf = (float*)(smem + 16384);                   // [64] chunk 0/1
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const int q0 = qt * 32;
@@ -549,6 +682,12 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
     qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
 
   const float scale2 = 0.125f * LOG2E;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
   float m_old = -1e30f, sum = 0.f;
   f32x16 oc[2];
   oc[0] = (f32x16)(0.f);
@@ -558,6 +697,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
     __syncthreads();
     stage_64<64>(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
     stage_c64_T(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vtlds);
+    if (mask) stage_mask(mask + (long long)b * S + c0, 64, maskf);
     __syncthreads();
 
     // S^T chunk = K_c Q^T (cols = q, rows = k within the chunk)
@@ -570,6 +710,13 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
         bf16x8 a = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
         acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
       }
+    }
+    if (mask) {
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          if (maskf[32 * t + crow(r, hi)] == 0.f) acc[t][r] = -INFINITY;
     }
     float mc = -1e30f;
 #pragma unroll
@@ -590,6 +737,17 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
     csum += __shfl_xor(csum, 32, 64);
     sum = sum * sf + csum;
     m_old = m_new;
+
+    if (DROP) {
+      // dropout is linear: scale the unnormalized e-values per chunk, the
+      // final 1/sum normalization distributes over the accumulated O
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          acc[t][r] *= drop_mult(seed, bh, S, q0 + lo31,
+                                 c0 + 32 * t + crow(r, hi), p_drop, inv_keep);
+    }
 
     // O rescale by sf (per q = O-tile ROW) then O += P_c V_c
 #pragma unroll
@@ -626,11 +784,25 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
   }
 }
 
-extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
+extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
+    const unsigned short* qkv, unsigned short* out, float* lse_out,
+    int B, int S, int nh, const unsigned char* mask) {
+  attn_fwd_big_body<false>(qkv, out, lse_out, B, S, nh, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big_drop(
+    const unsigned short* qkv, unsigned short* out, float* lse_out,
+    int B, int S, int nh, const unsigned char* mask,
+    const unsigned long long* seed, float p_drop) {
+  attn_fwd_big_body<true>(qkv, out, lse_out, B, S, nh, mask, seed, p_drop);
+}
+
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_q_big_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
     const unsigned short* __restrict__ dout, const float* __restrict__ lse_in,
     float* __restrict__ Dtab, unsigned short* __restrict__ dqkv, int B, int S,
-    int nh, int publish_d) {
+    int nh, int publish_d, const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   const int NT = S / 32;
   const int bh = blockIdx.x / NT, qt = blockIdx.x % NT;
@@ -642,6 +814,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
   unsigned short* Klds = (unsigned short*)smem;             // [64][64] swz
   unsigned short* Vlds = (unsigned short*)(smem + 8192);    // [64][64] swz
   unsigned short* Ktlds = (unsigned short*)(smem + 16384);  // [64][64] swz
+  float* maskf = (float*)(smem + 24576);                    // [64] chunk 0/1
 
   const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
   const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
@@ -674,6 +847,12 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
   }
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
   f32x16 dq[2];
   dq[0] = (f32x16)(0.f);
   dq[1] = (f32x16)(0.f);
@@ -683,6 +862,7 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
     stage_64<64>(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
     stage_64<64>(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vlds);
     stage_c64_T(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Ktlds);
+    if (mask) stage_mask(mask + (long long)b * S + c0, 64, maskf);
     __syncthreads();
 
     f32x16 acc[2], dacc[2];
@@ -699,8 +879,13 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
       }
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
-        dacc[t][r] = scale * p * (dacc[t][r] - D_q);  // dS^T chunk
+        const int kl = 32 * t + crow(r, hi);
+        float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+        if (mask && maskf[kl] == 0.f) p = 0.f;
+        float dp = dacc[t][r];
+        if (DROP)
+          dp *= drop_mult(seed, bh, S, q0 + lo31, c0 + kl, p_drop, inv_keep);
+        dacc[t][r] = scale * p * (dp - D_q);  // dS^T chunk
       }
     }
 #pragma unroll
@@ -723,10 +908,30 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
                     3 * H, hi, dq[dt]);
 }
 
-extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int S, int nh, int publish_d,
+    const unsigned char* mask) {
+  attn_bwd_q_big_body<false>(qkv, out, dout, lse_in, Dtab, dqkv, B, S, nh,
+                             publish_d, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big_drop(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int S, int nh, int publish_d,
+    const unsigned char* mask, const unsigned long long* seed, float p_drop) {
+  attn_bwd_q_big_body<true>(qkv, out, dout, lse_in, Dtab, dqkv, B, S, nh,
+                            publish_d, mask, seed, p_drop);
+}
+
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_kv_big_body(
     const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse_in, const float* __restrict__ Dtab,
-    unsigned short* __restrict__ dqkv, int B, int S, int nh) {
+    unsigned short* __restrict__ dqkv, int B, int S, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
   const int H = nh * ATTN_D;
   const int NT = S / 32;
   const int bh = blockIdx.x / NT, kt = blockIdx.x % NT;
@@ -756,6 +961,15 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
   }
 
   const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  const int k_lane = k0 + lo31;
+  const float kvalid =
+      mask ? (mask[(long long)b * S + k_lane] ? 1.f : 0.f) : 1.f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
   f32x16 dv_[2], dk_[2];
   dv_[0] = (f32x16)(0.f); dv_[1] = (f32x16)(0.f);
   dk_[0] = (f32x16)(0.f); dk_[1] = (f32x16)(0.f);
@@ -786,9 +1000,15 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        const float p = __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
-        p_qt[t][r] = p;
-        ds_qt[t][r] = scale * p * (dpacc[r] - dtab[q]);
+        const float p =
+            kvalid * __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+        float m = 1.f, dp = dpacc[r];
+        if (DROP) {
+          m = drop_mult(seed, bh, S, c0 + q, k_lane, p_drop, inv_keep);
+          dp *= m;
+        }
+        p_qt[t][r] = p * m;
+        ds_qt[t][r] = scale * p * (dp - dtab[q]);
       }
     }
 #pragma unroll
@@ -817,4 +1037,19 @@ extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
     write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
                     3 * H, hi, dk_[dt]);
   }
+}
+
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(
+    const unsigned short* qkv, const unsigned short* dout, const float* lse_in,
+    const float* Dtab, unsigned short* dqkv, int B, int S, int nh,
+    const unsigned char* mask) {
+  attn_bwd_kv_big_body<false>(qkv, dout, lse_in, Dtab, dqkv, B, S, nh, mask,
+                              nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big_drop(
+    const unsigned short* qkv, const unsigned short* dout, const float* lse_in,
+    const float* Dtab, unsigned short* dqkv, int B, int S, int nh,
+    const unsigned char* mask, const unsigned long long* seed, float p_drop) {
+  attn_bwd_kv_big_body<true>(qkv, dout, lse_in, Dtab, dqkv, B, S, nh, mask,
+                             seed, p_drop);
 }

@@ -74,23 +74,51 @@ extern "C" __global__ void k_emb3_fwd(const long long*, const long long*,
                                       const unsigned short*, unsigned short*,
                                       int, int);
 #define GA_DECL_ATTN(S)                                                        \
-  extern "C" __global__ void k_attn_fwd_##S(const unsigned short*,             \
-                                            unsigned short*, float*, int, int);\
+  extern "C" __global__ void k_attn_fwd_##S(                                   \
+      const unsigned short*, unsigned short*, float*, int, int,                \
+      const unsigned char*);                                                   \
+  extern "C" __global__ void k_attn_fwd_drop_##S(                              \
+      const unsigned short*, unsigned short*, float*, int, int,                \
+      const unsigned char*, const unsigned long long*, float);                 \
   extern "C" __global__ void k_attn_bwd_q_##S(                                 \
       const unsigned short*, const unsigned short*, const unsigned short*,     \
-      const float*, float*, unsigned short*, int, int, int);                   \
+      const float*, float*, unsigned short*, int, int, int,                    \
+      const unsigned char*);                                                   \
+  extern "C" __global__ void k_attn_bwd_q_drop_##S(                            \
+      const unsigned short*, const unsigned short*, const unsigned short*,     \
+      const float*, float*, unsigned short*, int, int, int,                    \
+      const unsigned char*, const unsigned long long*, float);                 \
   extern "C" __global__ void k_attn_bwd_kv_##S(                                \
       const unsigned short*, const unsigned short*, const float*,              \
-      const float*, unsigned short*, int, int);
+      const float*, unsigned short*, int, int, const unsigned char*);          \
+  extern "C" __global__ void k_attn_bwd_kv_drop_##S(                           \
+      const unsigned short*, const unsigned short*, const float*,              \
+      const float*, unsigned short*, int, int, const unsigned char*,           \
+      const unsigned long long*, float);
 GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
 extern "C" __global__ void k_attn_fwd_big(const unsigned short*, unsigned short*,
-                                           float*, int, int, int);
+                                           float*, int, int, int,
+                                           const unsigned char*);
+extern "C" __global__ void k_attn_fwd_big_drop(const unsigned short*, unsigned short*,
+                                               float*, int, int, int,
+                                               const unsigned char*,
+                                               const unsigned long long*, float);
 extern "C" __global__ void k_attn_bwd_q_big(const unsigned short*, const unsigned short*,
                                             const unsigned short*, const float*,
-                                            float*, unsigned short*, int, int, int, int);
+                                            float*, unsigned short*, int, int, int, int,
+                                            const unsigned char*);
+extern "C" __global__ void k_attn_bwd_q_big_drop(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, float*, unsigned short*, int, int, int, int,
+    const unsigned char*, const unsigned long long*, float);
 extern "C" __global__ void k_attn_bwd_kv_big(const unsigned short*, const unsigned short*,
                                              const float*, const float*,
-                                             unsigned short*, int, int, int);
+                                             unsigned short*, int, int, int,
+                                             const unsigned char*);
+extern "C" __global__ void k_attn_bwd_kv_big_drop(
+    const unsigned short*, const unsigned short*, const float*, const float*,
+    unsigned short*, int, int, int, const unsigned char*,
+    const unsigned long long*, float);
 extern "C" __global__ void k_attn_bwd_d(const unsigned short*, const unsigned short*,
                                         float*, int, int, int);
 #define WG_MAX_G 24
@@ -348,7 +376,28 @@ void colreduce_acc(at::Tensor partials, at::Tensor dest0,
 // captured graph the extra event edges + the standalone D kernel cost far
 // more than the ~4 us/layer of overlap they buy. The sequential launch
 // below (bwd_q computes and publishes D itself) is the fast path.
-std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
+static const unsigned char* mask_ptr_of(const c10::optional<at::Tensor>& mask,
+                                        int B, int S) {
+  if (!mask.has_value()) return nullptr;
+  const auto& m = mask.value();
+  TORCH_CHECK(m.is_cuda() && m.is_contiguous() && m.scalar_type() == at::kByte,
+              "attention mask must be contiguous u8 on device");
+  TORCH_CHECK(m.dim() == 2 && m.size(0) == B && m.size(1) == S,
+              "attention mask must be [B,S]");
+  return (const unsigned char*)m.data_ptr();
+}
+
+static const unsigned long long* seed_ptr_of(const c10::optional<at::Tensor>& seed) {
+  if (!seed.has_value()) return nullptr;
+  const auto& s = seed.value();
+  TORCH_CHECK(s.is_cuda() && s.scalar_type() == at::kLong && s.numel() == 1,
+              "dropout seed must be a device int64 scalar");
+  return (const unsigned long long*)s.data_ptr();
+}
+
+std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh,
+                                 c10::optional<at::Tensor> mask,
+                                 c10::optional<at::Tensor> seed, double p_drop) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() &&
               qkv.scalar_type() == at::kBFloat16, "qkv must be contiguous bf16");
   TORCH_CHECK(qkv.dim() == 4 && qkv.size(2) == 3, "qkv must be [B,S,3,H]");
@@ -356,77 +405,157 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh) {
   TORCH_CHECK(H == nh * 64, "head_dim must be 64");
   TORCH_CHECK((S <= 128 && S % 32 == 0) || S % 64 == 0,
               "attn kernel needs S%32==0 and (S<=128 or S%64==0)");
+  const bool drop = p_drop > 0.0;
+  TORCH_CHECK(!drop || seed.has_value(), "dropout needs a device seed scalar");
+  const unsigned char* mp = mask_ptr_of(mask, B, S);
+  const unsigned long long* sp = seed_ptr_of(seed);
+  const float pd = (float)p_drop;
   auto out = at::empty({B, S, H}, qkv.options());
   auto lse = at::empty({B, nh, S}, qkv.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
   if (S > 128) {
-    // chunked online-softmax variant: [64][64] K + V^T LDS panels
-    hipLaunchKernelGGL(k_attn_fwd_big, dim3(B * (int)nh * NT), dim3(64),
-                       16384, stream,
-                       (const unsigned short*)qkv.data_ptr(),
-                       (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
-                       B, S, (int)nh);
+    // chunked online-softmax variant: [64][64] K + V^T LDS panels (+mask)
+    if (drop)
+      hipLaunchKernelGGL(k_attn_fwd_big_drop, dim3(B * (int)nh * NT), dim3(64),
+                         16384 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                         B, S, (int)nh, mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_fwd_big, dim3(B * (int)nh * NT), dim3(64),
+                         16384 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                         B, S, (int)nh, mp);
     TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd_big launch failed");
     return {out, lse};
   }
-  const size_t lds = 16384 + 64 * 256;  // K + V^T panels (fixed offsets)
-  void (*fk)(const unsigned short*, unsigned short*, float*, int, int) =
-      S == 32 ? k_attn_fwd_32 : S == 64 ? k_attn_fwd_64
-      : S == 96 ? k_attn_fwd_96 : k_attn_fwd_128;
-  hipLaunchKernelGGL(fk, dim3(B * (int)nh * NT), dim3(64), lds, stream,
-                     (const unsigned short*)qkv.data_ptr(),
-                     (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
-                     B, (int)nh);
+  const size_t lds = 16384 + 64 * 256 + 512;  // K + V^T panels + mask table
+  if (drop) {
+    void (*fk)(const unsigned short*, unsigned short*, float*, int, int,
+               const unsigned char*, const unsigned long long*, float) =
+        S == 32 ? k_attn_fwd_drop_32 : S == 64 ? k_attn_fwd_drop_64
+        : S == 96 ? k_attn_fwd_drop_96 : k_attn_fwd_drop_128;
+    hipLaunchKernelGGL(fk, dim3(B * (int)nh * NT), dim3(64), lds, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                       B, (int)nh, mp, sp, pd);
+  } else {
+    void (*fk)(const unsigned short*, unsigned short*, float*, int, int,
+               const unsigned char*) =
+        S == 32 ? k_attn_fwd_32 : S == 64 ? k_attn_fwd_64
+        : S == 96 ? k_attn_fwd_96 : k_attn_fwd_128;
+    hipLaunchKernelGGL(fk, dim3(B * (int)nh * NT), dim3(64), lds, stream,
+                       (const unsigned short*)qkv.data_ptr(),
+                       (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                       B, (int)nh, mp);
+  }
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd launch failed");
   return {out, lse};
 }
 
 at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
-                    at::Tensor lse, int64_t nh) {
+                    at::Tensor lse, int64_t nh,
+                    c10::optional<at::Tensor> mask,
+                    c10::optional<at::Tensor> seed, double p_drop) {
   const int B = (int)qkv.size(0), S = (int)qkv.size(1), H = (int)qkv.size(3);
   TORCH_CHECK(dout.is_contiguous() && dout.scalar_type() == at::kBFloat16);
+  const bool drop = p_drop > 0.0;
+  TORCH_CHECK(!drop || seed.has_value(), "dropout needs a device seed scalar");
+  const unsigned char* mp = mask_ptr_of(mask, B, S);
+  const unsigned long long* sp = seed_ptr_of(seed);
+  const float pd = (float)p_drop;
   auto dqkv = at::empty_like(qkv);
   auto Dtab = at::empty({B, (long)nh, S}, lse.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
 
   if (S > 128) {
-    hipLaunchKernelGGL(k_attn_bwd_q_big, dim3(B * (int)nh * NT), dim3(64),
-                       24576, stream,
-                       (const unsigned short*)qkv.data_ptr(),
-                       (const unsigned short*)out.data_ptr(),
-                       (const unsigned short*)dout.data_ptr(),
-                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1);
-    hipLaunchKernelGGL(k_attn_bwd_kv_big, dim3(B * (int)nh * NT), dim3(64),
-                       16384 + 512, stream,
-                       (const unsigned short*)qkv.data_ptr(),
-                       (const unsigned short*)dout.data_ptr(),
-                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                       (unsigned short*)dqkv.data_ptr(), B, S, (int)nh);
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_q_big_drop, dim3(B * (int)nh * NT), dim3(64),
+                         24576 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_q_big, dim3(B * (int)nh * NT), dim3(64),
+                         24576 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1, mp);
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_kv_big_drop, dim3(B * (int)nh * NT), dim3(64),
+                         16384 + 512, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_kv_big, dim3(B * (int)nh * NT), dim3(64),
+                         16384 + 512, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, mp);
   } else {
-    const size_t lds_q = 16384 * 2 + 64 * 256;  // K + V + K^T (fixed offsets)
-    void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
-               const float*, float*, unsigned short*, int, int, int) =
-        S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
-        : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
-    hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
-                       (const unsigned short*)qkv.data_ptr(),
-                       (const unsigned short*)out.data_ptr(),
-                       (const unsigned short*)dout.data_ptr(),
-                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                       (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1);
+    const size_t lds_q = 16384 * 3 + 512;  // K + V + K^T + mask table
+    if (drop) {
+      void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
+                 const float*, float*, unsigned short*, int, int, int,
+                 const unsigned char*, const unsigned long long*, float) =
+          S == 32 ? k_attn_bwd_q_drop_32 : S == 64 ? k_attn_bwd_q_drop_64
+          : S == 96 ? k_attn_bwd_q_drop_96 : k_attn_bwd_q_drop_128;
+      hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1,
+                         mp, sp, pd);
+    } else {
+      void (*qk)(const unsigned short*, const unsigned short*, const unsigned short*,
+                 const float*, float*, unsigned short*, int, int, int,
+                 const unsigned char*) =
+          S == 32 ? k_attn_bwd_q_32 : S == 64 ? k_attn_bwd_q_64
+          : S == 96 ? k_attn_bwd_q_96 : k_attn_bwd_q_128;
+      hipLaunchKernelGGL(qk, dim3(B * (int)nh * NT), dim3(64), lds_q, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1, mp);
+    }
     const size_t lds_kv = (size_t)64 * 256 * 2 + 1024;  // dO^T + Q^T + tables
-    void (*kvk)(const unsigned short*, const unsigned short*, const float*,
-                const float*, unsigned short*, int, int) =
-        S == 32 ? k_attn_bwd_kv_32 : S == 64 ? k_attn_bwd_kv_64
-        : S == 96 ? k_attn_bwd_kv_96 : k_attn_bwd_kv_128;
-    hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
-                       (const unsigned short*)qkv.data_ptr(),
-                       (const unsigned short*)dout.data_ptr(),
-                       lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                       (unsigned short*)dqkv.data_ptr(), B, (int)nh);
+    if (drop) {
+      void (*kvk)(const unsigned short*, const unsigned short*, const float*,
+                  const float*, unsigned short*, int, int, const unsigned char*,
+                  const unsigned long long*, float) =
+          S == 32 ? k_attn_bwd_kv_drop_32 : S == 64 ? k_attn_bwd_kv_drop_64
+          : S == 96 ? k_attn_bwd_kv_drop_96 : k_attn_bwd_kv_drop_128;
+      hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh,
+                         mp, sp, pd);
+    } else {
+      void (*kvk)(const unsigned short*, const unsigned short*, const float*,
+                  const float*, unsigned short*, int, int, const unsigned char*) =
+          S == 32 ? k_attn_bwd_kv_32 : S == 64 ? k_attn_bwd_kv_64
+          : S == 96 ? k_attn_bwd_kv_96 : k_attn_bwd_kv_128;
+      hipLaunchKernelGGL(kvk, dim3(B * (int)nh * NT), dim3(64), lds_kv, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, mp);
+    }
   }
   TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_bwd launch failed");
   return dqkv;
@@ -713,10 +842,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           "one launch reducing every pending LN/GELU partial slab into accum");
   mod.def("embgrad_acc", &embgrad_acc,
           "scatter-add embedding grads into the flat fp32 accum slice");
-  mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)");
+  mod.def("attn_fwd", &attn_fwd, "fused MFMA attention fwd (packed qkv)",
+          pybind11::arg("qkv"), pybind11::arg("nh"),
+          pybind11::arg("mask") = c10::nullopt,
+          pybind11::arg("seed") = c10::nullopt, pybind11::arg("p_drop") = 0.0);
   mod.def("cls_head_fwd", &cls_head_fwd, "tanh+classifier+CE forward");
   mod.def("cls_head_bwd", &cls_head_bwd, "fused head backward -> d(pre-tanh)");
   mod.def("wgrad_mfma", &wgrad_mfma,
           "batched MFMA wgrad: accum_g += dy_g^T @ x_g over a tile table");
-  mod.def("attn_bwd", &attn_bwd, "fused MFMA attention bwd -> packed dqkv");
+  mod.def("attn_bwd", &attn_bwd, "fused MFMA attention bwd -> packed dqkv",
+          pybind11::arg("qkv"), pybind11::arg("out"), pybind11::arg("dout"),
+          pybind11::arg("lse"), pybind11::arg("nh"),
+          pybind11::arg("mask") = c10::nullopt,
+          pybind11::arg("seed") = c10::nullopt, pybind11::arg("p_drop") = 0.0);
 }

@@ -514,23 +514,32 @@ class DirectEmbedding(nn.Module):
 class _AttentionFn(torch.autograd.Function):
     """Hand-written MFMA flash attention over the packed QKV projection
     (ops/csrc/attn.hip): no permute copies, no dq/dk/dv zero-fills, packed
-    dqkv gradient. head_dim 64, no mask/dropout; S <= 128 single-pass or
-    any S %% 64 == 0 via the chunked online-softmax variants."""
+    dqkv gradient. head_dim 64; S <= 128 single-pass or any S %% 64 == 0
+    via the chunked online-softmax variants. Optional [B,S] u8 key-padding
+    mask and counter-based dropout (seed = device int64 scalar; both
+    backward kernels regenerate the identical keep mask from it)."""
 
     @staticmethod
-    def forward(ctx, qkv4, nh):
+    def forward(ctx, qkv4, nh, mask8, seed, p_drop):
         hip = require_hip()
-        out, lse = hip.attn_fwd(qkv4, nh)
-        ctx.save_for_backward(qkv4, out, lse)
+        out, lse = hip.attn_fwd(qkv4, nh, mask=mask8, seed=seed, p_drop=p_drop)
+        dummy = qkv4[:0, :0, :0, :0]
+        ctx.save_for_backward(qkv4, out, lse,
+                              mask8 if mask8 is not None else dummy,
+                              seed if seed is not None else dummy)
         ctx.nh = nh
+        ctx.p_drop = p_drop
         return out
 
     @staticmethod
     def backward(ctx, dout):
         hip = require_hip()
-        qkv4, out, lse = ctx.saved_tensors
-        dqkv = hip.attn_bwd(qkv4, out, dout.contiguous(), lse, ctx.nh)
-        return dqkv, None
+        qkv4, out, lse, mask8, seed = ctx.saved_tensors
+        mask8 = mask8 if mask8.numel() else None
+        seed = seed if seed.numel() else None
+        dqkv = hip.attn_bwd(qkv4, out, dout.contiguous(), lse, ctx.nh,
+                            mask=mask8, seed=seed, p_drop=ctx.p_drop)
+        return dqkv, None, None, None, None
 
 
 def fused_attention_supported(S: int, head_dim: int, training_extras: bool) -> bool:
@@ -541,15 +550,21 @@ def fused_attention_supported(S: int, head_dim: int, training_extras: bool) -> b
     if os.environ.get("GA_FUSED_ATTN", "1") == "0":  # A/B switch
         return False
     # S <= 128: fully-resident single-pass kernels; larger S (%64): the
-    # chunked online-softmax variants (k_attn_*_big)
+    # chunked online-softmax variants (k_attn_*_big). Key-padding masks and
+    # prob dropout are handled in-kernel (training_extras kept for
+    # signature compatibility; no longer a disqualifier).
     return (head_dim == 64 and ((S <= 128 and S % 32 == 0) or S % 64 == 0)
-            and not training_extras and hip_available())
+            and hip_available())
 
 
-def fused_attention(qkv: torch.Tensor, nh: int) -> torch.Tensor:
-    """qkv [B,S,3H] bf16 -> O [B,S,H]."""
+def fused_attention(qkv: torch.Tensor, nh: int, mask8=None, seed=None,
+                    p_drop: float = 0.0) -> torch.Tensor:
+    """qkv [B,S,3H] bf16 -> O [B,S,H]. mask8: [B,S] u8 (1=attend) or None;
+    dropout needs a device int64 seed scalar (vary it per step -- write it
+    OUTSIDE graph capture or with a captured RNG op)."""
     B, S, H3 = qkv.shape
-    return _AttentionFn.apply(qkv.view(B, S, 3, H3 // 3), nh)
+    return _AttentionFn.apply(qkv.view(B, S, 3, H3 // 3), nh, mask8, seed,
+                              float(p_drop))
 
 
 class _FFNFn(torch.autograd.Function):

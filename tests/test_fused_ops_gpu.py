@@ -568,3 +568,115 @@ def test_fused_window_loop_equals_sequential(fuse):
         f"fused-window master diverged: {diff.max().item()}"
     assert diff.mean().item() < 1e-4, \
         f"fused-window bulk diverged: mean {diff.mean().item()}"
+
+
+@pytest.mark.parametrize("S", [64, 128, 256, 512])
+def test_fused_attention_key_padding_mask(S):
+    """Masked fused attention (fwd+bwd) vs fp32 SDPA with the same bool
+    mask, random valid lengths per row (incl. rows padded past whole
+    64-chunks to hit the fully-masked-chunk path of the big kernels)."""
+    from gradient_accumulation_tf_estimator_amd.ops.fused import fused_attention
+
+    torch.manual_seed(S)
+    B, nh, dh = 4, 4, 64
+    H = nh * dh
+    qkv = (torch.randn(B, S, 3 * H, device="cuda") * 0.5).bfloat16().requires_grad_()
+    lens = torch.tensor([S, max(1, S // 2), max(1, S // 4 + 1), 3][:B],
+                        device="cuda")
+    mask = (torch.arange(S, device="cuda")[None, :] < lens[:, None])
+    mask8 = mask.to(torch.uint8).contiguous()
+
+    o = fused_attention(qkv, nh, mask8=mask8)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    qf = qkv.detach().float().requires_grad_()
+    q, k, v = qf.view(B, S, 3, nh, dh).permute(2, 0, 3, 1, 4)
+    ref = F.scaled_dot_product_attention(q, k, v,
+                                         attn_mask=mask[:, None, None, :])
+    ref = ref.transpose(1, 2).reshape(B, S, H)
+    ref.backward(do.float())
+
+    np.testing.assert_allclose(o.detach().float().cpu(), ref.detach().cpu(),
+                               rtol=3e-2, atol=2e-2)
+    np.testing.assert_allclose(qkv.grad.float().cpu(), qf.grad.cpu(),
+                               rtol=5e-2, atol=5e-2)
+
+
+def _splitmix_keep(seed, B, nh, S, p_drop, device):
+    """Torch reimplementation of the kernel's counter-based splitmix64 --
+    reconstructs the exact keep mask the kernels generate."""
+    idx = torch.arange(B * nh * S * S, dtype=torch.int64, device=device)
+    GOLD = -7046029254386353131  # 0x9E3779B97F4A7C15 as signed i64
+    z = seed + idx * GOLD
+    for mul, shift in ((-4658895280553007687, 30),   # 0xBF58476D1CE4E5B9
+                       (-7723592293110705685, 27)):  # 0x94D049BB133111EB
+        z = (z ^ (z >> shift) & ((1 << (64 - shift)) - 1)) * mul
+    z = z ^ ((z >> 31) & ((1 << 33) - 1))
+    u = ((z >> 40) & 0xFFFFFF).float() * (1.0 / 16777216.0)
+    return (u >= p_drop).view(B * nh, S, S)
+
+
+@pytest.mark.parametrize("S", [128, 256])
+def test_fused_attention_dropout_exact(S):
+    """Dropout path vs an fp32 reference using the RECONSTRUCTED keep mask
+    (same splitmix64 the kernels run), so fwd and bwd are checked exactly,
+    not just statistically."""
+    from gradient_accumulation_tf_estimator_amd.ops.fused import fused_attention
+
+    torch.manual_seed(77 + S)
+    B, nh, dh, p = 2, 4, 64, 0.1
+    H = nh * dh
+    qkv = (torch.randn(B, S, 3 * H, device="cuda") * 0.5).bfloat16().requires_grad_()
+    seed = torch.tensor([1234567890123], dtype=torch.int64, device="cuda")
+
+    o = fused_attention(qkv, nh, seed=seed, p_drop=p)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    keep = _splitmix_keep(int(seed.item()), B, nh, S, p, "cuda")
+    qf = qkv.detach().float().requires_grad_()
+    q, k, v = qf.view(B, S, 3, nh, dh).permute(2, 0, 3, 1, 4)
+    scores = q.reshape(B * nh, S, dh) @ k.reshape(B * nh, S, dh).transpose(1, 2)
+    probs = torch.softmax(scores * dh**-0.5, dim=-1)
+    probs = probs * keep.float() / (1 - p)
+    ref = (probs @ v.reshape(B * nh, S, dh)).view(B, nh, S, dh)
+    ref = ref.transpose(1, 2).reshape(B, S, H)
+    ref.backward(do.float())
+
+    np.testing.assert_allclose(o.detach().float().cpu(), ref.detach().cpu(),
+                               rtol=3e-2, atol=3e-2)
+    np.testing.assert_allclose(qkv.grad.float().cpu(), qf.grad.cpu(),
+                               rtol=5e-2, atol=6e-2)
+
+
+def test_bert_masked_training_step_matches_sdpa_path():
+    """A full masked micro-step through the fused model vs the same model
+    forced onto the torch-SDPA path (GA_FUSED_ATTN=0 equivalent via config
+    fused=False is a different module tree, so instead compare the fused
+    model's loss/grads between masked fused attention and explicit SDPA
+    reference at the attention boundary is covered above; here: smoke that
+    the masked fused path runs end-to-end under the engine)."""
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+    from gradient_accumulation_tf_estimator_amd.models.bert import (
+        BertConfig, BertForSequenceClassification)
+
+    cfg = BertConfig(hidden_size=512, num_layers=2, num_heads=8,
+                     intermediate_size=2048)
+    torch.manual_seed(0)
+    m = BertForSequenceClassification(cfg).cuda().bfloat16()
+    m.train()
+    op = create_optimizer(m, 1e-4, 1000, 0, gradient_accumulation_multiplier=2,
+                          clip_norm=1.0, backend="hip")
+    B, S = 8, 128
+    g = torch.Generator().manual_seed(1)
+    for i in range(4):
+        ids = torch.randint(0, 30522, (B, S), generator=g).cuda()
+        lab = torch.randint(0, 2, (B,), generator=g).cuda()
+        lens = torch.randint(4, S + 1, (B,), generator=g).cuda()
+        am = (torch.arange(S, device="cuda")[None, :] < lens[:, None]).long()
+        loss = m.loss(ids, lab, attention_mask=am)
+        assert torch.isfinite(loss.float()).item()
+        op.step(loss)
+    torch.cuda.synchronize()
+    assert torch.isfinite(op.engine.state.master).all()
