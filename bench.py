@@ -60,6 +60,9 @@ def parse_args():
                    help="issue the micro-step wgrads as one grouped hipBLASLt launch")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")
+    p.add_argument("--masked", default="off", choices=["on", "off"],
+                   help="random key-padding masks (valid lengths S/4..S) -- "
+                        "measures the masked fused-attention path")
     p.add_argument("--sdpa", default="auto",
                    choices=["auto", "flash", "efficient", "math"],
                    help="force a scaled_dot_product_attention backend")
@@ -141,12 +144,18 @@ def main():
     gen = torch.Generator(device="cpu").manual_seed(99 + rank)
     pool_ids = torch.randint(0, V, (POOL, B, S), generator=gen).to(device)
     pool_lab = torch.randint(0, cfg.num_labels, (POOL, B), generator=gen).to(device)
+    pool_msk = None
+    if args.masked == "on":
+        lens = torch.randint(S // 4, S + 1, (POOL, B), generator=gen)
+        pool_msk = (torch.arange(S)[None, None, :] < lens[:, :, None]) \
+            .to(torch.uint8).to(device)
 
     inv_world = 1.0 / world
 
     def eager_micro_step(i):
         ids, lab = pool_ids[i % POOL], pool_lab[i % POOL]
-        loss = model.loss(ids, lab)
+        msk = None if pool_msk is None else pool_msk[i % POOL]
+        loss = model.loss(ids, lab, attention_mask=msk)
         if world > 1:
             loss = loss * inv_world
         loss.backward()
@@ -172,7 +181,8 @@ def main():
     graphed = None
     if use_graphs and fuse > 1:
         try:
-            graphed = capture_fused(model, engine, pool_ids, pool_lab, world, fuse)
+            graphed = capture_fused(model, engine, pool_ids, pool_lab, world, fuse,
+                                    pool_msk=pool_msk)
         except Exception as e:
             print(f"[bench] fused-window capture failed, falling back: {e}",
                   file=sys.stderr)
@@ -180,7 +190,8 @@ def main():
     if use_graphs and graphed is None:
         try:
             graphed = capture_graphs(model, engine, pool_ids, pool_lab, inv_world,
-                                     world, window=args.graphs == "window")
+                                     world, window=args.graphs == "window",
+                                     pool_msk=pool_msk)
         except Exception as e:
             print(f"[bench] hipGraph capture failed, falling back to eager: {e}",
                   file=sys.stderr)
@@ -250,6 +261,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "graphs": graphed is not None,
                 "window_fuse": fuse,
+                "masked": args.masked == "on",
                 "peak_hbm_gb": round(peak_hbm_gb, 3),
             },
         }
@@ -259,23 +271,30 @@ def main():
         dist.destroy_process_group()
 
 
-def capture_fused(model, engine, pool_ids, pool_lab, world, fuse):
+def capture_fused(model, engine, pool_ids, pool_lab, world, fuse, pool_msk=None):
     """Window-fused hipGraph loop: each replay computes `fuse` micro-steps
     as one forward/backward over the concatenated [fuse*B, S] batch
     (engine/graphs.py FusedWindowLoop; exact by linearity)."""
     from gradient_accumulation_tf_estimator_amd.engine.graphs import FusedWindowLoop
 
     POOL, B, S = pool_ids.shape
-    pool_packed = torch.cat([pool_ids, pool_lab[:, :, None]], dim=2).contiguous()
+    cols = [pool_ids, pool_lab[:, :, None]]
+    if pool_msk is not None:
+        cols.append(pool_msk.long())
+    pool_packed = torch.cat(cols, dim=2).contiguous()
+    W = pool_packed.shape[2]
     # block b consumes pool micro-batches b*fuse .. b*fuse+fuse-1 (mod POOL)
     rolled = torch.stack(
-        [pool_packed[(torch.arange(fuse) + b) % POOL].reshape(fuse * B, S + 1)
+        [pool_packed[(torch.arange(fuse) + b) % POOL].reshape(fuse * B, W)
          for b in range(POOL)])
     static = rolled[0].clone()
     static_ids = static[:, :S]
     static_lab = static[:, S]
-    loop = FusedWindowLoop(engine, lambda: model.loss(static_ids, static_lab),
-                           n_micro=fuse, world=world)
+    static_msk = static[:, S + 1 :] if pool_msk is not None else None
+    loop = FusedWindowLoop(
+        engine,
+        lambda: model.loss(static_ids, static_lab, attention_mask=static_msk),
+        n_micro=fuse, world=world)
 
     def run(i):
         static.copy_(rolled[(i // fuse) % POOL])
@@ -285,25 +304,30 @@ def capture_fused(model, engine, pool_ids, pool_lab, world, fuse):
 
 
 def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world,
-                   window=False):
+                   window=False, pool_msk=None):
     """Wrap the framework's hipGraph-captured micro-batch loop
     (engine/graphs.py) with the bench's static input buffers."""
     from gradient_accumulation_tf_estimator_amd.engine.graphs import GraphedTrainLoop
 
-    # ids and labels travel as ONE packed [B, S+1] buffer so each step pays a
+    # ids, labels (and mask) travel as ONE packed buffer so each step pays a
     # single H2D-free device copy (two small copyBuffers cost ~9 us/step)
     B, S = pool_ids.shape[1], pool_ids.shape[2]
     POOL = pool_ids.shape[0]
-    pool_packed = torch.cat([pool_ids, pool_lab[:, :, None]], dim=2).contiguous()
+    cols = [pool_ids, pool_lab[:, :, None]]
+    if pool_msk is not None:
+        cols.append(pool_msk.long())
+    pool_packed = torch.cat(cols, dim=2).contiguous()
     K = engine.K
     nslots = K if (window and world == 1 and K > 1) else 1
     slots = [pool_packed[0].clone() for _ in range(nslots)]
 
+    def loss_of(slot):
+        msk = slot[:, S + 1 :] if pool_msk is not None else None
+        return model.loss(slot[:, :S], slot[:, S], attention_mask=msk)
+
     if nslots > 1:
-        loop = GraphedTrainLoop(
-            engine,
-            lambda k: model.loss(slots[k][:, :S], slots[k][:, S]),
-            world=world, window=True)
+        loop = GraphedTrainLoop(engine, lambda k: loss_of(slots[k]),
+                                world=world, window=True)
 
         def run(i):
             pos = i % K
@@ -318,10 +342,7 @@ def capture_graphs(model, engine, pool_ids, pool_lab, inv_world, world,
         return run
 
     static_packed = slots[0]
-    static_ids = static_packed[:, :S]
-    static_lab = static_packed[:, S]
-    loop = GraphedTrainLoop(engine, lambda: model.loss(static_ids, static_lab),
-                            world=world)
+    loop = GraphedTrainLoop(engine, lambda: loss_of(static_packed), world=world)
 
     def run(i):
         static_packed.copy_(pool_packed[i % POOL])
