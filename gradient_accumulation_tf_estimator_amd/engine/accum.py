@@ -65,13 +65,21 @@ class AccumEngine:
         strict_reference_semantics: bool = False,
         process_group=None,
         allreduce_bucket_mb: int = 64,
+        shard_apply: bool = True,
         backend: str = "auto",
         direct_names=(),
     ):
         if gradient_accumulation_multiplier < 1:
             raise ValueError("gradient_accumulation_multiplier must be >= 1")
+        self.group = process_group
+        init_world = self.world_size
+        # sharded DP boundary (reduce-scatter + 1/W apply + all-gather)
+        # needs the flat total to split into 64-elem-aligned equal shards
+        pad_to = 64 * init_world if (shard_apply and init_world > 1) else 64
         self.state = FlatState(list(named_params), exclude_from_weight_decay,
-                               direct_names)
+                               direct_names, pad_total_to=pad_to)
+        self.shard_apply = bool(shard_apply) and init_world > 1 and \
+            self.state.layout.total % (64 * init_world) == 0
         self.K = int(gradient_accumulation_multiplier)
         self.init_lr = float(init_lr)
         self.num_train_steps = int(num_train_steps)
@@ -268,6 +276,86 @@ class AccumEngine:
             )
         self.apply_count += 1
 
+    def boundary_apply(self, lr: Optional[float] = None) -> None:
+        """The window-boundary work: sum gradients across ranks, then the
+        fused normalize/clip/Adam apply.
+
+        world == 1 (or ``shard_apply`` off): bucketed all-reduce + full
+        apply on every rank (the reference's replica-redundant apply,
+        04:68-71, moved to the boundary by linearity).
+
+        world > 1 with ``shard_apply`` (default): ZeRO-style boundary --
+        reduce-scatter the accum buffer (each rank receives the SUM of its
+        1/W shard; comm (W-1)/W*N vs all-reduce's 2(W-1)/W*N), per-rank
+        shard sqnorm + scalar all-reduce for the global clip norm, fused
+        apply on the owned shard only (1/W the apply time), then all-gather
+        of the updated flat model params. Master/m/v stay sharded-valid;
+        ``state_dict`` gathers them (collective -- call on every rank).
+        """
+        if self.world_size <= 1 or not self.shard_apply:
+            self._allreduce_accum()
+            self.apply(lr)
+            return
+        self._sharded_apply(self.lr_at(self.global_step) if lr is None else lr)
+
+    def _sharded_apply(self, lr: float) -> None:
+        import torch.distributed as dist
+
+        st = self.state
+        self.last_lr = lr
+        eff = self._eff_lr(lr)
+        W = self.world_size
+        r = dist.get_rank(self.group) if self.group is not None else dist.get_rank()
+        n = st.accum.numel()
+        sh = n // W
+        lo, hi = r * sh, (r + 1) * sh
+        own = st.accum[lo:hi]
+        backend = str(dist.get_backend(self.group) if self.group is not None
+                      else dist.get_backend())
+        if "nccl" in backend:
+            # in-place RCCL reduce-scatter: recv shard aliases the send buffer
+            dist.reduce_scatter_tensor(own, st.accum, group=self.group)
+        else:
+            # gloo (CPU tests) has no reduce-scatter: all-reduce, then run
+            # the identical sharded apply on the owned shard
+            dist.all_reduce(st.accum, group=self.group)
+        if self.clip_norm is not None:
+            if self._hip is not None:
+                self._hip.sqnorm(own, self._sqnorm_dev)
+            else:
+                self._sqnorm_dev[0] = (own * own).sum()
+            dist.all_reduce(self._sqnorm_dev, group=self.group)
+        inv_k = 1.0 / self.K
+        b_rel = min(max(st.decay_boundary - lo, 0), sh)
+        model = None if st.master is st.model else st.model
+        if self._hip is not None:
+            self._lr_dev.fill_(eff)
+            self._hip.fused_apply(
+                own, st.m[lo:hi], st.v[lo:hi], st.master[lo:hi],
+                model[lo:hi] if model is not None else st.master[lo:hi],
+                model is not None,
+                self._lr_dev, self._sqnorm_dev, b_rel, inv_k,
+                -1.0 if self.clip_norm is None else float(self.clip_norm),
+                self.weight_decay, self.beta1, self.beta2, self.eps,
+                skip_norm=True)
+        else:
+            eager_ops.fused_apply(
+                own, st.m[lo:hi], st.v[lo:hi], st.master[lo:hi],
+                model[lo:hi] if model is not None else None,
+                self._sqnorm_dev[0] if self.clip_norm is not None else None,
+                b_rel, lr=eff, inv_k=inv_k, clip_norm=self.clip_norm,
+                weight_decay=self.weight_decay, beta1=self.beta1,
+                beta2=self.beta2, eps=self.eps)
+        # the apply zeroed the owned shard; clear the rest for the next window
+        if lo:
+            st.accum[:lo].zero_()
+        if hi < n:
+            st.accum[hi:].zero_()
+        # publish updated params: one all-gather of the flat model buffer
+        # (bf16 -> half the bytes of the fp32 all-reduce it replaces)
+        dist.all_gather_into_tensor(st.model, st.model[lo:hi], group=self.group)
+        self.apply_count += 1
+
     def micro_step(self) -> bool:
         """One reference session.run: accumulate, maybe apply, step += 1.
 
@@ -278,8 +366,7 @@ class AccumEngine:
         applied = self.is_apply_step()
         if applied:
             self._join_wgrad_stream()
-            self._allreduce_accum()
-            self.apply()
+            self.boundary_apply()
         self.global_step += 1
         return applied
 
@@ -333,14 +420,26 @@ class AccumEngine:
         applied = self.is_apply_step(s + n - 1)
         if applied:
             self._join_wgrad_stream()
-            self._allreduce_accum()
-            self.apply(lr=self.lr_at(s + n - 1))
+            self.boundary_apply(lr=self.lr_at(s + n - 1))
         self.global_step = s + n
         return applied
 
     # ---- checkpoint (accum + m/v + step all saved) ----
     def state_dict(self) -> Dict:
+        """COLLECTIVE when sharded DP is active: master/m/v are only valid
+        on their owning shard between boundaries, so every rank must call
+        this together (the estimator saves on every rank)."""
         self._join_wgrad_stream()
+        if self.shard_apply and self.world_size > 1 and self.apply_count > 0:
+            import torch.distributed as dist
+
+            st = self.state
+            W = self.world_size
+            r = dist.get_rank(self.group) if self.group is not None else dist.get_rank()
+            sh = st.master.numel() // W
+            for buf in (st.master, st.m, st.v):
+                dist.all_gather_into_tensor(buf, buf[r * sh : (r + 1) * sh],
+                                            group=self.group)
         d = self.state.state_dict()
         d["global_step"] = self.global_step
         d["apply_count"] = self.apply_count

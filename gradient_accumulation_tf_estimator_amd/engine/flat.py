@@ -80,6 +80,7 @@ def build_layout(
     named_params: Sequence[Tuple[str, torch.Tensor]],
     exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
     direct_names: Sequence[str] = (),
+    pad_total_to: int = ALIGN,
 ) -> FlatLayout:
     """Order params and assign aligned flat offsets.
 
@@ -115,6 +116,9 @@ def build_layout(
             decay_boundary = off
     if grad_lo is None:
         grad_lo = grad_hi = 0
+    # tail padding (zeros in every buffer, invariant under the update) so
+    # the total divides into aligned equal shards for reduce-scatter DP
+    off = _round_up(off, max(pad_total_to, ALIGN))
     lay = FlatLayout(slices=slices, total=off, decay_boundary=decay_boundary)
     lay.grad_lo, lay.grad_hi = grad_lo, grad_hi
     return lay
@@ -142,6 +146,7 @@ class FlatState:
         named_params: Sequence[Tuple[str, torch.Tensor]],
         exclude_from_weight_decay: Sequence[str] = DEFAULT_EXCLUDE_FROM_WEIGHT_DECAY,
         direct_names: Sequence[str] = (),
+        pad_total_to: int = ALIGN,
     ):
         named_params = [(n, p) for n, p in named_params if p.requires_grad]
         if not named_params:
@@ -158,7 +163,7 @@ class FlatState:
             raise ValueError(f"unsupported param dtype {self.dtype}")
 
         self.layout = build_layout(named_params, exclude_from_weight_decay,
-                                   direct_names)
+                                   direct_names, pad_total_to)
         N = self.layout.total
         dev = self.device
 
@@ -217,15 +222,27 @@ class FlatState:
             d["model"] = self.model
         return d
 
+    @staticmethod
+    def _copy_flat(dst: torch.Tensor, src: torch.Tensor) -> None:
+        """Copy tolerating different TAIL padding (a checkpoint written at a
+        different DP shard-alignment pads to a different total; the pad is
+        zeros by construction)."""
+        n = min(dst.numel(), src.numel())
+        dst[:n].copy_(src[:n])
+        if src.numel() > n and float(src[n:].abs().sum()) != 0.0:
+            raise ValueError("checkpoint longer than layout with non-zero tail")
+        if dst.numel() > n:
+            dst[n:].zero_()
+
     def load_state_dict(self, d: Dict) -> None:
         if d["names"] != self.layout.names:
             raise ValueError("checkpoint parameter layout does not match model")
-        self.master.copy_(d["master"])
-        self.m.copy_(d["m"])
-        self.v.copy_(d["v"])
-        self.accum.copy_(d["accum"])
+        self._copy_flat(self.master, d["master"])
+        self._copy_flat(self.m, d["m"])
+        self._copy_flat(self.v, d["v"])
+        self._copy_flat(self.accum, d["accum"])
         if self.master is not self.model:
             if "model" in d:
-                self.model.copy_(d["model"])
+                self._copy_flat(self.model, d["model"])
             else:
                 self.model.copy_(self.master.to(self.dtype))

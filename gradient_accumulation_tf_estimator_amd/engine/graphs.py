@@ -155,10 +155,10 @@ class GraphedTrainLoop:
                 engine.apply_count += 1
                 return self.loss_apply
             self.g_accum.replay()
-            engine._allreduce_accum()
-            engine.apply_from_device()
+            # world > 1: the boundary runs eagerly after the captured
+            # accumulate (all-reduce + apply, or RS + sharded apply + AG)
+            engine.boundary_apply(lr=engine.lr_at(engine.global_step))
             engine.global_step += 1
-            engine.apply_count += 1
             return self.loss_accum
         self.g_accum.replay()
         engine.global_step += 1
@@ -272,10 +272,8 @@ class FusedWindowLoop:
                 engine.apply_count += 1
                 return self.loss_apply
             self.g_accum.replay()
-            engine._allreduce_accum()
-            engine.apply_from_device()
+            engine.boundary_apply(lr=engine.lr_at(s + self.n - 1))
             engine.global_step = s + self.n
-            engine.apply_count += 1
             return self.loss_accum
         self.g_accum.replay()
         engine.global_step = s + self.n

@@ -89,6 +89,7 @@ def create_optimizer(
     strict_reference_semantics: bool = False,
     process_group=None,
     ddp_scale_loss: bool = True,
+    shard_apply: bool = True,
     backend: str = "auto",
 ) -> TrainOp:
     """Build the train op.
@@ -150,6 +151,7 @@ def create_optimizer(
         bias_correction=bias_correction,
         strict_reference_semantics=strict_reference_semantics,
         process_group=process_group,
+        shard_apply=shard_apply,
         backend=backend,
         direct_names=direct,
     )

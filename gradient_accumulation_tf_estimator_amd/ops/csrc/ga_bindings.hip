@@ -192,7 +192,10 @@ void fused_apply(at::Tensor accum, at::Tensor m, at::Tensor v, at::Tensor master
                  at::Tensor model, bool has_model,
                  at::Tensor lr_dev, at::Tensor sqnorm_ws,
                  int64_t decay_boundary, double inv_k, double clip,
-                 double weight_decay, double beta1, double beta2, double eps) {
+                 double weight_decay, double beta1, double beta2, double eps,
+                 bool skip_norm = false) {
+  // skip_norm: sqnorm_ws already holds the GLOBAL squared norm (sharded DP
+  // apply: per-rank shard sqnorm + scalar all-reduce) -- do not recompute
   check_flat(accum, "accum", at::kFloat);
   check_flat(m, "m", at::kFloat);
   check_flat(v, "v", at::kFloat);
@@ -204,7 +207,7 @@ void fused_apply(at::Tensor accum, at::Tensor m, at::Tensor v, at::Tensor master
   const long long n4 = n / 4;
   auto stream = c10::hip::getCurrentHIPStream().stream();
 
-  if (clip > 0.0) {
+  if (clip > 0.0 && !skip_norm) {
     hipMemsetAsync(sqnorm_ws.data_ptr<float>(), 0, sizeof(float), stream);
     hipLaunchKernelGGL(k_sqnorm, grid_for(n4), dim3(GA_THREADS), 0, stream,
                        (const float4*)accum.data_ptr<float>(), n4,
@@ -824,7 +827,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("accumulate", &accumulate, "accum += grad (fp32 upcast); grad = 0");
   mod.def("sqnorm", &sqnorm, "out[0] = sum(accum^2)");
   mod.def("fused_apply", &fused_apply,
-          "normalize + clip + AdamWeightDecay + bf16 write-back + zero accum");
+          "normalize + clip + AdamWeightDecay + bf16 write-back + zero accum",
+          pybind11::arg("accum"), pybind11::arg("m"), pybind11::arg("v"),
+          pybind11::arg("master"), pybind11::arg("model"),
+          pybind11::arg("has_model"), pybind11::arg("lr_dev"),
+          pybind11::arg("sqnorm_ws"), pybind11::arg("decay_boundary"),
+          pybind11::arg("inv_k"), pybind11::arg("clip"),
+          pybind11::arg("weight_decay"), pybind11::arg("beta1"),
+          pybind11::arg("beta2"), pybind11::arg("eps"),
+          pybind11::arg("skip_norm") = false);
   mod.def("addln_fwd", &addln_fwd, "fused residual+bias+LayerNorm forward");
   mod.def("addln_bwd", &addln_bwd, "fused LayerNorm backward -> dh + fp32 partials");
   mod.def("biasgelu_fwd", &biasgelu_fwd, "y = gelu_tanh(x + bias)");
