@@ -292,7 +292,16 @@ class AccumEngine:
         of the updated flat model params. Master/m/v stay sharded-valid;
         ``state_dict`` gathers them (collective -- call on every rank).
         """
-        if self.world_size <= 1 or not self.shard_apply:
+        use_shard = self.shard_apply and self.world_size > 1
+        if use_shard and self.state.accum.is_cuda:
+            import torch.distributed as dist
+
+            backend = str(dist.get_backend(self.group) if self.group is not None
+                          else dist.get_backend())
+            # gloo-on-GPU (the 1-device shakeout config) lacks the gather
+            # collectives; sharding needs RCCL there
+            use_shard = "nccl" in backend
+        if not use_shard:
             self._allreduce_accum()
             self.apply(lr)
             return
@@ -355,6 +364,7 @@ class AccumEngine:
         # (bf16 -> half the bytes of the fp32 all-reduce it replaces)
         dist.all_gather_into_tensor(st.model, st.model[lo:hi], group=self.group)
         self.apply_count += 1
+        self._shard_active = True
 
     def micro_step(self) -> bool:
         """One reference session.run: accumulate, maybe apply, step += 1.
@@ -430,7 +440,7 @@ class AccumEngine:
         on their owning shard between boundaries, so every rank must call
         this together (the estimator saves on every rank)."""
         self._join_wgrad_stream()
-        if self.shard_apply and self.world_size > 1 and self.apply_count > 0:
+        if getattr(self, "_shard_active", False) and self.world_size > 1:
             import torch.distributed as dist
 
             st = self.state
