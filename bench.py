@@ -60,6 +60,11 @@ def parse_args():
                    help="issue the micro-step wgrads as one grouped hipBLASLt launch")
     p.add_argument("--fused", default="on", choices=["on", "off"],
                    help="fused LN/GELU HIP modules (A/B switch)")
+    p.add_argument("--dropout", type=float, default=0.0,
+                   help="dropout prob (BERT default 0.1): attention-prob "
+                        "dropout runs in the fused kernels (counter-based "
+                        "RNG), hidden dropout as torch ops -- both draw "
+                        "fresh masks per hipGraph replay")
     p.add_argument("--masked", default="off", choices=["on", "off"],
                    help="random key-padding masks (valid lengths S/4..S) -- "
                         "measures the masked fused-attention path")
@@ -126,6 +131,7 @@ def main():
     torch.manual_seed(1234 + rank)
     cfg = CONFIGS[args.model]()
     cfg.fused = args.fused == "on"
+    cfg.dropout = args.dropout
     dtype = torch.bfloat16 if (args.dtype == "bf16" and use_cuda) else torch.float32
     model = BertForSequenceClassification(cfg).to(device=device, dtype=dtype)
     model.train()
@@ -262,6 +268,7 @@ def main():
                 "graphs": graphed is not None,
                 "window_fuse": fuse,
                 "masked": args.masked == "on",
+                "dropout": args.dropout,
                 "peak_hbm_gb": round(peak_hbm_gb, 3),
             },
         }

@@ -50,14 +50,30 @@ def test_fused_addln_forward_backward(H, with_res, with_pb, R):
     if with_res:
         np.testing.assert_allclose(res.grad.float().cpu(), rf.grad.cpu(),
                                    rtol=5e-2, atol=3e-2)
-    # param grads (unbound path -> returned as bf16): looser, they reduce R rows
-    np.testing.assert_allclose(mod.weight.grad.float().cpu(), wf.grad.cpu(),
-                               rtol=5e-2, atol=5e-1)
-    np.testing.assert_allclose(mod.bias.grad.float().cpu(), bf.grad.cpu(),
-                               rtol=5e-2, atol=5e-1)
+    # Param grads reduce R rows of bf16 products (fp32 partial slabs, final
+    # bf16 cast on the unbound path). Per-element error model: each bf16
+    # product carries relative rounding ~2^-9, partial sums accumulate in
+    # fp32, the output rounds once more -> bound_j = C*eps_b*sqrt(sum_r
+    # term_rj^2) + eps_b*|grad_j| with C a reduction-order margin. This
+    # scales with the actual per-shape term magnitudes instead of a flat
+    # atol.
+    EPS_B, C = 2.0**-9, 16.0
+    hn = h.detach()
+    mean = hn.mean(-1, keepdim=True)
+    xhat = (hn - mean) / torch.sqrt(hn.var(-1, unbiased=False, keepdim=True) + 1e-12)
+    dyf = dy.float()
+
+    def bound(terms2_sum, ref):
+        return (C * EPS_B * terms2_sum.sqrt() + EPS_B * ref.abs() + 1e-4).cpu().numpy()
+
+    dgamma_err = np.abs(mod.weight.grad.float().cpu().numpy() - wf.grad.cpu().numpy())
+    assert (dgamma_err <= bound(((dyf * xhat) ** 2).sum(0), wf.grad)).all()
+    dbeta_err = np.abs(mod.bias.grad.float().cpu().numpy() - bf.grad.cpu().numpy())
+    assert (dbeta_err <= bound((dyf ** 2).sum(0), bf.grad)).all()
     if with_pb:
-        np.testing.assert_allclose(mod.proj_bias.grad.float().cpu(), pf.grad.cpu(),
-                                   rtol=5e-2, atol=5e-1)
+        # proj-bias grad sums dh over rows; per-row dh == dL/dx rows
+        dpb_err = np.abs(mod.proj_bias.grad.float().cpu().numpy() - pf.grad.cpu().numpy())
+        assert (dpb_err <= bound((xf.grad ** 2).sum(0), pf.grad)).all()
 
 
 @pytest.mark.parametrize("H", [2048, 4096])
@@ -294,13 +310,23 @@ def test_wgrad_mfma_batched_matches_reference():
     finally:
         fops.set_grouped_wgrad(False)
     torch.cuda.synchronize()
-    for (K, N), a, ref, vb, vref in zip(shapes, accs, refs, vbs, vrefs):
-        np.testing.assert_allclose(a.view(N, K).cpu(), ref.cpu(),
-                                   rtol=2e-2, atol=3e-1,
-                                   err_msg=f"wgrad mismatch for K={K},N={N}")
+    # per-element fp32-accumulation error model (see the LN test): bf16
+    # products carry ~2^-9 relative rounding, summed over R rows in fp32
+    # MFMA accumulators -> bound_ij = C*eps_b*sqrt(sum_r (dy_ri*x_rj)^2)
+    EPS_B, C = 2.0**-9, 16.0
+    for (K, N), x, dy, a, ref, vb, vref in zip(shapes, xs, dys, accs, refs,
+                                               vbs, vrefs):
+        t2 = (dy.float() ** 2).t() @ (x.float() ** 2)  # [N,K] sum of term^2
+        bnd = (C * EPS_B * t2.sqrt() + EPS_B * ref.abs() + 1e-4).cpu().numpy()
+        err = np.abs(a.view(N, K).cpu().numpy() - ref.cpu().numpy())
+        assert (err <= bnd).all(), \
+            f"wgrad err beyond fp32-accum model for K={K},N={N}: " \
+            f"max {err.max()} vs bound {bnd.max()}"
         if vb is not None:
-            np.testing.assert_allclose(vb.cpu(), vref.cpu(), rtol=2e-2, atol=3e-1,
-                                       err_msg=f"dbias mismatch for N={N}")
+            t2b = (dy.float() ** 2).sum(0)
+            bndb = (C * EPS_B * t2b.sqrt() + EPS_B * vref.abs() + 1e-4).cpu().numpy()
+            errb = np.abs(vb.cpu().numpy() - vref.cpu().numpy())
+            assert (errb <= bndb).all(), f"dbias err beyond model for N={N}"
 
 
 def test_cls_head_matches_torch():
@@ -680,3 +706,45 @@ def test_bert_masked_training_step_matches_sdpa_path():
         op.step(loss)
     torch.cuda.synchronize()
     assert torch.isfinite(op.engine.state.master).all()
+
+
+def test_dropout_fresh_masks_across_graph_replays():
+    """The per-layer seed refresh is a captured RNG op: every hipGraph
+    replay must draw a fresh philox value, so two replays on identical
+    inputs produce different dropout masks (and a fixed seed reproduces)."""
+    from gradient_accumulation_tf_estimator_amd.ops.fused import fused_attention
+
+    torch.manual_seed(0)
+    B, nh, S, H = 2, 4, 128, 256
+    qkv = (torch.randn(B, S, 3 * H, device="cuda") * 0.5).bfloat16()
+    seeds = torch.zeros(1, dtype=torch.int64, device="cuda")
+
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(2):
+            seeds.random_()
+            out = fused_attention(qkv, nh, seed=seeds[0], p_drop=0.3)
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        seeds.random_()
+        out = fused_attention(qkv, nh, seed=seeds[0], p_drop=0.3)
+    g.replay()
+    torch.cuda.synchronize()
+    a = out.clone()
+    seed_a = seeds.clone()
+    g.replay()
+    torch.cuda.synchronize()
+    b = out.clone()
+    assert not torch.equal(seeds, seed_a), "captured RNG did not advance"
+    assert not torch.equal(a, b), "replays reused the same dropout mask"
+
+    # determinism: same seed value -> identical output
+    fixed = torch.tensor([42], dtype=torch.int64, device="cuda")
+    o1 = fused_attention(qkv, nh, seed=fixed[0], p_drop=0.3)
+    o2 = fused_attention(qkv, nh, seed=fixed[0], p_drop=0.3)
+    torch.cuda.synchronize()
+    assert torch.equal(o1, o2)

@@ -74,9 +74,15 @@ def single_rank_rccl():
     buf = torch.full((n,), 2.0, device="cuda:0")
     nb = bucketed_allreduce(buf)
     dist.broadcast(buf, src=0)
+    # the sharded-boundary collectives (engine/accum.py _sharded_apply):
+    # world-1 degenerate but still real RCCL enqueues of reduce-scatter /
+    # all-gather -- first-contact check before the driver's 8-GPU run
+    dist.reduce_scatter_tensor(buf[:n], buf)
+    model = torch.full((n,), 1.5, device="cuda:0", dtype=torch.bfloat16)
+    dist.all_gather_into_tensor(model, model[:n])
     torch.cuda.synchronize()
-    assert torch.all(buf == 2.0)
-    log(0, "single_rank_rccl", elems=n, buckets=nb, ok=True)
+    assert torch.all(buf == 2.0) and torch.all(model == 1.5)
+    log(0, "single_rank_rccl", elems=n, buckets=nb, rs_ag=True, ok=True)
     dist.destroy_process_group()
 
 
