@@ -36,9 +36,22 @@ setup(
                 os.path.join(CSRC, "ga_bindings.hip"),
             ],
             libraries=["hipblaslt"],
+            # GA_ASAN=1: instrument the HOST side (bindings, launch plumbing,
+            # hipBLASLt workspace management) with AddressSanitizer; device
+            # code is unchanged. Run via tools/asan_smoke.sh (LD_PRELOADs
+            # libasan). Device-side ASAN needs xnack+ and is not exercised
+            # on this pool.
+            # Compile-only instrumentation: the __asan_* references stay
+            # undefined in the .so (legal for shared objects) and resolve
+            # from the LD_PRELOADed clang runtime at run time -- the link
+            # step is g++, whose libasan must not be mixed in.
             extra_compile_args={
-                "cxx": ["-O3"],
-                "nvcc": ["-O3", "--offload-arch=gfx950"],
+                "cxx": ["-O3"] + (["-fsanitize=address", "-g"]
+                                  if os.environ.get("GA_ASAN") else []),
+                "nvcc": ["-O3", "--offload-arch=gfx950"]
+                + (["-fsanitize=address", "-g",
+                    "-Xarch_device", "-fno-sanitize=all"]
+                   if os.environ.get("GA_ASAN") else []),
             },
         )
     ],
