@@ -17,8 +17,14 @@ cp -r "$ROOT/gradient_accumulation_tf_estimator_amd" "$ROOT/setup.py" "$SCRATCH/
 rm -f "$SCRATCH"/gradient_accumulation_tf_estimator_amd/ops/_ga_hip*.so
 
 cd "$SCRATCH"
-GA_ASAN=1 PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace \
-    > build_asan.log 2>&1 || { tail -30 build_asan.log; exit 1; }
+if ls "$ROOT"/build_asan/_ga_hip*.so >/dev/null 2>&1; then
+    # prebuilt instrumented .so (built cross-compile off-box; travels with
+    # the repo snapshot) -- saves the on-box rebuild
+    cp "$ROOT"/build_asan/_ga_hip*.so gradient_accumulation_tf_estimator_amd/ops/
+else
+    GA_ASAN=1 PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace \
+        > build_asan.log 2>&1 || { tail -30 build_asan.log; exit 1; }
+fi
 
 # clang's asan runtime (hipcc is clang); detect_leaks off: python+torch leak
 # reports are noise for this smoke, we want memory-safety errors
