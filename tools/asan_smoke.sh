@@ -30,7 +30,9 @@ fi
 # reports are noise for this smoke, we want memory-safety errors
 ASAN_RT=$(/opt/rocm/lib/llvm/bin/clang --print-file-name=libclang_rt.asan-x86_64.so)
 export LD_PRELOAD="$ASAN_RT"
-export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1
+# allocator_may_return_null: libamdhip64 probes one huge allocation at
+# init; ASAN must hand it NULL instead of aborting (measured on MI355X)
+export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1:allocator_may_return_null=1
 export HSA_ENABLE_IPC_MODE_LEGACY=${HSA_ENABLE_IPC_MODE_LEGACY:-0}
 
 python - <<'EOF'
