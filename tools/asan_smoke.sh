@@ -30,9 +30,10 @@ fi
 # reports are noise for this smoke, we want memory-safety errors
 ASAN_RT=$(/opt/rocm/lib/llvm/bin/clang --print-file-name=libclang_rt.asan-x86_64.so)
 export LD_PRELOAD="$ASAN_RT"
-# allocator_may_return_null: libamdhip64 probes one huge allocation at
-# init; ASAN must hand it NULL instead of aborting (measured on MI355X)
-export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1:allocator_may_return_null=1
+# max_allocation_size_mb: libamdhip64 makes one huge allocation at init
+# that trips ASAN's default cap (returning it NULL instead SEGVs the
+# runtime -- both measured on MI355X); raise the cap
+export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1:max_allocation_size_mb=262144
 export HSA_ENABLE_IPC_MODE_LEGACY=${HSA_ENABLE_IPC_MODE_LEGACY:-0}
 
 python - <<'EOF'
