@@ -33,7 +33,9 @@ export LD_PRELOAD="$ASAN_RT"
 # max_allocation_size_mb: libamdhip64 makes one huge allocation at init
 # that trips ASAN's default cap (returning it NULL instead SEGVs the
 # runtime -- both measured on MI355X); raise the cap
-export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1:max_allocation_size_mb=262144
+# protect_shadow_gap=0: the HIP runtime must map device-visible memory
+# inside ASAN's shadow gap (the standard GPU-runtime/ASAN accommodation)
+export ASAN_OPTIONS=detect_leaks=0:halt_on_error=1:abort_on_error=1:max_allocation_size_mb=262144:protect_shadow_gap=0
 export HSA_ENABLE_IPC_MODE_LEGACY=${HSA_ENABLE_IPC_MODE_LEGACY:-0}
 
 python - <<'EOF'
