@@ -2,11 +2,16 @@
 // -> softmax cross-entropy, forward and backward as ONE kernel each.
 //
 // At the reference micro-batch (B = 8 rows, num_labels = 2) this tail is a
-// string of ~10 tiny torch kernels (tanh fwd/bwd, [8,512]x[512,2] GEMMs,
+// string of ~10 tiny torch kernels (tanh fwd/bwd, [B,512]x[512,2] GEMMs,
 // log_softmax, nll, scalar scale/fill) each paying the full launch/exec
-// floor. Shapes: H <= 1024, C (labels) <= 8, B <= 64 -- one workgroup does
-// everything; the classifier weight/bias gradients accumulate directly into
-// the engine's flat fp32 accum buffer (reference accum_grads semantics).
+// floor. Shapes: H <= 1024, C (labels) <= 8; B is row-parallel across
+// workgroups (one wave per row fwd, 8-row groups x column tiles bwd) --
+// the original one-workgroup version serialized at the window-fused B=32
+// (25+31 us/window); the row-parallel grids bring it to the small-kernel
+// floor. Loss is mean CE accumulated with one atomicAdd per row; the
+// classifier weight/bias gradients accumulate into the engine's flat fp32
+// accum buffer (reference accum_grads semantics) with per-column atomics
+// (contended only across the ceil(B/8) row groups).
 
 #include <hip/hip_runtime.h>
 
@@ -23,115 +28,107 @@ static inline __device__ float cls_bf2f(unsigned short u) {
   return c.f;
 }
 
-#define CLS_MAXB 64
 #define CLS_MAXC 8
+#define CLS_RG 8  // backward row-group size
 
 // fwd: pre [B,H] bf16 (pooler output, pre-tanh); Wc [C,H] bf16; bc [C] bf16;
-// labels [B] i64 -> loss [1] fp32 (mean CE), t [B,H] bf16 (tanh, saved),
-// probs [B,C] fp32 (saved).
+// labels [B] i64 -> loss [1] fp32 (mean CE, PRE-ZEROED by the binding),
+// t [B,H] bf16 (tanh, saved), probs [B,C] fp32 (saved).
+// grid: ceil(B/4) blocks, one wave64 per row.
 extern "C" __global__ __launch_bounds__(256) void k_cls_head_fwd(
     const unsigned short* __restrict__ pre, const unsigned short* __restrict__ Wc,
     const unsigned short* __restrict__ bc, const long long* __restrict__ labels,
     unsigned short* __restrict__ t_out, float* __restrict__ probs,
     float* __restrict__ loss, int B, int H, int C) {
-  __shared__ float logits[CLS_MAXB][CLS_MAXC];
-  __shared__ float red[256];
-
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
+  const int b = blockIdx.x * 4 + wid;
+  if (b >= B) return;
 
-  // tanh + per-(row, class) dot products: wave w handles rows w, w+4, ...
-  for (int b = wid; b < B; b += 4) {
-    // each lane: chunk of 8 columns
-    float dot[CLS_MAXC];
+  // tanh + per-class dot products; each lane owns 8-column chunks
+  float dot[CLS_MAXC];
 #pragma unroll
-    for (int c = 0; c < CLS_MAXC; ++c) dot[c] = 0.f;
-    for (int h0 = lane * 8; h0 < H; h0 += 64 * 8) {
-      bf16x8 pv = *(const bf16x8*)(pre + (long long)b * H + h0);
-      bf16x8 tv;
+  for (int c = 0; c < CLS_MAXC; ++c) dot[c] = 0.f;
+  for (int h0 = lane * 8; h0 < H; h0 += 64 * 8) {
+    bf16x8 pv = *(const bf16x8*)(pre + (long long)b * H + h0);
+    bf16x8 tv;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        float x = (float)pv[e];
-        // tanh via exp2 (native v_exp); |x| clamp keeps exp finite
-        float u = fminf(fmaxf(x, -9.f), 9.f);
-        float ex = __builtin_amdgcn_exp2f(u * 2.885390082f);  // e^{2u}
-        float th = (ex - 1.f) / (ex + 1.f);
-        tv[e] = (__bf16)th;
-        for (int c = 0; c < C; ++c)
-          dot[c] = fmaf(th, cls_bf2f(((const unsigned short*)Wc)[c * H + h0 + e]),
-                        dot[c]);
-      }
-      *(bf16x8*)(t_out + (long long)b * H + h0) = tv;
+    for (int e = 0; e < 8; ++e) {
+      float x = (float)pv[e];
+      // tanh via exp2 (native v_exp); |x| clamp keeps exp finite
+      float u = fminf(fmaxf(x, -9.f), 9.f);
+      float ex = __builtin_amdgcn_exp2f(u * 2.885390082f);  // e^{2u}
+      float th = (ex - 1.f) / (ex + 1.f);
+      tv[e] = (__bf16)th;
+      for (int c = 0; c < C; ++c)
+        dot[c] = fmaf(th, cls_bf2f(((const unsigned short*)Wc)[c * H + h0 + e]),
+                      dot[c]);
     }
-    for (int c = 0; c < C; ++c) {
-      float s = dot[c];
-      for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
-      if (lane == 0) logits[b][c] = s + cls_bf2f(bc[c]);
-    }
+    *(bf16x8*)(t_out + (long long)b * H + h0) = tv;
   }
-  __syncthreads();
-
-  // softmax CE (one thread per row) + mean reduce
-  float l = 0.f;
-  if (threadIdx.x < (unsigned)B) {
-    const int b = threadIdx.x;
+  float lg[CLS_MAXC];
+  for (int c = 0; c < C; ++c) {
+    float s = dot[c];
+    for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+    if (lane == 0) lg[c] = s + cls_bf2f(bc[c]);
+  }
+  if (lane == 0) {
     float m = -1e30f;
-    for (int c = 0; c < C; ++c) m = fmaxf(m, logits[b][c]);
+    for (int c = 0; c < C; ++c) m = fmaxf(m, lg[c]);
     float z = 0.f;
-    for (int c = 0; c < C; ++c) z += expf(logits[b][c] - m);
-    for (int c = 0; c < C; ++c)
-      probs[b * C + c] = expf(logits[b][c] - m) / z;
+    for (int c = 0; c < C; ++c) z += expf(lg[c] - m);
+    for (int c = 0; c < C; ++c) probs[b * C + c] = expf(lg[c] - m) / z;
     const int y = (int)labels[b];
-    l = -(logits[b][y] - m - logf(z));
-  }
-  red[threadIdx.x] = l;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float s = 0.f;
-    for (int i = 0; i < B; ++i) s += red[i];
-    loss[0] = s / B;
+    atomicAdd(loss, -(lg[y] - m - logf(z)) / B);
   }
 }
 
 // bwd: dloss (scalar) -> d_pre [B,H] bf16; Wc/bc grads ACCUMULATE into fp32
 // accum slices (acc_w [C*H], acc_b [C]) when given (non-null).
+// grid: (ceil(H/256), ceil(B/CLS_RG)); each block covers a column tile of
+// one row group.
 extern "C" __global__ __launch_bounds__(256) void k_cls_head_bwd(
     const float* __restrict__ dloss, const unsigned short* __restrict__ t_in,
     const float* __restrict__ probs, const long long* __restrict__ labels,
     const unsigned short* __restrict__ Wc,
     unsigned short* __restrict__ dpre, float* __restrict__ acc_w,
     float* __restrict__ acc_b, int B, int H, int C) {
-  __shared__ float dlog[CLS_MAXB][CLS_MAXC];
+  __shared__ float dlog[CLS_RG][CLS_MAXC];
   const float g = dloss[0] / B;
+  const int r0 = blockIdx.y * CLS_RG;
+  const int rows = min(CLS_RG, B - r0);
 
-  if (threadIdx.x < (unsigned)(B * C)) {
-    const int b = threadIdx.x / C, c = threadIdx.x % C;
-    dlog[b][c] = (probs[b * C + c] - (labels[b] == c ? 1.f : 0.f)) * g;
+  if (threadIdx.x < (unsigned)(rows * C)) {
+    const int br = threadIdx.x / C, c = threadIdx.x % C;
+    dlog[br][c] =
+        (probs[(r0 + br) * C + c] - (labels[r0 + br] == c ? 1.f : 0.f)) * g;
   }
   __syncthreads();
 
-  if (acc_b && threadIdx.x < (unsigned)C) {
+  if (acc_b && blockIdx.x == 0 && threadIdx.x < (unsigned)C) {
     float s = 0.f;
-    for (int b = 0; b < B; ++b) s += dlog[b][threadIdx.x];
-    acc_b[threadIdx.x] += s;
+    for (int br = 0; br < rows; ++br) s += dlog[br][threadIdx.x];
+    atomicAdd(&acc_b[threadIdx.x], s);
   }
 
-  // d_pre and dWc over H: thread -> column h (grid-stride by blockDim)
-  for (int h = threadIdx.x; h < H; h += blockDim.x) {
+  // d_pre and dWc over this block's column tile, all rows of the group
+  for (int h = blockIdx.x * blockDim.x + threadIdx.x; h < H;
+       h += gridDim.x * blockDim.x) {
     float wcol[CLS_MAXC];
     for (int c = 0; c < C; ++c) wcol[c] = cls_bf2f(Wc[c * H + h]);
     float dw[CLS_MAXC];
     for (int c = 0; c < CLS_MAXC; ++c) dw[c] = 0.f;
-    for (int b = 0; b < B; ++b) {
-      const float th = cls_bf2f(t_in[(long long)b * H + h]);
+    for (int br = 0; br < rows; ++br) {
+      const long long b = r0 + br;
+      const float th = cls_bf2f(t_in[b * H + h]);
       float dt = 0.f;
       for (int c = 0; c < C; ++c) {
-        dt = fmaf(dlog[b][c], wcol[c], dt);
-        dw[c] = fmaf(dlog[b][c], th, dw[c]);
+        dt = fmaf(dlog[br][c], wcol[c], dt);
+        dw[c] = fmaf(dlog[br][c], th, dw[c]);
       }
-      dpre[(long long)b * H + h] = cls_f2bf(dt * (1.f - th * th));
+      dpre[b * H + h] = cls_f2bf(dt * (1.f - th * th));
     }
     if (acc_w)
-      for (int c = 0; c < C; ++c) acc_w[c * H + h] += dw[c];
+      for (int c = 0; c < C; ++c) atomicAdd(&acc_w[c * H + h], dw[c]);
   }
 }

@@ -612,14 +612,14 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
 std::vector<at::Tensor> cls_head_fwd(at::Tensor pre, at::Tensor Wc, at::Tensor bc,
                                      at::Tensor labels) {
   const int B = (int)pre.size(0), H = (int)pre.size(1), C = (int)Wc.size(0);
-  TORCH_CHECK(B <= 64 && C <= 8 && H % 512 == 0 && H <= 1024,
-              "cls head: B<=64, C<=8, H in {512, 1024}");
+  TORCH_CHECK(B <= 4096 && C <= 8 && H % 512 == 0 && H <= 1024,
+              "cls head: B<=4096, C<=8, H in {512, 1024}");
   TORCH_CHECK(pre.is_contiguous() && pre.scalar_type() == at::kBFloat16);
   auto t = at::empty_like(pre);
   auto probs = at::empty({B, C}, pre.options().dtype(at::kFloat));
-  auto loss = at::empty({}, pre.options().dtype(at::kFloat));
+  auto loss = at::zeros({}, pre.options().dtype(at::kFloat));  // atomic target
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_cls_head_fwd, dim3(1), dim3(256), 0, stream,
+  hipLaunchKernelGGL(k_cls_head_fwd, dim3((B + 3) / 4), dim3(256), 0, stream,
                      (const unsigned short*)pre.data_ptr(),
                      (const unsigned short*)Wc.data_ptr(),
                      (const unsigned short*)bc.data_ptr(),
@@ -637,7 +637,8 @@ at::Tensor cls_head_bwd(at::Tensor dloss, at::Tensor t, at::Tensor probs,
   const int B = (int)t.size(0), H = (int)t.size(1), C = (int)Wc.size(0);
   auto dpre = at::empty_like(t);
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_cls_head_bwd, dim3(1), dim3(256), 0, stream,
+  hipLaunchKernelGGL(k_cls_head_bwd,
+                     dim3((H + 255) / 256, (B + 7) / 8), dim3(256), 0, stream,
                      dloss.data_ptr<float>(), (const unsigned short*)t.data_ptr(),
                      probs.data_ptr<float>(),
                      (const long long*)labels.data_ptr<int64_t>(),

@@ -679,8 +679,8 @@ class CEClassifier(nn.Module):
 
     def loss(self, pre, labels):
         if self._accum_view_w is not None:
-            if not (_use_hip(pre) and pre.shape[0] <= 64):
-                raise RuntimeError("bound CEClassifier needs bf16 GPU input, B<=64")
+            if not (_use_hip(pre) and pre.shape[0] <= 4096):
+                raise RuntimeError("bound CEClassifier needs bf16 GPU input, B<=4096")
             return _ClsHeadFn.apply(pre, self.weight, self.bias, labels, self)
         return F.cross_entropy(self.logits(pre).float(), labels)
 
