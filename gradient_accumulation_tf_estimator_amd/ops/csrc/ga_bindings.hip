@@ -131,7 +131,7 @@ struct WgArgs {
   int G;
 };
 extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
-extern "C" __global__ void k_wgrad_mfma256(WgArgs, int);
+extern "C" __global__ void k_wgrad_mfma256(WgArgs, int, int);
 extern "C" __global__ void k_cls_head_fwd(const unsigned short*, const unsigned short*,
                                           const unsigned short*, const long long*,
                                           unsigned short*, float*, float*, int, int, int);
@@ -600,8 +600,14 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     long long nt256 = 0;
     for (int g = 0; g < G; ++g)
       nt256 += (long long)(args.nk[g * 2] / 256) * (args.nk[g * 2 + 1] / 256);
-    hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)nt256), dim3(512), 65536,
-                       stream, args, (int)R);
+    // R-split so the launch reaches >= 2 workgroups per CU (window-fused
+    // R=4096 has only ~196 tiles); each split keeps >= 4 R-chunks
+    int splits = 1;
+    const long long chunks = R / 64;
+    while (nt256 * splits < 512 && splits < 8 && (long long)(splits * 2) * 4 <= chunks)
+      splits *= 2;
+    hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)(nt256 * splits)),
+                       dim3(512), 65536, stream, args, (int)R, splits);
   } else {
     hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
                        stream, args, (int)R);

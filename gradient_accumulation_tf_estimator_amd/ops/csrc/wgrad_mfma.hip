@@ -269,14 +269,31 @@ static __device__ __forceinline__ void stage512_write(unsigned short* lds,
   }
 }
 
+// splits > 1: R-split parallelism for the window-fused regime. At R = 4096
+// the tile count (~196 for bert-small) is below one workgroup per CU, so a
+// single-pass walk leaves the chip latency-bound (measured 322 us/window,
+// 12% MfmaUtil); `splits` blocks each reduce an R-slice of the same tile
+// and combine with fp32 atomics in the epilogue (extra output traffic
+// ~tiles*256KB*(splits-1), ~25 us at splits=4 -- far under the win).
+// splits == 1 keeps the deterministic read-modify-write epilogue.
 extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
-    WgArgs args, int R) {
+    WgArgs args, int R, int splits) {
   int id;
   {
     const int nwg = gridDim.x, orig = blockIdx.x;
     const int q = nwg / 8, r = nwg % 8, xcd = orig % 8;
     id = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig / 8;
   }
+  // split-major so each split's tile run keeps the XCD L2 locality
+  const int tiles_total = gridDim.x / splits;
+  const int split = id / tiles_total;
+  id = id % tiles_total;
+  // this block's R-slice, in whole 64-row chunks
+  const int nchunk = R / 64;
+  const int cper = (nchunk + splits - 1) / splits;
+  const int c_lo = split * cper;
+  const int c_hi = min(nchunk, c_lo + cper);
+  const int r_lo = c_lo * 64, r_hi = c_hi * 64;
   int g = 0, N = 0, K = 0;
   for (; g < args.G; ++g) {
     N = args.nk[g * 2 + 0];
@@ -312,14 +329,15 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
   const int brh = (threadIdx.x & 1) * 32;  // r half
   float bias_acc = 0.f;
 
+  if (r_lo >= r_hi) return;  // more splits than chunks (tiny R)
   StageRegs512 sdy, sx;
-  stage512_issue(dy, N, n0, sdy);
-  stage512_issue(x, K, k0, sx);
-  for (int r0 = 0; r0 < R; r0 += 64) {
+  stage512_issue(dy + (long long)r_lo * N, N, n0, sdy);
+  stage512_issue(x + (long long)r_lo * K, K, k0, sx);
+  for (int r0 = r_lo; r0 < r_hi; r0 += 64) {
     __syncthreads();  // previous chunk's readers done
     stage512_write(dyT, sdy);
     stage512_write(xT, sx);
-    if (r0 + 64 < R) {
+    if (r0 + 64 < r_hi) {
       stage512_issue(dy + (long long)(r0 + 64) * N, N, n0, sdy);
       stage512_issue(x + (long long)(r0 + 64) * K, K, k0, sx);
     }
@@ -353,8 +371,12 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
 
   if (do_bias) {
     bias_acc += __shfl_xor(bias_acc, 1, 64);
-    if ((threadIdx.x & 1) == 0)
-      ((float*)args.dbias[g])[n0 + bn] += bias_acc;
+    if ((threadIdx.x & 1) == 0) {
+      if (splits > 1)
+        atomicAdd(&((float*)args.dbias[g])[n0 + bn], bias_acc);
+      else
+        ((float*)args.dbias[g])[n0 + bn] += bias_acc;
+    }
   }
 
   // epilogue in 4 passes of 64 k-rows, bounced through a [64 k][256 n]
@@ -397,9 +419,16 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
                               ((n * 4) ^ (((k4 + 2) & 7) << 4)));
         v.w = *(const float*)((char*)ftile + (long long)(k4 + 3) * 1024 +
                               ((n * 4) ^ (((k4 + 3) & 7) << 4)));
-        float4 old = *(const float4*)(grow + c * 4);
-        old.x += v.x; old.y += v.y; old.z += v.z; old.w += v.w;
-        *(float4*)(grow + c * 4) = old;
+        if (splits > 1) {
+          atomicAdd(grow + c * 4 + 0, v.x);
+          atomicAdd(grow + c * 4 + 1, v.y);
+          atomicAdd(grow + c * 4 + 2, v.z);
+          atomicAdd(grow + c * 4 + 3, v.w);
+        } else {
+          float4 old = *(const float4*)(grow + c * 4);
+          old.x += v.x; old.y += v.y; old.z += v.z; old.w += v.w;
+          *(float4*)(grow + c * 4) = old;
+        }
       }
     }
   }

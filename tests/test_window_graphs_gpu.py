@@ -1,6 +1,7 @@
 """Window-sized hipGraph capture (K-1 accumulate steps in one graph) must
-match eager micro-steps bit-for-bit in losses and near-bit in master
-params (engine/graphs.py window mode)."""
+match eager micro-steps in losses (to the 1-ulp wobble of the row-parallel
+CLS head's atomic mean-CE sum; probs and grads stay deterministic) and
+near-bit in master params (engine/graphs.py window mode)."""
 
 import numpy as np
 import pytest
@@ -60,6 +61,10 @@ def test_window_graph_parity_small():
         lossesB.append(float(loop.step().detach().float()))
     torch.cuda.synchronize()
 
-    np.testing.assert_array_equal(np.array(lossesA), np.array(lossesB))
+    # the CLS head accumulates the mean CE with one atomicAdd per row:
+    # summation order differs run-to-run by ~1 ulp on the reported
+    # scalar only (cls_head.hip); everything downstream is exact
+    np.testing.assert_allclose(np.array(lossesA), np.array(lossesB),
+                               rtol=1e-6, atol=1e-7)
     dm = (opA.engine.state.master - opB.engine.state.master).abs().max().item()
     assert dm < 1e-6, f"master diverged by {dm}"
