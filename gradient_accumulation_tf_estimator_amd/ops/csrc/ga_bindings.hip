@@ -595,19 +595,31 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
   bool all256 = true;
   for (int g = 0; g < G; ++g)
     all256 &= (args.nk[g * 2] % 256 == 0) && (args.nk[g * 2 + 1] % 256 == 0);
+  // A/B knobs (re-measurement): GA_WGRAD_TILE=128 forces the 4-wave
+  // 128-tile kernel even when 256 tiles fit; GA_WGRAD_SPLITS=N forces
+  // R-splitting. Splits default OFF: the atomic scalar combine measured
+  // CATASTROPHIC at splits=4 on the fused bench (19.3k -> 8.4k samples/s;
+  // ~51M serialized fp32 atomics per window) -- a scratch+fixup combine
+  // would be needed to make R-splits pay.
+  static const int env_tile = [] {
+    const char* v = getenv("GA_WGRAD_TILE");
+    return v ? atoi(v) : 256;
+  }();
+  static const int env_splits = [] {
+    const char* v = getenv("GA_WGRAD_SPLITS");
+    return v ? atoi(v) : 1;
+  }();
+  all256 &= env_tile == 256;
   auto stream = c10::hip::getCurrentHIPStream().stream();
   if (all256) {
     long long nt256 = 0;
     for (int g = 0; g < G; ++g)
       nt256 += (long long)(args.nk[g * 2] / 256) * (args.nk[g * 2 + 1] / 256);
-    // R-split so the launch reaches >= 2 workgroups per CU (window-fused
-    // R=4096 has only ~196 tiles); each split keeps >= 4 R-chunks
-    int splits = 1;
-    const long long chunks = R / 64;
-    while (nt256 * splits < 512 && splits < 8 && (long long)(splits * 2) * 4 <= chunks)
-      splits *= 2;
+    int splits = env_splits;
+    while (splits > 1 && (long long)splits * 4 > R / 64) splits /= 2;
     hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)(nt256 * splits)),
-                       dim3(512), 65536, stream, args, (int)R, splits);
+                       dim3(512), 65536, stream, args, (int)R,
+                       splits < 1 ? 1 : splits);
   } else {
     hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
                        stream, args, (int)R);
