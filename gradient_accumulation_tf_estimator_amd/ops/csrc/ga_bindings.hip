@@ -618,7 +618,7 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     int splits = env_splits;
     while (splits > 1 && (long long)splits * 4 > R / 64) splits /= 2;
     hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)(nt256 * splits)),
-                       dim3(512), 131072, stream, args, (int)R,
+                       dim3(512), 65536, stream, args, (int)R,
                        splits < 1 ? 1 : splits);
   } else {
     hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,

@@ -315,17 +315,8 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
   const int wn = wave >> 1, wk = wave & 1;  // 4(n: 64 rows) x 2(k: 128 cols)
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // double-buffered [256][64 r] panels (2 x 64 KB of the 160 KB LDS): at
-  // the window-fused R=4096 there are fewer tiles than CUs, so the single
-  // resident workgroup must hide its own staging latency -- write chunk
-  // t+1 into the other buffer while the MFMAs read chunk t (one barrier
-  // per chunk instead of two)
-  unsigned short* dyT[2];
-  unsigned short* xT[2];
-  dyT[0] = (unsigned short*)smem;
-  xT[0] = (unsigned short*)(smem + 32768);
-  dyT[1] = (unsigned short*)(smem + 65536);
-  xT[1] = (unsigned short*)(smem + 98304);
+  unsigned short* dyT = (unsigned short*)smem;           // [256 n][64 r]
+  unsigned short* xT = (unsigned short*)(smem + 32768);  // [256 k][64 r]
 
   f32x16 acc[2][4];
 #pragma unroll
@@ -342,31 +333,31 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
   StageRegs512 sdy, sx;
   stage512_issue(dy + (long long)r_lo * N, N, n0, sdy);
   stage512_issue(x + (long long)r_lo * K, K, k0, sx);
-  stage512_write(dyT[0], sdy);
-  stage512_write(xT[0], sx);
-  int cur = 0;
   for (int r0 = r_lo; r0 < r_hi; r0 += 64) {
+    __syncthreads();  // previous chunk's readers done
+    stage512_write(dyT, sdy);
+    stage512_write(xT, sx);
     if (r0 + 64 < r_hi) {
       stage512_issue(dy + (long long)(r0 + 64) * N, N, n0, sdy);
       stage512_issue(x + (long long)(r0 + 64) * K, K, k0, sx);
     }
-    __syncthreads();  // buf[cur] fully written; buf[cur^1] free
+    __syncthreads();
     if (do_bias) {
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        bf16x8 v = *(const bf16x8*)((char*)dyT[cur] + wswz(bn, (brh + c * 8) * 2));
+        bf16x8 v = *(const bf16x8*)((char*)dyT + wswz(bn, (brh + c * 8) * 2));
 #pragma unroll
         for (int e = 0; e < 8; ++e) bias_acc += (float)v[e];
       }
     }
 #pragma unroll
     for (int s = 0; s < 4; ++s) {
-      bf16x8 a0 = *(const bf16x8*)((char*)dyT[cur] + wswz(wn * 64 + lo31, s * 32 + hi * 16));
-      bf16x8 a1 = *(const bf16x8*)((char*)dyT[cur] + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
-      bf16x8 b0 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 128 + lo31, s * 32 + hi * 16));
-      bf16x8 b1 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 128 + 32 + lo31, s * 32 + hi * 16));
-      bf16x8 b2 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 128 + 64 + lo31, s * 32 + hi * 16));
-      bf16x8 b3 = *(const bf16x8*)((char*)xT[cur] + wswz(wk * 128 + 96 + lo31, s * 32 + hi * 16));
+      bf16x8 a0 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 a1 = *(const bf16x8*)((char*)dyT + wswz(wn * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b0 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + lo31, s * 32 + hi * 16));
+      bf16x8 b1 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b2 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 64 + lo31, s * 32 + hi * 16));
+      bf16x8 b3 = *(const bf16x8*)((char*)xT + wswz(wk * 128 + 96 + lo31, s * 32 + hi * 16));
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
       acc[0][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b2, acc[0][2], 0, 0, 0);
@@ -376,13 +367,7 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
       acc[1][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b2, acc[1][2], 0, 0, 0);
       acc[1][3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b3, acc[1][3], 0, 0, 0);
     }
-    if (r0 + 64 < r_hi) {
-      stage512_write(dyT[cur ^ 1], sdy);
-      stage512_write(xT[cur ^ 1], sx);
-    }
-    cur ^= 1;
   }
-  __syncthreads();  // last chunk's MFMA readers done before ftile reuse
 
   if (do_bias) {
     bias_acc += __shfl_xor(bias_acc, 1, 64);
