@@ -131,7 +131,7 @@ struct WgArgs {
   int G;
 };
 extern "C" __global__ void k_wgrad_mfma(WgArgs, int);
-extern "C" __global__ void k_wgrad_mfma256(WgArgs, int, int);
+extern "C" __global__ void k_wgrad_mfma256(WgArgs, int, int, float*, int*);
 extern "C" __global__ void k_cls_head_fwd(const unsigned short*, const unsigned short*,
                                           const unsigned short*, const long long*,
                                           unsigned short*, float*, float*, int, int, int);
@@ -596,18 +596,19 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
   for (int g = 0; g < G; ++g)
     all256 &= (args.nk[g * 2] % 256 == 0) && (args.nk[g * 2 + 1] % 256 == 0);
   // A/B knobs (re-measurement): GA_WGRAD_TILE=128 forces the 4-wave
-  // 128-tile kernel even when 256 tiles fit; GA_WGRAD_SPLITS=N forces
-  // R-splitting. Splits default OFF: the atomic scalar combine measured
-  // CATASTROPHIC at splits=4 on the fused bench (19.3k -> 8.4k samples/s;
-  // ~51M serialized fp32 atomics per window) -- a scratch+fixup combine
-  // would be needed to make R-splits pay.
+  // 128-tile kernel even when 256 tiles fit; GA_WGRAD_SPLITS=N overrides
+  // the R-split factor (0/1 disables). Default: auto-split so the launch
+  // reaches >= 2 workgroups per CU, combined split-K-fixup style (scratch
+  // partials + last-arrival reduce). The naive scalar-atomic combine
+  // measured 19.3k -> 8.4k samples/s (~51M serialized fp32 atomics per
+  // window); the fixup pays only ~2x coalesced tile traffic.
   static const int env_tile = [] {
     const char* v = getenv("GA_WGRAD_TILE");
     return v ? atoi(v) : 256;
   }();
   static const int env_splits = [] {
     const char* v = getenv("GA_WGRAD_SPLITS");
-    return v ? atoi(v) : 1;
+    return v ? atoi(v) : -1;  // -1 = auto
   }();
   all256 &= env_tile == 256;
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -615,11 +616,35 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     long long nt256 = 0;
     for (int g = 0; g < G; ++g)
       nt256 += (long long)(args.nk[g * 2] / 256) * (args.nk[g * 2 + 1] / 256);
+    const long long chunks = R / 64;
     int splits = env_splits;
-    while (splits > 1 && (long long)splits * 4 > R / 64) splits /= 2;
+    if (splits < 0) {
+      splits = 1;
+      while (nt256 * splits < 512 && splits < 8 &&
+             chunks % (splits * 2) == 0 && chunks / (splits * 2) >= 4)
+        splits *= 2;
+    }
+    if (splits < 1) splits = 1;
+    while (splits > 1 && (long long)splits * 4 > chunks) splits /= 2;
+    float* scratch = nullptr;
+    int* counters = nullptr;
+    if (splits > 1) {
+      // cached device workspace (leaky holders: stable pointers for graph
+      // replays, no static-destructor-vs-context teardown hazard).
+      // Allocated during eager warmup; replays reuse the pointers.
+      static auto* scratch_t = new at::Tensor();
+      static auto* counters_t = new at::Tensor();
+      const long long need = nt256 * (long long)splits * 65536;
+      if (!scratch_t->defined() || scratch_t->numel() < need)
+        *scratch_t = at::empty({need}, accs[0].options().dtype(at::kFloat));
+      if (!counters_t->defined() || counters_t->numel() < nt256)
+        *counters_t = at::zeros({nt256}, accs[0].options().dtype(at::kInt));
+      scratch = scratch_t->data_ptr<float>();
+      counters = counters_t->data_ptr<int>();
+    }
     hipLaunchKernelGGL(k_wgrad_mfma256, dim3((unsigned)(nt256 * splits)),
-                       dim3(512), 65536, stream, args, (int)R,
-                       splits < 1 ? 1 : splits);
+                       dim3(512), 65536, stream, args, (int)R, splits,
+                       scratch, counters);
   } else {
     hipLaunchKernelGGL(k_wgrad_mfma, dim3((unsigned)ntiles), dim3(256), 65536,
                        stream, args, (int)R);

@@ -272,12 +272,19 @@ static __device__ __forceinline__ void stage512_write(unsigned short* lds,
 // splits > 1: R-split parallelism for the window-fused regime. At R = 4096
 // the tile count (~196 for bert-small) is below one workgroup per CU, so a
 // single-pass walk leaves the chip latency-bound (measured 322 us/window,
-// 12% MfmaUtil); `splits` blocks each reduce an R-slice of the same tile
-// and combine with fp32 atomics in the epilogue (extra output traffic
-// ~tiles*256KB*(splits-1), ~25 us at splits=4 -- far under the win).
-// splits == 1 keeps the deterministic read-modify-write epilogue.
+// 12% MfmaUtil). `splits` blocks each reduce an R-slice of the same tile.
+// Combine strategy matters enormously: scalar fp32 atomics on the output
+// measured CATASTROPHIC (19.3k -> 8.4k samples/s, ~51M serialized
+// atomics/window), so this uses the split-K FIXUP pattern instead: every
+// split stores its fp32 partial tile to `scratch` (coalesced float4),
+// bumps an arrival counter, and the LAST-arriving block reduces all
+// partials into the accum slice (one coalesced float4 RMW pass) and
+// resets the counter -- self-cleaning, so the launch is hipGraph-replay
+// safe with no host-side zeroing.
+// splits == 1 keeps the deterministic single-pass RMW epilogue.
 extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
-    WgArgs args, int R, int splits) {
+    WgArgs args, int R, int splits, float* __restrict__ scratch,
+    int* __restrict__ counters) {
   int id;
   {
     const int nwg = gridDim.x, orig = blockIdx.x;
@@ -288,6 +295,7 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
   const int tiles_total = gridDim.x / splits;
   const int split = id / tiles_total;
   id = id % tiles_total;
+  const int tile_gid = id;  // scratch/counter index for the fixup combine
   // this block's R-slice, in whole 64-row chunks
   const int nchunk = R / 64;
   const int cper = (nchunk + splits - 1) / splits;
@@ -329,7 +337,20 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
   const int brh = (threadIdx.x & 1) * 32;  // r half
   float bias_acc = 0.f;
 
-  if (r_lo >= r_hi) return;  // more splits than chunks (tiny R)
+  if (r_lo >= r_hi) {
+    // empty R-slice (uneven split override): the fixup still reads this
+    // split's partial and counts its arrival -- publish zeros and proceed
+    // to the combine protocol below
+    if (splits > 1) {
+      float* srow = scratch + ((long long)tile_gid * splits + split) * 65536;
+      const float4 z4 = make_float4(0.f, 0.f, 0.f, 0.f);
+      for (int e = threadIdx.x * 4; e < 65536; e += 512 * 4)
+        *(float4*)(srow + e) = z4;
+      goto arrival;
+    }
+    return;
+  }
+  {
   StageRegs512 sdy, sx;
   stage512_issue(dy + (long long)r_lo * N, N, n0, sdy);
   stage512_issue(x + (long long)r_lo * K, K, k0, sx);
@@ -407,6 +428,9 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
       const int n = threadIdx.x >> 1;
       const int kh = (threadIdx.x & 1) * 32;
       float* grow = acc_out + (long long)(n0 + n) * K + k0 + p * 64 + kh;
+      float* srow = scratch +
+                    ((long long)tile_gid * splits + split) * 65536 +
+                    (long long)n * 256 + p * 64 + kh;
 #pragma unroll
       for (int c = 0; c < 8; ++c) {
         const int k4 = kh + c * 4;
@@ -420,10 +444,7 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
         v.w = *(const float*)((char*)ftile + (long long)(k4 + 3) * 1024 +
                               ((n * 4) ^ (((k4 + 3) & 7) << 4)));
         if (splits > 1) {
-          atomicAdd(grow + c * 4 + 0, v.x);
-          atomicAdd(grow + c * 4 + 1, v.y);
-          atomicAdd(grow + c * 4 + 2, v.z);
-          atomicAdd(grow + c * 4 + 3, v.w);
+          *(float4*)(srow + c * 4) = v;  // coalesced partial store
         } else {
           float4 old = *(const float4*)(grow + c * 4);
           old.x += v.x; old.y += v.y; old.z += v.z; old.w += v.w;
@@ -432,4 +453,30 @@ extern "C" __global__ __launch_bounds__(512) void k_wgrad_mfma256(
       }
     }
   }
+  }  // compute + per-pass epilogue scope
+  if (splits <= 1) return;
+
+  // ---- fixup: last-arriving split reduces all partials into accum ----
+arrival:
+  __shared__ int is_last;
+  __threadfence();  // publish this block's partial before the arrival bump
+  __syncthreads();
+  if (threadIdx.x == 0)
+    is_last = (atomicAdd(&counters[tile_gid], 1) == splits - 1) ? 1 : 0;
+  __syncthreads();
+  if (!is_last) return;
+  __threadfence();  // acquire the other splits' partials
+  const float* base = scratch + (long long)tile_gid * splits * 65536;
+  for (int e = threadIdx.x * 4; e < 65536; e += 512 * 4) {
+    float4 s = *(const float4*)(base + e);
+    for (int sp = 1; sp < splits; ++sp) {
+      float4 v = *(const float4*)(base + (long long)sp * 65536 + e);
+      s.x += v.x; s.y += v.y; s.z += v.z; s.w += v.w;
+    }
+    float* g = acc_out + (long long)(n0 + (e >> 8)) * K + k0 + (e & 255);
+    float4 old = *(const float4*)g;
+    old.x += s.x; old.y += s.y; old.z += s.z; old.w += s.w;
+    *(float4*)g = old;
+  }
+  if (threadIdx.x == 0) counters[tile_gid] = 0;  // self-clean for replay
 }
