@@ -596,12 +596,15 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
   for (int g = 0; g < G; ++g)
     all256 &= (args.nk[g * 2] % 256 == 0) && (args.nk[g * 2 + 1] % 256 == 0);
   // A/B knobs (re-measurement): GA_WGRAD_TILE=128 forces the 4-wave
-  // 128-tile kernel even when 256 tiles fit; GA_WGRAD_SPLITS=N overrides
-  // the R-split factor (0/1 disables). Default: auto-split so the launch
-  // reaches >= 2 workgroups per CU, combined split-K-fixup style (scratch
-  // partials + last-arrival reduce). The naive scalar-atomic combine
-  // measured 19.3k -> 8.4k samples/s (~51M serialized fp32 atomics per
-  // window); the fixup pays only ~2x coalesced tile traffic.
+  // 128-tile kernel even when 256 tiles fit; GA_WGRAD_SPLITS=N forces
+  // R-splitting with the split-K-fixup combine (scratch partials +
+  // last-arrival reduce). Both combines MEASURED SLOWER at the fused
+  // bench shapes: scalar atomics 19.6k -> 8.4k (~51M serialized fp32
+  // atomics/window), coalesced fixup 19.6k -> 13.4k at auto-splits=4 --
+  // the kernel is operand-bandwidth-bound (panel re-reads ride L2), and
+  // splitting multiplies the in-flight working set past what L2 holds.
+  // Splits therefore default OFF; the machinery stays for re-measurement
+  // on future parts.
   static const int env_tile = [] {
     const char* v = getenv("GA_WGRAD_TILE");
     return v ? atoi(v) : 256;
@@ -617,14 +620,7 @@ void wgrad_mfma(std::vector<at::Tensor> xs, std::vector<at::Tensor> dys,
     for (int g = 0; g < G; ++g)
       nt256 += (long long)(args.nk[g * 2] / 256) * (args.nk[g * 2 + 1] / 256);
     const long long chunks = R / 64;
-    int splits = env_splits;
-    if (splits < 0) {
-      splits = 1;
-      while (nt256 * splits < 512 && splits < 8 &&
-             chunks % (splits * 2) == 0 && chunks / (splits * 2) >= 4)
-        splits *= 2;
-    }
-    if (splits < 1) splits = 1;
+    int splits = env_splits < 1 ? 1 : env_splits;
     while (splits > 1 && (long long)splits * 4 > chunks) splits /= 2;
     float* scratch = nullptr;
     int* counters = nullptr;
