@@ -305,7 +305,46 @@ class AccumEngine:
             self._allreduce_accum()
             self.apply(lr)
             return
-        self._sharded_apply(self.lr_at(self.global_step) if lr is None else lr)
+        lr = self.lr_at(self.global_step) if lr is None else lr
+        if not getattr(self, "_shard_active", False):
+            # first sharded boundary: probe the RS/AG collectives on tiny
+            # scratch tensors BEFORE touching real state -- if this stack
+            # rejects them, fall back to the replicated boundary permanently
+            # instead of failing the job (collective failures on real
+            # buffers still raise; no silent divergence).
+            import os
+
+            if os.environ.get("GA_SHARD_APPLY", "1") == "0" or \
+                    not self._probe_shard_collectives():
+                self.shard_apply = False
+                self._allreduce_accum()
+                self.apply(lr)
+                return
+        self._sharded_apply(lr)
+
+    def _probe_shard_collectives(self) -> bool:
+        import logging
+
+        import torch.distributed as dist
+
+        st = self.state
+        W = self.world_size
+        r = dist.get_rank(self.group) if self.group is not None else dist.get_rank()
+        try:
+            probe = torch.zeros(64 * W, device=st.accum.device, dtype=torch.float32)
+            dist.reduce_scatter_tensor(probe[r * 64 : (r + 1) * 64], probe,
+                                       group=self.group)
+            pm = torch.zeros(64 * W, device=st.model.device, dtype=st.model.dtype)
+            dist.all_gather_into_tensor(pm, pm[r * 64 : (r + 1) * 64],
+                                        group=self.group)
+            if st.accum.is_cuda:
+                torch.cuda.synchronize()
+            return True
+        except (RuntimeError, ValueError) as exc:
+            logging.getLogger("ga_amd.engine").warning(
+                "sharded boundary collectives unsupported on this stack "
+                "(%s); falling back to replicated all-reduce", exc)
+            return False
 
     def _sharded_apply(self, lr: float) -> None:
         import torch.distributed as dist
