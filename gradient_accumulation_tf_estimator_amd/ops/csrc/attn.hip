@@ -47,6 +47,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 #define ATTN_D 64
+#define H_OF(nh) ((nh) * ATTN_D)
 #define LOG2E 1.44269504088896340736f
 
 static inline __device__ unsigned short f2bf_rne(float f) {
@@ -295,6 +296,174 @@ __device__ __forceinline__ void attn_fwd_body(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 4-wave S=128 variants: ONE workgroup per (b,h) stages the operand panels
+// once and its four waves each run one 32-row q/k-tile -- 4x less staging
+// work and one barrier per block vs four 1-wave blocks re-staging the same
+// panels (the 1-wave geometry leaves only 4 lone waves per CU at the bench
+// shape). 256-thread staging helpers mirror stage_64/stage_64_T.
+// ---------------------------------------------------------------------------
+
+static __device__ __forceinline__ void stage128_w4(const unsigned short* g,
+                                                   int row_stride,
+                                                   unsigned short* lds) {
+  // [128][64] panel, 256 threads: 4 chunks of 16 B each
+  bf16x8 v[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int c = threadIdx.x + i * 256;
+    v[i] = *(const bf16x8*)(g + (long long)(c >> 3) * row_stride + ((c & 7) << 3));
+  }
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int c = threadIdx.x + i * 256;
+    *(bf16x8*)((char*)lds + swz(c >> 3, (c & 7) << 4)) = v[i];
+  }
+}
+
+static __device__ __forceinline__ void stage128_T_w4(const unsigned short* g,
+                                                     int row_stride,
+                                                     unsigned short* lds) {
+  // [128 s][64 d] -> LDS [64 d][128 s] (swz256 rows), 256 threads
+  bf16x8 r[4];
+  const int blk = threadIdx.x;  // 256 4s-x-8d sub-blocks exactly
+  const int s0 = (blk / 8) * 4;
+  const int d0 = (blk % 8) * 8;
+#pragma unroll
+  for (int t = 0; t < 4; ++t)
+    r[t] = *(const bf16x8*)(g + (long long)(s0 + t) * row_stride + d0);
+  const unsigned short* u0 = (const unsigned short*)&r[0];
+  const unsigned short* u1 = (const unsigned short*)&r[1];
+  const unsigned short* u2 = (const unsigned short*)&r[2];
+  const unsigned short* u3 = (const unsigned short*)&r[3];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    ush4v pack = {u0[c], u1[c], u2[c], u3[c]};
+    *(ush4v*)((char*)lds + swz256(d0 + c, s0 * 2)) = pack;
+  }
+}
+
+template <bool DROP>
+__device__ __forceinline__ void attn_fwd4_body(
+    const unsigned short* __restrict__ qkv, unsigned short* __restrict__ out,
+    float* __restrict__ lse_out, int B, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  constexpr int S = 128;
+  constexpr int NT = 4;
+  const int bh = blockIdx.x;
+  const int b = bh / nh, h = bh % nh;
+  const int qt = threadIdx.x >> 6;  // wave = q-tile
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;            // [128][64] swz
+  unsigned short* Vtlds = (unsigned short*)(smem + 16384); // [64][128] swz256
+  float* maskf = (float*)(smem + 16384 + 16384);           // [S] 0/1
+
+  const long long base = ((long long)b * S) * (3LL * H_OF(nh)) + (long long)h * ATTN_D;
+  stage128_w4(qkv + base + H_OF(nh), 3 * H_OF(nh), Klds);
+  stage128_T_w4(qkv + base + 2 * H_OF(nh), 3 * H_OF(nh), Vtlds);
+  if (mask) {
+    for (int i = threadIdx.x; i < S; i += 256)
+      maskf[i] = mask[(long long)b * S + i] ? 1.f : 0.f;
+  }
+  __syncthreads();
+
+  const int H = H_OF(nh);
+  const int q0 = qt * 32;
+  bf16x8 qf[4];
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk)
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+
+  f32x16 acc[NT];
+#pragma unroll
+  for (int t = 0; t < NT; ++t) acc[t] = (f32x16)(0.f);
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 a = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
+    }
+
+  if (mask) {
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r)
+        if (maskf[32 * t + crow(r, hi)] == 0.f) acc[t][r] = -INFINITY;
+  }
+
+  const float scale2 = 0.125f * LOG2E;
+  float m2 = -1e30f;
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) m2 = fmaxf(m2, acc[t][r]);
+  m2 = fmaxf(m2, __shfl_xor(m2, 32, 64)) * scale2;
+  float sum = 0.f;
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      acc[t][r] = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - m2);
+      sum += acc[t][r];
+    }
+  sum += __shfl_xor(sum, 32, 64);
+  const float inv_sum = 1.f / sum;
+  if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m2 + log2f(sum);
+
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      acc[t][r] *= inv_sum;
+      if (DROP)
+        acc[t][r] *= drop_mult(seed, bh, S, q0 + lo31, 32 * t + crow(r, hi),
+                               p_drop, inv_keep);
+    }
+
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    f32x16 oc = (f32x16)(0.f);
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int halfk = 0; halfk < 2; ++halfk) {
+        bf16x8 pa = cvt_swap(acc[t], halfk * 8);
+        bf16x8 bv = *(const bf16x8*)((char*)Vtlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfk + hi * 8) * 2));
+        oc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, bv, oc, 0, 0, 0);
+      }
+    write_tile_bf16(out + ((long long)b * S + q0) * H + h * ATTN_D + dt * 32 + lo31,
+                    H, hi, oc);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_fwd4_128(
+    const unsigned short* qkv, unsigned short* out, float* lse_out, int B,
+    int nh, const unsigned char* mask) {
+  attn_fwd4_body<false>(qkv, out, lse_out, B, nh, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_fwd4_drop_128(
+    const unsigned short* qkv, unsigned short* out, float* lse_out, int B,
+    int nh, const unsigned char* mask, const unsigned long long* seed,
+    float p_drop) {
+  attn_fwd4_body<true>(qkv, out, lse_out, B, nh, mask, seed, p_drop);
+}
+
 #define GA_ATTN_FWD_INST(S)                                                   \
   extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_##S(            \
       const unsigned short* qkv, unsigned short* out, float* lse_out, int B,  \
@@ -458,6 +627,140 @@ __device__ __forceinline__ void attn_bwd_q_body(
   }
 }
 
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_q4_body(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int nh, int publish_d,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  constexpr int S = 128;
+  constexpr int NT = 4;
+  const int bh = blockIdx.x;
+  const int b = bh / nh, h = bh % nh;
+  const int qt = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int H = H_OF(nh);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;             // [128][64] swz
+  unsigned short* Vlds = (unsigned short*)(smem + 16384);   // [128][64] swz
+  unsigned short* Ktlds = (unsigned short*)(smem + 32768);  // [64][128] swz256
+  float* maskf = (float*)(smem + 49152);                    // [S] 0/1
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  stage128_w4(qkv + base + H, 3 * H, Klds);
+  stage128_w4(qkv + base + 2 * H, 3 * H, Vlds);
+  stage128_T_w4(qkv + base + H, 3 * H, Ktlds);
+  if (mask) {
+    for (int i = threadIdx.x; i < S; i += 256)
+      maskf[i] = mask[(long long)b * S + i] ? 1.f : 0.f;
+  }
+  __syncthreads();
+
+  const int q0 = qt * 32;
+  const float lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
+
+  float D_q;
+  {
+    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
+    const unsigned short* orw = out + obase + (long long)(q0 + lo31) * H + hi * 32;
+    float sd = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
+      bf16x8 ov = *(const bf16x8*)(orw + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
+    }
+    D_q = sd + __shfl_xor(sd, 32, 64);
+    if (publish_d && hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+  }
+
+  bf16x8 qf[4], dof[4];
+  const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+  const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
+#pragma unroll
+  for (int kk = 0; kk < 4; ++kk) {
+    qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+    dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+  }
+
+  f32x16 acc[NT], dacc[NT];
+#pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    acc[t] = (f32x16)(0.f);
+    dacc[t] = (f32x16)(0.f);
+  }
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+      bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
+      acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
+      dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
+    }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int k = 32 * t + crow(r, hi);
+      float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+      if (mask && maskf[k] == 0.f) p = 0.f;
+      float dp = dacc[t][r];
+      if (DROP)
+        dp *= drop_mult(seed, bh, S, q0 + lo31, k, p_drop, inv_keep);
+      dacc[t][r] = scale * p * (dp - D_q);
+    }
+
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    f32x16 a = (f32x16)(0.f);
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int halfk = 0; halfk < 2; ++halfk) {
+        bf16x8 as = cvt_swap(dacc[t], halfk * 8);
+        bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfk + hi * 8) * 2));
+        a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bk, a, 0, 0, 0);
+      }
+    write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, a);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_q4_128(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int nh, int publish_d,
+    const unsigned char* mask) {
+  attn_bwd_q4_body<false>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh,
+                          publish_d, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_q4_drop_128(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int nh, int publish_d,
+    const unsigned char* mask, const unsigned long long* seed, float p_drop) {
+  attn_bwd_q4_body<true>(qkv, out, dout, lse_in, Dtab, dqkv, B, nh,
+                         publish_d, mask, seed, p_drop);
+}
+
 #define GA_ATTN_BWDQ_INST(S)                                                  \
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_##S(          \
       const unsigned short* qkv, const unsigned short* out,                   \
@@ -594,6 +897,130 @@ __device__ __forceinline__ void attn_bwd_kv_body(
     write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
                     3 * H, hi, ak_);
   }
+}
+
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_kv4_body(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  constexpr int S = 128;
+  constexpr int NT = 4;
+  const int bh = blockIdx.x;
+  const int b = bh / nh, h = bh % nh;
+  const int kt = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int H = H_OF(nh);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dOtlds = (unsigned short*)smem;           // [64][128] swz256
+  unsigned short* Qtlds = (unsigned short*)(smem + 16384);  // [64][128] swz256
+  float* lsetab = (float*)(smem + 32768);                   // [S]
+  float* dtab = (float*)(smem + 32768 + 512);               // [S]
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  stage128_T_w4(dout + obase, H, dOtlds);
+  stage128_T_w4(qkv + base, 3 * H, Qtlds);
+  for (int i = threadIdx.x; i < S; i += 256) {
+    lsetab[i] = lse_in[((long long)b * nh + h) * S + i];
+    dtab[i] = Dtab[((long long)b * nh + h) * S + i];
+  }
+  __syncthreads();
+
+  const int k0 = kt * 32;
+  bf16x8 kf[4], vf[4];
+  {
+    const unsigned short* krow = qkv + base + H + (long long)(k0 + lo31) * 3 * H;
+    const unsigned short* vrow = qkv + base + 2 * H + (long long)(k0 + lo31) * 3 * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      kf[kk] = *(const bf16x8*)(krow + kk * 16 + hi * 8);
+      vf[kk] = *(const bf16x8*)(vrow + kk * 16 + hi * 8);
+    }
+  }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  const int k_lane = k0 + lo31;
+  const float kvalid =
+      mask ? (mask[(long long)b * S + k_lane] ? 1.f : 0.f) : 1.f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+  f32x16 p_qt[NT], ds_qt[NT];
+#pragma unroll
+  for (int t = 0; t < NT; ++t) {
+    f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
+    const unsigned short* qrow = qkv + base + (long long)(t * 32 + lo31) * 3 * H;
+    const unsigned short* drow = dout + obase + (long long)(t * 32 + lo31) * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      bf16x8 aq = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+      bf16x8 ad = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[kk], sacc, 0, 0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[kk], dpacc, 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+      const float p =
+          kvalid * __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+      float m = 1.f, dp = dpacc[r];
+      if (DROP) {
+        m = drop_mult(seed, bh, S, q, k_lane, p_drop, inv_keep);
+        dp *= m;
+      }
+      p_qt[t][r] = p * m;
+      ds_qt[t][r] = scale * p * (dp - dtab[q]);
+    }
+  }
+
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    f32x16 av_ = (f32x16)(0.f), ak_ = (f32x16)(0.f);
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int halfq = 0; halfq < 2; ++halfq) {
+        bf16x8 ap = cvt_swap(p_qt[t], halfq * 8);
+        bf16x8 as = cvt_swap(ds_qt[t], halfq * 8);
+        bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfq + hi * 8) * 2));
+        bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
+                                     swz256(dt * 32 + lo31,
+                                            (32 * t + 16 * halfq + hi * 8) * 2));
+        av_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, av_, 0, 0, 0);
+        ak_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, ak_, 0, 0, 0);
+      }
+    write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, av_);
+    write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, ak_);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_kv4_128(
+    const unsigned short* qkv, const unsigned short* dout,
+    const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,
+    int nh, const unsigned char* mask) {
+  attn_bwd_kv4_body<false>(qkv, dout, lse_in, Dtab, dqkv, B, nh, mask,
+                           nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_kv4_drop_128(
+    const unsigned short* qkv, const unsigned short* dout,
+    const float* lse_in, const float* Dtab, unsigned short* dqkv, int B,
+    int nh, const unsigned char* mask, const unsigned long long* seed,
+    float p_drop) {
+  attn_bwd_kv4_body<true>(qkv, dout, lse_in, Dtab, dqkv, B, nh, mask,
+                          seed, p_drop);
 }
 
 #define GA_ATTN_BWDKV_INST(S)                                                 \

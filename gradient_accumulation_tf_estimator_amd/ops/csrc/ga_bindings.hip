@@ -96,6 +96,35 @@ extern "C" __global__ void k_emb3_fwd(const long long*, const long long*,
       const float*, unsigned short*, int, int, const unsigned char*,           \
       const unsigned long long*, float);
 GA_DECL_ATTN(32) GA_DECL_ATTN(64) GA_DECL_ATTN(96) GA_DECL_ATTN(128)
+extern "C" __global__ void k_attn_fwd4_128(
+    const unsigned short*, unsigned short*, float*, int, int,
+    const unsigned char*);
+extern "C" __global__ void k_attn_fwd4_drop_128(
+    const unsigned short*, unsigned short*, float*, int, int,
+    const unsigned char*, const unsigned long long*, float);
+extern "C" __global__ void k_attn_bwd_q4_128(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, float*, unsigned short*, int, int, int,
+    const unsigned char*);
+extern "C" __global__ void k_attn_bwd_q4_drop_128(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, float*, unsigned short*, int, int, int,
+    const unsigned char*, const unsigned long long*, float);
+extern "C" __global__ void k_attn_bwd_kv4_128(
+    const unsigned short*, const unsigned short*, const float*,
+    const float*, unsigned short*, int, int, const unsigned char*);
+extern "C" __global__ void k_attn_bwd_kv4_drop_128(
+    const unsigned short*, const unsigned short*, const float*,
+    const float*, unsigned short*, int, int, const unsigned char*,
+    const unsigned long long*, float);
+// 4-wave S=128 path A/B (one workgroup per (b,h), panels staged once)
+static inline bool attn_w4_on() {
+  static const bool on = [] {
+    const char* v = getenv("GA_ATTN_W4");
+    return !v || atoi(v) != 0;
+  }();
+  return on;
+}
 extern "C" __global__ void k_attn_fwd_big(const unsigned short*, unsigned short*,
                                            float*, int, int, int,
                                            const unsigned char*);
@@ -434,6 +463,21 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh,
     TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd_big launch failed");
     return {out, lse};
   }
+  if (S == 128 && attn_w4_on()) {
+    const size_t lds4 = 16384 + 16384 + 512;
+    if (drop)
+      hipLaunchKernelGGL(k_attn_fwd4_drop_128, dim3(B * (int)nh), dim3(256),
+                         lds4, stream, (const unsigned short*)qkv.data_ptr(),
+                         (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                         B, (int)nh, mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_fwd4_128, dim3(B * (int)nh), dim3(256),
+                         lds4, stream, (const unsigned short*)qkv.data_ptr(),
+                         (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                         B, (int)nh, mp);
+    TORCH_CHECK(hipGetLastError() == hipSuccess, "k_attn_fwd4 launch failed");
+    return {out, lse};
+  }
   const size_t lds = 16384 + 64 * 256 + 512;  // K + V^T panels + mask table
   if (drop) {
     void (*fk)(const unsigned short*, unsigned short*, float*, int, int,
@@ -507,6 +551,41 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
                          (const unsigned short*)dout.data_ptr(),
                          lse.data_ptr<float>(), Dtab.data_ptr<float>(),
                          (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, mp);
+  } else if (S == 128 && attn_w4_on()) {
+    const size_t lds_q4 = 16384 * 3 + 512;
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_q4_drop_128, dim3(B * (int)nh), dim3(256),
+                         lds_q4, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_q4_128, dim3(B * (int)nh), dim3(256),
+                         lds_q4, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1, mp);
+    const size_t lds_kv4 = 16384 * 2 + 1024;
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_kv4_drop_128, dim3(B * (int)nh), dim3(256),
+                         lds_kv4, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_kv4_128, dim3(B * (int)nh), dim3(256),
+                         lds_kv4, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, mp);
   } else {
     const size_t lds_q = 16384 * 3 + 512;  // K + V + K^T + mask table
     if (drop) {
