@@ -110,6 +110,13 @@ extern "C" __global__ void k_attn_bwd_q4_drop_128(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const float*, float*, unsigned short*, int, int, int,
     const unsigned char*, const unsigned long long*, float);
+extern "C" __global__ void k_attn_bwd8_128(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, unsigned short*, int, int, const unsigned char*);
+extern "C" __global__ void k_attn_bwd8_drop_128(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, unsigned short*, int, int, const unsigned char*,
+    const unsigned long long*, float);
 extern "C" __global__ void k_attn_bwd_kv4_128(
     const unsigned short*, const unsigned short*, const float*,
     const float*, unsigned short*, int, int, const unsigned char*);
@@ -552,39 +559,24 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
                          lse.data_ptr<float>(), Dtab.data_ptr<float>(),
                          (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, mp);
   } else if (S == 128 && attn_w4_on()) {
-    const size_t lds_q4 = 16384 * 3 + 512;
+    // merged 8-wave backward: one launch per layer, D table through LDS
+    const size_t lds8 = 16384 * 5 + 3 * 512;
     if (drop)
-      hipLaunchKernelGGL(k_attn_bwd_q4_drop_128, dim3(B * (int)nh), dim3(256),
-                         lds_q4, stream,
+      hipLaunchKernelGGL(k_attn_bwd8_drop_128, dim3(B * (int)nh), dim3(512),
+                         lds8, stream,
                          (const unsigned short*)qkv.data_ptr(),
                          (const unsigned short*)out.data_ptr(),
                          (const unsigned short*)dout.data_ptr(),
-                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1,
-                         mp, sp, pd);
-    else
-      hipLaunchKernelGGL(k_attn_bwd_q4_128, dim3(B * (int)nh), dim3(256),
-                         lds_q4, stream,
-                         (const unsigned short*)qkv.data_ptr(),
-                         (const unsigned short*)out.data_ptr(),
-                         (const unsigned short*)dout.data_ptr(),
-                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
-                         (unsigned short*)dqkv.data_ptr(), B, (int)nh, 1, mp);
-    const size_t lds_kv4 = 16384 * 2 + 1024;
-    if (drop)
-      hipLaunchKernelGGL(k_attn_bwd_kv4_drop_128, dim3(B * (int)nh), dim3(256),
-                         lds_kv4, stream,
-                         (const unsigned short*)qkv.data_ptr(),
-                         (const unsigned short*)dout.data_ptr(),
-                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         lse.data_ptr<float>(),
                          (unsigned short*)dqkv.data_ptr(), B, (int)nh,
                          mp, sp, pd);
     else
-      hipLaunchKernelGGL(k_attn_bwd_kv4_128, dim3(B * (int)nh), dim3(256),
-                         lds_kv4, stream,
+      hipLaunchKernelGGL(k_attn_bwd8_128, dim3(B * (int)nh), dim3(512),
+                         lds8, stream,
                          (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
                          (const unsigned short*)dout.data_ptr(),
-                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         lse.data_ptr<float>(),
                          (unsigned short*)dqkv.data_ptr(), B, (int)nh, mp);
   } else {
     const size_t lds_q = 16384 * 3 + 512;  // K + V + K^T + mask table
