@@ -102,26 +102,11 @@ extern "C" __global__ void k_attn_fwd4_128(
 extern "C" __global__ void k_attn_fwd4_drop_128(
     const unsigned short*, unsigned short*, float*, int, int,
     const unsigned char*, const unsigned long long*, float);
-extern "C" __global__ void k_attn_bwd_q4_128(
-    const unsigned short*, const unsigned short*, const unsigned short*,
-    const float*, float*, unsigned short*, int, int, int,
-    const unsigned char*);
-extern "C" __global__ void k_attn_bwd_q4_drop_128(
-    const unsigned short*, const unsigned short*, const unsigned short*,
-    const float*, float*, unsigned short*, int, int, int,
-    const unsigned char*, const unsigned long long*, float);
 extern "C" __global__ void k_attn_bwd8_128(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const float*, unsigned short*, int, int, const unsigned char*);
 extern "C" __global__ void k_attn_bwd8_drop_128(
     const unsigned short*, const unsigned short*, const unsigned short*,
-    const float*, unsigned short*, int, int, const unsigned char*,
-    const unsigned long long*, float);
-extern "C" __global__ void k_attn_bwd_kv4_128(
-    const unsigned short*, const unsigned short*, const float*,
-    const float*, unsigned short*, int, int, const unsigned char*);
-extern "C" __global__ void k_attn_bwd_kv4_drop_128(
-    const unsigned short*, const unsigned short*, const float*,
     const float*, unsigned short*, int, int, const unsigned char*,
     const unsigned long long*, float);
 // 4-wave S=128 path A/B (one workgroup per (b,h), panels staged once)
