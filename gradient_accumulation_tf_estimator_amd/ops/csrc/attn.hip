@@ -777,6 +777,232 @@ __device__ __forceinline__ void attn_bwd_kv_body(
   }
 }
 
+// ---------------------------------------------------------------------------
+// MERGED 8-wave S=128 backward: ONE workgroup per (b,h) does the whole
+// attention backward. Waves 0-3 are q-tiles (dQ; they compute the D table
+// and publish it through LDS -- no global Dtab traffic, no separate D
+// kernel), waves 4-7 are k-tiles (dK, dV). All five operand panels
+// (K, V, K^T, Q^T, dO^T) are staged ONCE for both roles (the two-kernel
+// path stages seven panels and bounces D through HBM). One launch per
+// layer instead of two.
+// ---------------------------------------------------------------------------
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd8_body(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, unsigned short* __restrict__ dqkv,
+    int B, int nh, const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  constexpr int S = 128;
+  constexpr int NT = 4;
+  const int bh = blockIdx.x;
+  const int b = bh / nh, h = bh % nh;
+  const int wave = threadIdx.x >> 6;  // 0-3 q-tiles, 4-7 k-tiles
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int H = H_OF(nh);
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;             // [128][64] swz
+  unsigned short* Vlds = (unsigned short*)(smem + 16384);   // [128][64] swz
+  unsigned short* Ktlds = (unsigned short*)(smem + 32768);  // [64][128] swz256
+  unsigned short* Qtlds = (unsigned short*)(smem + 49152);  // [64][128] swz256
+  unsigned short* dOtlds = (unsigned short*)(smem + 65536); // [64][128] swz256
+  float* lsetab = (float*)(smem + 81920);                   // [S]
+  float* dtab = (float*)(smem + 82432);                     // [S]
+  float* maskf = (float*)(smem + 82944);                    // [S]
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  // 512 threads stage five panels: the low half does the three bwd_q
+  // panels, the high half the two transposed kv panels
+  if (threadIdx.x < 256) {
+    stage128_w4t(qkv + base + H, 3 * H, Klds, threadIdx.x);
+    stage128_w4t(qkv + base + 2 * H, 3 * H, Vlds, threadIdx.x);
+    stage128_T_w4t(qkv + base + H, 3 * H, Ktlds, threadIdx.x);
+  } else {
+    const int tid = threadIdx.x - 256;
+    stage128_T_w4t(qkv + base, 3 * H, Qtlds, tid);
+    stage128_T_w4t(dout + obase, H, dOtlds, tid);
+  }
+  for (int i = threadIdx.x; i < S; i += 512)
+    lsetab[i] = lse_in[((long long)b * nh + h) * S + i];
+  if (mask) {
+    for (int i = threadIdx.x; i < S; i += 512)
+      maskf[i] = mask[(long long)b * S + i] ? 1.f : 0.f;
+  }
+
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+
+  // phase 1: q-waves compute + publish the D table; k-waves preload their
+  // K/V row fragments. ONE barrier at a single program point for all waves.
+  float D_q = 0.f;
+  bf16x8 kf[4], vf[4];
+  if (wave < 4) {
+    const int q0 = wave * 32;
+    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
+    const unsigned short* orw = out + obase + (long long)(q0 + lo31) * H + hi * 32;
+    float sd = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
+      bf16x8 ov = *(const bf16x8*)(orw + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
+    }
+    D_q = sd + __shfl_xor(sd, 32, 64);
+    if (hi == 0) dtab[q0 + lo31] = D_q;
+  } else {
+    const int k0 = (wave - 4) * 32;
+    const unsigned short* krow = qkv + base + H + (long long)(k0 + lo31) * 3 * H;
+    const unsigned short* vrow = qkv + base + 2 * H + (long long)(k0 + lo31) * 3 * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      kf[kk] = *(const bf16x8*)(krow + kk * 16 + hi * 8);
+      vf[kk] = *(const bf16x8*)(vrow + kk * 16 + hi * 8);
+    }
+  }
+  __syncthreads();  // panels + D table ready for every wave
+
+  if (wave < 4) {
+    // ---- q-tile role: dQ ----
+    const int qt = wave;
+    const int q0 = qt * 32;
+    const float lse2 = lsetab[q0 + lo31];
+    bf16x8 qf[4], dof[4];
+    const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+    const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+      dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+    }
+    f32x16 acc[NT], dacc[NT];
+#pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      acc[t] = (f32x16)(0.f);
+      dacc[t] = (f32x16)(0.f);
+    }
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
+        dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
+      }
+#pragma unroll
+    for (int t = 0; t < NT; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int k = 32 * t + crow(r, hi);
+        float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+        if (mask && maskf[k] == 0.f) p = 0.f;
+        float dp = dacc[t][r];
+        if (DROP)
+          dp *= drop_mult(seed, bh, S, q0 + lo31, k, p_drop, inv_keep);
+        dacc[t][r] = scale * p * (dp - D_q);
+      }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+      f32x16 a = (f32x16)(0.f);
+#pragma unroll
+      for (int t = 0; t < NT; ++t)
+#pragma unroll
+        for (int halfk = 0; halfk < 2; ++halfk) {
+          bf16x8 as = cvt_swap(dacc[t], halfk * 8);
+          bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
+                                       swz256(dt * 32 + lo31,
+                                              (32 * t + 16 * halfk + hi * 8) * 2));
+          a = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bk, a, 0, 0, 0);
+        }
+      write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
+                      3 * H, hi, a);
+    }
+  } else {
+    // ---- k-tile role: dK, dV (kf/vf preloaded in phase 1) ----
+    const int kt = wave - 4;
+    const int k0 = kt * 32;
+    const int k_lane = k0 + lo31;
+    const float kvalid =
+        mask ? (mask[(long long)b * S + k_lane] ? 1.f : 0.f) : 1.f;
+    f32x16 p_qt[NT], ds_qt[NT];
+#pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
+      const unsigned short* qrow = qkv + base + (long long)(t * 32 + lo31) * 3 * H;
+      const unsigned short* drow = dout + obase + (long long)(t * 32 + lo31) * H;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 aq = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+        bf16x8 ad = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[kk], sacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[kk], dpacc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float p =
+            kvalid * __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+        float m = 1.f, dp = dpacc[r];
+        if (DROP) {
+          m = drop_mult(seed, bh, S, q, k_lane, p_drop, inv_keep);
+          dp *= m;
+        }
+        p_qt[t][r] = p * m;
+        ds_qt[t][r] = scale * p * (dp - dtab[q]);
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+      f32x16 av_ = (f32x16)(0.f), ak_ = (f32x16)(0.f);
+#pragma unroll
+      for (int t = 0; t < NT; ++t)
+#pragma unroll
+        for (int halfq = 0; halfq < 2; ++halfq) {
+          bf16x8 ap = cvt_swap(p_qt[t], halfq * 8);
+          bf16x8 as = cvt_swap(ds_qt[t], halfq * 8);
+          bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
+                                       swz256(dt * 32 + lo31,
+                                              (32 * t + 16 * halfq + hi * 8) * 2));
+          bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
+                                       swz256(dt * 32 + lo31,
+                                              (32 * t + 16 * halfq + hi * 8) * 2));
+          av_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, av_, 0, 0, 0);
+          ak_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, ak_, 0, 0, 0);
+        }
+      write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                      3 * H, hi, av_);
+      write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                      3 * H, hi, ak_);
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(512) void k_attn_bwd8_128(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, unsigned short* dqkv,
+    int B, int nh, const unsigned char* mask) {
+  attn_bwd8_body<false>(qkv, out, dout, lse_in, dqkv, B, nh, mask,
+                        nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(512) void k_attn_bwd8_drop_128(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, unsigned short* dqkv,
+    int B, int nh, const unsigned char* mask,
+    const unsigned long long* seed, float p_drop) {
+  attn_bwd8_body<true>(qkv, out, dout, lse_in, dqkv, B, nh, mask,
+                       seed, p_drop);
+}
+
 #define GA_ATTN_BWDKV_INST(S)                                                 \
   extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_##S(         \
       const unsigned short* qkv, const unsigned short* dout,                  \
