@@ -171,7 +171,7 @@ def main():
     # ---- window fusion factor (exact reformulation: N micro-steps as one
     # fused fwd/bwd, engine.micro_step_many) ----
     fuse = 1
-    if use_cuda and args.graphs != "off" and args.fuse_micro != "off":
+    if use_cuda and args.fuse_micro != "off":
         if args.fuse_micro == "auto":
             for d in range(args.accum, 0, -1):
                 if args.accum % d == 0 and args.steps % d == 0 and \
@@ -202,12 +202,28 @@ def main():
             print(f"[bench] hipGraph capture failed, falling back to eager: {e}",
                   file=sys.stderr)
             graphed = None
-    if graphed is None:
-        fuse = 1
+    eager_fused = None
+    if graphed is None and fuse > 1:
+        # eager fused path (e.g. --graphs off for rocprof --pmc counter
+        # collection, which aborts under capture on this rocprofv3)
+        B_, S_ = args.micro_batch, args.seq_len
+        POOL_ = pool_ids.shape[0]
+
+        def eager_fused(i):
+            b = i // fuse
+            sel = [(b * fuse + j) % POOL_ for j in range(fuse)]
+            ids = pool_ids[sel].reshape(fuse * B_, S_)
+            lab = pool_lab[sel].reshape(fuse * B_)
+            msk = (None if pool_msk is None
+                   else pool_msk[sel].reshape(fuse * B_, S_))
+            loss = model.loss(ids, lab, attention_mask=msk)
+            return op.step_fused(loss, fuse)
 
     def step(i):
         if graphed is not None:
             return graphed(i)
+        if eager_fused is not None:
+            return eager_fused(i)
         return eager_micro_step(i)
 
     # ---- warmup ----

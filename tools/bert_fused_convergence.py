@@ -1,0 +1,167 @@
+#!/usr/bin/env python3
+"""End-to-end convergence equivalence: window-FUSED training vs the
+sequential micro-step chain on a LEARNABLE synthetic task.
+
+The reference's only success criterion is loss-curve overlap at equal
+effective batch (README.md:69-78, Loss_Step.png). Round 2's window fusion
+reformulates the window as one fused fwd/bwd -- exact by linearity -- and
+tests assert state equality over a few windows; this tool shows the claim
+holds over a real optimization trajectory: bert-small trained on a
+synthetic classification rule (label = parity of the sum of the first
+8 token ids -- linearly learnable from the embedding stream) with
+  A: sequential eager micro-steps (op.step per micro-batch)
+  B: window-fused graphed steps  (FusedWindowLoop, one replay per window)
+  C: window-fused with random key-padding masks (mask-aware rule)
+Writes loss-vs-update CSV + SVG and prints the tail-mean |A-B| gap.
+
+Usage (GPU): python tools/bert_fused_convergence.py --updates 300 --out profiles/
+"""
+
+import argparse
+import csv
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), ".."))
+
+import torch
+
+from gradient_accumulation_tf_estimator_amd import create_optimizer
+from gradient_accumulation_tf_estimator_amd.engine.graphs import FusedWindowLoop
+from gradient_accumulation_tf_estimator_amd.models.bert import (
+    CONFIGS, BertForSequenceClassification)
+
+K, B, S, V = 4, 8, 128, 30522
+SEED = 19830610
+
+
+def batches(n_micro, seed, masked=False):
+    g = torch.Generator().manual_seed(seed)
+    out = []
+    for _ in range(n_micro):
+        ids = torch.randint(0, V, (B, S), generator=g)
+        lab = (ids[:, :8].sum(1) % 2).long()
+        msk = None
+        if masked:
+            lens = torch.randint(16, S + 1, (B,), generator=g)
+            msk = (torch.arange(S)[None, :] < lens[:, None]).to(torch.uint8)
+        out.append((ids.cuda(), lab.cuda(),
+                    msk.cuda() if msk is not None else None))
+    return out
+
+
+def make(lr):
+    torch.manual_seed(SEED)
+    m = BertForSequenceClassification(CONFIGS["bert-small"]()).cuda().bfloat16()
+    m.train()
+    op = create_optimizer(m, lr, 10**6, 200,
+                          gradient_accumulation_multiplier=K, clip_norm=1.0,
+                          backend="hip")
+    return m, op
+
+
+def run_sequential(data, lr):
+    m, op = make(lr)
+    losses = []
+    acc = 0.0
+    for i, (ids, lab, _) in enumerate(data):
+        loss = m.loss(ids, lab)
+        acc += float(loss.detach().float())
+        if op.step(loss):
+            losses.append(acc / K)
+            acc = 0.0
+    return losses
+
+
+def run_fused(data, lr, masked=False):
+    m, op = make(lr)
+    sid = torch.zeros(K * B, S, dtype=torch.long, device="cuda")
+    slab = torch.zeros(K * B, dtype=torch.long, device="cuda")
+    smsk = torch.ones(K * B, S, dtype=torch.long, device="cuda") if masked else None
+
+    def loss_fn():
+        return m.loss(sid, slab, attention_mask=smsk)
+
+    loop = FusedWindowLoop(op.engine, loss_fn, n_micro=K, world=1)
+    losses = []
+    for w in range(len(data) // K):
+        blk = data[w * K : (w + 1) * K]
+        sid.copy_(torch.cat([b[0] for b in blk]))
+        slab.copy_(torch.cat([b[1] for b in blk]))
+        if masked:
+            smsk.copy_(torch.cat([b[2] for b in blk]).long())
+        losses.append(float(loop.step().detach().float()))
+    return losses
+
+
+def write_svg(path, curves):
+    import math
+
+    W, H = 860, 420
+    all_y = [y for _, ys in curves for y in ys]
+    y0, y1 = min(all_y), max(all_y)
+    pad = 0.05 * (y1 - y0 + 1e-9)
+    y0, y1 = y0 - pad, y1 + pad
+    n = max(len(ys) for _, ys in curves)
+    colors = ["#1f77b4", "#d62728", "#2ca02c"]
+
+    def pt(i, y):
+        return (40 + (W - 60) * i / max(n - 1, 1),
+                H - 30 - (H - 60) * (y - y0) / (y1 - y0))
+
+    parts = [f'<svg xmlns="http://www.w3.org/2000/svg" width="{W}" height="{H}">',
+             f'<rect width="{W}" height="{H}" fill="white"/>']
+    for ci, (name, ys) in enumerate(curves):
+        d = " ".join(f"{'M' if i == 0 else 'L'}{pt(i, y)[0]:.1f},{pt(i, y)[1]:.1f}"
+                     for i, y in enumerate(ys))
+        parts.append(f'<path d="{d}" fill="none" stroke="{colors[ci % 3]}" '
+                     f'stroke-width="1.2" opacity="0.85"/>')
+        parts.append(f'<text x="60" y="{20 + 16 * ci}" fill="{colors[ci % 3]}" '
+                     f'font-size="13">{name}</text>')
+    parts.append(f'<text x="{W//2}" y="{H-8}" font-size="12">optimizer update</text>')
+    parts.append("</svg>")
+    with open(path, "w") as f:
+        f.write("".join(parts))
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--updates", type=int, default=300)
+    p.add_argument("--lr", type=float, default=5e-5)
+    p.add_argument("--out", default="profiles")
+    args = p.parse_args()
+
+    data = batches(args.updates * K, seed=7)
+    data_m = batches(args.updates * K, seed=7, masked=True)
+
+    seq = run_sequential(data, args.lr)
+    fus = run_fused(data, args.lr)
+    fus_m = run_fused(data_m, args.lr, masked=True)
+
+    n = min(len(seq), len(fus))
+    tail = slice(n // 2, n)
+    gap = sum(abs(a - b) for a, b in zip(seq[tail], fus[tail])) / max(n - n // 2, 1)
+    tail_seq = sum(seq[tail]) / max(n - n // 2, 1)
+    tail_fus = sum(fus[tail]) / max(n - n // 2, 1)
+    tail_msk = sum(fus_m[tail]) / max(n - n // 2, 1)
+    print(f"updates={n} tail-mean loss: sequential={tail_seq:.4f} "
+          f"fused={tail_fus:.4f} fused+mask={tail_msk:.4f} "
+          f"tail mean|seq-fused|={gap:.4f}")
+
+    os.makedirs(args.out, exist_ok=True)
+    with open(os.path.join(args.out, "bert_fused_convergence.csv"), "w",
+              newline="") as f:
+        w = csv.writer(f)
+        w.writerow(["update", "sequential", "fused", "fused_masked"])
+        for i in range(n):
+            w.writerow([i, seq[i], fus[i],
+                        fus_m[i] if i < len(fus_m) else ""])
+    write_svg(os.path.join(args.out, "bert_fused_convergence.svg"),
+              [("sequential micro-steps", seq[:n]),
+               ("window-fused", fus[:n]),
+               ("window-fused + masks", fus_m[:n])])
+    print(f"wrote {args.out}/bert_fused_convergence.csv and .svg")
+
+
+if __name__ == "__main__":
+    main()
