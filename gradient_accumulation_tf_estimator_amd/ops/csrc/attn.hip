@@ -1191,6 +1191,198 @@ __device__ __forceinline__ void attn_fwd_big_body(
   }
 }
 
+// 256-thread chunk staging for the 4-wave big-S variants
+static __device__ __forceinline__ void stage64_b4(const unsigned short* g,
+                                                  int row_stride,
+                                                  unsigned short* lds) {
+  // [64][64] chunk, 256 threads: 2 x 16 B chunks each
+  bf16x8 v[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int c = threadIdx.x + i * 256;
+    v[i] = *(const bf16x8*)(g + (long long)(c >> 3) * row_stride + ((c & 7) << 3));
+  }
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int c = threadIdx.x + i * 256;
+    *(bf16x8*)((char*)lds + swz(c >> 3, (c & 7) << 4)) = v[i];
+  }
+}
+
+static __device__ __forceinline__ void stage_c64_T_b4(const unsigned short* g,
+                                                      int row_stride,
+                                                      unsigned short* lds) {
+  // [64 s][64 d] -> LDS [64 d][64 s]: 128 4s-x-8d sub-blocks; threads >=128 idle
+  if (threadIdx.x >= 128) return;
+  bf16x8 r[4];
+  const int s0 = (threadIdx.x / 8) * 4;
+  const int d0 = (threadIdx.x % 8) * 8;
+#pragma unroll
+  for (int t = 0; t < 4; ++t)
+    r[t] = *(const bf16x8*)(g + (long long)(s0 + t) * row_stride + d0);
+  const unsigned short* u0 = (const unsigned short*)&r[0];
+  const unsigned short* u1 = (const unsigned short*)&r[1];
+  const unsigned short* u2 = (const unsigned short*)&r[2];
+  const unsigned short* u3 = (const unsigned short*)&r[3];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    ush4v pack = {u0[c], u1[c], u2[c], u3[c]};
+    *(ush4v*)((char*)lds + swz(d0 + c, s0 * 2)) = pack;
+  }
+}
+
+// 4-wave chunked forward (any S % 64 == 0): one workgroup per
+// (b, h, 4-q-tile group); each 64-row K/V chunk is staged ONCE for the
+// four q-tile waves instead of once per 1-wave tile.
+template <bool DROP>
+__device__ __forceinline__ void attn_fwd_big4_body(
+    const unsigned short* __restrict__ qkv, unsigned short* __restrict__ out,
+    float* __restrict__ lse_out, int B, int S, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int ngrp = (NT + 3) / 4;
+  const int bh = blockIdx.x / ngrp, grp = blockIdx.x % ngrp;
+  const int b = bh / nh, h = bh % nh;
+  const int wave = threadIdx.x >> 6;
+  const int qt = grp * 4 + wave;
+  const bool active = qt < NT;
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;            // [64][64] swz
+  unsigned short* Vtlds = (unsigned short*)(smem + 8192);  // [64][64] swz
+  float* maskf = (float*)(smem + 16384);                   // [64] chunk
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const int q0 = qt * 32;
+  bf16x8 qf[4];
+  if (active) {
+    const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+  }
+
+  const float scale2 = 0.125f * LOG2E;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+  float m_old = -1e30f, sum = 0.f;
+  f32x16 oc[2];
+  oc[0] = (f32x16)(0.f);
+  oc[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage64_b4(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
+    stage_c64_T_b4(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vtlds);
+    if (mask) {
+      for (int i = threadIdx.x; i < 64; i += 256)
+        maskf[i] = mask[(long long)b * S + c0 + i] ? 1.f : 0.f;
+    }
+    __syncthreads();
+    if (!active) continue;
+
+    f32x16 acc[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      acc[t] = (f32x16)(0.f);
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 a = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kk], acc[t], 0, 0, 0);
+      }
+    }
+    if (mask) {
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          if (maskf[32 * t + crow(r, hi)] == 0.f) acc[t][r] = -INFINITY;
+    }
+    float mc = -1e30f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) mc = fmaxf(mc, acc[t][r]);
+    mc = fmaxf(mc, __shfl_xor(mc, 32, 64)) * scale2;
+    const float m_new = fmaxf(m_old, mc);
+    const float sf = __builtin_amdgcn_exp2f(m_old - m_new);
+    float csum = 0.f;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        acc[t][r] = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - m_new);
+        csum += acc[t][r];
+      }
+    csum += __shfl_xor(csum, 32, 64);
+    sum = sum * sf + csum;
+    m_old = m_new;
+
+    if (DROP) {
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int r = 0; r < 16; ++r)
+          acc[t][r] *= drop_mult(seed, bh, S, q0 + lo31,
+                                 c0 + 32 * t + crow(r, hi), p_drop, inv_keep);
+    }
+
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow_r = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        oc[dt][r] *= __shfl(sf, qrow_r, 64);
+      }
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfk = 0; halfk < 2; ++halfk) {
+          bf16x8 pa = cvt_swap(acc[t], halfk * 8);
+          bf16x8 bv = *(const bf16x8*)((char*)Vtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfk + hi * 8) * 2));
+          oc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, bv, oc[dt], 0, 0, 0);
+        }
+    }
+  }
+
+  if (!active) return;
+  const float inv_sum = 1.f / sum;
+  if (hi == 0) lse_out[((long long)b * nh + h) * S + q0 + lo31] = m_old + log2f(sum);
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow_r = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      oc[dt][r] *= __shfl(inv_sum, qrow_r, 64);
+    }
+    write_tile_bf16(out + ((long long)b * S + q0) * H + h * ATTN_D + dt * 32 + lo31,
+                    H, hi, oc[dt]);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_fwd_big4(
+    const unsigned short* qkv, unsigned short* out, float* lse_out,
+    int B, int S, int nh, const unsigned char* mask) {
+  attn_fwd_big4_body<false>(qkv, out, lse_out, B, S, nh, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_fwd_big4_drop(
+    const unsigned short* qkv, unsigned short* out, float* lse_out,
+    int B, int S, int nh, const unsigned char* mask,
+    const unsigned long long* seed, float p_drop) {
+  attn_fwd_big4_body<true>(qkv, out, lse_out, B, S, nh, mask, seed, p_drop);
+}
+
 extern "C" __global__ __launch_bounds__(64) void k_attn_fwd_big(
     const unsigned short* qkv, unsigned short* out, float* lse_out,
     int B, int S, int nh, const unsigned char* mask) {
@@ -1313,6 +1505,145 @@ __device__ __forceinline__ void attn_bwd_q_big_body(
   for (int dt = 0; dt < 2; ++dt)
     write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
                     3 * H, hi, dq[dt]);
+}
+
+// 4-wave chunked backward, q-tiles (any S % 64 == 0): chunk panels staged
+// once for four q-tile waves.
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_q_big4_body(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ dout, const float* __restrict__ lse_in,
+    float* __restrict__ Dtab, unsigned short* __restrict__ dqkv, int B, int S,
+    int nh, int publish_d, const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int ngrp = (NT + 3) / 4;
+  const int bh = blockIdx.x / ngrp, grp = blockIdx.x % ngrp;
+  const int b = bh / nh, h = bh % nh;
+  const int wave = threadIdx.x >> 6;
+  const int qt = grp * 4 + wave;
+  const bool active = qt < NT;
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* Klds = (unsigned short*)smem;             // [64][64] swz
+  unsigned short* Vlds = (unsigned short*)(smem + 8192);    // [64][64] swz
+  unsigned short* Ktlds = (unsigned short*)(smem + 16384);  // [64][64] swz
+  float* maskf = (float*)(smem + 24576);                    // [64] chunk
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  const int q0 = qt * 32;
+  float lse2 = 0.f, D_q = 0.f;
+  bf16x8 qf[4], dof[4];
+  if (active) {
+    lse2 = lse_in[((long long)b * nh + h) * S + q0 + lo31];
+    const unsigned short* dor = dout + obase + (long long)(q0 + lo31) * H + hi * 32;
+    const unsigned short* orw = out + obase + (long long)(q0 + lo31) * H + hi * 32;
+    float sd = 0.f;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 dv = *(const bf16x8*)(dor + c * 8);
+      bf16x8 ov = *(const bf16x8*)(orw + c * 8);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sd += (float)dv[e] * (float)ov[e];
+    }
+    D_q = sd + __shfl_xor(sd, 32, 64);
+    if (publish_d && hi == 0) Dtab[((long long)b * nh + h) * S + q0 + lo31] = D_q;
+    const unsigned short* qrow = qkv + base + (long long)(q0 + lo31) * 3 * H;
+    const unsigned short* drow = dout + obase + (long long)(q0 + lo31) * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      qf[kk] = *(const bf16x8*)(qrow + kk * 16 + hi * 8);
+      dof[kk] = *(const bf16x8*)(drow + kk * 16 + hi * 8);
+    }
+  }
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+  f32x16 dq[2];
+  dq[0] = (f32x16)(0.f);
+  dq[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage64_b4(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Klds);
+    stage64_b4(qkv + base + 2 * H + (long long)c0 * 3 * H, 3 * H, Vlds);
+    stage_c64_T_b4(qkv + base + H + (long long)c0 * 3 * H, 3 * H, Ktlds);
+    if (mask) {
+      for (int i = threadIdx.x; i < 64; i += 256)
+        maskf[i] = mask[(long long)b * S + c0 + i] ? 1.f : 0.f;
+    }
+    __syncthreads();
+    if (!active) continue;
+
+    f32x16 acc[2], dacc[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      acc[t] = (f32x16)(0.f);
+      dacc[t] = (f32x16)(0.f);
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 ak = *(const bf16x8*)((char*)Klds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        bf16x8 av = *(const bf16x8*)((char*)Vlds + swz(32 * t + lo31, kk * 32 + hi * 16));
+        acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, qf[kk], acc[t], 0, 0, 0);
+        dacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(av, dof[kk], dacc[t], 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kl = 32 * t + crow(r, hi);
+        float p = __builtin_amdgcn_exp2f(acc[t][r] * scale2 - lse2);
+        if (mask && maskf[kl] == 0.f) p = 0.f;
+        float dp = dacc[t][r];
+        if (DROP)
+          dp *= drop_mult(seed, bh, S, q0 + lo31, c0 + kl, p_drop, inv_keep);
+        dacc[t][r] = scale * p * (dp - D_q);
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfk = 0; halfk < 2; ++halfk) {
+          bf16x8 as = cvt_swap(dacc[t], halfk * 8);
+          bf16x8 bk = *(const bf16x8*)((char*)Ktlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfk + hi * 8) * 2));
+          dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bk, dq[dt], 0, 0, 0);
+        }
+  }
+
+  if (!active) return;
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt)
+    write_tile_bf16(dqkv + base + (long long)q0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dq[dt]);
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_q_big4(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int S, int nh, int publish_d,
+    const unsigned char* mask) {
+  attn_bwd_q_big4_body<false>(qkv, out, dout, lse_in, Dtab, dqkv, B, S, nh,
+                              publish_d, mask, nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_q_big4_drop(
+    const unsigned short* qkv, const unsigned short* out,
+    const unsigned short* dout, const float* lse_in, float* Dtab,
+    unsigned short* dqkv, int B, int S, int nh, int publish_d,
+    const unsigned char* mask, const unsigned long long* seed, float p_drop) {
+  attn_bwd_q_big4_body<true>(qkv, out, dout, lse_in, Dtab, dqkv, B, S, nh,
+                             publish_d, mask, seed, p_drop);
 }
 
 extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_q_big(
@@ -1444,6 +1775,143 @@ __device__ __forceinline__ void attn_bwd_kv_big_body(
     write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
                     3 * H, hi, dk_[dt]);
   }
+}
+
+// 4-wave chunked backward, k-tiles (any S % 64 == 0)
+template <bool DROP>
+__device__ __forceinline__ void attn_bwd_kv_big4_body(
+    const unsigned short* __restrict__ qkv, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse_in, const float* __restrict__ Dtab,
+    unsigned short* __restrict__ dqkv, int B, int S, int nh,
+    const unsigned char* __restrict__ mask,
+    const unsigned long long* __restrict__ seed_ptr, float p_drop) {
+  const int H = nh * ATTN_D;
+  const int NT = S / 32;
+  const int ngrp = (NT + 3) / 4;
+  const int bh = blockIdx.x / ngrp, grp = blockIdx.x % ngrp;
+  const int b = bh / nh, h = bh % nh;
+  const int wave = threadIdx.x >> 6;
+  const int kt = grp * 4 + wave;
+  const bool active = kt < NT;
+  const int lane = threadIdx.x & 63;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* dOtlds = (unsigned short*)smem;          // [64][64] swz
+  unsigned short* Qtlds = (unsigned short*)(smem + 8192);  // [64][64] swz
+  float* lsetab = (float*)(smem + 16384);                  // [64]
+  float* dtab = (float*)(smem + 16384 + 256);              // [64]
+
+  const long long base = ((long long)b * S) * (3LL * H) + (long long)h * ATTN_D;
+  const long long obase = ((long long)b * S) * H + (long long)h * ATTN_D;
+  const int k0 = kt * 32;
+
+  bf16x8 kf[4], vf[4];
+  float kvalid = 1.f;
+  if (active) {
+    const unsigned short* krow = qkv + base + H + (long long)(k0 + lo31) * 3 * H;
+    const unsigned short* vrow = qkv + base + 2 * H + (long long)(k0 + lo31) * 3 * H;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      kf[kk] = *(const bf16x8*)(krow + kk * 16 + hi * 8);
+      vf[kk] = *(const bf16x8*)(vrow + kk * 16 + hi * 8);
+    }
+    if (mask) kvalid = mask[(long long)b * S + k0 + lo31] ? 1.f : 0.f;
+  }
+  const int k_lane = k0 + lo31;
+
+  const float scale2 = 0.125f * LOG2E, scale = 0.125f;
+  unsigned long long seed = 0;
+  float inv_keep = 1.f;
+  if (DROP) {
+    seed = seed_ptr[0];
+    inv_keep = 1.f / (1.f - p_drop);
+  }
+  f32x16 dv_[2], dk_[2];
+  dv_[0] = (f32x16)(0.f); dv_[1] = (f32x16)(0.f);
+  dk_[0] = (f32x16)(0.f); dk_[1] = (f32x16)(0.f);
+
+  for (int c0 = 0; c0 < S; c0 += 64) {
+    __syncthreads();
+    stage_c64_T_b4(dout + obase + (long long)c0 * H, H, dOtlds);
+    stage_c64_T_b4(qkv + base + (long long)c0 * 3 * H, 3 * H, Qtlds);
+    for (int i = threadIdx.x; i < 64; i += 256) {
+      lsetab[i] = lse_in[((long long)b * nh + h) * S + c0 + i];
+      dtab[i] = Dtab[((long long)b * nh + h) * S + c0 + i];
+    }
+    __syncthreads();
+    if (!active) continue;
+
+    f32x16 p_qt[2], ds_qt[2];
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {
+      f32x16 sacc = (f32x16)(0.f), dpacc = (f32x16)(0.f);
+      const unsigned short* qr = qkv + base + (long long)(c0 + t * 32 + lo31) * 3 * H;
+      const unsigned short* dr = dout + obase + (long long)(c0 + t * 32 + lo31) * H;
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        bf16x8 aq = *(const bf16x8*)(qr + kk * 16 + hi * 8);
+        bf16x8 ad = *(const bf16x8*)(dr + kk * 16 + hi * 8);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, kf[kk], sacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ad, vf[kk], dpacc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int q = t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        const float p =
+            kvalid * __builtin_amdgcn_exp2f(sacc[r] * scale2 - lsetab[q]);
+        float m = 1.f, dp = dpacc[r];
+        if (DROP) {
+          m = drop_mult(seed, bh, S, c0 + q, k_lane, p_drop, inv_keep);
+          dp *= m;
+        }
+        p_qt[t][r] = p * m;
+        ds_qt[t][r] = scale * p * (dp - dtab[q]);
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < 2; ++dt)
+#pragma unroll
+      for (int t = 0; t < 2; ++t)
+#pragma unroll
+        for (int halfq = 0; halfq < 2; ++halfq) {
+          bf16x8 ap = cvt_swap(p_qt[t], halfq * 8);
+          bf16x8 as = cvt_swap(ds_qt[t], halfq * 8);
+          bf16x8 bd = *(const bf16x8*)((char*)dOtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfq + hi * 8) * 2));
+          bf16x8 bq = *(const bf16x8*)((char*)Qtlds +
+                                       swz(dt * 32 + lo31,
+                                           (32 * t + 16 * halfq + hi * 8) * 2));
+          dv_[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ap, bd, dv_[dt], 0, 0, 0);
+          dk_[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(as, bq, dk_[dt], 0, 0, 0);
+        }
+  }
+
+  if (!active) return;
+#pragma unroll
+  for (int dt = 0; dt < 2; ++dt) {
+    write_tile_bf16(dqkv + base + 2 * H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dv_[dt]);
+    write_tile_bf16(dqkv + base + H + (long long)k0 * 3 * H + dt * 32 + lo31,
+                    3 * H, hi, dk_[dt]);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_kv_big4(
+    const unsigned short* qkv, const unsigned short* dout, const float* lse_in,
+    const float* Dtab, unsigned short* dqkv, int B, int S, int nh,
+    const unsigned char* mask) {
+  attn_bwd_kv_big4_body<false>(qkv, dout, lse_in, Dtab, dqkv, B, S, nh, mask,
+                               nullptr, 0.f);
+}
+extern "C" __global__ __launch_bounds__(256) void k_attn_bwd_kv_big4_drop(
+    const unsigned short* qkv, const unsigned short* dout, const float* lse_in,
+    const float* Dtab, unsigned short* dqkv, int B, int S, int nh,
+    const unsigned char* mask, const unsigned long long* seed, float p_drop) {
+  attn_bwd_kv_big4_body<true>(qkv, dout, lse_in, Dtab, dqkv, B, S, nh, mask,
+                              seed, p_drop);
 }
 
 extern "C" __global__ __launch_bounds__(64) void k_attn_bwd_kv_big(

@@ -109,6 +109,27 @@ extern "C" __global__ void k_attn_bwd8_drop_128(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const float*, unsigned short*, int, int, const unsigned char*,
     const unsigned long long*, float);
+extern "C" __global__ void k_attn_fwd_big4(
+    const unsigned short*, unsigned short*, float*, int, int, int,
+    const unsigned char*);
+extern "C" __global__ void k_attn_fwd_big4_drop(
+    const unsigned short*, unsigned short*, float*, int, int, int,
+    const unsigned char*, const unsigned long long*, float);
+extern "C" __global__ void k_attn_bwd_q_big4(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, float*, unsigned short*, int, int, int, int,
+    const unsigned char*);
+extern "C" __global__ void k_attn_bwd_q_big4_drop(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const float*, float*, unsigned short*, int, int, int, int,
+    const unsigned char*, const unsigned long long*, float);
+extern "C" __global__ void k_attn_bwd_kv_big4(
+    const unsigned short*, const unsigned short*, const float*,
+    const float*, unsigned short*, int, int, int, const unsigned char*);
+extern "C" __global__ void k_attn_bwd_kv_big4_drop(
+    const unsigned short*, const unsigned short*, const float*,
+    const float*, unsigned short*, int, int, int, const unsigned char*,
+    const unsigned long long*, float);
 // 4-wave S=128 path A/B (one workgroup per (b,h), panels staged once)
 static inline bool attn_w4_on() {
   static const bool on = [] {
@@ -439,8 +460,24 @@ std::vector<at::Tensor> attn_fwd(at::Tensor qkv, int64_t nh,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
   if (S > 128) {
-    // chunked online-softmax variant: [64][64] K + V^T LDS panels (+mask)
-    if (drop)
+    // chunked online-softmax variants: [64][64] K + V^T LDS panels (+mask).
+    // Default 4-wave grouping (chunks staged once per four q-tile waves);
+    // GA_ATTN_W4=0 reverts to the 1-wave-per-tile kernels.
+    if (attn_w4_on()) {
+      const int ngrp = (NT + 3) / 4;
+      if (drop)
+        hipLaunchKernelGGL(k_attn_fwd_big4_drop, dim3(B * (int)nh * ngrp),
+                           dim3(256), 16384 + 256, stream,
+                           (const unsigned short*)qkv.data_ptr(),
+                           (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                           B, S, (int)nh, mp, sp, pd);
+      else
+        hipLaunchKernelGGL(k_attn_fwd_big4, dim3(B * (int)nh * ngrp),
+                           dim3(256), 16384 + 256, stream,
+                           (const unsigned short*)qkv.data_ptr(),
+                           (unsigned short*)out.data_ptr(), lse.data_ptr<float>(),
+                           B, S, (int)nh, mp);
+    } else if (drop)
       hipLaunchKernelGGL(k_attn_fwd_big_drop, dim3(B * (int)nh * NT), dim3(64),
                          16384 + 256, stream,
                          (const unsigned short*)qkv.data_ptr(),
@@ -510,7 +547,41 @@ at::Tensor attn_bwd(at::Tensor qkv, at::Tensor out, at::Tensor dout,
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int NT = S / 32;
 
-  if (S > 128) {
+  if (S > 128 && attn_w4_on()) {
+    const int ngrp = (NT + 3) / 4;
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_q_big4_drop, dim3(B * (int)nh * ngrp),
+                         dim3(256), 24576 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_q_big4, dim3(B * (int)nh * ngrp),
+                         dim3(256), 24576 + 256, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)out.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, 1, mp);
+    if (drop)
+      hipLaunchKernelGGL(k_attn_bwd_kv_big4_drop, dim3(B * (int)nh * ngrp),
+                         dim3(256), 16384 + 512, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh,
+                         mp, sp, pd);
+    else
+      hipLaunchKernelGGL(k_attn_bwd_kv_big4, dim3(B * (int)nh * ngrp),
+                         dim3(256), 16384 + 512, stream,
+                         (const unsigned short*)qkv.data_ptr(),
+                         (const unsigned short*)dout.data_ptr(),
+                         lse.data_ptr<float>(), Dtab.data_ptr<float>(),
+                         (unsigned short*)dqkv.data_ptr(), B, S, (int)nh, mp);
+  } else if (S > 128) {
     if (drop)
       hipLaunchKernelGGL(k_attn_bwd_q_big_drop, dim3(B * (int)nh * NT), dim3(64),
                          24576 + 256, stream,
