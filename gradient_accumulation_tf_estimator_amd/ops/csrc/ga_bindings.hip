@@ -21,6 +21,22 @@ extern "C" __global__ void k_apply_bf16(float4*, float4*, float4*, float4*, usho
                                         const float*, const float*,
                                         long long, long long,
                                         float, float, float, float, float, float);
+extern "C" __global__ void k_apply_f32_g2(float4*, float4*, float4*, float4*,
+                                          const float*, const float*,
+                                          long long, long long,
+                                          float, float, float, float, float, float);
+extern "C" __global__ void k_apply_bf16_g2(float4*, float4*, float4*, float4*, ushort4*,
+                                           const float*, const float*,
+                                           long long, long long,
+                                           float, float, float, float, float, float);
+// GA_APPLY_G2=1 selects the 2-group-ILP apply (A/B vs the 4-group default)
+static inline bool apply_g2() {
+  static const bool on = [] {
+    const char* v = getenv("GA_APPLY_G2");
+    return v && atoi(v) != 0;
+  }();
+  return on;
+}
 #define GA_DECL_LN(EPL)                                                                  \
   extern "C" __global__ void k_addln_fwd_##EPL(                                          \
       const unsigned short*, const unsigned short*, const unsigned short*,               \
@@ -258,7 +274,8 @@ void fused_apply(at::Tensor accum, at::Tensor m, at::Tensor v, at::Tensor master
   if (has_model) {
     TORCH_CHECK(model.scalar_type() == at::kBFloat16 && model.numel() == n,
                 "model buffer must be flat bf16 of same length");
-    hipLaunchKernelGGL(k_apply_bf16, grid_for(n4), dim3(GA_THREADS), 0, stream,
+    hipLaunchKernelGGL(apply_g2() ? k_apply_bf16_g2 : k_apply_bf16,
+                       grid_for(n4), dim3(GA_THREADS), 0, stream,
                        (float4*)accum.data_ptr<float>(), (float4*)m.data_ptr<float>(),
                        (float4*)v.data_ptr<float>(), (float4*)master.data_ptr<float>(),
                        (ushort4*)model.data_ptr(),
@@ -267,7 +284,8 @@ void fused_apply(at::Tensor accum, at::Tensor m, at::Tensor v, at::Tensor master
                        (float)inv_k, (float)clip, (float)weight_decay,
                        (float)beta1, (float)beta2, (float)eps);
   } else {
-    hipLaunchKernelGGL(k_apply_f32, grid_for(n4), dim3(GA_THREADS), 0, stream,
+    hipLaunchKernelGGL(apply_g2() ? k_apply_f32_g2 : k_apply_f32,
+                       grid_for(n4), dim3(GA_THREADS), 0, stream,
                        (float4*)accum.data_ptr<float>(), (float4*)m.data_ptr<float>(),
                        (float4*)v.data_ptr<float>(), (float4*)master.data_ptr<float>(),
                        lr_dev.data_ptr<float>(), sqnorm_ws.data_ptr<float>(),

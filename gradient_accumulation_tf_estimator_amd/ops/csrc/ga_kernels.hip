@@ -200,6 +200,90 @@ static __device__ void apply_body(float4* __restrict__ accum,
   }
 }
 
+template <bool HAS_MODEL>
+static __device__ void apply_body_g2(float4* __restrict__ accum,
+                                  float4* __restrict__ m,
+                                  float4* __restrict__ v,
+                                  float4* __restrict__ p,
+                                  ushort4* __restrict__ model,
+                                  const float* __restrict__ lr_p,
+                                  const float* __restrict__ sq_p,
+                                  long long n4, long long boundary4,
+                                  float inv_k, float clip, float wd,
+                                  float b1, float b2, float eps) {
+  const float lr = lr_p[0];
+  float coef = 1.f;
+  if (clip > 0.f) {
+    float norm = sqrtf(sq_p[0]) * inv_k;  // norm of accum/K (clip AFTER normalize)
+    coef = clip / fmaxf(norm, clip);      // tf.clip_by_global_norm scale
+  }
+  const float s = inv_k * coef;
+  const float omb1 = 1.f - b1, omb2 = 1.f - b2;
+  const float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
+  // two float4 groups per iteration: independent mm/vv/u chains double the
+  // in-flight loads and hide the sqrt+div latency; the bf16 model write
+  // becomes one 16-byte store per pair
+  long long stride = (long long)gridDim.x * blockDim.x * 2;
+  long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  for (; i < n4; i += stride) {
+    const bool two = (i + 1) < n4;
+    float4 a = accum[i], mm = m[i], vv = v[i], pp = p[i];
+    float4 a2, mm2, vv2, pp2;
+    if (two) { a2 = accum[i + 1]; mm2 = m[i + 1]; vv2 = v[i + 1]; pp2 = p[i + 1]; }
+    const float dw = (i < boundary4) ? wd : 0.f;
+    const float dw2 = ((i + 1) < boundary4) ? wd : 0.f;
+#define GA_C2(av, mv, vvv, pv, c, dww)                  \
+    {                                                  \
+      float g = av.c * s;                              \
+      mv.c = fmaf(b1, mv.c, omb1 * g);                 \
+      vvv.c = fmaf(b2, vvv.c, omb2 * g * g);           \
+      float u = mv.c / (sqrtf(vvv.c) + eps);           \
+      u = fmaf(dww, pv.c, u);                          \
+      pv.c = fmaf(-lr, u, pv.c);                       \
+    }
+    GA_C2(a, mm, vv, pp, x, dw) GA_C2(a, mm, vv, pp, y, dw)
+    GA_C2(a, mm, vv, pp, z, dw) GA_C2(a, mm, vv, pp, w, dw)
+    if (two) {
+      GA_C2(a2, mm2, vv2, pp2, x, dw2) GA_C2(a2, mm2, vv2, pp2, y, dw2)
+      GA_C2(a2, mm2, vv2, pp2, z, dw2) GA_C2(a2, mm2, vv2, pp2, w, dw2)
+    }
+#undef GA_C2
+    m[i] = mm; v[i] = vv; p[i] = pp; accum[i] = z;
+    if (two) { m[i + 1] = mm2; v[i + 1] = vv2; p[i + 1] = pp2; accum[i + 1] = z; }
+    if (HAS_MODEL) {
+      if (two) {
+        union { ushort4 u4[2]; uint4 u16; } pk;
+        pk.u4[0] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
+                                f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+        pk.u4[1] = make_ushort4(f32_to_bf16(pp2.x), f32_to_bf16(pp2.y),
+                                f32_to_bf16(pp2.z), f32_to_bf16(pp2.w));
+        *(uint4*)(model + i) = pk.u16;
+      } else {
+        model[i] = make_ushort4(f32_to_bf16(pp.x), f32_to_bf16(pp.y),
+                                f32_to_bf16(pp.z), f32_to_bf16(pp.w));
+      }
+    }
+  }
+}
+
+extern "C" __global__ void k_apply_f32_g2(float4* accum, float4* m, float4* v, float4* p,
+                                          const float* lr_p, const float* sq_p,
+                                          long long n4, long long boundary4,
+                                          float inv_k, float clip, float wd,
+                                          float b1, float b2, float eps) {
+  apply_body_g2<false>(accum, m, v, p, nullptr, lr_p, sq_p, n4, boundary4,
+                       inv_k, clip, wd, b1, b2, eps);
+}
+extern "C" __global__ void k_apply_bf16_g2(float4* accum, float4* m, float4* v, float4* p,
+                                           ushort4* model,
+                                           const float* lr_p, const float* sq_p,
+                                           long long n4, long long boundary4,
+                                           float inv_k, float clip, float wd,
+                                           float b1, float b2, float eps) {
+  apply_body_g2<true>(accum, m, v, p, model, lr_p, sq_p, n4, boundary4,
+                      inv_k, clip, wd, b1, b2, eps);
+}
+
 extern "C" __global__ void k_apply_f32(float4* accum, float4* m, float4* v, float4* p,
                                        const float* lr_p, const float* sq_p,
                                        long long n4, long long boundary4,
