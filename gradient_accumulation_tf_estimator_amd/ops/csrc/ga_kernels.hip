@@ -143,11 +143,11 @@ static __device__ void apply_body(float4* __restrict__ accum,
   const float s = inv_k * coef;
   const float omb1 = 1.f - b1, omb2 = 1.f - b2;
   const float4 z = make_float4(0.f, 0.f, 0.f, 0.f);
-  // four float4 groups per iteration (flat totals are 64-element aligned,
-  // so n4 % 4 == 0 and every group is complete): 16 loads in flight per
-  // buffer sweep hide the sqrt+div latency; the bf16 model write becomes
-  // two 16-byte stores. (The 2-group version measured ~5.5 TB/s of the
-  // ~7 TB/s this RMW mix can reach.)
+  // four float4 groups per iteration -- MEASURED SLOWER than the 2-group
+  // variant (19.6-19.8k vs 20.8-21.1k samples/s same-box interleaved):
+  // the extra in-flight registers cost occupancy on this 4-stream RMW.
+  // Kept behind GA_APPLY_G4 for re-measurement; apply_body_g2 below is
+  // the default.
   long long stride = (long long)gridDim.x * blockDim.x * 4;
   long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   for (; i < n4; i += stride) {
