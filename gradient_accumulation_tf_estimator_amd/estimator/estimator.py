@@ -77,6 +77,12 @@ class RunConfig:
     tf_random_seed: Optional[int] = None  # reference name (01:77); seeds torch
     device: Optional[str] = None
     train_distribute: Optional[Any] = None  # process group / True for default
+    # window fusion at the estimator level: gather a whole accumulation
+    # window's micro-batches and run them as ONE fused fwd/bwd
+    # (TrainOp.step_fused -- exact by linearity). Falls back to per-micro-
+    # batch stepping for ragged tails, strict semantics, or when a step/
+    # checkpoint limit lands mid-window.
+    window_fuse: bool = False
 
 
 @dataclass
@@ -98,6 +104,20 @@ def _call_input_fn(input_fn, mode):
     if "mode" in sig.parameters:
         kwargs["mode"] = mode
     return input_fn(**kwargs)
+
+
+def _same_shape(a, b):
+    if torch.is_tensor(a) and torch.is_tensor(b):
+        return a.shape == b.shape
+    if isinstance(a, dict) and isinstance(b, dict):
+        return a.keys() == b.keys() and all(_same_shape(a[k], b[k]) for k in a)
+    return False
+
+
+def _cat_batches(parts):
+    if torch.is_tensor(parts[0]):
+        return torch.cat(parts)
+    return {k: torch.cat([p[k] for p in parts]) for k in parts[0]}
 
 
 def _to_device(x, device):
@@ -217,34 +237,89 @@ class Estimator:
                 except StopIteration:
                     break
         last_loss = None
+        eng = op.engine
+        can_fuse = (cfg.window_fuse and eng.K > 1 and not eng.strict)
+
+        def limit_left(step):
+            left = None
+            if max_steps is not None:
+                left = max_steps - step
+            if steps is not None:
+                r = steps - done_this_call
+                left = r if left is None else min(left, r)
+            return left
+
+        def next_batch():
+            nonlocal pending
+            if pending is not None:
+                b, pending = pending, None
+                return b
+            f, l = next(it)
+            return _to_device(f, device), _to_device(l, device)
+
+        def cadence(prev, new, loss):
+            nonlocal t_last, s_last
+            L = cfg.log_step_count_steps
+            if L and new // L > prev // L:
+                now = time.perf_counter()
+                rate = (new - s_last) / max(now - t_last, 1e-9)
+                self._logger.log(step=new, loss=float(loss.detach().float()),
+                                 lr=op.last_lr, steps_per_sec=round(rate, 3))
+                t_last, s_last = now, new
+            C = cfg.save_checkpoints_steps
+            if C and new // C > prev // C:
+                self._save(spec, new)
+
         while True:
             step = op.global_step
-            if max_steps is not None and step >= max_steps:
+            left = limit_left(step)
+            if left is not None and left <= 0:
                 break
-            if steps is not None and done_this_call >= steps:
-                break
-            if pending is None:
+            fuse_n = 0
+            if can_fuse:
+                n = eng.fused_block_sizes()[0]
+                if n > 1 and (left is None or left >= n):
+                    fuse_n = n
+            if fuse_n:
+                # gather a whole block; ragged tails step singly below
+                blk = []
                 try:
-                    f, l = next(it)
+                    while len(blk) < fuse_n:
+                        blk.append(next_batch())
                 except StopIteration:
+                    pass
+                same = len(blk) == fuse_n and all(
+                    _same_shape(b[0], blk[0][0]) and _same_shape(b[1], blk[0][1])
+                    for b in blk[1:])
+                if same:
+                    f = _cat_batches([b[0] for b in blk])
+                    l = _cat_batches([b[1] for b in blk])
+                    loss = spec.loss_fn(f, l)
+                    op.step_fused(loss, fuse_n)
+                    done_this_call += fuse_n
+                    last_loss = loss
+                    cadence(step, op.global_step, loss)
+                    continue
+                if not blk:
                     break
-                features, labels = _to_device(f, device), _to_device(l, device)
-            else:
-                features, labels = pending
-                pending = None
+                # fall back: run the gathered batches one micro-step each
+                for f, l in blk:
+                    prev = op.global_step
+                    loss = spec.loss_fn(f, l)
+                    op.step(loss)
+                    done_this_call += 1
+                    last_loss = loss
+                    cadence(prev, op.global_step, loss)
+                continue
+            try:
+                features, labels = next_batch()
+            except StopIteration:
+                break
             loss = spec.loss_fn(features, labels)
             op.step(loss)
             done_this_call += 1
             last_loss = loss
-            step = op.global_step
-            if cfg.log_step_count_steps and step % cfg.log_step_count_steps == 0:
-                now = time.perf_counter()
-                rate = (step - s_last) / max(now - t_last, 1e-9)
-                self._logger.log(step=step, loss=float(loss.detach().float()),
-                                 lr=op.last_lr, steps_per_sec=round(rate, 3))
-                t_last, s_last = now, step
-            if cfg.save_checkpoints_steps and step % cfg.save_checkpoints_steps == 0:
-                self._save(spec, step)
+            cadence(step, op.global_step, loss)
         self._save(spec, op.global_step)
         return {
             "global_step": op.global_step,

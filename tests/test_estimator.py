@@ -197,3 +197,40 @@ def test_safetensors_export_roundtrip(tmp_path):
     m2.load_state_dict(back)
     for a, b in zip(m.parameters(), m2.parameters()):
         assert torch.equal(a, b)
+
+
+def test_estimator_window_fuse_equivalence(tmp_path):
+    """RunConfig(window_fuse=True) must train identically to per-micro-batch
+    stepping (linearity; fp32 CPU -> near-exact), including step counting,
+    and fall back cleanly on ragged tails."""
+    res = {}
+    for fuse in (False, True):
+        cfg = RunConfig(model_dir=str(tmp_path / f"m{int(fuse)}"),
+                        log_step_count_steps=8, tf_random_seed=19830610,
+                        window_fuse=fuse)
+        est = Estimator(mnist_model_fn, cfg,
+                        params={"learning_rate": 1e-3,
+                                "gradient_accumulation_multiplier": 4, "seed": 0})
+        r = est.train(train_input_fn, max_steps=24)
+        sd = est._train_spec.train_op.state_dict()
+        res[fuse] = (r["global_step"], sd["master"].clone(),
+                     sd["global_step"], sd["apply_count"])
+    assert res[False][0] == res[True][0] == 24
+    assert res[False][2] == res[True][2]
+    assert res[False][3] == res[True][3]
+    import numpy as np
+    np.testing.assert_allclose(res[True][1].numpy(), res[False][1].numpy(),
+                               rtol=2e-5, atol=2e-6)
+
+
+def test_estimator_window_fuse_mid_window_limit(tmp_path):
+    """A max_steps that lands mid-window forces the single-step fallback for
+    the final partial window; counts must still be exact."""
+    cfg = RunConfig(model_dir=str(tmp_path / "m"), tf_random_seed=1,
+                    window_fuse=True)
+    est = Estimator(mnist_model_fn, cfg,
+                    params={"learning_rate": 1e-3,
+                            "gradient_accumulation_multiplier": 4, "seed": 0})
+    r = est.train(train_input_fn, max_steps=10)  # 2 fused windows + 2 singles
+    assert r["global_step"] == 10
+    assert est._train_spec.train_op.engine.apply_count == 2
