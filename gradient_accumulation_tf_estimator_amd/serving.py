@@ -31,12 +31,20 @@ class GraphedPredictor:
     """
 
     def __init__(self, model: torch.nn.Module, example: torch.Tensor,
-                 fwd_fn: Optional[Callable] = None, warmup_iters: int = 3):
+                 fwd_fn: Optional[Callable] = None, warmup_iters: int = 3,
+                 example_mask: Optional[torch.Tensor] = None):
+        """``example_mask`` ([B,S] 1/0) captures the key-padding-mask path
+        (real serving batches are padded); requests then pass ``mask=``."""
         self.model = model
-        self.fwd = fwd_fn or model
+        base = fwd_fn or model
+        if example_mask is not None:
+            self.fwd = lambda x, m: base(x, attention_mask=m)
+        else:
+            self.fwd = lambda x, m: base(x)
         self.graphed = example.is_cuda
         model.eval()
         self._static_in = example.clone()
+        self._static_mask = None if example_mask is None else example_mask.clone()
         if not self.graphed:
             self._graph = None
             return
@@ -44,23 +52,30 @@ class GraphedPredictor:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s), torch.no_grad():
             for _ in range(warmup_iters):
-                self.fwd(self._static_in)
+                self.fwd(self._static_in, self._static_mask)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self._graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self._graph), torch.no_grad():
-            self._static_out = self.fwd(self._static_in)
+            self._static_out = self.fwd(self._static_in, self._static_mask)
         torch.cuda.synchronize()
 
     @torch.no_grad()
-    def __call__(self, inputs: torch.Tensor) -> torch.Tensor:
+    def __call__(self, inputs: torch.Tensor,
+                 mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+        if (mask is None) != (self._static_mask is None):
+            raise ValueError("predictor was captured "
+                             + ("WITH" if self._static_mask is not None else "WITHOUT")
+                             + " a mask; pass requests the same way")
         if self._graph is None:
-            return self.fwd(inputs)
+            return self.fwd(inputs, mask)
         if inputs.shape != self._static_in.shape:
             raise ValueError(
                 f"GraphedPredictor captured shape {tuple(self._static_in.shape)}"
                 f", got {tuple(inputs.shape)} -- pad the batch or capture a "
                 "second predictor for this shape")
         self._static_in.copy_(inputs)
+        if mask is not None:
+            self._static_mask.copy_(mask)
         self._graph.replay()
         return self._static_out
