@@ -354,11 +354,14 @@ std::vector<at::Tensor> addln_bwd(at::Tensor dy, at::Tensor h, at::Tensor gamma,
   static const bool want_split = std::getenv("GA_LN_SPLIT2") != nullptr;
   const bool split2 = want_split && (H == 512 || H == 1024) && R >= 256;
   const int units = split2 ? R * 2 : R;
-  // one row per wave up to 1024 blocks (the old 256 cap left the fused
-  // R=4096 regime at a single 4-wave workgroup per CU with 4 serial rows
-  // per wave; the partials the cap was protecting cost ~6 MB and a few us
-  // of colreduce)
-  int blocks = std::max(1, std::min((units + 3) / 4, 1024));
+  // block cap: GA_LN_BWD_BLOCKS overrides for A/B (256 = one 4-wave
+  // workgroup per CU with 4 serial rows/wave at fused R=4096; 1024 = one
+  // row per wave, 4x the partial slabs)
+  static const int bwd_cap = [] {
+    const char* v = getenv("GA_LN_BWD_BLOCKS");
+    return v ? atoi(v) : 256;
+  }();
+  int blocks = std::max(1, std::min((units + 3) / 4, bwd_cap));
   auto partials = at::empty({blocks, 3, H}, dy.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream().stream();
   size_t lds = (size_t)4 * 3 * (split2 ? H / 2 : H) * sizeof(float);
