@@ -128,7 +128,9 @@ def main():
 
         fused_ops.set_grouped_wgrad(True)
 
-    torch.manual_seed(1234 + rank)
+    # model init must be rank-IDENTICAL for DP (replicas of one model);
+    # only the data stream is rank-dependent (generator below)
+    torch.manual_seed(1234)
     cfg = CONFIGS[args.model]()
     cfg.fused = args.fused == "on"
     cfg.dropout = args.dropout
