@@ -120,6 +120,20 @@ def _cat_batches(parts):
     return {k: torch.cat([p[k] for p in parts]) for k in parts[0]}
 
 
+def _clone_batch(x):
+    if torch.is_tensor(x):
+        return x.clone()
+    return {k: v.clone() for k, v in x.items()}
+
+
+def _copy_batch(dst, src):
+    if torch.is_tensor(dst):
+        dst.copy_(src)
+        return
+    for k in dst:
+        dst[k].copy_(src[k])
+
+
 def _to_device(x, device):
     if device is None:
         return x
@@ -141,6 +155,10 @@ class Estimator:
         self._train_spec: Optional[EstimatorSpec] = None
         self._train_iter = None
         self._train_iter_key = None
+        # window-fusion hipGraph capture (built lazily on the first fused
+        # block when on the HIP engine; None = eager, False = don't retry)
+        self._fused_loop = None
+        self._fused_static = None
         self._logger = StepLogger(self.config.model_dir)
         if self.config.tf_random_seed is not None:
             torch.manual_seed(self.config.tf_random_seed)
@@ -214,6 +232,9 @@ class Estimator:
         skip = 0
         if self._train_spec is None:
             self._train_spec = self._build_spec(ModeKeys.TRAIN, features, labels)
+            # any captured fusion graph belongs to the previous spec's engine
+            self._fused_loop = None
+            self._fused_static = None
             restored = self._restore(self._train_spec, with_engine=True)
             if fresh_iter and restored:
                 # deterministic input replay on resume: the checkpointed
@@ -294,8 +315,7 @@ class Estimator:
                 if same:
                     f = _cat_batches([b[0] for b in blk])
                     l = _cat_batches([b[1] for b in blk])
-                    loss = spec.loss_fn(f, l)
-                    op.step_fused(loss, fuse_n)
+                    loss = self._fused_step(spec, op, f, l, fuse_n)
                     done_this_call += fuse_n
                     last_loss = loss
                     cadence(step, op.global_step, loss)
@@ -325,6 +345,43 @@ class Estimator:
             "global_step": op.global_step,
             "loss": float(last_loss.detach().float()) if last_loss is not None else None,
         }
+
+    def _fused_step(self, spec, op, f, l, n):
+        """One fused block. On the HIP engine, capture the block as a
+        FusedWindowLoop hipGraph on first use (static input buffers +
+        replay) so estimator-API training runs at bench speed; anything
+        that does not fit the captured shape falls back to the eager
+        ``step_fused`` (same math)."""
+        eng = op.engine
+        if self._fused_loop is None and eng.backend == "hip" and \
+                not eng.strict and eng.K % n == 0 and \
+                eng.global_step % eng.K == 0 and torch.is_tensor(l):
+            try:
+                from ..engine.graphs import FusedWindowLoop
+
+                sf = _clone_batch(f)
+                sl = l.clone()
+                loop = FusedWindowLoop(
+                    eng, lambda: spec.loss_fn(sf, sl), n_micro=n,
+                    world=eng.world_size)
+                self._fused_loop = (loop, n)
+                self._fused_static = (sf, sl)
+            except Exception as exc:  # capture unsupported -> stay eager
+                import logging
+
+                logging.getLogger("ga_amd.estimator").warning(
+                    "window-fusion capture failed (%s); running eager", exc)
+                self._fused_loop = False
+        if self._fused_loop not in (None, False):
+            loop, ln = self._fused_loop
+            sf, sl = self._fused_static
+            if ln == n and _same_shape(f, sf) and l.shape == sl.shape:
+                _copy_batch(sf, f)
+                sl.copy_(l)
+                return loop.step()
+        loss = spec.loss_fn(f, l)
+        op.step_fused(loss, n)
+        return loss
 
     def evaluate(self, input_fn, steps: Optional[int] = None,
                  checkpoint_path: Optional[str] = None,

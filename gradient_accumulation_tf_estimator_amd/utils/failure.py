@@ -85,6 +85,7 @@ def _reset_to_checkpoint(estimator) -> None:
     """Drop the live train spec/iterator so the next ``train()`` rebuilds
     from the latest checkpoint instead of continuing a poisoned in-memory
     state (stale grads in the flat buffer after a mid-step exception)."""
-    for attr in ("_train_spec", "_train_iter", "_train_iter_key"):
+    for attr in ("_train_spec", "_train_iter", "_train_iter_key",
+                 "_fused_loop", "_fused_static"):
         if hasattr(estimator, attr):
             setattr(estimator, attr, None)

@@ -81,6 +81,11 @@ def test_estimator_window_fuse_gpu(tmp_path):
                                   window_fuse=fuse))
         r = est.train(lambda mode=None: input_fn_iterator(
             ds, 16, num_epochs=None, seed=2), max_steps=16)
+        if fuse:
+            # the estimator must have CAPTURED the fused window (hipGraph),
+            # not just run it eagerly
+            assert est._fused_loop not in (None, False), \
+                "window-fusion capture did not engage on the HIP engine"
         sd = est._train_spec.train_op.state_dict()
         res[fuse] = (r["global_step"], sd["apply_count"], sd["master"].clone())
     assert res[False][0] == res[True][0] == 16
