@@ -79,10 +79,15 @@ class RunConfig:
     train_distribute: Optional[Any] = None  # process group / True for default
     # window fusion at the estimator level: gather a whole accumulation
     # window's micro-batches and run them as ONE fused fwd/bwd
-    # (TrainOp.step_fused -- exact by linearity). Falls back to per-micro-
-    # batch stepping for ragged tails, strict semantics, or when a step/
-    # checkpoint limit lands mid-window.
+    # (TrainOp.step_fused -- exact by linearity; hipGraph-captured on the
+    # HIP engine). Falls back to per-micro-batch stepping for ragged
+    # tails, strict semantics, or when a step/checkpoint limit lands
+    # mid-window.
     window_fuse: bool = False
+    # background-thread input prefetch depth (0 = off, matching the
+    # reference input_fn which has no prefetch -- SURVEY C12); overlaps
+    # host batch assembly + H2D copies with training
+    prefetch: int = 0
 
 
 @dataclass
@@ -219,7 +224,12 @@ class Estimator:
         # (train_and_evaluate) continue the stream instead of restarting it
         fresh_iter = self._train_iter is None or self._train_iter_key is not input_fn
         if fresh_iter:
-            self._train_iter = iter(_call_input_fn(input_fn, ModeKeys.TRAIN))
+            it0 = iter(_call_input_fn(input_fn, ModeKeys.TRAIN))
+            if self.config.prefetch and device:
+                from ..data.input_fn import DevicePrefetcher
+
+                it0 = DevicePrefetcher(it0, device, depth=self.config.prefetch)
+            self._train_iter = it0
             self._train_iter_key = input_fn
         it = self._train_iter
         try:

@@ -234,3 +234,42 @@ def test_estimator_window_fuse_mid_window_limit(tmp_path):
     r = est.train(train_input_fn, max_steps=10)  # 2 fused windows + 2 singles
     assert r["global_step"] == 10
     assert est._train_spec.train_op.engine.apply_count == 2
+
+
+def test_device_prefetcher_stream_identical():
+    from gradient_accumulation_tf_estimator_amd.data.input_fn import (
+        DevicePrefetcher)
+
+    def gen():
+        for i in range(7):
+            yield (torch.full((2, 3), float(i)), torch.tensor([i]))
+
+    ref = list(gen())
+    got = list(DevicePrefetcher(gen(), torch.device("cpu"), depth=3))
+    assert len(got) == len(ref)
+    for (fa, la), (fb, lb) in zip(ref, got):
+        assert torch.equal(fa, fb) and torch.equal(la, lb)
+
+    # exceptions propagate
+    def bad():
+        yield (torch.zeros(1), torch.zeros(1))
+        raise RuntimeError("producer blew up")
+
+    it = DevicePrefetcher(bad(), torch.device("cpu"))
+    next(it)
+    with pytest.raises(RuntimeError, match="blew up"):
+        next(it)
+
+
+def test_estimator_prefetch_equivalence(tmp_path):
+    res = {}
+    for depth in (0, 3):
+        cfg = RunConfig(model_dir=str(tmp_path / f"p{depth}"),
+                        tf_random_seed=19830610, device="cpu",
+                        prefetch=depth)
+        est = Estimator(mnist_model_fn, cfg,
+                        params={"learning_rate": 1e-3,
+                                "gradient_accumulation_multiplier": 2, "seed": 0})
+        est.train(train_input_fn, max_steps=16)
+        res[depth] = est._train_spec.train_op.state_dict()["master"].clone()
+    assert torch.equal(res[0], res[3])

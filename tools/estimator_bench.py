@@ -41,8 +41,9 @@ def input_fn(mode=None):
                torch.randint(0, 2, (B,), generator=g))
 
 
-def run(fuse, steps, warmup):
+def run(fuse, steps, warmup, prefetch=0):
     est = Estimator(model_fn, RunConfig(device="cuda", window_fuse=fuse,
+                                        prefetch=prefetch,
                                         log_step_count_steps=0))
     est.train(input_fn, max_steps=warmup)
     torch.cuda.synchronize()
@@ -60,9 +61,11 @@ def main():
     args = p.parse_args()
     eager = run(False, args.steps, args.warmup)
     fused = run(True, args.steps, args.warmup)
+    fusedp = run(True, args.steps, args.warmup, prefetch=3)
     print(f"estimator API, bert-small mb{B} K={K}: "
           f"per-micro-batch {eager:.0f} samples/s, "
-          f"window_fuse=True {fused:.0f} samples/s ({fused/eager:.2f}x)")
+          f"window_fuse=True {fused:.0f} ({fused/eager:.2f}x), "
+          f"+prefetch=3 {fusedp:.0f} ({fusedp/eager:.2f}x)")
 
 
 if __name__ == "__main__":
