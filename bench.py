@@ -174,6 +174,17 @@ def main():
     # fused fwd/bwd, engine.micro_step_many) ----
     fuse = 1
     if use_cuda and args.fuse_micro != "off":
+        if args.fused == "off" and args.graphs != "off" and \
+                args.fuse_micro == "auto":
+            # KNOWN ROCm/torch-capture instability (measured 2026-09-14):
+            # capturing the UNFUSED torch module tree at >= 4096 rows
+            # aborts asynchronously with HSA_STATUS_ERROR_MEMORY_APERTURE
+            # _VIOLATION (fused modules are clean at 20k steps, eager
+            # unfused is clean, <= 2048-row captures are clean; the fault
+            # surfaces only after ~100 steps of queue progress --
+            # evidence matrix in docs/NEXT_STEPS.md). Cap the A/B debug
+            # path's fused rows; explicit --fuse-micro overrides.
+            args.max_fuse_rows = min(args.max_fuse_rows, 2048)
         if args.fuse_micro == "auto":
             for d in range(args.accum, 0, -1):
                 if args.accum % d == 0 and args.steps % d == 0 and \
