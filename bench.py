@@ -176,14 +176,13 @@ def main():
     if use_cuda and args.fuse_micro != "off":
         if args.fused == "off" and args.graphs != "off" and \
                 args.fuse_micro == "auto":
-            # KNOWN ROCm/torch-capture instability (measured 2026-09-14):
-            # capturing the UNFUSED torch module tree at >= 4096 rows
-            # aborts asynchronously with HSA_STATUS_ERROR_MEMORY_APERTURE
-            # _VIOLATION (fused modules are clean at 20k steps, eager
-            # unfused is clean, <= 2048-row captures are clean; the fault
-            # surfaces only after ~100 steps of queue progress --
-            # evidence matrix in docs/NEXT_STEPS.md). Cap the A/B debug
-            # path's fused rows; explicit --fuse-micro overrides.
+            # KNOWN ROCm/torch bug (attributed, tools/capture_bug_bisect
+            # .py + docs/NEXT_STEPS.md): torch's nn.Embedding BACKWARD
+            # captured with >= 4096 indices that vary between replays
+            # aborts asynchronously (APERTURE_VIOLATION, surfacing ~100
+            # steps late). The fused path's DirectEmbedding is value-
+            # independent and immune; cap the unfused A/B path's fused
+            # rows at 2048. Explicit --fuse-micro overrides.
             args.max_fuse_rows = min(args.max_fuse_rows, 2048)
         if args.fuse_micro == "auto":
             for d in range(args.accum, 0, -1):

@@ -82,13 +82,13 @@ class RunConfig:
     # (TrainOp.step_fused -- exact by linearity; hipGraph-captured on the
     # HIP engine). Falls back to per-micro-batch stepping for ragged
     # tails, strict semantics, or when a step/checkpoint limit lands
-    # mid-window. NOTE: on this ROCm/torch build, capturing PLAIN-torch
-    # module trees at >= 4096 rows can abort asynchronously (known
-    # platform issue, docs/NEXT_STEPS.md); the in-house fused modules are
-    # unaffected. Keep fused windows under ~2048 rows for plain-torch
-    # models, or they will run eagerly anyway via the capture-failure
-    # fallback only if the capture FAILS (this fault does not fail the
-    # capture -- it aborts later), so prefer smaller batches there.
+    # mid-window. NOTE: on this ROCm/torch build, torch's nn.Embedding
+    # BACKWARD aborts asynchronously when captured with >= 4096 indices
+    # that change between replays (attributed via
+    # tools/capture_bug_bisect.py; docs/NEXT_STEPS.md). The in-house
+    # DirectEmbedding scatter is value-independent and unaffected; keep
+    # fused windows under ~2048 rows for plain-torch embedding models
+    # (the fault does NOT fail capture -- it aborts later).
     window_fuse: bool = False
     # background-thread input prefetch depth (0 = off, matching the
     # reference input_fn which has no prefetch -- SURVEY C12); overlaps
