@@ -171,3 +171,51 @@ def _dp_worker(rank, world, q):
     if rank == 0:
         q.put((rank, op.engine.state.master.numpy()))
     dist.destroy_process_group()
+
+
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=10, deadline=None, derandomize=True)
+@given(
+    K=st.integers(2, 6),
+    max_micro=st.integers(1, 6),
+    steps=st.integers(4, 12),
+    opt=st.sampled_from(["adamw", "adam"]),
+    strict=st.booleans(),
+    seed=st.integers(0, 10_000),
+)
+def test_random_fused_partitions_equal_sequential(K, max_micro, steps, opt,
+                                                  strict, seed):
+    """Any block partition fused_block_sizes(max_micro) produces must equal
+    the sequential chain -- property-swept over K, partition granularity,
+    optimizer flavor, and the strict predicate."""
+    torch.manual_seed(seed)
+    batches = data(steps, B=5, seed=seed + 1)
+
+    net_a, op_a = make(K, strict=strict, optimizer=opt)
+    for x, y in batches:
+        op_a.step(net_a.loss(x, y))
+
+    net_b, op_b = make(K, strict=strict, optimizer=opt)
+    i = 0
+    while i < len(batches):
+        advanced = False
+        for n in op_b.engine.fused_block_sizes(max_micro=max_micro):
+            n = min(n, len(batches) - i)
+            if n <= 0:
+                break
+            blk = batches[i : i + n]
+            x = torch.cat([b[0] for b in blk])
+            y = torch.cat([b[1] for b in blk])
+            op_b.step_fused(net_b.loss(x, y), n)
+            i += n
+            advanced = True
+        if not advanced:
+            break
+    assert op_b.engine.global_step == op_a.engine.global_step
+    assert op_b.engine.apply_count == op_a.engine.apply_count
+    for (na, pa), (_, pb) in zip(net_a.named_parameters(),
+                                 net_b.named_parameters()):
+        np.testing.assert_allclose(pb.detach().numpy(), pa.detach().numpy(),
+                                   rtol=5e-5, atol=5e-6, err_msg=na)
