@@ -116,6 +116,25 @@ def main():
     else:
         device = torch.device("cpu")
 
+    # experimental: sweep rocBLAS/hipBLASLt GEMM solutions for the bench's
+    # exact shapes (torch TunableOp).  GA_TUNABLEOP=tune records winners to
+    # GA_TUNABLEOP_FILE during the eager capture-warmup (every GEMM shape
+    # runs eagerly before hipGraph capture); =replay loads them read-only.
+    tunable = None
+    if use_cuda and os.environ.get("GA_TUNABLEOP") in ("tune", "replay"):
+        import torch.cuda.tunable as tunable
+
+        tunable.enable(True)
+        tunable.set_filename(
+            os.environ.get("GA_TUNABLEOP_FILE", "gpurun_out/tunableop.csv"),
+            insert_device_ordinal=False)
+        if os.environ["GA_TUNABLEOP"] == "tune":
+            tunable.tuning_enable(True)
+        else:
+            tunable.tuning_enable(False)
+            if not tunable.read_file():
+                print("tunableop: read_file failed", file=sys.stderr)
+
     from gradient_accumulation_tf_estimator_amd import create_optimizer
     from gradient_accumulation_tf_estimator_amd.models.bert import CONFIGS, BertForSequenceClassification
 
@@ -301,6 +320,13 @@ def main():
             },
         }
         print(json.dumps(out))
+
+    if tunable is not None and tunable.tuning_is_enabled():
+        # results file is written by the C++ side at process exit
+        os.makedirs(os.path.dirname(tunable.get_filename()) or ".",
+                    exist_ok=True)
+        print(f"tunableop: {len(tunable.get_results())} tuned results -> "
+              f"{tunable.get_filename()} (written at exit)", file=sys.stderr)
 
     if dist:
         dist.destroy_process_group()
