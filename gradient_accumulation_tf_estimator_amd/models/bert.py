@@ -19,6 +19,7 @@ north star.
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass
 
 import torch
@@ -32,6 +33,7 @@ from ..ops.fused import (
     FusedAddLayerNorm,
     FusedBiasGelu,
     FusedFFN,
+    ffn_mfma_supported,
     fused_attention,
     fused_attention_supported,
 )
@@ -157,17 +159,29 @@ class BertLayer(nn.Module):
             self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
             self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
         self.dropout = nn.Dropout(cfg.dropout)
+        self._custom_ffn = (cfg.fused
+                            and os.environ.get("GA_CUSTOM_FFN", "0") == "1"
+                            and ffn_mfma_supported(cfg.hidden_size,
+                                                   cfg.intermediate_size))
         if cfg.fused:
             self.attention_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
-            # NOTE: FusedFFN (GEMM-fused GELU epilogues) stays experimental:
-            # this hipblaslt build returns no GELU_AUX solutions, so the FFN
-            # uses DirectLinear + the fused bias+GELU kernels instead
-            self.intermediate = DirectLinear(cfg.hidden_size,
-                                             cfg.intermediate_size, bias=False)
-            self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
-            self.output = DirectLinear(cfg.intermediate_size, cfg.hidden_size,
-                                       bias=False)
+            if self._custom_ffn:
+                # GA_CUSTOM_FFN=1: both FFN boundary GEMMs with the GELU in
+                # the MFMA epilogue (ops/csrc/ffn_mfma.hip) -- no standalone
+                # activation kernels, one fewer [R,I] HBM round-trip per
+                # direction
+                self.ffn = FusedFFN(cfg.hidden_size, cfg.intermediate_size)
+            else:
+                # default FFN: this hipblaslt build returns no GELU_AUX
+                # solutions, so the activation runs as the fused
+                # bias+GELU kernels between DirectLinear GEMMs
+                self.intermediate = DirectLinear(cfg.hidden_size,
+                                                 cfg.intermediate_size,
+                                                 bias=False)
+                self.intermediate_act = FusedBiasGelu(cfg.intermediate_size)
+                self.output = DirectLinear(cfg.intermediate_size,
+                                           cfg.hidden_size, bias=False)
             self.output_LayerNorm = FusedAddLayerNorm(
                 cfg.hidden_size, eps=cfg.layer_norm_eps, proj_bias=True)
         else:
@@ -177,6 +191,12 @@ class BertLayer(nn.Module):
     def _bind_direct_extras(self, engine):
         # the intermediate Linear's wgrad colsum over d(pre-gelu) IS the
         # gelu bias gradient: delegate it so gelu backward is elementwise
+        if self.fused and self._custom_ffn:
+            # FusedFFN owns its bias-grad wgrad entry; only the deferred
+            # residual grad needs wiring (rides the FFN dx dgrad epilogue)
+            self.attention_LayerNorm._defer_residual_to = self.attention.qkv
+            self.output_LayerNorm._defer_residual_to = self.ffn
+            return
         if self.fused:
             self.intermediate._accum_view_b = engine.state.accum_view(
                 self.intermediate_act.bias)
@@ -194,7 +214,10 @@ class BertLayer(nn.Module):
         a = self.attention(x, attn_mask, mask8=mask8, seed=seed)
         if self.fused:
             x = self.attention_LayerNorm(self.dropout(self.attention_output(a)), residual=x)
-            h = self.output(self.intermediate_act(self.intermediate(x)))
+            if self._custom_ffn:
+                h = self.ffn(x)
+            else:
+                h = self.output(self.intermediate_act(self.intermediate(x)))
             return self.output_LayerNorm(self.dropout(h), residual=x)
         x = self.attention_LayerNorm(x + self.dropout(self.attention_output(a)))
         h = self.output(F.gelu(self.intermediate(x), approximate="tanh"))

@@ -1013,11 +1013,13 @@ void embgrad_acc(at::Tensor dy, at::Tensor ids, at::Tensor accum_slice, int64_t 
 void register_blas_acc(pybind11::module_& mod);
 void register_lt_gemm(pybind11::module_& mod);
 void register_grouped_wgrad(pybind11::module_& mod);
+void register_ffn_mfma(pybind11::module_& mod);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   register_blas_acc(mod);
   register_lt_gemm(mod);
   register_grouped_wgrad(mod);
+  register_ffn_mfma(mod);
   mod.def("accumulate", &accumulate, "accum += grad (fp32 upcast); grad = 0");
   mod.def("sqnorm", &sqnorm, "out[0] = sum(accum^2)");
   mod.def("fused_apply", &fused_apply,

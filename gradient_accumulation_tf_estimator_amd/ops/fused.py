@@ -568,19 +568,24 @@ def fused_attention(qkv: torch.Tensor, nh: int, mask8=None, seed=None,
 
 
 class _FFNFn(torch.autograd.Function):
-    """intermediate GEMM + bias + GELU (GELU_AUX_BIAS epilogue) -> output
-    GEMM, with the backward's dgelu fused into the output dgrad (DGELU
-    epilogue) and both wgrads + the intermediate bias grad routed through
-    the direct-accum wgrad path. Replaces the standalone bias+GELU kernels
-    entirely on the bound GPU path."""
+    """intermediate GEMM with bias+GELU fused as the MFMA epilogue
+    (ops/csrc/ffn_mfma.hip k_ffn_fwd; this hipblaslt build exposes no
+    GELU_AUX solutions, so the epilogue lives in a hand-written kernel) ->
+    output GEMM, with the backward's dgelu fused into the output dgrad
+    (k_ffn_dgrad_dgelu) and both wgrads + the intermediate bias grad
+    routed through the direct-accum wgrad path. Replaces the standalone
+    bias+GELU kernels entirely on the bound GPU path; the FFN-input
+    residual grad (deferred by FusedAddLayerNorm) rides the dx dgrad
+    epilogue exactly as DirectLinear's does."""
 
     @staticmethod
     def forward(ctx, x, wi, bi, wo, module):
         from . import gemm
 
+        hip = require_hip()
         x = x.contiguous()
         x2d = x.reshape(-1, x.shape[-1])
-        h, aux = gemm.linear_gelu(x2d, wi, bi)
+        h, aux = hip.ffn_fwd(x2d, wi, bi.contiguous())
         y = gemm.linear_fwd(h, wo, None)
         ctx.save_for_backward(x, aux, h, wi, wo)
         ctx.module = module
@@ -590,10 +595,11 @@ class _FFNFn(torch.autograd.Function):
     def backward(ctx, dy):
         from . import gemm
 
+        hip = require_hip()
         x, aux, h, wi, wo = ctx.saved_tensors
         mod = ctx.module
         dy2d = dy.contiguous().reshape(-1, dy.shape[-1])
-        d_h = gemm.dgrad_dgelu(dy2d, wo, aux)
+        d_h = hip.ffn_dgrad_dgelu(dy2d, wo, aux)
         x2d = x.reshape(-1, x.shape[-1])
         if _GROUPED_WGRAD:
             _pending_wgrads.append((h, dy2d, mod._accum_view_wo, None))
@@ -602,7 +608,13 @@ class _FFNFn(torch.autograd.Function):
             gemm.wgrad_acc(h, dy2d, mod._accum_view_wo)
             gemm.wgrad_acc(x2d, d_h, mod._accum_view_wi)
             mod._accum_view_bi.add_(d_h.sum(0, dtype=torch.float32))
-        dx = gemm.dgrad(d_h, wi).reshape(x.shape)
+        addend = getattr(mod, "_pending_dres_add", None)
+        if addend is not None:
+            mod._pending_dres_add = None
+            dx = gemm.dgrad_add(d_h, wi,
+                                addend.reshape(-1, addend.shape[-1])).reshape(x.shape)
+        else:
+            dx = gemm.dgrad(d_h, wi).reshape(x.shape)
         return dx, None, None, None, None
 
 
@@ -624,15 +636,30 @@ class FusedFFN(nn.Module):
         self._accum_view_wi = None
         self._accum_view_bi = None
         self._accum_view_wo = None
+        self._pending_dres_add = None
 
     def forward(self, x):
+        self._pending_dres_add = None  # drop any unconsumed deferred grad
         if self._accum_view_wi is not None:
             if not _use_hip(x):
                 raise RuntimeError("bound FusedFFN got non-bf16/non-GPU input")
-            return _FFNFn.apply(x, self.weight_in, self.bias_in, self.weight_out,
-                                self)
+            rows = x.numel() // x.shape[-1]
+            if rows % 256 == 0:
+                return _FFNFn.apply(x, self.weight_in, self.bias_in,
+                                    self.weight_out, self)
+            if torch.is_grad_enabled():
+                raise RuntimeError(
+                    f"FusedFFN training rows must be a multiple of 256, got "
+                    f"{rows} -- pad the batch or disable GA_CUSTOM_FFN")
+            # eval/predict with an odd row count: plain ops, no grads needed
         h = F.gelu(F.linear(x, self.weight_in, self.bias_in), approximate="tanh")
         return F.linear(h, self.weight_out)
+
+
+def ffn_mfma_supported(hidden: int, intermediate: int) -> bool:
+    """Shape gate for the k_ffn_* kernels (row count is checked at call
+    time; all BERT encoder shapes qualify)."""
+    return hidden % 64 == 0 and intermediate % 128 == 0 and hip_available()
 
 
 class _ClsHeadFn(torch.autograd.Function):

@@ -31,6 +31,7 @@ setup(
                 os.path.join(CSRC, "grouped_wgrad.hip"),
                 os.path.join(CSRC, "attn.hip"),
                 os.path.join(CSRC, "wgrad_mfma.hip"),
+                os.path.join(CSRC, "ffn_mfma.hip"),
                 os.path.join(CSRC, "cls_head.hip"),
                 os.path.join(CSRC, "linear_small.hip"),
                 os.path.join(CSRC, "ga_bindings.hip"),
