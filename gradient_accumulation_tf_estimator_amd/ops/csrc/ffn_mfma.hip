@@ -99,36 +99,36 @@ static __device__ __forceinline__ void dstage_write(unsigned short* lds,
 }
 
 // ---- transpose staging (wgrad idiom): [64 h][128 c] global -> LDS
-// [128 c][64 h]; 4-row packs so LDS writes are 8 B. 512 threads. ----
+// [128 c][64 h]; 2-row packs: 64r x 128c / (2r x 8c) = 512 sub-blocks,
+// exactly one per thread; LDS writes are 4 B. ----
+typedef __attribute__((ext_vector_type(2))) unsigned short ush2f;
+
 struct TStage {
-  bf16x8 r[4];  // 4 h-rows x 8 c each, half the chunk per 512 threads
+  bf16x8 r[2];  // 2 h-rows x 8 c per thread
 };
 
 static __device__ __forceinline__ void tstage_issue(const unsigned short* g,
                                                     long long ld, int c0,
                                                     TStage& s) {
-  // 64r x 128c / (4r x 8c) = 512 sub-blocks exactly
   const int blk = threadIdx.x;
-  const int r0 = (blk >> 4) * 4;
+  const int r0 = (blk >> 4) * 2;
   const int cc = (blk & 15) * 8;
 #pragma unroll
-  for (int t = 0; t < 4; ++t)
+  for (int t = 0; t < 2; ++t)
     s.r[t] = *(const bf16x8*)(g + (long long)(r0 + t) * ld + c0 + cc);
 }
 
 static __device__ __forceinline__ void tstage_write(unsigned short* lds,
                                                     const TStage& s) {
   const int blk = threadIdx.x;
-  const int r0 = (blk >> 4) * 4;
+  const int r0 = (blk >> 4) * 2;
   const int cc = (blk & 15) * 8;
   const unsigned short* u0 = (const unsigned short*)&s.r[0];
   const unsigned short* u1 = (const unsigned short*)&s.r[1];
-  const unsigned short* u2 = (const unsigned short*)&s.r[2];
-  const unsigned short* u3 = (const unsigned short*)&s.r[3];
 #pragma unroll
   for (int c = 0; c < 8; ++c) {
-    ush4f pack = {u0[c], u1[c], u2[c], u3[c]};
-    *(ush4f*)((char*)lds + fswz(cc + c, r0 * 2)) = pack;
+    ush2f pack = {u0[c], u1[c]};
+    *(ush2f*)((char*)lds + fswz(cc + c, r0 * 2)) = pack;
   }
 }
 
