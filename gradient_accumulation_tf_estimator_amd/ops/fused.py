@@ -659,6 +659,8 @@ class FusedFFN(nn.Module):
 def ffn_mfma_supported(hidden: int, intermediate: int) -> bool:
     """Shape gate for the k_ffn_* kernels (row count is checked at call
     time; all BERT encoder shapes qualify)."""
+    from . import hip_available
+
     return hidden % 64 == 0 and intermediate % 128 == 0 and hip_available()
 
 
