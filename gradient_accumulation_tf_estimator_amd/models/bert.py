@@ -159,8 +159,17 @@ class BertLayer(nn.Module):
             self.intermediate = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
             self.output = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
         self.dropout = nn.Dropout(cfg.dropout)
+        # GA_CUSTOM_FFN: "1" force, "0" off, "auto" (default) = on for the
+        # shapes where the k_ffn_* MFMA-epilogue kernels measured faster
+        # (H=512/I=2048 at fused-window rows; bert-base/large shapes still
+        # favor hipBLASLt + the standalone gelu kernels -- see
+        # docs/KERNELS.md). Row-count dispatch is inside _FFNFn.
+        _ffn_env = os.environ.get("GA_CUSTOM_FFN", "auto")
         self._custom_ffn = (cfg.fused
-                            and os.environ.get("GA_CUSTOM_FFN", "0") == "1"
+                            and (_ffn_env == "1"
+                                 or (_ffn_env == "auto"
+                                     and cfg.hidden_size == 512
+                                     and cfg.intermediate_size == 2048))
                             and ffn_mfma_supported(cfg.hidden_size,
                                                    cfg.intermediate_size))
         if cfg.fused:
@@ -187,6 +196,16 @@ class BertLayer(nn.Module):
         else:
             self.attention_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
             self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kw):
+        if self._custom_ffn:
+            remap = {"intermediate.weight": "ffn.weight_in",
+                     "intermediate_act.bias": "ffn.bias_in",
+                     "output.weight": "ffn.weight_out"}
+            for old, new in remap.items():
+                if prefix + old in state_dict:
+                    state_dict[prefix + new] = state_dict.pop(prefix + old)
+        super()._load_from_state_dict(state_dict, prefix, *args, **kw)
 
     def _bind_direct_extras(self, engine):
         # the intermediate Linear's wgrad colsum over d(pre-gelu) IS the

@@ -578,6 +578,13 @@ class _FFNFn(torch.autograd.Function):
     residual grad (deferred by FusedAddLayerNorm) rides the dx dgrad
     epilogue exactly as DirectLinear's does."""
 
+    # the MFMA tile kernels win at large fused-window row counts (measured
+    # +0.9% headline at R=4096; they LOSE at R=1024 where 64 workgroups
+    # under-fill the 256-CU chip) -- smaller/ragged batches take the
+    # composite route (hipBLASLt GEMM + the standalone bias+GELU kernels)
+    # through the same autograd Function
+    KERNEL_MIN_ROWS = 2048
+
     @staticmethod
     def forward(ctx, x, wi, bi, wo, module):
         from . import gemm
@@ -585,10 +592,17 @@ class _FFNFn(torch.autograd.Function):
         hip = require_hip()
         x = x.contiguous()
         x2d = x.reshape(-1, x.shape[-1])
-        h, aux = hip.ffn_fwd(x2d, wi, bi.contiguous())
+        use_kernel = (x2d.shape[0] % 256 == 0
+                      and x2d.shape[0] >= _FFNFn.KERNEL_MIN_ROWS)
+        if use_kernel:
+            h, aux = hip.ffn_fwd(x2d, wi, bi.contiguous())
+        else:
+            aux = gemm.linear_fwd(x2d, wi, bi)  # pre-activation
+            h = hip.biasgelu_fwd(aux, module._zero_bias())
         y = gemm.linear_fwd(h, wo, None)
         ctx.save_for_backward(x, aux, h, wi, wo)
         ctx.module = module
+        ctx.use_kernel = use_kernel
         return y.reshape(*x.shape[:-1], wo.shape[0])
 
     @staticmethod
@@ -599,7 +613,11 @@ class _FFNFn(torch.autograd.Function):
         x, aux, h, wi, wo = ctx.saved_tensors
         mod = ctx.module
         dy2d = dy.contiguous().reshape(-1, dy.shape[-1])
-        d_h = hip.ffn_dgrad_dgelu(dy2d, wo, aux)
+        if ctx.use_kernel:
+            d_h = hip.ffn_dgrad_dgelu(dy2d, wo, aux)
+        else:
+            d_h = hip.biasgelu_bwd_ew(gemm.dgrad(dy2d, wo), aux,
+                                      mod._zero_bias())
         x2d = x.reshape(-1, x.shape[-1])
         if _GROUPED_WGRAD:
             _pending_wgrads.append((h, dy2d, mod._accum_view_wo, None))
@@ -637,21 +655,22 @@ class FusedFFN(nn.Module):
         self._accum_view_bi = None
         self._accum_view_wo = None
         self._pending_dres_add = None
+        self._zbias = None
+
+    def _zero_bias(self):
+        # the composite route reuses the bias+GELU kernels with the bias
+        # already folded into the saved pre-activation
+        if self._zbias is None or self._zbias.device != self.bias_in.device:
+            self._zbias = torch.zeros_like(self.bias_in)
+        return self._zbias
 
     def forward(self, x):
         self._pending_dres_add = None  # drop any unconsumed deferred grad
         if self._accum_view_wi is not None:
             if not _use_hip(x):
                 raise RuntimeError("bound FusedFFN got non-bf16/non-GPU input")
-            rows = x.numel() // x.shape[-1]
-            if rows % 256 == 0:
-                return _FFNFn.apply(x, self.weight_in, self.bias_in,
-                                    self.weight_out, self)
-            if torch.is_grad_enabled():
-                raise RuntimeError(
-                    f"FusedFFN training rows must be a multiple of 256, got "
-                    f"{rows} -- pad the batch or disable GA_CUSTOM_FFN")
-            # eval/predict with an odd row count: plain ops, no grads needed
+            return _FFNFn.apply(x, self.weight_in, self.bias_in,
+                                self.weight_out, self)
         h = F.gelu(F.linear(x, self.weight_in, self.bias_in), approximate="tanh")
         return F.linear(h, self.weight_out)
 
