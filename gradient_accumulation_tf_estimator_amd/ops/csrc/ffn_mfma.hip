@@ -263,6 +263,175 @@ static __device__ __forceinline__ void ffn_tile(
   }
 }
 
+// ---------------------------------------------------------------------------
+// 256x256-tile variant (GA_FFN_TILE=256): halves the W-panel re-reads
+// (16 -> 8 r-tiles share each staged i-slice) at the cost of half the
+// workgroups (128 at the bert-small fused window vs 256 CUs). acc[2][4]
+// per wave as in wgrad_mfma.hip's 256 kernel; 64 KB LDS (two [256][64]
+// panels), epilogue bounced through a [64 r][256 i] fp32 tile.
+// ---------------------------------------------------------------------------
+struct TStage256 {
+  bf16x8 r[2][2];  // 2 iters x 2 h-rows x 8 c
+};
+
+static __device__ __forceinline__ void tstage256_issue(const unsigned short* g,
+                                                       long long ld,
+                                                       TStage256& s) {
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int blk = threadIdx.x + it * 512;
+    const int r0 = (blk >> 5) * 2;
+    const int cc = (blk & 31) * 8;
+#pragma unroll
+    for (int t = 0; t < 2; ++t)
+      s.r[it][t] = *(const bf16x8*)(g + (long long)(r0 + t) * ld + cc);
+  }
+}
+
+static __device__ __forceinline__ void tstage256_write(unsigned short* lds,
+                                                       const TStage256& s) {
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    const int blk = threadIdx.x + it * 512;
+    const int r0 = (blk >> 5) * 2;
+    const int cc = (blk & 31) * 8;
+    const unsigned short* u0 = (const unsigned short*)&s.r[it][0];
+    const unsigned short* u1 = (const unsigned short*)&s.r[it][1];
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      ush2f pack = {u0[c], u1[c]};
+      *(ush2f*)((char*)lds + fswz(cc + c, r0 * 2)) = pack;
+    }
+  }
+}
+
+template <int MODE>
+static __device__ __forceinline__ void ffn_tile256(
+    const unsigned short* a, const unsigned short* b,
+    const unsigned short* bias, const unsigned short* aux_in,
+    unsigned short* out0, unsigned short* out1, int H, long long ldI) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo31 = lane & 31;
+  const int hi = (lane >> 5) & 1;
+  const int wr = wave >> 1;  // 4 x 64-row quadrant
+  const int wk = wave & 1;   // 2 x 128-col half
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  unsigned short* As = (unsigned short*)smem;           // [256 r][64 h]
+  unsigned short* Bs = (unsigned short*)(smem + 32768); // [256 i][64 h]
+
+  f32x16 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x16)(0.f);
+
+  DStage<256> sa, sbf;
+  TStage256 sbt;
+  dstage_issue<256>(a, H, sa);
+  if (MODE == 0)
+    dstage_issue<256>(b, H, sbf);
+  else
+    tstage256_issue(b, ldI, sbt);
+  for (int h0 = 0; h0 < H; h0 += 64) {
+    __syncthreads();
+    dstage_write<256>(As, sa);
+    if (MODE == 0)
+      dstage_write<256>(Bs, sbf);
+    else
+      tstage256_write(Bs, sbt);
+    if (h0 + 64 < H) {
+      dstage_issue<256>(a + h0 + 64, H, sa);
+      if (MODE == 0)
+        dstage_issue<256>(b + h0 + 64, H, sbf);
+      else
+        tstage256_issue(b + (long long)(h0 + 64) * ldI, ldI, sbt);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {
+      bf16x8 a0 = *(const bf16x8*)((char*)As + fswz(wr * 64 + lo31, s * 32 + hi * 16));
+      bf16x8 a1 = *(const bf16x8*)((char*)As + fswz(wr * 64 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b0 = *(const bf16x8*)((char*)Bs + fswz(wk * 128 + lo31, s * 32 + hi * 16));
+      bf16x8 b1 = *(const bf16x8*)((char*)Bs + fswz(wk * 128 + 32 + lo31, s * 32 + hi * 16));
+      bf16x8 b2 = *(const bf16x8*)((char*)Bs + fswz(wk * 128 + 64 + lo31, s * 32 + hi * 16));
+      bf16x8 b3 = *(const bf16x8*)((char*)Bs + fswz(wk * 128 + 96 + lo31, s * 32 + hi * 16));
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b1, acc[0][1], 0, 0, 0);
+      acc[0][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b2, acc[0][2], 0, 0, 0);
+      acc[0][3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b3, acc[0][3], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc[1][1], 0, 0, 0);
+      acc[1][2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b2, acc[1][2], 0, 0, 0);
+      acc[1][3] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b3, acc[1][3], 0, 0, 0);
+    }
+  }
+
+  // 4 passes over the 64-r quadrants; ftile [64 r][256 i] fp32 = 64 KB
+  float* ftile = (float*)smem;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    __syncthreads();
+    if (wr == p) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int col = wk * 128 + j * 32 + lo31;
+#pragma unroll
+          for (int rq = 0; rq < 4; ++rq) {
+            const int rb = i * 32 + 8 * rq + 4 * hi;
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+              *(float*)((char*)ftile + (long long)(rb + e) * 1024 +
+                        ((col * 4) ^ (((rb + e) & 7) << 4))) =
+                  acc[i][j][rq * 4 + e];
+          }
+        }
+    }
+    __syncthreads();
+    {
+      const int r = threadIdx.x >> 3;
+      const int c32 = (threadIdx.x & 7) * 32;
+      const long long gr = (long long)(p * 64 + r) * ldI + c32;
+      unsigned short ybuf[32];
+      unsigned short xbuf[32];
+      bf16x8 auxv[4];
+      if (MODE == 1) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q) auxv[q] = *(const bf16x8*)(aux_in + gr + q * 8);
+      }
+#pragma unroll
+      for (int e = 0; e < 32; ++e) {
+        const int col = c32 + e;
+        float v = *(const float*)((char*)ftile + (long long)r * 1024 +
+                                  ((col * 4) ^ ((r & 7) << 4)));
+        if (MODE == 0) {
+          __bf16 bb = ((const __bf16*)bias)[col];
+          const float pre = v + (float)bb;
+          const __bf16 preb = (__bf16)pre;
+          xbuf[e] = *(const unsigned short*)&preb;
+          const __bf16 yb = (__bf16)fgelu_fwd1(pre);
+          ybuf[e] = *(const unsigned short*)&yb;
+        } else {
+          const float h = (float)((const __bf16*)&auxv[e >> 3])[e & 7];
+          const __bf16 db = (__bf16)fgelu_bwd1(h, v);
+          ybuf[e] = *(const unsigned short*)&db;
+        }
+      }
+#pragma unroll
+      for (int q = 0; q < 4; ++q)
+        *(bf16x8*)(out0 + gr + q * 8) = *(const bf16x8*)&ybuf[q * 8];
+      if (MODE == 0) {
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          *(bf16x8*)(out1 + gr + q * 8) = *(const bf16x8*)&xbuf[q * 8];
+      }
+    }
+  }
+}
+
 // XCD-aware bijective remap (wgrad idiom): consecutive remapped ids share
 // the X panel (same r-stripe), so its re-reads stay in one XCD's L2.
 static inline __device__ int ffn_remap() {
@@ -293,6 +462,28 @@ extern "C" __global__ __launch_bounds__(512) void k_ffn_dgrad_dgelu(
               dpre + r0 * I + i0, nullptr, H, I);
 }
 
+extern "C" __global__ __launch_bounds__(512) void k_ffn_fwd_t256(
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ w,
+    const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
+    unsigned short* __restrict__ aux, int H, int I) {
+  const int id = ffn_remap();
+  const int ti = id % (I / 256), tr = id / (I / 256);
+  const long long r0 = (long long)tr * 256, i0 = (long long)ti * 256;
+  ffn_tile256<0>(x + r0 * H, w + i0 * H, bias + i0, nullptr,
+                 y + r0 * I + i0, aux + r0 * I + i0, H, I);
+}
+
+extern "C" __global__ __launch_bounds__(512) void k_ffn_dgrad_dgelu_t256(
+    const unsigned short* __restrict__ dy, const unsigned short* __restrict__ wo,
+    const unsigned short* __restrict__ auxp, unsigned short* __restrict__ dpre,
+    int H, int I) {
+  const int id = ffn_remap();
+  const int ti = id % (I / 256), tr = id / (I / 256);
+  const long long r0 = (long long)tr * 256, i0 = (long long)ti * 256;
+  ffn_tile256<1>(dy + r0 * H, wo + i0, nullptr, auxp + r0 * I + i0,
+                 dpre + r0 * I + i0, nullptr, H, I);
+}
+
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
@@ -302,6 +493,16 @@ void check_ffn(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
                   t.scalar_type() == at::kBFloat16,
               "ffn_mfma: ", name, " must be contiguous CUDA bf16");
+}
+
+// GA_FFN_TILE=256 selects the wide-tile variant (halved W re-reads, half
+// the workgroups); default 128. Read once.
+bool use_t256() {
+  static const bool v = [] {
+    const char* e = getenv("GA_FFN_TILE");
+    return e && atoi(e) == 256;
+  }();
+  return v;
 }
 
 std::vector<at::Tensor> ffn_fwd(at::Tensor x2d, at::Tensor w, at::Tensor bias) {
@@ -316,12 +517,20 @@ std::vector<at::Tensor> ffn_fwd(at::Tensor x2d, at::Tensor w, at::Tensor bias) {
   auto y = at::empty({R, I}, x2d.options());
   auto aux = at::empty({R, I}, x2d.options());
   auto stream = at::cuda::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_ffn_fwd, dim3((R / 256) * (I / 128)), dim3(512), 49152,
-                     stream, (const unsigned short*)x2d.data_ptr(),
-                     (const unsigned short*)w.data_ptr(),
-                     (const unsigned short*)bias.data_ptr(),
-                     (unsigned short*)y.data_ptr(),
-                     (unsigned short*)aux.data_ptr(), (int)H, (int)I);
+  if (use_t256() && I % 256 == 0)
+    hipLaunchKernelGGL(k_ffn_fwd_t256, dim3((R / 256) * (I / 256)), dim3(512),
+                       65536, stream, (const unsigned short*)x2d.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (const unsigned short*)bias.data_ptr(),
+                       (unsigned short*)y.data_ptr(),
+                       (unsigned short*)aux.data_ptr(), (int)H, (int)I);
+  else
+    hipLaunchKernelGGL(k_ffn_fwd, dim3((R / 256) * (I / 128)), dim3(512), 49152,
+                       stream, (const unsigned short*)x2d.data_ptr(),
+                       (const unsigned short*)w.data_ptr(),
+                       (const unsigned short*)bias.data_ptr(),
+                       (unsigned short*)y.data_ptr(),
+                       (unsigned short*)aux.data_ptr(), (int)H, (int)I);
   return {y, aux};
 }
 
@@ -337,11 +546,20 @@ at::Tensor ffn_dgrad_dgelu(at::Tensor dy2d, at::Tensor wo, at::Tensor aux) {
               "x", H);
   auto dpre = at::empty({R, I}, dy2d.options());
   auto stream = at::cuda::getCurrentHIPStream().stream();
-  hipLaunchKernelGGL(k_ffn_dgrad_dgelu, dim3((R / 256) * (I / 128)), dim3(512),
-                     49152, stream, (const unsigned short*)dy2d.data_ptr(),
-                     (const unsigned short*)wo.data_ptr(),
-                     (const unsigned short*)aux.data_ptr(),
-                     (unsigned short*)dpre.data_ptr(), (int)H, (int)I);
+  if (use_t256() && I % 256 == 0)
+    hipLaunchKernelGGL(k_ffn_dgrad_dgelu_t256, dim3((R / 256) * (I / 256)),
+                       dim3(512), 65536, stream,
+                       (const unsigned short*)dy2d.data_ptr(),
+                       (const unsigned short*)wo.data_ptr(),
+                       (const unsigned short*)aux.data_ptr(),
+                       (unsigned short*)dpre.data_ptr(), (int)H, (int)I);
+  else
+    hipLaunchKernelGGL(k_ffn_dgrad_dgelu, dim3((R / 256) * (I / 128)),
+                       dim3(512), 49152, stream,
+                       (const unsigned short*)dy2d.data_ptr(),
+                       (const unsigned short*)wo.data_ptr(),
+                       (const unsigned short*)aux.data_ptr(),
+                       (unsigned short*)dpre.data_ptr(), (int)H, (int)I);
   return dpre;
 }
 
