@@ -34,17 +34,15 @@ def main():
                       f"rocpd_info_kernel_symbol_{sfx}")}
     kd = f"rocpd_kernel_dispatch_{sfx}"
     ks = f"rocpd_info_kernel_symbol_{sfx}"
-    # counter name table, if present
-    pd = next((t for t in tables if "pmc_description" in t or "info_pmc" in t
-               or "counter" in t.lower()), None)
+    pd = next((t for t in tables if t.startswith("rocpd_info_pmc_")), None)
     try:
         q = f"""
-        SELECT s.display_name, e.name, AVG(e.value), COUNT(*)
-        FROM (SELECT p.event_id AS eid, d.kernel_id AS kid, p.value AS value,
-                     p.name AS name
-              FROM {pmc} p JOIN {kd} d ON p.event_id = d.event_id) e
-        JOIN {ks} s ON e.kid = s.id
-        GROUP BY s.display_name, e.name"""
+        SELECT s.display_name, i.name, AVG(p.value), COUNT(*)
+        FROM {pmc} p
+        JOIN {kd} d ON p.event_id = d.event_id
+        JOIN {ks} s ON d.kernel_id = s.id
+        JOIN {pd} i ON p.pmc_id = i.id
+        GROUP BY s.display_name, i.name"""
         rows = list(con.execute(q))
     except sqlite3.OperationalError:
         # layout differs -- dump schema so the query can be adapted
