@@ -330,7 +330,13 @@ std::vector<at::Tensor> addln_fwd(at::Tensor x, c10::optional<at::Tensor> res,
   static const bool want_split = std::getenv("GA_LN_SPLIT2") != nullptr;
   const bool split2 = want_split && (H == 512 || H == 1024) && R >= 256;
   const int units = split2 ? R * 2 : R;
-  int blocks = std::min((units + 3) / 4, 2048);
+  // GA_LN_FWD_BLOCKS: like the backward's 256-cap (serial rows/wave
+  // amortize the gamma/beta loads), A/B'd on the forward
+  static const int fwd_cap = [] {
+    const char* v = getenv("GA_LN_FWD_BLOCKS");
+    return v ? atoi(v) : 2048;
+  }();
+  int blocks = std::max(1, std::min((units + 3) / 4, fwd_cap));
   void (*kern)(const unsigned short*, const unsigned short*, const unsigned short*,
                const unsigned short*, const unsigned short*, unsigned short*,
                unsigned short*, float*, float*, int, int, float) =
