@@ -99,36 +99,40 @@ static __device__ __forceinline__ void dstage_write(unsigned short* lds,
 }
 
 // ---- transpose staging (wgrad idiom): [64 h][128 c] global -> LDS
-// [128 c][64 h]; 2-row packs: 64r x 128c / (2r x 8c) = 512 sub-blocks,
-// exactly one per thread; LDS writes are 4 B. ----
+// [128 c][64 h]; 4-row x 4-col packs (64/4 x 128/4 = 512 sub-blocks, one
+// per thread) so the LDS writes are 8 B 4-row packs as in wgrad_mfma.hip
+// (a 2-row/4 B variant measured 2.5 LdsBankConflict per dispatch). ----
 typedef __attribute__((ext_vector_type(2))) unsigned short ush2f;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 
 struct TStage {
-  bf16x8 r[2];  // 2 h-rows x 8 c per thread
+  bf16x4 r[4];  // 4 h-rows x 4 c per thread
 };
 
 static __device__ __forceinline__ void tstage_issue(const unsigned short* g,
                                                     long long ld, int c0,
                                                     TStage& s) {
   const int blk = threadIdx.x;
-  const int r0 = (blk >> 4) * 2;
-  const int cc = (blk & 15) * 8;
+  const int r0 = (blk >> 5) * 4;
+  const int cc = (blk & 31) * 4;
 #pragma unroll
-  for (int t = 0; t < 2; ++t)
-    s.r[t] = *(const bf16x8*)(g + (long long)(r0 + t) * ld + c0 + cc);
+  for (int t = 0; t < 4; ++t)
+    s.r[t] = *(const bf16x4*)(g + (long long)(r0 + t) * ld + c0 + cc);
 }
 
 static __device__ __forceinline__ void tstage_write(unsigned short* lds,
                                                     const TStage& s) {
   const int blk = threadIdx.x;
-  const int r0 = (blk >> 4) * 2;
-  const int cc = (blk & 15) * 8;
+  const int r0 = (blk >> 5) * 4;
+  const int cc = (blk & 31) * 4;
   const unsigned short* u0 = (const unsigned short*)&s.r[0];
   const unsigned short* u1 = (const unsigned short*)&s.r[1];
+  const unsigned short* u2 = (const unsigned short*)&s.r[2];
+  const unsigned short* u3 = (const unsigned short*)&s.r[3];
 #pragma unroll
-  for (int c = 0; c < 8; ++c) {
-    ush2f pack = {u0[c], u1[c]};
-    *(ush2f*)((char*)lds + fswz(cc + c, r0 * 2)) = pack;
+  for (int c = 0; c < 4; ++c) {
+    ush4f pack = {u0[c], u1[c], u2[c], u3[c]};
+    *(ush4f*)((char*)lds + fswz(cc + c, r0 * 2)) = pack;
   }
 }
 
