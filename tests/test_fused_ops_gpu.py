@@ -167,10 +167,13 @@ def test_direct_accum_equals_grad_path():
         assert (abs(a).sum() > 0) or "bias" in sA.name
 
 
-def test_bertlayer_delegated_gelu_bias():
-    """A bound fused BertLayer delegates the gelu bias gradient to the
-    intermediate Linear's wgrad colsum (models/bert.py _bind_direct_extras);
-    the accumulated bias grad must match an unbound twin's .grad path."""
+def test_bertlayer_delegated_gelu_bias(monkeypatch):
+    """A bound fused BertLayer on the COMPOSITE FFN path (GA_CUSTOM_FFN=0;
+    the custom-FFN path is covered by test_custom_ffn_layer_parity)
+    delegates the gelu bias gradient to the intermediate Linear's wgrad
+    colsum (models/bert.py _bind_direct_extras); the accumulated bias grad
+    must match an unbound twin's .grad path."""
+    monkeypatch.setenv("GA_CUSTOM_FFN", "0")
     from gradient_accumulation_tf_estimator_amd.engine.accum import AccumEngine
     from gradient_accumulation_tf_estimator_amd.models.bert import BertConfig, BertLayer
     from gradient_accumulation_tf_estimator_amd.ops.fused import (
