@@ -219,3 +219,27 @@ def test_random_fused_partitions_equal_sequential(K, max_micro, steps, opt,
                                  net_b.named_parameters()):
         np.testing.assert_allclose(pb.detach().numpy(), pa.detach().numpy(),
                                    rtol=5e-5, atol=5e-6, err_msg=na)
+
+
+def test_fused_ffn_cpu_fallback_matches_torch():
+    """Unbound FusedFFN (CPU eager path) == the explicit torch expression,
+    including through autograd -- guards the fallback branch the GPU
+    kernels shadow."""
+    from gradient_accumulation_tf_estimator_amd.ops.fused import FusedFFN
+
+    torch.manual_seed(0)
+    ffn = FusedFFN(32, 128)
+    x = torch.randn(6, 32, requires_grad=True)
+    y = ffn(x)
+    y.square().mean().backward()
+
+    xf = x.detach().clone().requires_grad_()
+    h = torch.nn.functional.gelu(
+        torch.nn.functional.linear(xf, ffn.weight_in, ffn.bias_in),
+        approximate="tanh")
+    yref = torch.nn.functional.linear(h, ffn.weight_out)
+    yref.square().mean().backward()
+    np.testing.assert_allclose(y.detach().numpy(), yref.detach().numpy(),
+                               rtol=1e-6, atol=1e-7)
+    np.testing.assert_allclose(x.grad.numpy(), xf.grad.numpy(),
+                               rtol=1e-6, atol=1e-7)
