@@ -28,11 +28,12 @@ def make_net():
     return nn.Sequential(nn.Linear(IN_DIM, 16), nn.ReLU(), nn.Linear(16, 1))
 
 
-def make_data():
+def make_data(n=None):
     g = torch.Generator().manual_seed(5)
-    # 2 windows x WORLD ranks x K micro-steps
-    X = torch.randn(2 * WORLD * K, B, IN_DIM, generator=g)
-    y = torch.randn(2 * WORLD * K, B, 1, generator=g)
+    # 2 windows x WORLD ranks x K micro-steps (or n explicit micro-batches)
+    n = n if n is not None else 2 * WORLD * K
+    X = torch.randn(n, B, IN_DIM, generator=g)
+    y = torch.randn(n, B, 1, generator=g)
     return X, y
 
 
@@ -145,3 +146,57 @@ def test_sharded_apply_equals_replicated_bf16(tmp_path):
     b = torch.load(tmp_path / "sd_0.pt", weights_only=False)
     for key in ("master", "m", "v", "model", "accum"):
         assert torch.equal(a[key], b[key]), f"{key} diverged under sharding"
+
+
+def _worker_w(rank, world, tmpdir):
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+
+    dist.init_process_group(
+        "gloo", init_method=f"file://{tmpdir}/store4", rank=rank,
+        world_size=world)
+    net = make_net()
+    op = create_optimizer(net, 1e-2, 10**9, 0,
+                          gradient_accumulation_multiplier=K, clip_norm=1.0)
+    X, y = make_data(2 * world * K)
+    for w in range(2):
+        for k in range(K):
+            i = (w * K + k) * world + rank
+            op.step(((net(X[i]) - y[i]) ** 2).mean())
+    if rank == 0:
+        torch.save(op.engine.state.master.clone(),
+                   os.path.join(tmpdir, "dp4.pt"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_dp4_equals_single_process_k8(tmp_path):
+    """World 4 (the driver's intermediate SCALE point): DP=4 x K=2 must
+    equal single-process K=8 -- catches any world-size-dependent bug the
+    world-2 test can't (padding divisors, bucket chunking, 1/W scaling)."""
+    world = 4
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_w, args=(r, world, str(tmp_path)))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+
+    from gradient_accumulation_tf_estimator_amd import create_optimizer
+
+    net = make_net()
+    op = create_optimizer(net, 1e-2, 10**9, 0,
+                          gradient_accumulation_multiplier=world * K,
+                          clip_norm=1.0)
+    X, y = make_data(2 * world * K)
+    for i in range(2 * world * K):
+        op.step(((net(X[i]) - y[i]) ** 2).mean())
+
+    dp_master = torch.load(tmp_path / "dp4.pt", weights_only=True)
+    # world>1 pads the flat layout to 64*W (sharded-boundary alignment):
+    # compare the real-parameter prefix only
+    n = sum(pp.numel() for pp in net.parameters())
+    np.testing.assert_allclose(
+        dp_master[:n].numpy(), op.engine.state.master[:n].numpy(), rtol=1e-5,
+        atol=1e-7, err_msg="DP=4 x K=2 diverged from single-process K=8")
