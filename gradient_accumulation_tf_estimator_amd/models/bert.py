@@ -198,13 +198,16 @@ class BertLayer(nn.Module):
             self.output_LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
 
     def _load_from_state_dict(self, state_dict, prefix, *args, **kw):
-        if self._custom_ffn:
-            remap = {"intermediate.weight": "ffn.weight_in",
-                     "intermediate_act.bias": "ffn.bias_in",
-                     "output.weight": "ffn.weight_out"}
-            for old, new in remap.items():
-                if prefix + old in state_dict:
-                    state_dict[prefix + new] = state_dict.pop(prefix + old)
+        # checkpoints are loadable across the GA_CUSTOM_FFN setting: the
+        # FFN params are the same tensors under either module layout
+        remap = {"intermediate.weight": "ffn.weight_in",
+                 "intermediate_act.bias": "ffn.bias_in",
+                 "output.weight": "ffn.weight_out"}
+        if not self._custom_ffn:
+            remap = {v: k for k, v in remap.items()}
+        for old, new in remap.items():
+            if prefix + old in state_dict:
+                state_dict[prefix + new] = state_dict.pop(prefix + old)
         super()._load_from_state_dict(state_dict, prefix, *args, **kw)
 
     def _bind_direct_extras(self, engine):

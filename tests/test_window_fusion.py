@@ -243,3 +243,30 @@ def test_fused_ffn_cpu_fallback_matches_torch():
                                rtol=1e-6, atol=1e-7)
     np.testing.assert_allclose(x.grad.numpy(), xf.grad.numpy(),
                                rtol=1e-6, atol=1e-7)
+
+
+def test_ffn_checkpoint_keys_remap_into_composite(monkeypatch):
+    """A checkpoint saved under the GA_CUSTOM_FFN layout (ffn.weight_in /
+    ffn.bias_in / ffn.weight_out) loads into a composite fused layer: the
+    inverse key remap in BertLayer._load_from_state_dict converts them to
+    intermediate.weight / intermediate_act.bias / output.weight."""
+    from gradient_accumulation_tf_estimator_amd.models.bert import (
+        BertConfig, BertLayer)
+
+    monkeypatch.setenv("GA_CUSTOM_FFN", "0")
+    cfg = BertConfig(fused=True)  # fused modules construct fine on CPU
+    torch.manual_seed(9)
+    lay = BertLayer(cfg)
+    sd = lay.state_dict()
+    fwd = {"intermediate.weight": "ffn.weight_in",
+           "intermediate_act.bias": "ffn.bias_in",
+           "output.weight": "ffn.weight_out"}
+    custom_sd = {fwd.get(k, k): v for k, v in sd.items()}
+    assert "ffn.weight_in" in custom_sd
+
+    torch.manual_seed(10)
+    lay2 = BertLayer(cfg)
+    lay2.load_state_dict(custom_sd)
+    sd2 = lay2.state_dict()
+    for k, v in sd.items():
+        assert torch.equal(sd2[k], v), k
