@@ -144,9 +144,11 @@ class BertLayer(nn.Module):
 
     The out-projection / FFN-output biases fold into the following fused
     residual+LayerNorm kernel; the intermediate bias folds into the fused
-    bias+GELU kernel -- one launch each where eager PyTorch runs 2-3, and
-    their backwards feed parameter grads straight into the engine's fp32
-    accum buffer (ops/fused.py).
+    bias+GELU kernel (or, for bert-small shapes under GA_CUSTOM_FFN=auto,
+    into the intermediate GEMM's own MFMA epilogue -- ops/csrc/ffn_mfma.hip)
+    -- one launch each where eager PyTorch runs 2-3, and their backwards
+    feed parameter grads straight into the engine's fp32 accum buffer
+    (ops/fused.py).
     """
 
     def __init__(self, cfg: BertConfig):
